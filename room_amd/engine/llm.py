@@ -1,0 +1,307 @@
+"""In-process GPU inference engine with continuous batching.
+
+Replaces the reference's Ollama sidecar + HTTP hop (src/shared/local-model.ts,
+agent-executor.ts:327-338). Agents call chat() from their own threads; a
+single scheduler thread owns the GPU and fuses concurrent agents' work into
+batched prefill/decode kernel launches — the control inversion SURVEY §7
+calls out: per-agent cycle semantics unchanged, per-GPU kernels batched.
+
+Sessions: chat() reuses the sequence slot (and its paged KV blocks) when the
+new prompt extends the previous token stream — session-as-KV-cache, with the
+agent_sessions SQLite row as the durable fallback.
+"""
+from __future__ import annotations
+
+import os
+import queue
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+
+from .. import ops
+from ..models.qwen3_moe import Qwen3MoEConfig, Qwen3MoEModel
+from . import tokenizer as tok
+from .kv_cache import BLOCK_SIZE, PagedKVCache
+from .types import AgentExecutionOptions, ToolDef
+
+PREFILL_CHUNK = 4096          # max tokens per prefill forward
+DEFAULT_MAX_SEQS = 64
+
+
+@dataclass
+class GenRequest:
+    prompt_tokens: list[int]
+    max_new_tokens: int = 128
+    temperature: float = 0.7
+    top_p: float = 0.95
+    top_k: int = 40
+    session_key: Optional[str] = None
+    done: threading.Event = field(default_factory=threading.Event)
+    out_tokens: list[int] = field(default_factory=list)
+    error: Optional[str] = None
+    prefill_tokens_run: int = 0   # actually-prefilled (after cache reuse)
+    # scheduler state
+    slot: int = -1
+    pos: int = 0                  # next position to write
+    pending_prefill: list[int] = field(default_factory=list)
+    last_token: int = -1
+
+
+class _Session:
+    __slots__ = ("slot", "tokens")
+
+    def __init__(self, slot: int, tokens: list[int]):
+        self.slot = slot
+        self.tokens = tokens
+
+
+class LocalEngine:
+    """ChatEngine implementation running qwen3-coder-30b on this GPU."""
+
+    def __init__(self, cfg: Qwen3MoEConfig | None = None,
+                 device: str = "cuda", seed: int = 1234,
+                 kv_gb: float | None = None, max_seqs: int = DEFAULT_MAX_SEQS):
+        if not torch.cuda.is_available():
+            raise RuntimeError("LocalEngine requires a GPU (MI355X)")
+        if cfg is None:
+            cfg = (Qwen3MoEConfig.tiny()
+                   if os.environ.get("ROOMAMD_MODEL_CONFIG") == "tiny"
+                   else Qwen3MoEConfig.qwen3_coder_30b())
+        self.cfg = cfg
+        self.device = torch.device(device)
+        t0 = time.time()
+        self.model = Qwen3MoEModel(cfg, self.device, seed=seed)
+        self.load_seconds = time.time() - t0
+
+        bpb = PagedKVCache.bytes_per_block(cfg.num_layers, cfg.num_kv_heads,
+                                           cfg.head_dim)
+        if kv_gb is None:
+            free_b, _total = torch.cuda.mem_get_info(self.device)
+            kv_bytes = max(int(free_b - 8e9), 2 << 30)  # leave 8 GB headroom
+        else:
+            kv_bytes = int(kv_gb * (1 << 30))
+        max_blocks_per_seq = cfg.max_position // BLOCK_SIZE
+        num_blocks = min(kv_bytes // bpb, max_seqs * max_blocks_per_seq + 1)
+        self.cache = PagedKVCache(cfg.num_layers, cfg.num_kv_heads, cfg.head_dim,
+                                  int(num_blocks), max_seqs, max_blocks_per_seq,
+                                  self.device)
+        self.sessions: dict[str, _Session] = {}
+        self._queue: "queue.Queue[GenRequest]" = queue.Queue()
+        self._active: list[GenRequest] = []
+        self._lock = threading.Lock()
+        self._stop = False
+        self.stats = {"decode_steps": 0, "decode_tokens": 0, "prefill_tokens": 0,
+                      "decode_time": 0.0, "prefill_time": 0.0}
+        self._thread = threading.Thread(target=self._scheduler_loop, daemon=True,
+                                        name="room-amd-engine")
+        self._thread.start()
+
+    # ------------------------------------------------------------ public API
+
+    def generate(self, prompt_tokens: list[int], max_new_tokens: int = 128,
+                 temperature: float = 0.7, top_p: float = 0.95, top_k: int = 40,
+                 session_key: str | None = None,
+                 timeout: float = 600.0) -> GenRequest:
+        req = GenRequest(prompt_tokens=list(prompt_tokens),
+                         max_new_tokens=max_new_tokens, temperature=temperature,
+                         top_p=top_p, top_k=top_k, session_key=session_key)
+        self._queue.put(req)
+        if not req.done.wait(timeout):
+            req.error = "generation timeout"
+            raise TimeoutError("generation timed out")
+        if req.error:
+            raise RuntimeError(req.error)
+        return req
+
+    def chat(self, messages: list[dict], tools: list[ToolDef],
+             options: AgentExecutionOptions) -> tuple[str, int, int]:
+        """ChatEngine protocol (engine/providers.py)."""
+        prompt = tok.encode_chat(messages)
+        skey = f"w{options.worker_id}" if options.worker_id is not None else None
+        req = self.generate(prompt, max_new_tokens=options.max_new_tokens,
+                            temperature=options.temperature, top_p=options.top_p,
+                            top_k=options.top_k, session_key=skey)
+        text = tok.decode(req.out_tokens)
+        return text, len(prompt), len(req.out_tokens)
+
+    def release_session(self, session_key: str) -> None:
+        with self._lock:
+            s = self.sessions.pop(session_key, None)
+            if s is not None:
+                self.cache.free_seq(s.slot)
+
+    def shutdown(self) -> None:
+        self._stop = True
+        self._thread.join(timeout=5)
+
+    # ------------------------------------------------------------ scheduler
+
+    def _admit(self, req: GenRequest) -> None:
+        cache = self.cache
+        sess = self.sessions.get(req.session_key) if req.session_key else None
+        if sess is not None:
+            # reuse the longest common token prefix: the session keeps its slot
+            # and blocks; position rolls back to the divergence point.
+            cached = sess.tokens
+            common = 0
+            limit = min(len(cached), len(req.prompt_tokens) - 1)
+            while common < limit and cached[common] == req.prompt_tokens[common]:
+                common += 1
+            if common > 0:
+                req.slot = sess.slot
+                req.pos = common
+                req.pending_prefill = req.prompt_tokens[common:]
+            else:
+                cache.free_seq(sess.slot)
+                self.sessions.pop(req.session_key, None)
+                sess = None
+        if req.slot < 0:
+            req.slot = cache.alloc_seq()
+            req.pos = 0
+            req.pending_prefill = list(req.prompt_tokens)
+            if req.session_key:
+                self.sessions[req.session_key] = _Session(req.slot, [])
+        req.prefill_tokens_run = len(req.pending_prefill)
+        if not req.pending_prefill:
+            # prompt identical to cache (rare): re-run last token for logits
+            req.pos = max(0, req.pos - 1)
+            req.pending_prefill = [req.prompt_tokens[-1]]
+
+    def _scheduler_loop(self) -> None:
+        torch.cuda.set_device(self.device)
+        while not self._stop:
+            try:
+                self._scheduler_iteration()
+            except Exception as e:  # engine errors resolve all active futures
+                for r in self._active:
+                    r.error = f"engine error: {e}"
+                    r.done.set()
+                self._active.clear()
+
+    def _scheduler_iteration(self) -> None:
+        # admit new requests
+        block = not self._active
+        try:
+            req = self._queue.get(timeout=0.05 if block else 0)
+            with self._lock:
+                self._admit(req)
+            self._active.append(req)
+            while True:  # drain whatever else is queued
+                req = self._queue.get_nowait()
+                with self._lock:
+                    self._admit(req)
+                self._active.append(req)
+        except queue.Empty:
+            pass
+        if not self._active:
+            return
+
+        # phase 1: prefill pending prompts (batched across requests, chunked)
+        pre = [r for r in self._active if r.pending_prefill]
+        if pre:
+            self._prefill_step(pre)
+            return  # re-check queue between chunks
+
+        # phase 2: one decode step for all active sequences
+        self._decode_step([r for r in self._active if not r.pending_prefill])
+
+    def _prefill_step(self, reqs: list[GenRequest]) -> None:
+        t0 = time.time()
+        budget = PREFILL_CHUNK
+        tokens, seq_ids, q_pos, last_rows, sampled_reqs = [], [], [], [], []
+        for r in reqs:
+            if budget <= 0:
+                break
+            take = min(budget, len(r.pending_prefill))
+            chunk = r.pending_prefill[:take]
+            r.pending_prefill = r.pending_prefill[take:]
+            self.cache.ensure_capacity(r.slot, r.pos + take)
+            tokens.extend(chunk)
+            seq_ids.extend([r.slot] * take)
+            q_pos.extend(range(r.pos, r.pos + take))
+            r.pos += take
+            budget -= take
+            if not r.pending_prefill:  # prompt complete → sample first token
+                last_rows.append(len(tokens) - 1)
+                sampled_reqs.append(r)
+        dev = self.device
+        tokens_t = torch.tensor(tokens, dtype=torch.int64, device=dev)
+        seq_t = torch.tensor(seq_ids, dtype=torch.int32, device=dev)
+        pos_t = torch.tensor(q_pos, dtype=torch.int32, device=dev)
+        rows_t = (torch.tensor(last_rows, dtype=torch.int64, device=dev)
+                  if sampled_reqs else None)
+        logits = self.model.forward(tokens_t, seq_t, pos_t, self.cache.block_table,
+                                    self.cache.kcaches, self.cache.vcaches,
+                                    logits_rows=rows_t)
+        self.stats["prefill_tokens"] += len(tokens)
+        if sampled_reqs:
+            self._sample_and_append(sampled_reqs, logits)
+        self.stats["prefill_time"] += time.time() - t0
+
+    def _decode_step(self, reqs: list[GenRequest]) -> None:
+        if not reqs:
+            return
+        t0 = time.time()
+        dev = self.device
+        for r in reqs:
+            self.cache.ensure_capacity(r.slot, r.pos + 1)
+        tokens_t = torch.tensor([r.last_token for r in reqs], dtype=torch.int64,
+                                device=dev)
+        seq_t = torch.tensor([r.slot for r in reqs], dtype=torch.int32, device=dev)
+        pos_t = torch.tensor([r.pos for r in reqs], dtype=torch.int32, device=dev)
+        for r in reqs:
+            r.pos += 1
+        logits = self.model.forward(tokens_t, seq_t, pos_t, self.cache.block_table,
+                                    self.cache.kcaches, self.cache.vcaches)
+        self._sample_and_append(reqs, logits)
+        self.stats["decode_steps"] += 1
+        self.stats["decode_tokens"] += len(reqs)
+        self.stats["decode_time"] += time.time() - t0
+
+    def _sample_and_append(self, reqs: list[GenRequest], logits: torch.Tensor) -> None:
+        dev = self.device
+        temps = {(r.temperature, r.top_p, r.top_k) for r in reqs}
+        # all agents share sampling params in practice; take the first
+        temperature, top_p, top_k = next(iter(temps))
+        seeds = torch.randint(1, 2**62, (logits.size(0),), dtype=torch.int64,
+                              device=dev)
+        toks = ops.sample_tokens(logits, seeds, top_k=top_k,
+                                 temperature=temperature, top_p=top_p)
+        toks_host = toks.tolist()  # one small sync per step
+        finished = []
+        for r, t in zip(reqs, toks_host):
+            r.last_token = int(t)
+            r.out_tokens.append(int(t))
+            if (len(r.out_tokens) >= r.max_new_tokens
+                    or t in (tok.EOS, tok.IM_END)):
+                finished.append(r)
+        for r in finished:
+            # the sampled token at r.pos is NOT yet in KV; it will be written
+            # if the session continues (prompt extension re-runs it)
+            if r.session_key and r.session_key in self.sessions:
+                self.sessions[r.session_key].tokens = (
+                    r.prompt_tokens + r.out_tokens[:-1])
+            self._active.remove(r)
+            r.done.set()
+
+
+# ------------------------------------------------------------ singleton
+
+_engine: LocalEngine | None = None
+_engine_lock = threading.Lock()
+
+
+def get_local_engine(model: str = "qwen3-coder-30b") -> LocalEngine:
+    global _engine
+    with _engine_lock:
+        if _engine is None:
+            _engine = LocalEngine()
+        return _engine
+
+
+def set_local_engine(engine: LocalEngine) -> None:
+    global _engine
+    _engine = engine

@@ -1,0 +1,196 @@
+"""Qwen3-MoE (qwen3-coder-30b = Qwen3-30B-A3B shape) — MI355X-native forward.
+
+The reference pinned `qwen3-coder:30b` served by an external Ollama sidecar
+(src/shared/local-model.ts:3-5); here the model runs in-process: weights are
+bf16 tensors resident in HBM3E, the hot ops are the CDNA4 HIP kernels in
+room_amd/ops (RMSNorm, fused QK-norm+RoPE, paged attention, MoE router/GEMV/
+grouped-MFMA-GEMM, fused sampling), and the plain projections (QKV/O/router/
+lm_head) go through hipBLASLt via torch.nn.functional.linear.
+
+Inference-only (no autograd); a training path is out of scope for the
+reference's semantics (it never trains).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn.functional as F
+
+from .. import ops
+from ..ops.reference import rope_tables
+
+
+@dataclass
+class Qwen3MoEConfig:
+    vocab_size: int = 151936
+    hidden_size: int = 2048
+    num_layers: int = 48
+    num_q_heads: int = 32
+    num_kv_heads: int = 4
+    head_dim: int = 128
+    num_experts: int = 128
+    num_experts_per_tok: int = 8
+    moe_intermediate_size: int = 768
+    rms_eps: float = 1e-6
+    rope_theta: float = 10_000_000.0
+    max_position: int = 16384
+    # GEMV→grouped-GEMM switchover: below this many tokens the decode GEMV
+    # path wins (per-pair weight streaming); above it, grouped MFMA GEMM.
+    moe_grouped_threshold: int = 16
+
+    @staticmethod
+    def qwen3_coder_30b() -> "Qwen3MoEConfig":
+        return Qwen3MoEConfig()
+
+    @staticmethod
+    def tiny(vocab: int = 4096) -> "Qwen3MoEConfig":
+        """Small config for smoke tests / CPU-free validation on one GPU call."""
+        return Qwen3MoEConfig(vocab_size=vocab, hidden_size=512, num_layers=4,
+                              num_q_heads=8, num_kv_heads=1, head_dim=128,
+                              num_experts=16, num_experts_per_tok=4,
+                              moe_intermediate_size=256, max_position=4096)
+
+
+class Qwen3MoELayer:
+    def __init__(self, cfg: Qwen3MoEConfig, device: torch.device, gen: torch.Generator):
+        h, d = cfg.hidden_size, cfg.head_dim
+        qdim = cfg.num_q_heads * d
+        kvdim = cfg.num_kv_heads * d
+        std = 0.02
+
+        def rnd(*shape):
+            return torch.empty(*shape, dtype=torch.bfloat16, device=device).normal_(
+                0, std, generator=gen)
+
+        self.input_norm_w = torch.ones(h, dtype=torch.bfloat16, device=device)
+        self.post_attn_norm_w = torch.ones(h, dtype=torch.bfloat16, device=device)
+        self.q_norm_w = torch.ones(d, dtype=torch.bfloat16, device=device)
+        self.k_norm_w = torch.ones(d, dtype=torch.bfloat16, device=device)
+        self.wqkv = rnd(qdim + 2 * kvdim, h)      # fused QKV projection
+        self.wo = rnd(h, qdim)
+        self.router_w = rnd(cfg.num_experts, h)
+        self.w13 = rnd(cfg.num_experts, 2 * cfg.moe_intermediate_size, h)
+        self.w2 = rnd(cfg.num_experts, h, cfg.moe_intermediate_size)
+
+
+class Qwen3MoEModel:
+    """Flat-tensor model (no nn.Module overhead on the decode path)."""
+
+    def __init__(self, cfg: Qwen3MoEConfig, device: str | torch.device = "cuda",
+                 seed: int = 1234):
+        self.cfg = cfg
+        self.device = torch.device(device)
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(seed)
+        h = cfg.hidden_size
+        self.embed = torch.empty(cfg.vocab_size, h, dtype=torch.bfloat16,
+                                 device=self.device).normal_(0, 0.02, generator=gen)
+        self.layers = [Qwen3MoELayer(cfg, self.device, gen)
+                       for _ in range(cfg.num_layers)]
+        self.final_norm_w = torch.ones(h, dtype=torch.bfloat16, device=self.device)
+        self.lm_head = torch.empty(cfg.vocab_size, h, dtype=torch.bfloat16,
+                                   device=self.device).normal_(0, 0.02, generator=gen)
+        cos_t, sin_t = rope_tables(cfg.max_position, cfg.head_dim, cfg.rope_theta)
+        self.cos_t = cos_t.to(self.device)
+        self.sin_t = sin_t.to(self.device)
+        self.scale = cfg.head_dim ** -0.5
+
+    def num_params(self) -> int:
+        cfg = self.cfg
+        per_layer = (self.layers[0].wqkv.numel() + self.layers[0].wo.numel()
+                     + self.layers[0].router_w.numel() + self.layers[0].w13.numel()
+                     + self.layers[0].w2.numel())
+        return self.embed.numel() + self.lm_head.numel() + cfg.num_layers * per_layer
+
+    @torch.inference_mode()
+    def forward(self, tokens: torch.Tensor, seq_ids: torch.Tensor,
+                q_pos: torch.Tensor, block_table: torch.Tensor,
+                kcaches: list[torch.Tensor], vcaches: list[torch.Tensor],
+                logits_rows: torch.Tensor | None = None) -> torch.Tensor:
+        """tokens/seq_ids/q_pos: [T] on device. kcaches/vcaches: one pair per
+        layer. logits_rows: row indices to compute logits for (default: all).
+        Returns fp32 logits [R, vocab]."""
+        cfg = self.cfg
+        T = tokens.numel()
+        x = self.embed[tokens.long()]              # [T, H] bf16 (residual stream)
+        hbuf = torch.empty_like(x)                 # normed activations
+        moe_out = None                             # pending delta for fused add
+
+        for li, layer in enumerate(self.layers):
+            # --- attention block
+            if moe_out is None:
+                ops.rmsnorm(hbuf, x, layer.input_norm_w, cfg.rms_eps)
+            else:
+                ops.fused_add_rmsnorm(hbuf, x, moe_out, layer.input_norm_w,
+                                      cfg.rms_eps)
+            qkv = F.linear(hbuf, layer.wqkv)       # hipBLASLt GEMM
+            qdim = cfg.num_q_heads * cfg.head_dim
+            kvdim = cfg.num_kv_heads * cfg.head_dim
+            q = qkv[:, :qdim].reshape(T, cfg.num_q_heads, cfg.head_dim).contiguous()
+            k = qkv[:, qdim:qdim + kvdim].reshape(T, cfg.num_kv_heads,
+                                                  cfg.head_dim).contiguous()
+            v = qkv[:, qdim + kvdim:].reshape(T, cfg.num_kv_heads,
+                                              cfg.head_dim).contiguous()
+            ops.qk_norm_rope(q, k, layer.q_norm_w, layer.k_norm_w, self.cos_t,
+                             self.sin_t, q_pos, cfg.num_q_heads, cfg.num_kv_heads,
+                             cfg.head_dim, cfg.rms_eps)
+            ops.write_kv(kcaches[li], vcaches[li], k, v, block_table, seq_ids, q_pos)
+            attn = torch.empty_like(q)
+            ops.paged_attention(attn, q, kcaches[li], vcaches[li], block_table,
+                                seq_ids, q_pos, self.scale)
+            o = F.linear(attn.reshape(T, qdim), layer.wo)
+
+            # --- MoE block
+            ops.fused_add_rmsnorm(hbuf, x, o, layer.post_attn_norm_w, cfg.rms_eps)
+            router_logits = F.linear(hbuf.float(), layer.router_w.float())
+            topk_ids, topk_w = ops.moe_router(router_logits, cfg.num_experts_per_tok)
+            moe_out_f32 = self._moe(hbuf, layer, topk_ids, topk_w)
+            moe_out = moe_out_f32.to(torch.bfloat16)
+
+        # final residual add + norm
+        ops.fused_add_rmsnorm(hbuf, x, moe_out, self.final_norm_w, cfg.rms_eps)
+        sel = hbuf if logits_rows is None else hbuf[logits_rows.long()]
+        logits = F.linear(sel, self.lm_head).float()
+        return logits
+
+    def _moe(self, hbuf: torch.Tensor, layer: Qwen3MoELayer,
+             topk_ids: torch.Tensor, topk_w: torch.Tensor) -> torch.Tensor:
+        cfg = self.cfg
+        T = hbuf.size(0)
+        K = cfg.num_experts_per_tok
+        H, I = cfg.hidden_size, cfg.moe_intermediate_size
+        out = torch.zeros(T, H, dtype=torch.float32, device=hbuf.device)
+
+        if T < cfg.moe_grouped_threshold:
+            # decode path: GEMV per (token, expert) pair
+            pair_token = torch.arange(T, device=hbuf.device,
+                                      dtype=torch.int32).repeat_interleave(K)
+            pair_expert = topk_ids.flatten().contiguous()
+            pair_w = topk_w.flatten().contiguous()
+            P = pair_token.numel()
+            h = torch.empty(P, I, dtype=torch.bfloat16, device=hbuf.device)
+            ops.moe_gemv_h(h, hbuf, layer.w13, pair_token, pair_expert)
+            ops.moe_gemv_down(out, h, layer.w2, pair_w, pair_token, pair_expert)
+            return out
+
+        # prefill path: sort pairs by expert, grouped MFMA GEMMs
+        flat_expert = topk_ids.flatten()
+        order = torch.argsort(flat_expert).int()
+        pair_expert = flat_expert[order.long()].contiguous()
+        pair_token = (torch.arange(T, device=hbuf.device, dtype=torch.int32)
+                      .repeat_interleave(K))[order.long()].contiguous()
+        pair_w = topk_w.flatten()[order.long()].contiguous()
+        P = pair_token.numel()
+        desc1 = ops.build_moe_tile_desc(pair_expert, (2 * I) // 64).to(hbuf.device)
+        gateup = torch.empty(P, 2 * I, dtype=torch.bfloat16, device=hbuf.device)
+        ops.moe_grouped_gemm(gateup, hbuf, layer.w13, pair_token, desc1)
+        h = torch.empty(P, I, dtype=torch.bfloat16, device=hbuf.device)
+        ops.silu_mul(h, gateup)
+        desc2 = ops.build_moe_tile_desc(pair_expert, H // 64).to(hbuf.device)
+        z = torch.empty(P, H, dtype=torch.bfloat16, device=hbuf.device)
+        ops.moe_grouped_gemm(z, h, layer.w2,
+                             torch.arange(P, device=hbuf.device, dtype=torch.int32),
+                             desc2)
+        ops.moe_combine(out, z, pair_w, pair_token)
+        return out

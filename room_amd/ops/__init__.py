@@ -1,0 +1,146 @@
+"""CDNA4 kernel library bindings.
+
+On a GPU box the extension MUST load — there is no eager/PyTorch fallback for
+the compute path (a silent fallback would fake GPU test results). CPU-only
+environments (CI here) can import this module; calling any op raises.
+"""
+from __future__ import annotations
+
+import torch
+
+_C = None
+_import_error: Exception | None = None
+try:
+    from room_amd import _C  # type: ignore
+except Exception as e:  # pragma: no cover
+    _import_error = e
+
+if _C is None and torch.cuda.is_available():
+    raise ImportError(
+        "room_amd._C HIP extension missing on a GPU machine — build it with "
+        "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace`. "
+        f"Import error: {_import_error}")
+
+
+def _require():
+    if _C is None:
+        raise RuntimeError(
+            f"room_amd._C not available (CPU-only environment?): {_import_error}")
+    return _C
+
+
+def rmsnorm(out: torch.Tensor, x: torch.Tensor, weight: torch.Tensor,
+            eps: float = 1e-6) -> torch.Tensor:
+    _require().rmsnorm(out, x, weight, eps)
+    return out
+
+
+def fused_add_rmsnorm(out: torch.Tensor, residual: torch.Tensor, x: torch.Tensor,
+                      weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    _require().fused_add_rmsnorm(out, residual, x, weight, eps)
+    return out
+
+
+def qk_norm_rope(q: torch.Tensor, k: torch.Tensor, q_w: torch.Tensor,
+                 k_w: torch.Tensor, cos_t: torch.Tensor, sin_t: torch.Tensor,
+                 positions: torch.Tensor, n_qheads: int, n_kvheads: int,
+                 head_dim: int = 128, eps: float = 1e-6) -> None:
+    _require().qk_norm_rope(q, k, q_w, k_w, cos_t, sin_t, positions,
+                            n_qheads, n_kvheads, head_dim, eps)
+
+
+def silu_mul(out: torch.Tensor, gateup: torch.Tensor) -> torch.Tensor:
+    _require().silu_mul(out, gateup)
+    return out
+
+
+def paged_attention(out: torch.Tensor, q: torch.Tensor, kcache: torch.Tensor,
+                    vcache: torch.Tensor, block_table: torch.Tensor,
+                    seq_ids: torch.Tensor, q_pos: torch.Tensor,
+                    scale: float) -> torch.Tensor:
+    _require().paged_attention(out, q, kcache, vcache, block_table, seq_ids,
+                               q_pos, scale)
+    return out
+
+
+def write_kv(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
+             v: torch.Tensor, block_table: torch.Tensor, seq_ids: torch.Tensor,
+             q_pos: torch.Tensor) -> None:
+    _require().write_kv(kcache, vcache, k, v, block_table, seq_ids, q_pos)
+
+
+def moe_router(logits: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:
+    T = logits.size(0)
+    ids = torch.empty(T, k, dtype=torch.int32, device=logits.device)
+    w = torch.empty(T, k, dtype=torch.float32, device=logits.device)
+    _require().moe_router(ids, w, logits, k)
+    return ids, w
+
+
+def moe_gemv_h(h: torch.Tensor, x: torch.Tensor, w13: torch.Tensor,
+               pair_token: torch.Tensor, pair_expert: torch.Tensor) -> torch.Tensor:
+    _require().moe_gemv_h(h, x, w13, pair_token, pair_expert)
+    return h
+
+
+def moe_gemv_down(out: torch.Tensor, h: torch.Tensor, w2: torch.Tensor,
+                  pair_w: torch.Tensor, pair_token: torch.Tensor,
+                  pair_expert: torch.Tensor) -> torch.Tensor:
+    _require().moe_gemv_down(out, h, w2, pair_w, pair_token, pair_expert)
+    return out
+
+
+def moe_grouped_gemm(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
+                     pair_token: torch.Tensor, tile_desc: torch.Tensor) -> torch.Tensor:
+    _require().moe_grouped_gemm(out, x, w, pair_token, tile_desc)
+    return out
+
+
+def moe_combine(out: torch.Tensor, z: torch.Tensor, pair_w: torch.Tensor,
+                pair_token: torch.Tensor) -> torch.Tensor:
+    _require().moe_combine(out, z, pair_w, pair_token)
+    return out
+
+
+def sample_tokens(logits: torch.Tensor, seeds: torch.Tensor, top_k: int = 40,
+                  temperature: float = 0.7, top_p: float = 0.95) -> torch.Tensor:
+    out = torch.empty(logits.size(0), dtype=torch.int32, device=logits.device)
+    _require().sample_tokens(out, logits, seeds, top_k, temperature, top_p)
+    return out
+
+
+def vs_topk(mat: torch.Tensor, query: torch.Tensor, k: int,
+            nblocks: int = 512) -> tuple[torch.Tensor, torch.Tensor]:
+    """Cosine top-k over an HBM-resident [N, 384] bf16 matrix (rows L2-normed)."""
+    dev = mat.device
+    nblocks = min(nblocks, max(1, (mat.size(0) + 255) // 256))
+    cand_v = torch.empty(nblocks, k, dtype=torch.float32, device=dev)
+    cand_i = torch.empty(nblocks, k, dtype=torch.int32, device=dev)
+    out_v = torch.empty(k, dtype=torch.float32, device=dev)
+    out_i = torch.empty(k, dtype=torch.int64, device=dev)
+    _require().vs_topk(out_v, out_i, cand_v, cand_i, mat, query, k)
+    return out_v, out_i
+
+
+def build_moe_tile_desc(pair_expert_sorted: torch.Tensor, n_tiles_n: int,
+                        bm: int = 16) -> torch.Tensor:
+    """Host-side tile descriptors for the grouped MFMA GEMM: one entry
+    (expert, row_start, m_size, n_tile) per (M-tile × N-tile). pair_expert_sorted
+    must be ascending (tokens grouped by expert)."""
+    pe = pair_expert_sorted.cpu()
+    desc = []
+    P = pe.numel()
+    i = 0
+    while i < P:
+        e = int(pe[i])
+        j = i
+        while j < P and int(pe[j]) == e:
+            j += 1
+        for m0 in range(i, j, bm):
+            msize = min(bm, j - m0)
+            for nt in range(n_tiles_n):
+                desc.append((e, m0, msize, nt))
+        i = j
+    if not desc:
+        return torch.zeros(0, 4, dtype=torch.int32)
+    return torch.tensor(desc, dtype=torch.int32)

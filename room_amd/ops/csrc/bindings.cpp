@@ -1,0 +1,50 @@
+// Python bindings for room_amd._C — the CDNA4 kernel library.
+#include <torch/extension.h>
+
+void rmsnorm(torch::Tensor out, torch::Tensor in, torch::Tensor weight, double eps);
+void fused_add_rmsnorm(torch::Tensor out, torch::Tensor residual, torch::Tensor in,
+                       torch::Tensor weight, double eps);
+void qk_norm_rope(torch::Tensor q, torch::Tensor k, torch::Tensor q_w,
+                  torch::Tensor k_w, torch::Tensor cos_t, torch::Tensor sin_t,
+                  torch::Tensor positions, int64_t n_qheads, int64_t n_kvheads,
+                  int64_t head_dim, double eps);
+void silu_mul(torch::Tensor out, torch::Tensor gateup);
+void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
+                     torch::Tensor vcache, torch::Tensor block_table,
+                     torch::Tensor seq_ids, torch::Tensor q_pos, double scale);
+void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
+              torch::Tensor v, torch::Tensor block_table, torch::Tensor seq_ids,
+              torch::Tensor q_pos);
+void moe_router(torch::Tensor topk_ids, torch::Tensor topk_w, torch::Tensor logits,
+                int64_t K);
+void moe_gemv_h(torch::Tensor h, torch::Tensor x, torch::Tensor w13,
+                torch::Tensor pair_token, torch::Tensor pair_expert);
+void moe_gemv_down(torch::Tensor out, torch::Tensor h, torch::Tensor w2,
+                   torch::Tensor pair_w, torch::Tensor pair_token,
+                   torch::Tensor pair_expert);
+void moe_grouped_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                      torch::Tensor pair_token, torch::Tensor tile_desc);
+void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
+                 torch::Tensor pair_token);
+void sample_tokens(torch::Tensor out_tokens, torch::Tensor logits,
+                   torch::Tensor seeds, int64_t top_k, double temperature,
+                   double top_p);
+void vs_topk(torch::Tensor out_v, torch::Tensor out_i, torch::Tensor cand_v,
+             torch::Tensor cand_i, torch::Tensor mat, torch::Tensor query,
+             int64_t K);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "residual += x; rmsnorm(residual)");
+  m.def("qk_norm_rope", &qk_norm_rope, "fused per-head QK RMSNorm + RoPE");
+  m.def("silu_mul", &silu_mul, "silu(gate) * up");
+  m.def("paged_attention", &paged_attention, "paged causal attention (GQA 8:1)");
+  m.def("write_kv", &write_kv, "scatter K/V rows into paged cache");
+  m.def("moe_router", &moe_router, "softmax top-k router");
+  m.def("moe_gemv_h", &moe_gemv_h, "MoE gate/up GEMV + silu-mul (decode)");
+  m.def("moe_gemv_down", &moe_gemv_down, "MoE down GEMV + weighted scatter-add");
+  m.def("moe_grouped_gemm", &moe_grouped_gemm, "grouped MFMA GEMM (prefill)");
+  m.def("moe_combine", &moe_combine, "weighted scatter-add combine");
+  m.def("sample_tokens", &sample_tokens, "fused temperature/top-k/top-p sampling");
+  m.def("vs_topk", &vs_topk, "vector-store cosine top-k over bf16 matrix");
+}

@@ -1,0 +1,312 @@
+// Mixture-of-Experts kernels for Qwen3-30B-A3B on CDNA4 (gfx950).
+//   E=128 experts, top-8, hidden H=2048, expert intermediate I=768.
+//
+// Weight layouts (we own them):
+//   W13: [E, 2*I, H]  — gate rows [0,I), up rows [I,2I); each row contiguous
+//                       over H (GEMV-friendly streaming, MFMA B-operand rows)
+//   W2:  [E, H, I]    — down-proj rows contiguous over I
+//
+// Decode path (few tokens): wave-per-output GEMV — each (token,expert) pair
+//   is bandwidth-bound on expert weights; 64-lane dot with 16 B/lane loads.
+// Prefill path (many tokens): grouped MFMA GEMM (16x16x32 bf16) with LDS
+//   tiles and +16B row padding against bank conflicts; tokens pre-sorted by
+//   expert on host, tile descriptors built host-side.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include "common.h"
+
+// ---------------------------------------------------------------- router
+// softmax over E=128 logits, pick top-K=8, renormalize their probs.
+// one wave per token; lane l owns logits[2l], [2l+1].
+__global__ void moe_router_kernel(int* __restrict__ topk_ids,      // [T, K]
+                                  float* __restrict__ topk_w,      // [T, K]
+                                  const float* __restrict__ logits,  // [T, E]
+                                  int E, int K) {
+  const int t = blockIdx.x;
+  const int lane = threadIdx.x;
+  const float l0 = (2 * lane < E) ? logits[(long)t * E + 2 * lane] : -INFINITY;
+  const float l1 = (2 * lane + 1 < E) ? logits[(long)t * E + 2 * lane + 1] : -INFINITY;
+
+  float m = wave_reduce_max(fmaxf(l0, l1));
+  float e0 = __expf(l0 - m), e1 = __expf(l1 - m);
+  float denom = wave_reduce_sum(e0 + e1);
+
+  // iterative top-K selection by masking
+  float v0 = l0, v1 = l1;
+  float picked_sum = 0.f;
+  for (int k = 0; k < K; ++k) {
+    float best = fmaxf(v0, v1);
+    float gmax = wave_reduce_max(best);
+    // first lane holding gmax wins; prefer slot 0
+    int cand_id = (v0 == gmax) ? 2 * lane : ((v1 == gmax) ? 2 * lane + 1 : INT_MAX);
+    int winner = cand_id;
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      winner = min(winner, __shfl_xor(winner, off, WAVE));
+    float prob = __expf(gmax - m) / denom;
+    if (lane == 0) {
+      topk_ids[(long)t * K + k] = winner;
+      topk_w[(long)t * K + k] = prob;
+    }
+    picked_sum += prob;  // same value in all lanes
+    if (winner == 2 * lane) v0 = -INFINITY;
+    if (winner == 2 * lane + 1) v1 = -INFINITY;
+  }
+  // renormalize among top-K (Qwen3 norm_topk_prob)
+  if (lane < K) {
+    float w = topk_w[(long)t * K + lane];
+    topk_w[(long)t * K + lane] = w / picked_sum;
+  }
+}
+
+// ---------------------------------------------------------------- GEMV path
+// h[p][j] = silu(dot(x_t, Wg_row_j)) * dot(x_t, Wu_row_j)
+// grid: (npairs, I/4); block 256 = 4 waves; wave w computes output j.
+__global__ __launch_bounds__(256)
+void moe_gemv_h_kernel(short* __restrict__ h,             // [P, I]
+                       const short* __restrict__ x,        // [T, H]
+                       const short* __restrict__ w13,      // [E, 2I, H]
+                       const int* __restrict__ pair_token,  // [P]
+                       const int* __restrict__ pair_expert, // [P]
+                       int H, int I) {
+  const int p = blockIdx.x;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int j = blockIdx.y * 4 + wid;
+  if (j >= I) return;
+  const int t = pair_token[p];
+  const int e = pair_expert[p];
+  const short* xrow = x + (long)t * H;
+  const short* grow = w13 + ((long)e * 2 * I + j) * H;
+  const short* urow = w13 + ((long)e * 2 * I + I + j) * H;
+
+  float dg = 0.f, du = 0.f;
+  for (int base = lane * 8; base < H; base += WAVE * 8) {
+    bf16x8 xv = *reinterpret_cast<const bf16x8*>(xrow + base);
+    bf16x8 gv = *reinterpret_cast<const bf16x8*>(grow + base);
+    bf16x8 uv = *reinterpret_cast<const bf16x8*>(urow + base);
+    #pragma unroll
+    for (int q_ = 0; q_ < 8; ++q_) {
+      float xf = bf2f(xv[q_]);
+      dg += xf * bf2f(gv[q_]);
+      du += xf * bf2f(uv[q_]);
+    }
+  }
+  dg = wave_reduce_sum(dg);
+  du = wave_reduce_sum(du);
+  if (lane == 0) {
+    float s = dg / (1.0f + __expf(-dg));
+    h[(long)p * I + j] = f2bf(s * du);
+  }
+}
+
+// z[t] += w_p * (h_p @ W2_e^T): wave per output dim o; atomic f32 add.
+__global__ __launch_bounds__(256)
+void moe_gemv_down_kernel(float* __restrict__ out,         // [T, H] f32
+                          const short* __restrict__ h,      // [P, I]
+                          const short* __restrict__ w2,     // [E, H, I]
+                          const float* __restrict__ pair_w,  // [P]
+                          const int* __restrict__ pair_token,
+                          const int* __restrict__ pair_expert,
+                          int H, int I) {
+  const int p = blockIdx.x;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int o = blockIdx.y * 4 + wid;
+  if (o >= H) return;
+  const int t = pair_token[p];
+  const int e = pair_expert[p];
+  const short* hrow = h + (long)p * I;
+  const short* wrow = w2 + ((long)e * H + o) * I;
+
+  float d = 0.f;
+  for (int base = lane * 8; base < I; base += WAVE * 8) {
+    bf16x8 hv = *reinterpret_cast<const bf16x8*>(hrow + base);
+    bf16x8 wv = *reinterpret_cast<const bf16x8*>(wrow + base);
+    #pragma unroll
+    for (int q_ = 0; q_ < 8; ++q_) d += bf2f(hv[q_]) * bf2f(wv[q_]);
+  }
+  d = wave_reduce_sum(d);
+  if (lane == 0)
+    atomicAdd(out + (long)t * H + o, d * pair_w[p]);
+}
+
+// ------------------------------------------------ grouped MFMA GEMM (prefill)
+// C[P_sorted, N] = X[token(p), :] @ W[e, :, :]^T for tokens grouped by expert.
+// Tile: BM=16 rows (pairs), BN=64 (4 waves × 16), BK=64.
+// tile_desc per workgroup: (expert, pair_row_start, m_size, n_tile)
+//
+// MFMA fragment layouts (gfx950 mfma_f32_16x16x32_bf16, verified vs torch in
+// tests/test_kernels_gpu.py):
+//   A (16×32): lane l holds A[l&15][(l>>4)*8 .. +8]
+//   B (32×16): lane l holds B[(l>>4)*8 .. +8][l&15]
+//   C (16×16): lane l, reg r ↦ row (l>>4)*4 + r, col l&15
+typedef __attribute__((ext_vector_type(4))) float cfrag_t;
+
+#define GG_BM 16
+#define GG_BN 64
+#define GG_BK 64
+#define GG_PAD 8   // bf16 elements of row padding in LDS (16B, keeps b128 alignment)
+
+__global__ __launch_bounds__(256)
+void moe_grouped_gemm_kernel(short* __restrict__ out,       // [P, N] bf16
+                             const short* __restrict__ x,    // [T, H]
+                             const short* __restrict__ w,    // [E, N, H]
+                             const int* __restrict__ pair_token,   // [P] sorted
+                             const int* __restrict__ tile_desc,    // [G, 4]
+                             int H, int N) {
+  const int g = blockIdx.x;
+  const int e = tile_desc[g * 4 + 0];
+  const int row0 = tile_desc[g * 4 + 1];
+  const int msize = tile_desc[g * 4 + 2];
+  const int n0 = tile_desc[g * 4 + 3] * GG_BN;
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+
+  __shared__ short xs[GG_BM][GG_BK + GG_PAD];
+  __shared__ short ws[GG_BN][GG_BK + GG_PAD];
+
+  const short* wbase = w + (long)e * N * H;
+
+  cfrag_t acc = {0.f, 0.f, 0.f, 0.f};  // each wave: one 16×16 C fragment at n0+wid*16
+
+  for (int k0 = 0; k0 < H; k0 += GG_BK) {
+    // stage X tile: 16 rows × 64 k = 128 × 8-vec; 256 threads → first 128
+    {
+      const int nvec = GG_BM * GG_BK / 8;  // 128
+      if (tid < nvec) {
+        const int r = (tid * 8) / GG_BK;
+        const int c = (tid * 8) % GG_BK;
+        bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (r < msize) {
+          const int tok = pair_token[row0 + r];
+          v = *reinterpret_cast<const bf16x8*>(x + (long)tok * H + k0 + c);
+        }
+        *reinterpret_cast<bf16x8*>(&xs[r][c]) = v;
+      }
+    }
+    // stage W tile: 64 rows × 64 k = 512 × 8-vec; 256 threads → 2 each
+    {
+      #pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        const int idx = tid + it * 256;
+        const int r = (idx * 8) / GG_BK;
+        const int c = (idx * 8) % GG_BK;
+        bf16x8 v = *reinterpret_cast<const bf16x8*>(
+            wbase + (long)(n0 + r) * H + k0 + c);
+        *reinterpret_cast<bf16x8*>(&ws[r][c]) = v;
+      }
+    }
+    __syncthreads();
+
+    // 2 MFMA per K-tile (K=32 each)
+    #pragma unroll
+    for (int kk = 0; kk < GG_BK / 32; ++kk) {
+      const int arow = lane & 15;
+      const int akoff = (lane >> 4) * 8;
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(&xs[arow][kk * 32 + akoff]);
+      const int bcol = wid * 16 + (lane & 15);
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(&ws[bcol][kk * 32 + akoff]);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C fragment — lane l, reg r → row (l>>4)*4+r, col l&15
+  const int crow = (lane >> 4) * 4;
+  const int ccol = wid * 16 + (lane & 15);
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = crow + r;
+    if (m < msize && n0 + ccol < N)
+      out[(long)(row0 + m) * N + n0 + ccol] = f2bf(acc[r]);
+  }
+}
+
+// combine: out[t] += w_p * z[p]  (vectorized scatter-add over sorted pairs)
+__global__ void moe_combine_kernel(float* __restrict__ out,     // [T, H] f32
+                                   const short* __restrict__ z,  // [P, H]
+                                   const float* __restrict__ pair_w,
+                                   const int* __restrict__ pair_token,
+                                   int H) {
+  const int p = blockIdx.x;
+  const int t = pair_token[p];
+  const float wp = pair_w[p];
+  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+    bf16x8 zv = *reinterpret_cast<const bf16x8*>(z + (long)p * H + i);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j)
+      atomicAdd(out + (long)t * H + i + j, wp * bf2f(zv[j]));
+  }
+}
+
+// ---------------------------------------------------------------- wrappers
+
+void moe_router(torch::Tensor topk_ids, torch::Tensor topk_w, torch::Tensor logits,
+                int64_t K) {
+  const int T = logits.size(0), E = logits.size(1);
+  TORCH_CHECK(E <= 128, "router kernel handles E<=128");
+  TORCH_CHECK(logits.dtype() == torch::kFloat32);
+  dim3 grid(T), block(WAVE);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(moe_router_kernel, grid, block, 0, s,
+                     topk_ids.data_ptr<int>(), topk_w.data_ptr<float>(),
+                     logits.data_ptr<float>(), E, (int)K);
+  HIP_CHECK_KERNEL();
+}
+
+void moe_gemv_h(torch::Tensor h, torch::Tensor x, torch::Tensor w13,
+                torch::Tensor pair_token, torch::Tensor pair_expert) {
+  const int P = pair_token.size(0);
+  const int H = x.size(-1), I = h.size(-1);
+  dim3 grid(P, (I + 3) / 4), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(moe_gemv_h_kernel, grid, block, 0, s,
+                     (short*)h.data_ptr(), (const short*)x.data_ptr(),
+                     (const short*)w13.data_ptr(), pair_token.data_ptr<int>(),
+                     pair_expert.data_ptr<int>(), H, I);
+  HIP_CHECK_KERNEL();
+}
+
+void moe_gemv_down(torch::Tensor out, torch::Tensor h, torch::Tensor w2,
+                   torch::Tensor pair_w, torch::Tensor pair_token,
+                   torch::Tensor pair_expert) {
+  const int P = pair_token.size(0);
+  const int H = out.size(-1), I = h.size(-1);
+  TORCH_CHECK(out.dtype() == torch::kFloat32);
+  dim3 grid(P, (H + 3) / 4), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(moe_gemv_down_kernel, grid, block, 0, s,
+                     out.data_ptr<float>(), (const short*)h.data_ptr(),
+                     (const short*)w2.data_ptr(), pair_w.data_ptr<float>(),
+                     pair_token.data_ptr<int>(), pair_expert.data_ptr<int>(), H, I);
+  HIP_CHECK_KERNEL();
+}
+
+void moe_grouped_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                      torch::Tensor pair_token, torch::Tensor tile_desc) {
+  const int H = x.size(-1), N = out.size(-1);
+  const int G = tile_desc.size(0);
+  TORCH_CHECK(H % GG_BK == 0 && N % GG_BN == 0);
+  dim3 grid(G), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(moe_grouped_gemm_kernel, grid, block, 0, s,
+                     (short*)out.data_ptr(), (const short*)x.data_ptr(),
+                     (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
+                     tile_desc.data_ptr<int>(), H, N);
+  HIP_CHECK_KERNEL();
+}
+
+void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
+                 torch::Tensor pair_token) {
+  const int P = pair_token.size(0), H = out.size(-1);
+  TORCH_CHECK(out.dtype() == torch::kFloat32);
+  dim3 grid(P), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(moe_combine_kernel, grid, block, 0, s,
+                     out.data_ptr<float>(), (const short*)z.data_ptr(),
+                     pair_w.data_ptr<float>(), pair_token.data_ptr<int>(), H);
+  HIP_CHECK_KERNEL();
+}

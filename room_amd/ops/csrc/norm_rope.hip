@@ -1,0 +1,220 @@
+// RMSNorm (+fused residual-add) and fused per-head QK-RMSNorm + RoPE.
+//
+// The reference delegated all model math to Ollama (src/shared/local-model.ts);
+// these are the in-process CDNA4 replacements. All memory-bound: bf16 loads
+// vectorized 8-wide (16 B/lane), fp32 accumulation, one workgroup per row.
+// cos/sin tables are host-precomputed (on-device trig turns memory-bound into
+// VALU-bound — guide Appendix B).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include "common.h"
+
+// ---------------------------------------------------------------- rmsnorm
+// y = x / rms(x) * w.   rows × cols, cols % 8 == 0, cols <= 8192.
+
+__global__ void rmsnorm_kernel(short* __restrict__ out,
+                               const short* __restrict__ in,
+                               const short* __restrict__ weight,
+                               int cols, float eps) {
+  __shared__ float red[16];
+  const long row = blockIdx.x;
+  const short* x = in + row * (long)cols;
+  short* y = out + row * (long)cols;
+
+  float sumsq = 0.f;
+  const int vecs = cols / 8;
+  for (int i = threadIdx.x; i < vecs; i += blockDim.x) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) { float f = bf2f(v[j]); sumsq += f * f; }
+  }
+  float total = block_reduce_sum(sumsq, red);
+  float scale = rsqrtf(total / cols + eps);
+
+  for (int i = threadIdx.x; i < vecs; i += blockDim.x) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
+    bf16x8 w = *reinterpret_cast<const bf16x8*>(weight + i * 8);
+    bf16x8 o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(bf2f(v[j]) * scale * bf2f(w[j]));
+    *reinterpret_cast<bf16x8*>(y + i * 8) = o;
+  }
+}
+
+// residual = residual + x;  y = rmsnorm(residual) * w   (fused: one HBM pass)
+__global__ void fused_add_rmsnorm_kernel(short* __restrict__ out,
+                                         short* __restrict__ residual,
+                                         const short* __restrict__ in,
+                                         const short* __restrict__ weight,
+                                         int cols, float eps) {
+  __shared__ float red[16];
+  const long row = blockIdx.x;
+  const short* x = in + row * (long)cols;
+  short* r = residual + row * (long)cols;
+  short* y = out + row * (long)cols;
+
+  float sumsq = 0.f;
+  const int vecs = cols / 8;
+  for (int i = threadIdx.x; i < vecs; i += blockDim.x) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
+    bf16x8 rv = *reinterpret_cast<const bf16x8*>(r + i * 8);
+    bf16x8 nr;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(v[j]) + bf2f(rv[j]);
+      nr[j] = f2bf(f);
+      float g = bf2f(nr[j]);         // norm over the *stored* bf16 residual
+      sumsq += g * g;
+    }
+    *reinterpret_cast<bf16x8*>(r + i * 8) = nr;
+  }
+  float total = block_reduce_sum(sumsq, red);
+  float scale = rsqrtf(total / cols + eps);
+
+  for (int i = threadIdx.x; i < vecs; i += blockDim.x) {
+    bf16x8 rv = *reinterpret_cast<const bf16x8*>(r + i * 8);
+    bf16x8 w = *reinterpret_cast<const bf16x8*>(weight + i * 8);
+    bf16x8 o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(bf2f(rv[j]) * scale * bf2f(w[j]));
+    *reinterpret_cast<bf16x8*>(y + i * 8) = o;
+  }
+}
+
+// ------------------------------------------------ fused QK-norm + RoPE
+// Qwen3 applies per-head RMSNorm to Q and K, then rotary embedding.
+// One fused kernel: grid = (tokens, q_heads + kv_heads); each workgroup is
+// TWO waves (head_dim=128 → one f32/lane per half). rotate-half (NeoX) form:
+//   out[i]      = x[i]   * cos[i] - x[i+D/2] * sin[i]
+//   out[i+D/2]  = x[i+D/2] * cos[i] + x[i]   * sin[i]
+// cos/sin: [max_pos, D/2] fp32, positions: [tokens] int32.
+
+__global__ void qk_norm_rope_kernel(short* __restrict__ q,      // [T, Hq*D]
+                                    short* __restrict__ k,      // [T, Hk*D]
+                                    const short* __restrict__ q_w,  // [D]
+                                    const short* __restrict__ k_w,  // [D]
+                                    const float* __restrict__ cos_t,  // [P, D/2]
+                                    const float* __restrict__ sin_t,
+                                    const int* __restrict__ pos,      // [T]
+                                    int n_qheads, int n_kvheads,
+                                    int head_dim, float eps) {
+  const int t = blockIdx.x;
+  const int h = blockIdx.y;
+  const bool is_q = h < n_qheads;
+  short* base = is_q ? (q + ((long)t * n_qheads + h) * head_dim)
+                     : (k + ((long)t * n_kvheads + (h - n_qheads)) * head_dim);
+  const short* w = is_q ? q_w : k_w;
+
+  // 128 threads, head_dim = 128: lane i owns element i.
+  const int i = threadIdx.x;
+  float x = bf2f(base[i]);
+  // RMS over the head: two-wave reduce via LDS
+  __shared__ float red[16];
+  float total = block_reduce_sum(x * x, red);
+  float scale = rsqrtf(total / head_dim + eps);
+  float xn = x * scale * bf2f(w[i]);
+
+  // exchange halves for rotate-half through LDS
+  __shared__ float sh[128];
+  sh[i] = xn;
+  __syncthreads();
+  const int half = head_dim / 2;
+  const int p = pos[t];
+  float out;
+  if (i < half) {
+    float c = cos_t[(long)p * half + i];
+    float s = sin_t[(long)p * half + i];
+    out = xn * c - sh[i + half] * s;
+  } else {
+    float c = cos_t[(long)p * half + (i - half)];
+    float s = sin_t[(long)p * half + (i - half)];
+    out = xn * c + sh[i - half] * s;
+  }
+  base[i] = f2bf(out);
+}
+
+// ---------------------------------------------------------------- silu_mul
+// h = silu(gate) * up, operating on packed [rows, 2*inter] where first half
+// is gate, second is up (MoE activation). Memory-bound, 8-wide.
+__global__ void silu_mul_kernel(short* __restrict__ out,        // [rows, inter]
+                                const short* __restrict__ gateup,  // [rows, 2*inter]
+                                int inter) {
+  const long row = blockIdx.x;
+  const short* g = gateup + row * (long)(2 * inter);
+  const short* u = g + inter;
+  short* y = out + row * (long)inter;
+  const int vecs = inter / 8;
+  for (int i = threadIdx.x; i < vecs; i += blockDim.x) {
+    bf16x8 gv = *reinterpret_cast<const bf16x8*>(g + i * 8);
+    bf16x8 uv = *reinterpret_cast<const bf16x8*>(u + i * 8);
+    bf16x8 o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(gv[j]);
+      float s = gf / (1.0f + __expf(-gf));
+      o[j] = f2bf(s * bf2f(uv[j]));
+    }
+    *reinterpret_cast<bf16x8*>(y + i * 8) = o;
+  }
+}
+
+// ---------------------------------------------------------------- host wrappers
+
+void rmsnorm(torch::Tensor out, torch::Tensor in, torch::Tensor weight, double eps) {
+  TORCH_CHECK(in.is_cuda() && in.dtype() == torch::kBFloat16);
+  TORCH_CHECK(in.is_contiguous() && out.is_contiguous());
+  long rows = in.numel() / in.size(-1);
+  int cols = in.size(-1);
+  TORCH_CHECK(cols % 8 == 0);
+  dim3 grid(rows), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(rmsnorm_kernel, grid, block, 0, s,
+                     (short*)out.data_ptr(), (const short*)in.data_ptr(),
+                     (const short*)weight.data_ptr(), cols, (float)eps);
+  HIP_CHECK_KERNEL();
+}
+
+void fused_add_rmsnorm(torch::Tensor out, torch::Tensor residual, torch::Tensor in,
+                       torch::Tensor weight, double eps) {
+  TORCH_CHECK(in.is_cuda() && in.dtype() == torch::kBFloat16);
+  long rows = in.numel() / in.size(-1);
+  int cols = in.size(-1);
+  TORCH_CHECK(cols % 8 == 0);
+  dim3 grid(rows), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(fused_add_rmsnorm_kernel, grid, block, 0, s,
+                     (short*)out.data_ptr(), (short*)residual.data_ptr(),
+                     (const short*)in.data_ptr(), (const short*)weight.data_ptr(),
+                     cols, (float)eps);
+  HIP_CHECK_KERNEL();
+}
+
+void qk_norm_rope(torch::Tensor q, torch::Tensor k, torch::Tensor q_w,
+                  torch::Tensor k_w, torch::Tensor cos_t, torch::Tensor sin_t,
+                  torch::Tensor positions, int64_t n_qheads, int64_t n_kvheads,
+                  int64_t head_dim, double eps) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
+  TORCH_CHECK(head_dim == 128, "kernel assumes head_dim=128");
+  int T = positions.size(0);
+  dim3 grid(T, n_qheads + n_kvheads), block(head_dim);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(qk_norm_rope_kernel, grid, block, 0, s,
+                     (short*)q.data_ptr(), (short*)k.data_ptr(),
+                     (const short*)q_w.data_ptr(), (const short*)k_w.data_ptr(),
+                     cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
+                     positions.data_ptr<int>(), (int)n_qheads, (int)n_kvheads,
+                     (int)head_dim, (float)eps);
+  HIP_CHECK_KERNEL();
+}
+
+void silu_mul(torch::Tensor out, torch::Tensor gateup) {
+  TORCH_CHECK(gateup.is_cuda() && gateup.dtype() == torch::kBFloat16);
+  long rows = gateup.numel() / gateup.size(-1);
+  int inter = gateup.size(-1) / 2;
+  TORCH_CHECK(inter % 8 == 0);
+  dim3 grid(rows), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(silu_mul_kernel, grid, block, 0, s,
+                     (short*)out.data_ptr(), (const short*)gateup.data_ptr(), inter);
+  HIP_CHECK_KERNEL();
+}

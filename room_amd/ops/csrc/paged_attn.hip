@@ -1,0 +1,193 @@
+// Paged attention for GQA (Qwen3-30B-A3B: 32 q heads / 4 kv heads / head_dim 128).
+//
+// One unified kernel serves prefill AND decode: every query token attends to
+// the paged KV cache of its sequence up to its own position (causal). Decode
+// is the T == batch case. This keeps the agent-swarm engine simple: sessions
+// are KV-cache residency (SURVEY §5 "session-as-KV-cache") and both phases
+// read the same cache.
+//
+// Layouts (chosen for this kernel, we own the cache):
+//   K,V cache: [num_blocks, kv_heads, BLOCK_SIZE, head_dim] bf16
+//   block_table: [num_seqs, max_blocks] int32
+//
+// Workgroup = (query token, kv head): 256 threads = 4 waves.
+//   score phase: thread (h = t>>5, j = t&31) computes dot(q[h], k[pos_j])
+//                (full 128-dim dot per thread; rows stream through L1)
+//   value phase: thread (h = t>>5, d4 = (t&31)*4) accumulates 4 output dims;
+//                V reads are 32×8B consecutive = coalesced per head-group.
+// Online softmax with per-head running max/sum (flash-style, fp32).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include "common.h"
+
+#define BLOCK_SIZE 16
+#define CHUNK 32          // kv positions per iteration
+#define QH_PER_KV 8       // GQA group width (32/4)
+#define HEAD_DIM 128
+
+__global__ __launch_bounds__(256)
+void paged_attn_kernel(short* __restrict__ out,          // [T, Hq, D]
+                       const short* __restrict__ q,      // [T, Hq, D]
+                       const short* __restrict__ kcache, // [NB, Hk, BS, D]
+                       const short* __restrict__ vcache,
+                       const int* __restrict__ block_table,  // [S, MB]
+                       const int* __restrict__ seq_ids,      // [T]
+                       const int* __restrict__ q_pos,        // [T] absolute pos
+                       int n_kvheads, int max_blocks, float scale) {
+  const int t = blockIdx.x;       // query token
+  const int hk = blockIdx.y;      // kv head
+  const int tid = threadIdx.x;
+  const int h = tid >> 5;         // 0..7: q-head within group
+  const int sub = tid & 31;       // score: position lane / value: dim quarter
+  const int seq = seq_ids[t];
+  const int bound = q_pos[t] + 1; // causal: attend to [0, bound)
+
+  const int hq = hk * QH_PER_KV + h;
+  const int n_qheads = n_kvheads * QH_PER_KV;
+
+  __shared__ float q_s[QH_PER_KV][HEAD_DIM];
+  __shared__ float p_s[QH_PER_KV][CHUNK];
+
+  // load q for the 8 heads of this group into LDS (fp32)
+  for (int i = tid; i < QH_PER_KV * HEAD_DIM; i += blockDim.x) {
+    int hh = i / HEAD_DIM, dd = i % HEAD_DIM;
+    q_s[hh][dd] = bf2f(q[((long)t * n_qheads + hk * QH_PER_KV + hh) * HEAD_DIM + dd]);
+  }
+  __syncthreads();
+
+  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  float m_run = -INFINITY, l_run = 0.f;
+
+  const long kv_stride_block = (long)n_kvheads * BLOCK_SIZE * HEAD_DIM;
+  const int* btab = block_table + (long)seq * max_blocks;
+
+  for (int base = 0; base < bound; base += CHUNK) {
+    // ---- score phase: this thread scores position base+sub for head h
+    {
+      const int pos = base + sub;
+      float s = -INFINITY;
+      if (pos < bound) {
+        const int blk = btab[pos / BLOCK_SIZE];
+        const short* krow = kcache + (long)blk * kv_stride_block
+                            + ((long)hk * BLOCK_SIZE + (pos % BLOCK_SIZE)) * HEAD_DIM;
+        float dot = 0.f;
+        #pragma unroll
+        for (int v8 = 0; v8 < HEAD_DIM / 8; ++v8) {
+          bf16x8 kv = *reinterpret_cast<const bf16x8*>(krow + v8 * 8);
+          #pragma unroll
+          for (int j = 0; j < 8; ++j) dot += q_s[h][v8 * 8 + j] * bf2f(kv[j]);
+        }
+        s = dot * scale;
+      }
+      p_s[h][sub] = s;
+    }
+    __syncthreads();
+
+    // ---- softmax + value phase: thread owns dims [sub*4, sub*4+4) of head h
+    {
+      float chunk_max = -INFINITY;
+      #pragma unroll
+      for (int j = 0; j < CHUNK; ++j) chunk_max = fmaxf(chunk_max, p_s[h][j]);
+      const float m_new = fmaxf(m_run, chunk_max);
+      if (m_new != -INFINITY) {
+        const float rescale = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+        acc[0] *= rescale; acc[1] *= rescale; acc[2] *= rescale; acc[3] *= rescale;
+        l_run *= rescale;
+        const int lim = min(CHUNK, bound - base);
+        for (int j = 0; j < lim; ++j) {
+          const float w = __expf(p_s[h][j] - m_new);
+          l_run += w;
+          const int pos = base + j;
+          const int blk = btab[pos / BLOCK_SIZE];
+          const short* vrow = vcache + (long)blk * kv_stride_block
+                              + ((long)hk * BLOCK_SIZE + (pos % BLOCK_SIZE)) * HEAD_DIM
+                              + sub * 4;
+          bf16x4 vv = *reinterpret_cast<const bf16x4*>(vrow);
+          acc[0] += w * bf2f(vv[0]);
+          acc[1] += w * bf2f(vv[1]);
+          acc[2] += w * bf2f(vv[2]);
+          acc[3] += w * bf2f(vv[3]);
+        }
+        m_run = m_new;
+      }
+    }
+    __syncthreads();
+  }
+
+  const float inv_l = (l_run > 0.f) ? 1.0f / l_run : 0.f;
+  short* orow = out + ((long)t * n_qheads + hq) * HEAD_DIM + sub * 4;
+  bf16x4 o;
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) o[j] = f2bf(acc[j] * inv_l);
+  *reinterpret_cast<bf16x4*>(orow) = o;
+
+  // NOTE: the value phase recomputes l_run identically in all 32 threads of a
+  // head (each walks the same 32 scores) — redundant VALU, zero extra HBM.
+}
+
+// ------------------------------------------------ KV cache scatter
+// Write new K/V rows for T tokens into their paged slots.
+__global__ void write_kv_kernel(short* __restrict__ kcache,
+                                short* __restrict__ vcache,
+                                const short* __restrict__ k,   // [T, Hk, D]
+                                const short* __restrict__ v,
+                                const int* __restrict__ block_table,
+                                const int* __restrict__ seq_ids,
+                                const int* __restrict__ q_pos,
+                                int n_kvheads, int max_blocks) {
+  const int t = blockIdx.x;
+  const int seq = seq_ids[t];
+  const int pos = q_pos[t];
+  const int blk = block_table[(long)seq * max_blocks + pos / BLOCK_SIZE];
+  const long dst_base = ((long)blk * n_kvheads) * BLOCK_SIZE * HEAD_DIM
+                        + (long)(pos % BLOCK_SIZE) * HEAD_DIM;
+  // threads cover Hk * D elements in 8-wide vectors
+  const int total_vec = n_kvheads * HEAD_DIM / 8;
+  for (int i = threadIdx.x; i < total_vec; i += blockDim.x) {
+    const int hh = (i * 8) / HEAD_DIM;
+    const int dd = (i * 8) % HEAD_DIM;
+    const long dst = dst_base + (long)hh * BLOCK_SIZE * HEAD_DIM + dd;
+    *reinterpret_cast<bf16x8*>(kcache + dst) =
+        *reinterpret_cast<const bf16x8*>(k + ((long)t * n_kvheads + hh) * HEAD_DIM + dd);
+    *reinterpret_cast<bf16x8*>(vcache + dst) =
+        *reinterpret_cast<const bf16x8*>(v + ((long)t * n_kvheads + hh) * HEAD_DIM + dd);
+  }
+}
+
+// ---------------------------------------------------------------- wrappers
+
+void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
+                     torch::Tensor vcache, torch::Tensor block_table,
+                     torch::Tensor seq_ids, torch::Tensor q_pos, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
+  TORCH_CHECK(q.size(-1) == HEAD_DIM, "head_dim must be 128");
+  const int T = q.size(0);
+  const int n_kvheads = kcache.size(1);
+  TORCH_CHECK(q.size(1) == n_kvheads * QH_PER_KV, "GQA group must be 8");
+  TORCH_CHECK(kcache.size(2) == BLOCK_SIZE);
+  const int max_blocks = block_table.size(1);
+  dim3 grid(T, n_kvheads), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(paged_attn_kernel, grid, block, 0, s,
+                     (short*)out.data_ptr(), (const short*)q.data_ptr(),
+                     (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
+                     block_table.data_ptr<int>(), seq_ids.data_ptr<int>(),
+                     q_pos.data_ptr<int>(), n_kvheads, max_blocks, (float)scale);
+  HIP_CHECK_KERNEL();
+}
+
+void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
+              torch::Tensor v, torch::Tensor block_table, torch::Tensor seq_ids,
+              torch::Tensor q_pos) {
+  const int T = k.size(0);
+  const int n_kvheads = kcache.size(1);
+  const int max_blocks = block_table.size(1);
+  dim3 grid(T), block(128);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(write_kv_kernel, grid, block, 0, s,
+                     (short*)kcache.data_ptr(), (short*)vcache.data_ptr(),
+                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                     block_table.data_ptr<int>(), seq_ids.data_ptr<int>(),
+                     q_pos.data_ptr<int>(), n_kvheads, max_blocks);
+  HIP_CHECK_KERNEL();
+}

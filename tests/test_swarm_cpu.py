@@ -1,0 +1,66 @@
+"""Multi-process swarm-collective tests on CPU (gloo, world_size 2) — the
+distributed path the driver scales to 8 GPUs must be correct by construction."""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from room_amd.parallel.swarm import SwarmContext
+
+
+def _find_free_port() -> int:
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _worker(rank: int, world: int, port: int, fail_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world)
+        os.environ["LOCAL_RANK"] = str(rank)
+        ctx = SwarmContext.from_env(device=torch.device("cpu"))
+        assert ctx.is_distributed
+
+        # quorum all-gather: rank 0 votes [1, 1], rank 1 votes [-1, 0]
+        votes = torch.tensor([1, 1] if rank == 0 else [-1, 0], dtype=torch.int8)
+        tally = ctx.quorum_tally(votes)
+        assert tally == {"yes": 2, "no": 1, "abstain": 1, "total": 4}, tally
+
+        # goal broadcast from queen rank
+        obj = {"goal": "ship it", "skills": ["deploy"]} if rank == 0 else None
+        got = ctx.broadcast_blob(obj, src=0)
+        assert got == {"goal": "ship it", "skills": ["deploy"]}, got
+
+        # memory top-k merge with globally-offset ids
+        local_v = torch.tensor([0.9, 0.5] if rank == 0 else [0.8, 0.7])
+        local_i = torch.tensor([0, 1] if rank == 0 else [100, 101])
+        v, i = ctx.topk_merge(local_v, local_i, k=3)
+        assert i.tolist() == [0, 100, 101], i.tolist()
+
+        ctx.barrier()
+        torch.distributed.destroy_process_group()
+    except Exception as e:  # propagate to parent
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}")
+        raise
+
+
+def test_swarm_collectives_world2():
+    port = _find_free_port()
+    fail_q = mp.get_context("spawn").SimpleQueue()
+    mp.spawn(_worker, args=(2, port, fail_q), nprocs=2, join=True)
+    assert fail_q.empty()
+
+
+def test_single_rank_fallbacks():
+    ctx = SwarmContext(0, 1, torch.device("cpu"))
+    votes = torch.tensor([1, -1, 0], dtype=torch.int8)
+    t = ctx.quorum_tally(votes)
+    assert t == {"yes": 1, "no": 1, "abstain": 1, "total": 3}
+    assert ctx.broadcast_blob({"a": 1}) == {"a": 1}
+    v, i = ctx.topk_merge(torch.tensor([0.1, 0.9]), torch.tensor([5, 6]), k=1)
+    assert i.tolist() == [6]
