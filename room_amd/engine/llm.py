@@ -104,7 +104,9 @@ class LocalEngine:
     def generate(self, prompt_tokens: list[int], max_new_tokens: int = 128,
                  temperature: float = 0.7, top_p: float = 0.95, top_k: int = 40,
                  session_key: str | None = None,
-                 timeout: float = 600.0) -> GenRequest:
+                 timeout: float | None = None) -> GenRequest:
+        if timeout is None:
+            timeout = float(os.environ.get("ROOMAMD_GEN_TIMEOUT", "600"))
         req = GenRequest(prompt_tokens=list(prompt_tokens),
                          max_new_tokens=max_new_tokens, temperature=temperature,
                          top_p=top_p, top_k=top_k, session_key=session_key)
@@ -182,18 +184,20 @@ class LocalEngine:
                 self._active.clear()
 
     def _scheduler_iteration(self) -> None:
-        # admit new requests
+        # admit new requests; an admit failure must resolve that future, never
+        # strand it (the caller blocks on req.done)
         block = not self._active
         try:
-            req = self._queue.get(timeout=0.05 if block else 0)
-            with self._lock:
-                self._admit(req)
-            self._active.append(req)
-            while True:  # drain whatever else is queued
-                req = self._queue.get_nowait()
-                with self._lock:
-                    self._admit(req)
-                self._active.append(req)
+            while True:
+                req = self._queue.get(timeout=0.05 if block else 0)
+                block = False
+                try:
+                    with self._lock:
+                        self._admit(req)
+                    self._active.append(req)
+                except Exception as e:
+                    req.error = f"admit failed: {e}"
+                    req.done.set()
         except queue.Empty:
             pass
         if not self._active:
