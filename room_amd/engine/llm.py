@@ -173,11 +173,16 @@ class LocalEngine:
             req.pending_prefill = [req.prompt_tokens[-1]]
 
     def _scheduler_loop(self) -> None:
-        torch.cuda.set_device(self.device)
+        idx = self.device.index if self.device.index is not None \
+            else torch.cuda.current_device()
+        torch.cuda.set_device(idx)
         while not self._stop:
             try:
                 self._scheduler_iteration()
             except Exception as e:  # engine errors resolve all active futures
+                import sys
+                import traceback
+                traceback.print_exc(file=sys.stderr)
                 for r in self._active:
                     r.error = f"engine error: {e}"
                     r.done.set()
