@@ -143,6 +143,11 @@ class LocalEngine:
 
     def _admit(self, req: GenRequest) -> None:
         cache = self.cache
+        # map token ids into this model's vocab (the synthetic tokenizer emits
+        # Qwen-range ids; reduced test configs have a smaller embedding table —
+        # out-of-range ids would be an out-of-bounds gather on the GPU)
+        V = self.cfg.vocab_size
+        req.prompt_tokens = [t % V for t in req.prompt_tokens]
         sess = self.sessions.get(req.session_key) if req.session_key else None
         if sess is not None:
             # reuse the longest common token prefix: the session keeps its slot

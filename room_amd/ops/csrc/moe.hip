@@ -31,9 +31,11 @@ __global__ void moe_router_kernel(int* __restrict__ topk_ids,      // [T, K]
   float e0 = __expf(l0 - m), e1 = __expf(l1 - m);
   float denom = wave_reduce_sum(e0 + e1);
 
-  // iterative top-K selection by masking
+  // iterative top-K selection by masking (K ≤ 8)
   float v0 = l0, v1 = l1;
   float picked_sum = 0.f;
+  float probs[8];
+  int winners[8];
   for (int k = 0; k < K; ++k) {
     float best = fmaxf(v0, v1);
     float gmax = wave_reduce_max(best);
@@ -44,18 +46,19 @@ __global__ void moe_router_kernel(int* __restrict__ topk_ids,      // [T, K]
     for (int off = 32; off > 0; off >>= 1)
       winner = min(winner, __shfl_xor(winner, off, WAVE));
     float prob = __expf(gmax - m) / denom;
-    if (lane == 0) {
-      topk_ids[(long)t * K + k] = winner;
-      topk_w[(long)t * K + k] = prob;
-    }
-    picked_sum += prob;  // same value in all lanes
+    probs[k] = prob;       // every lane tracks the same values
+    winners[k] = winner;
+    picked_sum += prob;
     if (winner == 2 * lane) v0 = -INFINITY;
     if (winner == 2 * lane + 1) v1 = -INFINITY;
   }
-  // renormalize among top-K (Qwen3 norm_topk_prob)
-  if (lane < K) {
-    float w = topk_w[(long)t * K + lane];
-    topk_w[(long)t * K + lane] = w / picked_sum;
+  // lane 0 writes ids + renormalized weights (Qwen3 norm_topk_prob) — single
+  // writer, no cross-lane visibility assumptions on global memory
+  if (lane == 0) {
+    for (int k = 0; k < K; ++k) {
+      topk_ids[(long)t * K + k] = winners[k];
+      topk_w[(long)t * K + k] = probs[k] / picked_sum;
+    }
   }
 }
 
