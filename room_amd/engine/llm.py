@@ -148,6 +148,15 @@ class LocalEngine:
         # out-of-range ids would be an out-of-bounds gather on the GPU)
         V = self.cfg.vocab_size
         req.prompt_tokens = [t % V for t in req.prompt_tokens]
+        # context-window guard: middle-truncate prompts that cannot fit (keep
+        # the head [system prompt] and the recent tail), mirroring the
+        # reference's context-overflow recovery (agent-loop.ts:773-782)
+        budget = self.cfg.max_position - req.max_new_tokens - 8
+        if len(req.prompt_tokens) > budget:
+            head = budget // 4
+            tail = budget - head
+            req.prompt_tokens = (req.prompt_tokens[:head]
+                                 + req.prompt_tokens[-tail:])
         sess = self.sessions.get(req.session_key) if req.session_key else None
         if sess is not None:
             # reuse the longest common token prefix: the session keeps its slot
