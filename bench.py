@@ -146,6 +146,16 @@ def main() -> None:
     value = total_cycles / elapsed
     p50 = statistics.median(queen_latencies[args.warmup:] or queen_latencies)
 
+    if engine is not None:
+        s = engine.stats
+        print(f"[rank {ctx.rank}] engine stats: "
+              f"prefill {s['prefill_tokens']} tok in {s['prefill_time']:.2f}s "
+              f"({s['prefill_tokens']/max(s['prefill_time'],1e-9):.0f} tok/s), "
+              f"decode {s['decode_tokens']} tok in {s['decode_steps']} steps "
+              f"{s['decode_time']:.2f}s "
+              f"({s['decode_tokens']/max(s['decode_time'],1e-9):.0f} tok/s)",
+              file=sys.stderr)
+
     if ctx.rank == 0:
         out = {
             "metric": "agent-cycles/sec",

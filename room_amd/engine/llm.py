@@ -254,8 +254,7 @@ class LocalEngine:
         tokens_t = torch.tensor(tokens, dtype=torch.int64, device=dev)
         seq_t = torch.tensor(seq_ids, dtype=torch.int32, device=dev)
         pos_t = torch.tensor(q_pos, dtype=torch.int32, device=dev)
-        rows_t = (torch.tensor(last_rows, dtype=torch.int64, device=dev)
-                  if sampled_reqs else None)
+        rows_t = torch.tensor(last_rows, dtype=torch.int64, device=dev)
         logits = self.model.forward(tokens_t, seq_t, pos_t, self.cache.block_table,
                                     self.cache.kcaches, self.cache.vcaches,
                                     logits_rows=rows_t)

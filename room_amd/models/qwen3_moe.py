@@ -148,7 +148,11 @@ class Qwen3MoEModel:
             moe_out_f32 = self._moe(hbuf, layer, topk_ids, topk_w)
             moe_out = moe_out_f32.to(torch.bfloat16)
 
-        # final residual add + norm
+        # final residual add + norm; skip lm_head entirely for logits-free
+        # prefill chunks (an empty logits_rows means "no sampling this chunk")
+        if logits_rows is not None and logits_rows.numel() == 0:
+            return torch.empty(0, cfg.vocab_size, dtype=torch.float32,
+                               device=x.device)
         ops.fused_add_rmsnorm(hbuf, x, moe_out, self.final_norm_w, cfg.rms_eps)
         sel = hbuf if logits_rows is None else hbuf[logits_rows.long()]
         logits = F.linear(sel, self.lm_head).float()
@@ -182,12 +186,12 @@ class Qwen3MoEModel:
                       .repeat_interleave(K))[order.long()].contiguous()
         pair_w = topk_w.flatten()[order.long()].contiguous()
         P = pair_token.numel()
-        desc1 = ops.build_moe_tile_desc(pair_expert, (2 * I) // 64).to(hbuf.device)
+        desc1 = ops.build_moe_tile_desc(pair_expert, (2 * I) // 64, cfg.num_experts)
         gateup = torch.empty(P, 2 * I, dtype=torch.bfloat16, device=hbuf.device)
         ops.moe_grouped_gemm(gateup, hbuf, layer.w13, pair_token, desc1)
         h = torch.empty(P, I, dtype=torch.bfloat16, device=hbuf.device)
         ops.silu_mul(h, gateup)
-        desc2 = ops.build_moe_tile_desc(pair_expert, H // 64).to(hbuf.device)
+        desc2 = ops.build_moe_tile_desc(pair_expert, H // 64, cfg.num_experts)
         z = torch.empty(P, H, dtype=torch.bfloat16, device=hbuf.device)
         ops.moe_grouped_gemm(z, h, layer.w2,
                              torch.arange(P, device=hbuf.device, dtype=torch.int32),

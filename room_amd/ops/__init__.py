@@ -123,24 +123,24 @@ def vs_topk(mat: torch.Tensor, query: torch.Tensor, k: int,
 
 
 def build_moe_tile_desc(pair_expert_sorted: torch.Tensor, n_tiles_n: int,
-                        bm: int = 16) -> torch.Tensor:
-    """Host-side tile descriptors for the grouped MFMA GEMM: one entry
-    (expert, row_start, m_size, n_tile) per (M-tile × N-tile). pair_expert_sorted
-    must be ascending (tokens grouped by expert)."""
-    pe = pair_expert_sorted.cpu()
-    desc = []
-    P = pe.numel()
-    i = 0
-    while i < P:
-        e = int(pe[i])
-        j = i
-        while j < P and int(pe[j]) == e:
-            j += 1
-        for m0 in range(i, j, bm):
-            msize = min(bm, j - m0)
-            for nt in range(n_tiles_n):
-                desc.append((e, m0, msize, nt))
-        i = j
-    if not desc:
-        return torch.zeros(0, 4, dtype=torch.int32)
-    return torch.tensor(desc, dtype=torch.int32)
+                        num_experts: int, bm: int = 16) -> torch.Tensor:
+    """Tile descriptors for the grouped MFMA GEMM: one row
+    (expert, row_start, m_size, n_tile) per (M-tile × N-tile). Fully
+    vectorized on the device — no host sync (runs once per layer per step)."""
+    dev = pair_expert_sorted.device
+    counts = torch.bincount(pair_expert_sorted.long(), minlength=num_experts)
+    mtiles = (counts + bm - 1) // bm                      # [E]
+    e_ids = torch.repeat_interleave(
+        torch.arange(num_experts, device=dev), mtiles)    # [G_m]
+    starts = torch.cumsum(counts, 0) - counts             # row offset per expert
+    mt_off = torch.cumsum(mtiles, 0) - mtiles             # first tile idx per expert
+    g = e_ids.numel()
+    tile_in_e = torch.arange(g, device=dev) - mt_off[e_ids]
+    row_start = starts[e_ids] + bm * tile_in_e
+    m_size = torch.minimum(counts[e_ids] - bm * tile_in_e,
+                           torch.full_like(row_start, bm))
+    desc_m = torch.stack([e_ids, row_start, m_size], dim=1)  # [G_m, 3]
+    # cross-product with the N tiles
+    desc = desc_m.repeat_interleave(n_tiles_n, dim=0)
+    nt = torch.arange(n_tiles_n, device=dev).repeat(g).unsqueeze(1)
+    return torch.cat([desc, nt], dim=1).int().contiguous()

@@ -183,7 +183,7 @@ def test_moe_grouped_gemm_vs_matmul():
     order = torch.argsort(pair_expert)
     pair_expert = pair_expert[order].contiguous()
     pair_token = order.int().contiguous()
-    tile_desc = ops.build_moe_tile_desc(pair_expert, N // 64).to(DEV)
+    tile_desc = ops.build_moe_tile_desc(pair_expert, N // 64, E)
     out = torch.empty(T, N, dtype=torch.bfloat16, device=DEV)
     ops.moe_grouped_gemm(out, x, w, pair_token, tile_desc)
     for p in range(T):
@@ -207,12 +207,12 @@ def test_moe_grouped_path_full():
     P = pair_token.numel()
 
     gateup = torch.empty(P, 2 * I, dtype=torch.bfloat16, device=DEV)
-    desc1 = ops.build_moe_tile_desc(pair_expert, (2 * I) // 64).to(DEV)
+    desc1 = ops.build_moe_tile_desc(pair_expert, (2 * I) // 64, 16)
     ops.moe_grouped_gemm(gateup, x, w13, pair_token, desc1)
     h = torch.empty(P, I, dtype=torch.bfloat16, device=DEV)
     ops.silu_mul(h, gateup)
     z = torch.empty(P, H, dtype=torch.bfloat16, device=DEV)
-    desc2 = ops.build_moe_tile_desc(pair_expert, H // 64).to(DEV)
+    desc2 = ops.build_moe_tile_desc(pair_expert, H // 64, 16)
     # down gemm reads h rows per sorted pair: identity token mapping
     ops.moe_grouped_gemm(z, h, w2, torch.arange(P, device=DEV).int(), desc2)
     out = torch.zeros(T, H, dtype=torch.float32, device=DEV)
