@@ -29,6 +29,7 @@ from .types import AgentExecutionOptions, ToolDef
 
 PREFILL_CHUNK = 4096          # max tokens per prefill forward
 DEFAULT_MAX_SEQS = 64
+DECODE_BUCKETS = (1, 2, 4, 8, 16, 32, 64)  # hipGraph capture sizes
 
 
 @dataclass
@@ -56,6 +57,64 @@ class _Session:
     def __init__(self, slot: int, tokens: list[int]):
         self.slot = slot
         self.tokens = tokens
+
+
+class _DecodeGraph:
+    """One captured hipGraph per (batch bucket, sampling params).
+
+    The decode step is ~1500 tiny kernel launches across 48 layers; capturing
+    it collapses launch overhead to a single hipGraphLaunch. Inputs are
+    persistent device buffers; the sampler RNG state lives on-device so
+    replays draw fresh randomness with no host work. Padded lanes point at a
+    reserved pad sequence slot (their KV writes land in a scrap block)."""
+
+    def __init__(self, engine: "LocalEngine", bucket: int,
+                 temperature: float, top_p: float, top_k: int):
+        self.bucket = bucket
+        self.pad_slot = engine.pad_slot
+        dev = engine.device
+        self.tok_in = torch.zeros(bucket, dtype=torch.int64, device=dev)
+        self.seq_in = torch.full((bucket,), engine.pad_slot, dtype=torch.int32,
+                                 device=dev)
+        self.pos_in = torch.zeros(bucket, dtype=torch.int32, device=dev)
+        self.seeds = torch.randint(1, 2**62, (bucket,), dtype=torch.int64,
+                                   device=dev)
+        model, cache = engine.model, engine.cache
+        from .. import ops as _ops
+
+        def run():
+            logits = model.forward(self.tok_in, self.seq_in, self.pos_in,
+                                   cache.block_table, cache.kcaches,
+                                   cache.vcaches)
+            toks = torch.empty(bucket, dtype=torch.int32, device=dev)
+            _ops._require().sample_tokens(toks, logits, self.seeds, top_k,
+                                          temperature, top_p)
+            return toks
+
+        # warmup on a side stream (required before capture)
+        s = torch.cuda.Stream(dev)
+        s.wait_stream(torch.cuda.current_stream(dev))
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                run()
+        torch.cuda.current_stream(dev).wait_stream(s)
+        torch.cuda.synchronize(dev)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.out_tokens = run()
+
+    def replay(self, tokens: list[int], slots: list[int],
+               positions: list[int]) -> torch.Tensor:
+        n, b = len(tokens), self.bucket
+        pad = b - n
+        self.tok_in.copy_(torch.tensor(tokens + [0] * pad, dtype=torch.int64),
+                          non_blocking=True)
+        self.seq_in.copy_(torch.tensor(slots + [self.pad_slot] * pad,
+                                       dtype=torch.int32), non_blocking=True)
+        self.pos_in.copy_(torch.tensor(positions + [0] * pad, dtype=torch.int32),
+                          non_blocking=True)
+        self.graph.replay()
+        return self.out_tokens[:n]
 
 
 class LocalEngine:
@@ -88,13 +147,21 @@ class LocalEngine:
         self.cache = PagedKVCache(cfg.num_layers, cfg.num_kv_heads, cfg.head_dim,
                                   int(num_blocks), max_seqs, max_blocks_per_seq,
                                   self.device)
+        # pad slot for hipGraph batch padding (one scrap block absorbs the
+        # dummy lanes' KV writes)
+        self.pad_slot = self.cache.alloc_seq()
+        self.cache.ensure_capacity(self.pad_slot, 1)
+        self.graphs_enabled = (os.environ.get("ROOMAMD_NO_GRAPHS") != "1")
+        self._graphs: dict[tuple, _DecodeGraph] = {}
+        self._graphs_broken = False
         self.sessions: dict[str, _Session] = {}
         self._queue: "queue.Queue[GenRequest]" = queue.Queue()
         self._active: list[GenRequest] = []
         self._lock = threading.Lock()
         self._stop = False
         self.stats = {"decode_steps": 0, "decode_tokens": 0, "prefill_tokens": 0,
-                      "decode_time": 0.0, "prefill_time": 0.0}
+                      "decode_time": 0.0, "prefill_time": 0.0,
+                      "graphs_captured": 0}
         self._thread = threading.Thread(target=self._scheduler_loop, daemon=True,
                                         name="room-amd-engine")
         self._thread.start()
@@ -270,15 +337,48 @@ class LocalEngine:
         dev = self.device
         for r in reqs:
             self.cache.ensure_capacity(r.slot, r.pos + 1)
-        tokens_t = torch.tensor([r.last_token for r in reqs], dtype=torch.int64,
-                                device=dev)
-        seq_t = torch.tensor([r.slot for r in reqs], dtype=torch.int32, device=dev)
-        pos_t = torch.tensor([r.pos for r in reqs], dtype=torch.int32, device=dev)
+        tokens = [r.last_token for r in reqs]
+        slots = [r.slot for r in reqs]
+        positions = [r.pos for r in reqs]
         for r in reqs:
             r.pos += 1
-        logits = self.model.forward(tokens_t, seq_t, pos_t, self.cache.block_table,
-                                    self.cache.kcaches, self.cache.vcaches)
-        self._sample_and_append(reqs, logits)
+
+        toks_t = None
+        if self.graphs_enabled and not self._graphs_broken and len(reqs) <= 64:
+            r0 = reqs[0]
+            bucket = next(b for b in DECODE_BUCKETS if b >= len(reqs))
+            key = (bucket, round(r0.temperature, 3), round(r0.top_p, 3), r0.top_k)
+            uniform = all((r.temperature, r.top_p, r.top_k)
+                          == (r0.temperature, r0.top_p, r0.top_k) for r in reqs)
+            if uniform:
+                try:
+                    g = self._graphs.get(key)
+                    if g is None:
+                        g = _DecodeGraph(self, bucket, r0.temperature, r0.top_p,
+                                         r0.top_k)
+                        self._graphs[key] = g
+                        self.stats["graphs_captured"] = len(self._graphs)
+                    toks_t = g.replay(tokens, slots, positions)
+                except Exception as e:
+                    import sys
+                    print(f"[room_amd] hipGraph decode disabled: {e}",
+                          file=sys.stderr)
+                    self._graphs_broken = True
+                    toks_t = None
+
+        if toks_t is None:  # eager fallback
+            tokens_t = torch.tensor(tokens, dtype=torch.int64, device=dev)
+            seq_t = torch.tensor(slots, dtype=torch.int32, device=dev)
+            pos_t = torch.tensor(positions, dtype=torch.int32, device=dev)
+            logits = self.model.forward(tokens_t, seq_t, pos_t,
+                                        self.cache.block_table,
+                                        self.cache.kcaches, self.cache.vcaches)
+            seeds = torch.randint(1, 2**62, (logits.size(0),), dtype=torch.int64,
+                                  device=dev)
+            r0 = reqs[0]
+            toks_t = ops.sample_tokens(logits, seeds, top_k=r0.top_k,
+                                       temperature=r0.temperature, top_p=r0.top_p)
+        self._finish_tokens(reqs, toks_t)
         self.stats["decode_steps"] += 1
         self.stats["decode_tokens"] += len(reqs)
         self.stats["decode_time"] += time.time() - t0
@@ -292,6 +392,9 @@ class LocalEngine:
                               device=dev)
         toks = ops.sample_tokens(logits, seeds, top_k=top_k,
                                  temperature=temperature, top_p=top_p)
+        self._finish_tokens(reqs, toks)
+
+    def _finish_tokens(self, reqs: list[GenRequest], toks: torch.Tensor) -> None:
         toks_host = toks.tolist()  # one small sync per step
         finished = []
         for r, t in zip(reqs, toks_host):

@@ -27,7 +27,7 @@ __device__ __forceinline__ float xorshift_unit(uint64_t* state) {
 __global__ __launch_bounds__(SMP_THREADS, 1)
 void sample_kernel(int* __restrict__ out_tokens,      // [B]
                    const float* __restrict__ logits,   // [B, V]
-                   const uint64_t* __restrict__ seeds,  // [B]
+                   uint64_t* __restrict__ seeds,        // [B] in/out RNG state
                    int V, int K, float temperature, float top_p) {
   const int b = blockIdx.x;
   const float* row = logits + (long)b * V;
@@ -104,8 +104,11 @@ void sample_kernel(int* __restrict__ out_tokens,      // [B]
     }
     float denom2 = 0.f;
     for (int i = 0; i < cut; ++i) denom2 += probs[i];
+    // stateful on-device RNG: the state advances in place, so graph replays
+    // (hipGraph decode capture) draw fresh randomness with zero host work
     uint64_t st = seeds[b] | 1ull;
     float r = xorshift_unit(&st) * denom2;
+    seeds[b] = st;
     float acc = 0.f;
     int pick = ci[cut - 1];
     for (int i = 0; i < cut; ++i) {
@@ -126,7 +129,7 @@ void sample_tokens(torch::Tensor out_tokens, torch::Tensor logits,
   hipStream_t s = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(sample_kernel, grid, block, 0, s,
                      out_tokens.data_ptr<int>(), logits.data_ptr<float>(),
-                     (const uint64_t*)seeds.data_ptr(), V, (int)top_k,
+                     (uint64_t*)seeds.data_ptr(), V, (int)top_k,
                      (float)temperature, (float)top_p);
   HIP_CHECK_KERNEL();
 }
