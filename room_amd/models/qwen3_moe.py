@@ -113,9 +113,25 @@ class Qwen3MoEModel:
         Returns fp32 logits [R, vocab]."""
         cfg = self.cfg
         T = tokens.numel()
+        dev = self.device
+        # decode-mode (T ≤ 8): dense projections use the hand-written GEMV
+        # (hipBLASLt ~1.1 TB/s on M≤8 skinny shapes) and attention uses the
+        # split-KV flash-decode kernels so the tiny grid still fills the chip.
+        decode = T <= 8
+        qdim = cfg.num_q_heads * cfg.head_dim
+        kvdim = cfg.num_kv_heads * cfg.head_dim
         x = self.embed[tokens.long()]              # [T, H] bf16 (residual stream)
         hbuf = torch.empty_like(x)                 # normed activations
         moe_out = None                             # pending delta for fused add
+        if decode:
+            qkv = torch.empty(T, qdim + 2 * kvdim, dtype=torch.bfloat16, device=dev)
+            obuf = torch.empty(T, cfg.hidden_size, dtype=torch.bfloat16, device=dev)
+            router_logits = torch.empty(T, cfg.num_experts, dtype=torch.float32,
+                                        device=dev)
+            part = torch.empty(T, cfg.num_q_heads, 16, cfg.head_dim,
+                               dtype=torch.float32, device=dev)
+            part_ml = torch.empty(T, cfg.num_q_heads, 16, 2, dtype=torch.float32,
+                                  device=dev)
 
         for li, layer in enumerate(self.layers):
             # --- attention block
@@ -124,9 +140,10 @@ class Qwen3MoEModel:
             else:
                 ops.fused_add_rmsnorm(hbuf, x, moe_out, layer.input_norm_w,
                                       cfg.rms_eps)
-            qkv = F.linear(hbuf, layer.wqkv)       # hipBLASLt GEMM
-            qdim = cfg.num_q_heads * cfg.head_dim
-            kvdim = cfg.num_kv_heads * cfg.head_dim
+            if decode:
+                ops.gemv(qkv, hbuf, layer.wqkv)
+            else:
+                qkv = F.linear(hbuf, layer.wqkv)   # hipBLASLt GEMM
             q = qkv[:, :qdim].reshape(T, cfg.num_q_heads, cfg.head_dim).contiguous()
             k = qkv[:, qdim:qdim + kvdim].reshape(T, cfg.num_kv_heads,
                                                   cfg.head_dim).contiguous()
@@ -137,13 +154,23 @@ class Qwen3MoEModel:
                              cfg.head_dim, cfg.rms_eps)
             ops.write_kv(kcaches[li], vcaches[li], k, v, block_table, seq_ids, q_pos)
             attn = torch.empty_like(q)
-            ops.paged_attention(attn, q, kcaches[li], vcaches[li], block_table,
-                                seq_ids, q_pos, self.scale)
-            o = F.linear(attn.reshape(T, qdim), layer.wo)
+            if decode:
+                ops.paged_attention_split(attn, q, kcaches[li], vcaches[li],
+                                          block_table, seq_ids, q_pos, part,
+                                          part_ml, self.scale)
+                ops.gemv(obuf, attn.reshape(T, qdim), layer.wo)
+                o = obuf
+            else:
+                ops.paged_attention(attn, q, kcaches[li], vcaches[li], block_table,
+                                    seq_ids, q_pos, self.scale)
+                o = F.linear(attn.reshape(T, qdim), layer.wo)
 
             # --- MoE block
             ops.fused_add_rmsnorm(hbuf, x, o, layer.post_attn_norm_w, cfg.rms_eps)
-            router_logits = F.linear(hbuf.float(), layer.router_w.float())
+            if decode:
+                ops.gemv(router_logits, hbuf, layer.router_w)
+            else:
+                router_logits = F.linear(hbuf, layer.router_w).float()
             topk_ids, topk_w = ops.moe_router(router_logits, cfg.num_experts_per_tok)
             moe_out_f32 = self._moe(hbuf, layer, topk_ids, topk_w)
             moe_out = moe_out_f32.to(torch.bfloat16)
@@ -155,7 +182,12 @@ class Qwen3MoEModel:
                                device=x.device)
         ops.fused_add_rmsnorm(hbuf, x, moe_out, self.final_norm_w, cfg.rms_eps)
         sel = hbuf if logits_rows is None else hbuf[logits_rows.long()]
-        logits = F.linear(sel, self.lm_head).float()
+        if sel.size(0) <= 8:
+            logits = torch.empty(sel.size(0), cfg.vocab_size, dtype=torch.float32,
+                                 device=dev)
+            ops.gemv(logits, sel.contiguous(), self.lm_head)
+        else:
+            logits = F.linear(sel, self.lm_head).float()
         return logits
 
     def _moe(self, hbuf: torch.Tensor, layer: Qwen3MoELayer,

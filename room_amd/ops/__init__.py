@@ -69,6 +69,23 @@ def write_kv(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
     _require().write_kv(kcache, vcache, k, v, block_table, seq_ids, q_pos)
 
 
+def paged_attention_split(out: torch.Tensor, q: torch.Tensor, kcache: torch.Tensor,
+                          vcache: torch.Tensor, block_table: torch.Tensor,
+                          seq_ids: torch.Tensor, q_pos: torch.Tensor,
+                          part: torch.Tensor, part_ml: torch.Tensor,
+                          scale: float) -> torch.Tensor:
+    """Split-KV flash-decode: part [T, Hq, 16, 128] f32, part_ml [T, Hq, 16, 2]."""
+    _require().paged_attention_split(out, q, kcache, vcache, block_table,
+                                     seq_ids, q_pos, part, part_ml, scale)
+    return out
+
+
+def gemv(y: torch.Tensor, x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """y[B,N] = x[B,H] @ w[N,H]^T for B<=8 (decode projections)."""
+    _require().gemv(y, x, w)
+    return y
+
+
 def moe_router(logits: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:
     T = logits.size(0)
     ids = torch.empty(T, k, dtype=torch.int32, device=logits.device)

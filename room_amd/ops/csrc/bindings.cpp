@@ -29,6 +29,12 @@ void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
 void sample_tokens(torch::Tensor out_tokens, torch::Tensor logits,
                    torch::Tensor seeds, int64_t top_k, double temperature,
                    double top_p);
+void paged_attention_split(torch::Tensor out, torch::Tensor q,
+                           torch::Tensor kcache, torch::Tensor vcache,
+                           torch::Tensor block_table, torch::Tensor seq_ids,
+                           torch::Tensor q_pos, torch::Tensor part,
+                           torch::Tensor part_ml, double scale);
+void gemv(torch::Tensor y, torch::Tensor x, torch::Tensor w);
 void vs_topk(torch::Tensor out_v, torch::Tensor out_i, torch::Tensor cand_v,
              torch::Tensor cand_i, torch::Tensor mat, torch::Tensor query,
              int64_t K);
@@ -46,5 +52,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_grouped_gemm", &moe_grouped_gemm, "grouped MFMA GEMM (prefill)");
   m.def("moe_combine", &moe_combine, "weighted scatter-add combine");
   m.def("sample_tokens", &sample_tokens, "fused temperature/top-k/top-p sampling");
+  m.def("paged_attention_split", &paged_attention_split,
+        "split-KV flash-decode paged attention");
+  m.def("gemv", &gemv, "dense skinny-batch GEMV (decode projections)");
   m.def("vs_topk", &vs_topk, "vector-store cosine top-k over bf16 matrix");
 }

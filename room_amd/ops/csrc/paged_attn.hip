@@ -191,3 +191,170 @@ void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
                      q_pos.data_ptr<int>(), n_kvheads, max_blocks);
   HIP_CHECK_KERNEL();
 }
+
+// ------------------------------------------------ split-KV flash-decode
+// Decode grids are tiny (T = batch ≈ swarm width, 4 kv heads → ~20 workgroups
+// for 256 CUs). Split the KV range across NSPLITS workgroups per (token, kv
+// head); each writes an unnormalized partial (acc, m, l); a merge kernel
+// combines. NSPLITS is static so the decode step stays hipGraph-capturable.
+
+#define NSPLITS 16
+
+__global__ __launch_bounds__(256)
+void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
+                             float* __restrict__ part_ml, // [T, Hq, NSPLITS, 2]
+                             const short* __restrict__ q,
+                             const short* __restrict__ kcache,
+                             const short* __restrict__ vcache,
+                             const int* __restrict__ block_table,
+                             const int* __restrict__ seq_ids,
+                             const int* __restrict__ q_pos,
+                             int n_kvheads, int max_blocks, float scale) {
+  const int t = blockIdx.x;
+  const int hk = blockIdx.y;
+  const int split = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int h = tid >> 5;
+  const int sub = tid & 31;
+  const int seq = seq_ids[t];
+  const int bound = q_pos[t] + 1;
+  const int n_qheads = n_kvheads * QH_PER_KV;
+  const int hq = hk * QH_PER_KV + h;
+
+  // split range, CHUNK-aligned
+  const int span = ((bound + NSPLITS - 1) / NSPLITS + CHUNK - 1) & ~(CHUNK - 1);
+  const int lo = split * span;
+  const int hi = min(bound, lo + span);
+
+  float* ml = part_ml + (((long)t * n_qheads + hq) * NSPLITS + split) * 2;
+  float* acc_out = part + (((long)t * n_qheads + hq) * NSPLITS + split) * HEAD_DIM;
+
+  if (lo >= hi) {  // empty split still writes a neutral partial
+    if (h < QH_PER_KV) {
+      if (sub == 0) { ml[0] = -INFINITY; ml[1] = 0.f; }
+      *reinterpret_cast<f32x4*>(acc_out + sub * 4) = f32x4{0.f, 0.f, 0.f, 0.f};
+    }
+    return;
+  }
+
+  __shared__ float q_s[QH_PER_KV][HEAD_DIM];
+  __shared__ float p_s[QH_PER_KV][CHUNK];
+  for (int i = tid; i < QH_PER_KV * HEAD_DIM; i += blockDim.x) {
+    int hh = i / HEAD_DIM, dd = i % HEAD_DIM;
+    q_s[hh][dd] = bf2f(q[((long)t * n_qheads + hk * QH_PER_KV + hh) * HEAD_DIM + dd]);
+  }
+  __syncthreads();
+
+  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  float m_run = -INFINITY, l_run = 0.f;
+  const long kv_stride_block = (long)n_kvheads * BLOCK_SIZE * HEAD_DIM;
+  const int* btab = block_table + (long)seq * max_blocks;
+
+  for (int base = lo; base < hi; base += CHUNK) {
+    {
+      const int pos = base + sub;
+      float s = -INFINITY;
+      if (pos < hi) {
+        const int blk = btab[pos / BLOCK_SIZE];
+        const short* krow = kcache + (long)blk * kv_stride_block
+                            + ((long)hk * BLOCK_SIZE + (pos % BLOCK_SIZE)) * HEAD_DIM;
+        float dot = 0.f;
+        #pragma unroll
+        for (int v8 = 0; v8 < HEAD_DIM / 8; ++v8) {
+          bf16x8 kv = *reinterpret_cast<const bf16x8*>(krow + v8 * 8);
+          #pragma unroll
+          for (int j = 0; j < 8; ++j) dot += q_s[h][v8 * 8 + j] * bf2f(kv[j]);
+        }
+        s = dot * scale;
+      }
+      p_s[h][sub] = s;
+    }
+    __syncthreads();
+    {
+      float chunk_max = -INFINITY;
+      #pragma unroll
+      for (int j = 0; j < CHUNK; ++j) chunk_max = fmaxf(chunk_max, p_s[h][j]);
+      const float m_new = fmaxf(m_run, chunk_max);
+      if (m_new != -INFINITY) {
+        const float rescale = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+        acc[0] *= rescale; acc[1] *= rescale; acc[2] *= rescale; acc[3] *= rescale;
+        l_run *= rescale;
+        const int lim = min(CHUNK, hi - base);
+        for (int j = 0; j < lim; ++j) {
+          const float w = __expf(p_s[h][j] - m_new);
+          l_run += w;
+          const int pos = base + j;
+          const int blk = btab[pos / BLOCK_SIZE];
+          const short* vrow = vcache + (long)blk * kv_stride_block
+                              + ((long)hk * BLOCK_SIZE + (pos % BLOCK_SIZE)) * HEAD_DIM
+                              + sub * 4;
+          bf16x4 vv = *reinterpret_cast<const bf16x4*>(vrow);
+          acc[0] += w * bf2f(vv[0]);
+          acc[1] += w * bf2f(vv[1]);
+          acc[2] += w * bf2f(vv[2]);
+          acc[3] += w * bf2f(vv[3]);
+        }
+        m_run = m_new;
+      }
+    }
+    __syncthreads();
+  }
+
+  if (sub == 0) { ml[0] = m_run; ml[1] = l_run; }
+  *reinterpret_cast<f32x4*>(acc_out + sub * 4) =
+      f32x4{acc[0], acc[1], acc[2], acc[3]};
+}
+
+// merge: one wave per (t, hq); lane d-pairs combine the NSPLITS partials
+__global__ __launch_bounds__(128)
+void paged_attn_merge_kernel(short* __restrict__ out,       // [T, Hq, D]
+                             const float* __restrict__ part,
+                             const float* __restrict__ part_ml,
+                             int n_qheads) {
+  const int t = blockIdx.x;
+  const int hq = blockIdx.y;
+  const int d = threadIdx.x;  // 128 threads = one dim each
+  const float* ml = part_ml + (((long)t * n_qheads + hq) * NSPLITS) * 2;
+  const float* pacc = part + (((long)t * n_qheads + hq) * NSPLITS) * HEAD_DIM;
+
+  float m_star = -INFINITY;
+  #pragma unroll
+  for (int s = 0; s < NSPLITS; ++s) m_star = fmaxf(m_star, ml[2 * s]);
+  float l_tot = 0.f, a = 0.f;
+  #pragma unroll
+  for (int s = 0; s < NSPLITS; ++s) {
+    const float ms = ml[2 * s];
+    if (ms == -INFINITY) continue;
+    const float f = __expf(ms - m_star);
+    l_tot += ml[2 * s + 1] * f;
+    a += pacc[(long)s * HEAD_DIM + d] * f;
+  }
+  out[((long)t * n_qheads + hq) * HEAD_DIM + d] =
+      f2bf(l_tot > 0.f ? a / l_tot : 0.f);
+}
+
+void paged_attention_split(torch::Tensor out, torch::Tensor q,
+                           torch::Tensor kcache, torch::Tensor vcache,
+                           torch::Tensor block_table, torch::Tensor seq_ids,
+                           torch::Tensor q_pos, torch::Tensor part,
+                           torch::Tensor part_ml, double scale) {
+  const int T = q.size(0);
+  const int n_kvheads = kcache.size(1);
+  const int n_qheads = n_kvheads * QH_PER_KV;
+  const int max_blocks = block_table.size(1);
+  TORCH_CHECK(part.size(0) >= T && part.size(2) == NSPLITS);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  dim3 g1(T, n_kvheads, NSPLITS);
+  hipLaunchKernelGGL(paged_attn_split_kernel, g1, dim3(256), 0, s,
+                     part.data_ptr<float>(), part_ml.data_ptr<float>(),
+                     (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
+                     (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
+                     seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
+                     n_kvheads, max_blocks, (float)scale);
+  HIP_CHECK_KERNEL();
+  dim3 g2(T, n_qheads);
+  hipLaunchKernelGGL(paged_attn_merge_kernel, g2, dim3(HEAD_DIM), 0, s,
+                     (short*)out.data_ptr(), part.data_ptr<float>(),
+                     part_ml.data_ptr<float>(), n_qheads);
+  HIP_CHECK_KERNEL();
+}

@@ -1,15 +1,18 @@
 // Fused sampling: temperature + top-k + top-p + multinomial draw, one kernel.
 //
-// The reference's local-model path delegated sampling to Ollama; here it is a
-// CDNA4 kernel so a decode step never leaves the GPU (no logits→host copy).
-// Vocab ≈ 152k. Scheme:
-//   1) each of 256 threads keeps a sorted local top-K over its strided slice
-//      (insertion guarded by the current min → ~O(1) amortized per element);
-//   2) tournament merge in LDS: 8 rounds of pairwise sorted-list merges
-//      (keep top-K), leaving the global top-K sorted at list 0;
-//   3) thread 0 applies temperature softmax, top-p cut, and draws.
-// LDS: 256 lists × K(≤64) × 8 B = ≤128 KiB (fits the 160 KiB/CU budget; this
-// kernel runs at low occupancy by design — it is launch-latency bound).
+// V ≈ 152k, B small (decode batch). Everything stays on-GPU (hipGraph-
+// capturable; RNG state advances on-device so replays draw fresh randomness).
+//
+// Design (v2 — v1 kept per-thread sorted lists in scratch memory, which is
+// dynamic-indexed local array = scratch traffic, ~5 ms/step; guide common-
+// mistake #20):
+//   1) each of 256 threads scans its strided slice keeping an UNSORTED
+//      top-K set in its LDS row via replace-min (register min guard →
+//      insertions are rare, each costs one K-element LDS rescan);
+//   2) block tournament: K iterations of block-wide argmax; only the winning
+//      thread rescans its row. Produces the global top-K sorted descending;
+//   3) thread 0: temperature softmax, top-p nucleus cut, multinomial draw.
+// LDS: 256 rows × K × 8 B ≤ 128 KiB (K ≤ 64) — deliberate 1-block/CU.
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include "common.h"
@@ -32,68 +35,84 @@ void sample_kernel(int* __restrict__ out_tokens,      // [B]
   const int b = blockIdx.x;
   const float* row = logits + (long)b * V;
   const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
 
   __shared__ float cv[SMP_THREADS * SMP_MAXK];
   __shared__ int ci[SMP_THREADS * SMP_MAXK];
+  __shared__ float sel_v[SMP_MAXK];
+  __shared__ int sel_i[SMP_MAXK];
+  __shared__ float wmax[4];
+  __shared__ int wwin[4];
 
-  // 1) local sorted-descending top-K over strided slice
-  float lv[SMP_MAXK];
-  int li[SMP_MAXK];
-  for (int i = 0; i < K; ++i) { lv[i] = -INFINITY; li[i] = -1; }
-  float lmin = -INFINITY;  // register copy of lv[K-1] (lv lives in scratch)
-  for (int v = tid; v < V; v += SMP_THREADS) {
-    float x = row[v];
-    if (x <= lmin) continue;
-    int pos = K - 1;
-    while (pos > 0 && lv[pos - 1] < x) {
-      lv[pos] = lv[pos - 1]; li[pos] = li[pos - 1]; --pos;
-    }
-    lv[pos] = x; li[pos] = v;
-    lmin = lv[K - 1];
-  }
   float* mycv = cv + tid * SMP_MAXK;
   int* myci = ci + tid * SMP_MAXK;
-  for (int i = 0; i < K; ++i) { mycv[i] = lv[i]; myci[i] = li[i]; }
+
+  // 1) strided scan, replace-min into LDS row
+  for (int i = 0; i < K; ++i) { mycv[i] = -INFINITY; myci[i] = -1; }
+  float lmin = -INFINITY;
+  int min_slot = 0;
+  for (int v = tid; v < V; v += SMP_THREADS) {
+    const float x = row[v];
+    if (x <= lmin) continue;
+    mycv[min_slot] = x;
+    myci[min_slot] = v;
+    // rescan for the new min (K LDS reads; insertions are rare after warmup)
+    float nm = mycv[0]; int ns = 0;
+    for (int i = 1; i < K; ++i) {
+      if (mycv[i] < nm) { nm = mycv[i]; ns = i; }
+    }
+    lmin = nm; min_slot = ns;
+  }
+  // local max (kept in registers for the tournament)
+  float my_max = -INFINITY; int my_slot = 0;
+  for (int i = 0; i < K; ++i)
+    if (mycv[i] > my_max) { my_max = mycv[i]; my_slot = i; }
   __syncthreads();
 
-  // 2) tournament merge: 8 rounds; thread t < half merges list[t+half] into
-  //    list[t], both sorted desc → sorted top-K kept in registers then stored.
-  for (int half = SMP_THREADS / 2; half >= 1; half >>= 1) {
-    if (tid < half) {
-      const float* av = cv + tid * SMP_MAXK;
-      const int* ai = ci + tid * SMP_MAXK;
-      const float* bv = cv + (tid + half) * SMP_MAXK;
-      const int* bi = ci + (tid + half) * SMP_MAXK;
-      float mv[SMP_MAXK];
-      int mi[SMP_MAXK];
-      int pa = 0, pb = 0;
-      for (int i = 0; i < K; ++i) {
-        if (pb >= K || (pa < K && av[pa] >= bv[pb])) {
-          mv[i] = av[pa]; mi[i] = ai[pa]; ++pa;
-        } else {
-          mv[i] = bv[pb]; mi[i] = bi[pb]; ++pb;
-        }
+  // 2) block tournament: K rounds of argmax over 256 candidates
+  for (int k = 0; k < K; ++k) {
+    // wave argmax (value, tid)
+    float v = my_max; int who = tid;
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      float ov = __shfl_xor(v, off, WAVE);
+      int ow = __shfl_xor(who, off, WAVE);
+      if (ov > v || (ov == v && ow < who)) { v = ov; who = ow; }
+    }
+    if (lane == 0) { wmax[wid] = v; wwin[wid] = who; }
+    __syncthreads();
+    // cross-wave (4 entries) resolved by every thread identically
+    float bv = wmax[0]; int bw = wwin[0];
+    #pragma unroll
+    for (int w = 1; w < 4; ++w)
+      if (wmax[w] > bv || (wmax[w] == bv && wwin[w] < bw)) {
+        bv = wmax[w]; bw = wwin[w];
       }
-      float* ov = cv + tid * SMP_MAXK;
-      int* oi = ci + tid * SMP_MAXK;
-      for (int i = 0; i < K; ++i) { ov[i] = mv[i]; oi[i] = mi[i]; }
+    if (tid == bw) {
+      sel_v[k] = my_max;
+      sel_i[k] = myci[my_slot];
+      mycv[my_slot] = -INFINITY;
+      my_max = -INFINITY;
+      for (int i = 0; i < K; ++i)
+        if (mycv[i] > my_max) { my_max = mycv[i]; my_slot = i; }
     }
     __syncthreads();
+    if (bv == -INFINITY) break;
   }
 
-  // 3) thread 0: temperature softmax over global top-K (sorted desc at list 0),
-  //    top-p nucleus cut, multinomial draw.
+  // 3) thread 0: softmax over the (sorted-desc) top-K, top-p cut, draw
   if (tid == 0) {
     int n = 0;
-    while (n < K && ci[n] >= 0 && cv[n] != -INFINITY) ++n;
+    while (n < K && sel_i[n] >= 0 && sel_v[n] != -INFINITY) ++n;
     if (n == 0) { out_tokens[b] = 0; return; }
-    if (temperature <= 1e-5f) { out_tokens[b] = ci[0]; return; }  // greedy
+    if (temperature <= 1e-5f) { out_tokens[b] = sel_i[0]; return; }  // greedy
     const float invt = 1.0f / temperature;
-    const float m = cv[0];
+    const float m = sel_v[0];
     float probs[SMP_MAXK];
     float denom = 0.f;
     for (int i = 0; i < n; ++i) {
-      probs[i] = __expf((cv[i] - m) * invt);
+      probs[i] = __expf((sel_v[i] - m) * invt);
       denom += probs[i];
     }
     float cum = 0.f;
@@ -104,16 +123,14 @@ void sample_kernel(int* __restrict__ out_tokens,      // [B]
     }
     float denom2 = 0.f;
     for (int i = 0; i < cut; ++i) denom2 += probs[i];
-    // stateful on-device RNG: the state advances in place, so graph replays
-    // (hipGraph decode capture) draw fresh randomness with zero host work
     uint64_t st = seeds[b] | 1ull;
     float r = xorshift_unit(&st) * denom2;
-    seeds[b] = st;
+    seeds[b] = st;  // on-device state advance (graph-replay safe)
     float acc = 0.f;
-    int pick = ci[cut - 1];
+    int pick = sel_i[cut - 1];
     for (int i = 0; i < cut; ++i) {
       acc += probs[i];
-      if (r <= acc) { pick = ci[i]; break; }
+      if (r <= acc) { pick = sel_i[i]; break; }
     }
     out_tokens[b] = pick;
   }

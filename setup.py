@@ -16,6 +16,7 @@ SOURCES = [
     "room_amd/ops/csrc/norm_rope.hip",
     "room_amd/ops/csrc/paged_attn.hip",
     "room_amd/ops/csrc/moe.hip",
+    "room_amd/ops/csrc/gemv.hip",
     "room_amd/ops/csrc/sampling.hip",
     "room_amd/ops/csrc/vector_store.hip",
 ]

@@ -264,3 +264,35 @@ def test_vector_store_topk():
     assert torch.allclose(v.cpu(), v_ref.cpu(), atol=1e-2)
     overlap = len(set(i.tolist()) & set(i_ref.tolist()))
     assert overlap >= k - 2
+
+
+def test_gemv_vs_matmul():
+    torch.manual_seed(13)
+    for B, H, N in [(1, 2048, 5120), (5, 4096, 2048), (8, 2048, 128)]:
+        x = torch.randn(B, H, dtype=torch.bfloat16, device=DEV)
+        w = torch.randn(N, H, dtype=torch.bfloat16, device=DEV) * 0.05
+        y = torch.empty(B, N, dtype=torch.bfloat16, device=DEV)
+        ops.gemv(y, x, w)
+        expect = (x.float() @ w.float().T)
+        assert bf16_close(y, expect.to(torch.bfloat16), atol=5e-2, rtol=5e-2)
+        yf = torch.empty(B, N, dtype=torch.float32, device=DEV)
+        ops.gemv(yf, x, w)
+        assert torch.allclose(yf, expect, atol=5e-2, rtol=5e-2)
+
+
+def test_paged_attention_split_matches_ref():
+    torch.manual_seed(14)
+    B, Hq, Hk, D = 4, 32, 4, 128
+    lens = [1, 33, 250, 500]
+    kcache, vcache, bt = _setup_cache(B, 512)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+    seq_ids = torch.arange(B, dtype=torch.int32, device=DEV)
+    q_pos = torch.tensor([l - 1 for l in lens], dtype=torch.int32, device=DEV)
+    out = torch.empty_like(q)
+    part = torch.empty(B, Hq, 16, D, dtype=torch.float32, device=DEV)
+    part_ml = torch.empty(B, Hq, 16, 2, dtype=torch.float32, device=DEV)
+    scale = D ** -0.5
+    ops.paged_attention_split(out, q, kcache, vcache, bt, seq_ids, q_pos,
+                              part, part_ml, scale)
+    expect = ref.paged_attention_ref(q, kcache, vcache, bt, seq_ids, q_pos, scale)
+    assert bf16_close(out, expect, atol=3e-2)
