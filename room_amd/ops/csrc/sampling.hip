@@ -38,15 +38,17 @@ void sample_kernel(int* __restrict__ out_tokens,      // [B]
   const int lane = tid & 63;
   const int wid = tid >> 6;
 
-  __shared__ float cv[SMP_THREADS * SMP_MAXK];
-  __shared__ int ci[SMP_THREADS * SMP_MAXK];
+  // row stride K+1: stride 64 would put a wave's 64 lanes on one bank
+  // (32-way conflict on every access — guide G4); +1 spreads them.
+  __shared__ float cv[SMP_THREADS * (SMP_MAXK + 1)];
+  __shared__ int ci[SMP_THREADS * (SMP_MAXK + 1)];
   __shared__ float sel_v[SMP_MAXK];
   __shared__ int sel_i[SMP_MAXK];
   __shared__ float wmax[4];
   __shared__ int wwin[4];
 
-  float* mycv = cv + tid * SMP_MAXK;
-  int* myci = ci + tid * SMP_MAXK;
+  float* mycv = cv + tid * (SMP_MAXK + 1);
+  int* myci = ci + tid * (SMP_MAXK + 1);
 
   // 1) strided scan, replace-min into LDS row
   for (int i = 0; i < K; ++i) { mycv[i] = -INFINITY; myci[i] = -1; }

@@ -56,8 +56,11 @@ res = x.clone()
 total += bench("fused_add_rmsnorm",
                lambda: ops.fused_add_rmsnorm(out, res, x, norm_w),
                bytes_moved=4 * B * H * 2)
-total += bench("qkv linear (hipBLASLt)", lambda: F.linear(x, wqkv),
+qkv_out = torch.empty(B, QD + 2 * KVD, dtype=torch.bfloat16, device=DEV)
+total += bench("qkv gemv", lambda: ops.gemv(qkv_out, x, wqkv),
                bytes_moved=wqkv.numel() * 2)
+bench("qkv linear (hipBLASLt)", lambda: F.linear(x, wqkv),
+      bytes_moved=wqkv.numel() * 2)
 
 q = torch.randn(B, 32, 128, dtype=torch.bfloat16, device=DEV)
 k = torch.randn(B, 4, 128, dtype=torch.bfloat16, device=DEV)
@@ -77,15 +80,26 @@ qpos = torch.full((B,), 512, dtype=torch.int32, device=DEV)
 total += bench("write_kv",
                lambda: ops.write_kv(kcache, vcache, k, k, bt, seq_ids, qpos))
 attn = torch.empty_like(q)
-total += bench("paged_attention seq=512",
-               lambda: ops.paged_attention(attn, q, kcache, vcache, bt, seq_ids,
-                                           qpos, 0.0884),
+part = torch.empty(B, 32, 16, 128, dtype=torch.float32, device=DEV)
+part_ml = torch.empty(B, 32, 16, 2, dtype=torch.float32, device=DEV)
+total += bench("paged_attn_split seq=512",
+               lambda: ops.paged_attention_split(attn, q, kcache, vcache, bt,
+                                                 seq_ids, qpos, part, part_ml,
+                                                 0.0884),
                bytes_moved=B * 4 * 512 * 128 * 2 * 2)
-total += bench("o linear", lambda: F.linear(attn.reshape(B, QD), wo),
+bench("paged_attention(old) seq=512",
+      lambda: ops.paged_attention(attn, q, kcache, vcache, bt, seq_ids,
+                                  qpos, 0.0884),
+      bytes_moved=B * 4 * 512 * 128 * 2 * 2)
+o_out = torch.empty(B, H, dtype=torch.bfloat16, device=DEV)
+attn_flat = attn.reshape(B, QD).contiguous()
+total += bench("o gemv", lambda: ops.gemv(o_out, attn_flat, wo),
                bytes_moved=wo.numel() * 2)
-total += bench("router linear f32", lambda: F.linear(x.float(), wr),
-               bytes_moved=wr.numel() * 4)
-rl = F.linear(x.float(), wr)
+wr_bf = wr.to(torch.bfloat16)
+r_out = torch.empty(B, E, dtype=torch.float32, device=DEV)
+total += bench("router gemv f32out", lambda: ops.gemv(r_out, x, wr_bf),
+               bytes_moved=wr_bf.numel() * 2)
+rl = r_out
 total += bench("moe_router", lambda: ops.moe_router(rl, K))
 
 ids, w = ops.moe_router(rl, K)
@@ -104,9 +118,12 @@ t_d = bench("moe_gemv_down",
 total += t_h + t_d
 
 print(f"{'—'*50}\nper-layer total ≈ {total:.0f} µs → 48 layers = {total*48/1000:.2f} ms")
-lm = bench("lm_head (151936×2048)", lambda: F.linear(x, lm_head),
+lm_out = torch.empty(B, V, dtype=torch.float32, device=DEV)
+lm = bench("lm_head gemv", lambda: ops.gemv(lm_out, x, lm_head),
            bytes_moved=lm_head.numel() * 2)
-smp_logits = F.linear(x, lm_head).float()
+bench("lm_head hipBLASLt", lambda: F.linear(x, lm_head),
+      bytes_moved=lm_head.numel() * 2)
+smp_logits = lm_out
 seeds = torch.randint(1, 2**62, (B,), dtype=torch.int64, device=DEV)
 bench("sample_tokens", lambda: ops.sample_tokens(smp_logits, seeds))
 print(f"estimated decode step = {(total*48 + lm)/1000:.2f} ms (+sampling)")
