@@ -29,7 +29,8 @@ from .types import AgentExecutionOptions, ToolDef
 
 PREFILL_CHUNK = 4096          # max tokens per prefill forward
 DEFAULT_MAX_SEQS = 64
-DECODE_BUCKETS = (1, 2, 4, 8, 16, 32, 64)  # hipGraph capture sizes
+DECODE_BUCKETS = (1, 2, 3, 4, 5, 6, 7, 8, 16, 32, 64)  # hipGraph capture sizes
+# ≤8 buckets are exact (padding inflates MoE pair traffic ~linearly)
 
 
 @dataclass

@@ -35,6 +35,10 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
                            torch::Tensor q_pos, torch::Tensor part,
                            torch::Tensor part_ml, double scale);
 void gemv(torch::Tensor y, torch::Tensor x, torch::Tensor w);
+void sample_tokens_v3(torch::Tensor out_tokens, torch::Tensor logits,
+                      torch::Tensor seeds, int64_t top_k, double temperature,
+                      double top_p);
+void sample_scan_probe(torch::Tensor out, torch::Tensor logits);
 void vs_topk(torch::Tensor out_v, torch::Tensor out_i, torch::Tensor cand_v,
              torch::Tensor cand_i, torch::Tensor mat, torch::Tensor query,
              int64_t K);
@@ -55,5 +59,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_split", &paged_attention_split,
         "split-KV flash-decode paged attention");
   m.def("gemv", &gemv, "dense skinny-batch GEMV (decode projections)");
+  m.def("sample_tokens_v3", &sample_tokens_v3, "register top-8 sampler");
+  m.def("sample_scan_probe", &sample_scan_probe, "scan-cost probe");
   m.def("vs_topk", &vs_topk, "vector-store cosine top-k over bf16 matrix");
 }

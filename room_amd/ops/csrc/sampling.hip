@@ -152,3 +152,164 @@ void sample_tokens(torch::Tensor out_tokens, torch::Tensor logits,
                      (float)temperature, (float)top_p);
   HIP_CHECK_KERNEL();
 }
+
+// ---- v3: register-resident per-thread top-8 + block tournament ----
+// Per-thread top-8 kept in VGPRs via a fully-static compare/shift chain (no
+// dynamic indexing → no scratch, no LDS in the scan). The global top-K
+// (K ≤ 64) is contained in the union of per-thread top-8 unless one thread's
+// strided slice holds ≥9 of the global top-K: P ≈ C(64,9)/256⁸ < 1e-9 for
+// random position assignment — negligible for sampling purposes.
+#define V3_KEEP 8
+
+__global__ __launch_bounds__(SMP_THREADS, 1)
+void sample_v3_kernel(int* __restrict__ out_tokens,
+                      const float* __restrict__ logits,
+                      uint64_t* __restrict__ seeds,
+                      int V, int K, float temperature, float top_p) {
+  const int b = blockIdx.x;
+  const float* row = logits + (long)b * V;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+
+  float r0 = -INFINITY, r1 = -INFINITY, r2 = -INFINITY, r3 = -INFINITY,
+        r4 = -INFINITY, r5 = -INFINITY, r6 = -INFINITY, r7 = -INFINITY;
+  int i0 = -1, i1 = -1, i2 = -1, i3 = -1, i4 = -1, i5 = -1, i6 = -1, i7 = -1;
+
+  for (int v = tid; v < V; v += SMP_THREADS) {
+    const float x = row[v];
+    if (x <= r7) continue;
+    if (x > r0) {
+      r7=r6; i7=i6; r6=r5; i6=i5; r5=r4; i5=i4; r4=r3; i4=i3;
+      r3=r2; i3=i2; r2=r1; i2=i1; r1=r0; i1=i0; r0=x; i0=v;
+    } else if (x > r1) {
+      r7=r6; i7=i6; r6=r5; i6=i5; r5=r4; i5=i4; r4=r3; i4=i3;
+      r3=r2; i3=i2; r2=r1; i2=i1; r1=x; i1=v;
+    } else if (x > r2) {
+      r7=r6; i7=i6; r6=r5; i6=i5; r5=r4; i5=i4; r4=r3; i4=i3;
+      r3=r2; i3=i2; r2=x; i2=v;
+    } else if (x > r3) {
+      r7=r6; i7=i6; r6=r5; i6=i5; r5=r4; i5=i4; r4=r3; i4=i3; r3=x; i3=v;
+    } else if (x > r4) {
+      r7=r6; i7=i6; r6=r5; i6=i5; r5=r4; i5=i4; r4=x; i4=v;
+    } else if (x > r5) {
+      r7=r6; i7=i6; r6=r5; i6=i5; r5=x; i5=v;
+    } else if (x > r6) {
+      r7=r6; i7=i6; r6=x; i6=v;
+    } else {
+      r7=x; i7=v;
+    }
+  }
+
+  __shared__ float cv[SMP_THREADS * (V3_KEEP + 1)];
+  __shared__ int ci[SMP_THREADS * (V3_KEEP + 1)];
+  __shared__ float sel_v[SMP_MAXK];
+  __shared__ int sel_i[SMP_MAXK];
+  __shared__ float wmax[4];
+  __shared__ int wwin[4];
+  float* mycv = cv + tid * (V3_KEEP + 1);
+  int* myci = ci + tid * (V3_KEEP + 1);
+  mycv[0]=r0; mycv[1]=r1; mycv[2]=r2; mycv[3]=r3;
+  mycv[4]=r4; mycv[5]=r5; mycv[6]=r6; mycv[7]=r7;
+  myci[0]=i0; myci[1]=i1; myci[2]=i2; myci[3]=i3;
+  myci[4]=i4; myci[5]=i5; myci[6]=i6; myci[7]=i7;
+  float my_max = r0;  // rows are sorted desc
+  int my_slot = 0;
+  __syncthreads();
+
+  for (int k = 0; k < K; ++k) {
+    float v = my_max; int who = tid;
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      float ov = __shfl_xor(v, off, WAVE);
+      int ow = __shfl_xor(who, off, WAVE);
+      if (ov > v || (ov == v && ow < who)) { v = ov; who = ow; }
+    }
+    if (lane == 0) { wmax[wid] = v; wwin[wid] = who; }
+    __syncthreads();
+    float bv = wmax[0]; int bw = wwin[0];
+    #pragma unroll
+    for (int w = 1; w < 4; ++w)
+      if (wmax[w] > bv || (wmax[w] == bv && wwin[w] < bw)) {
+        bv = wmax[w]; bw = wwin[w];
+      }
+    if (tid == bw) {
+      sel_v[k] = my_max;
+      sel_i[k] = myci[my_slot];
+      // advance to next candidate in the sorted row
+      ++my_slot;
+      my_max = (my_slot < V3_KEEP) ? mycv[my_slot] : -INFINITY;
+    }
+    __syncthreads();
+    if (bv == -INFINITY) break;
+  }
+
+  if (tid == 0) {
+    int n = 0;
+    while (n < K && sel_i[n] >= 0 && sel_v[n] != -INFINITY) ++n;
+    if (n == 0) { out_tokens[b] = 0; return; }
+    if (temperature <= 1e-5f) { out_tokens[b] = sel_i[0]; return; }
+    const float invt = 1.0f / temperature;
+    const float m = sel_v[0];
+    float denom = 0.f;
+    for (int i = 0; i < n; ++i) denom += __expf((sel_v[i] - m) * invt);
+    uint64_t st = seeds[b] | 1ull;
+    // top-p cut + draw in one pass (sel sorted desc)
+    float target = top_p * denom;
+    float cum = 0.f;
+    int cut = n;
+    for (int i = 0; i < n; ++i) {
+      cum += __expf((sel_v[i] - m) * invt);
+      if (cum >= target) { cut = i + 1; break; }
+    }
+    float denom2 = 0.f;
+    for (int i = 0; i < cut; ++i) denom2 += __expf((sel_v[i] - m) * invt);
+    float r = xorshift_unit(&st) * denom2;
+    seeds[b] = st;
+    float acc = 0.f;
+    int pick = sel_i[cut - 1];
+    for (int i = 0; i < cut; ++i) {
+      acc += __expf((sel_v[i] - m) * invt);
+      if (r <= acc) { pick = sel_i[i]; break; }
+    }
+    out_tokens[b] = pick;
+  }
+}
+
+// scan-only probe: block max (isolates the logits-scan cost)
+__global__ __launch_bounds__(SMP_THREADS)
+void sample_scan_probe_kernel(int* __restrict__ out, const float* __restrict__ logits,
+                              int V) {
+  const int b = blockIdx.x;
+  const float* row = logits + (long)b * V;
+  float m = -INFINITY; int mi = 0;
+  for (int v = threadIdx.x; v < V; v += SMP_THREADS) {
+    float x = row[v];
+    if (x > m) { m = x; mi = v; }
+  }
+  __shared__ float red[16];
+  float gm = block_reduce_max(m, red);
+  if (m == gm && threadIdx.x % 64 == 0) out[b] = mi;
+}
+
+void sample_tokens_v3(torch::Tensor out_tokens, torch::Tensor logits,
+                      torch::Tensor seeds, int64_t top_k, double temperature,
+                      double top_p) {
+  const int B = logits.size(0), V = logits.size(1);
+  TORCH_CHECK(logits.dtype() == torch::kFloat32);
+  TORCH_CHECK(top_k >= 1 && top_k <= SMP_MAXK);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(sample_v3_kernel, dim3(B), dim3(SMP_THREADS), 0, s,
+                     out_tokens.data_ptr<int>(), logits.data_ptr<float>(),
+                     (uint64_t*)seeds.data_ptr(), V, (int)top_k,
+                     (float)temperature, (float)top_p);
+  HIP_CHECK_KERNEL();
+}
+
+void sample_scan_probe(torch::Tensor out, torch::Tensor logits) {
+  const int B = logits.size(0), V = logits.size(1);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(sample_scan_probe_kernel, dim3(B), dim3(SMP_THREADS), 0, s,
+                     out.data_ptr<int>(), logits.data_ptr<float>(), V);
+  HIP_CHECK_KERNEL();
+}

@@ -126,4 +126,11 @@ bench("lm_head hipBLASLt", lambda: F.linear(x, lm_head),
 smp_logits = lm_out
 seeds = torch.randint(1, 2**62, (B,), dtype=torch.int64, device=DEV)
 bench("sample_tokens", lambda: ops.sample_tokens(smp_logits, seeds))
+out_i = torch.empty(B, dtype=torch.int32, device=DEV)
+bench("sample_v3", lambda: ops._require().sample_tokens_v3(out_i, smp_logits, seeds, 40, 0.7, 0.95))
+bench("sample_scan_probe", lambda: ops._require().sample_scan_probe(out_i, smp_logits))
+# v3 greedy correctness inline check
+ops._require().sample_tokens_v3(out_i, smp_logits, seeds, 40, 0.0, 1.0)
+assert torch.equal(out_i.long().cpu(), smp_logits.argmax(-1).cpu()), "v3 greedy mismatch"
+print("v3 greedy OK")
 print(f"estimated decode step = {(total*48 + lm)/1000:.2f} ms (+sampling)")
