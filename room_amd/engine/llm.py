@@ -88,7 +88,7 @@ class _DecodeGraph:
                                    cache.block_table, cache.kcaches,
                                    cache.vcaches)
             toks = torch.empty(bucket, dtype=torch.int32, device=dev)
-            _ops._require().sample_tokens(toks, logits, self.seeds, top_k,
+            _ops._require().sample_tokens_v3(toks, logits, self.seeds, top_k,
                                           temperature, top_p)
             return toks
 

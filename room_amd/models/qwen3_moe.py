@@ -215,15 +215,17 @@ class Qwen3MoEModel:
                       .repeat_interleave(K))[order.long()].contiguous()
         pair_w = topk_w.flatten()[order.long()].contiguous()
         P = pair_token.numel()
-        desc1 = ops.build_moe_tile_desc(pair_expert, (2 * I) // 64, cfg.num_experts)
+        desc1 = ops.build_moe_tile_desc(pair_expert, (2 * I) // 64,
+                                        cfg.num_experts, bm=128)
         gateup = torch.empty(P, 2 * I, dtype=torch.bfloat16, device=hbuf.device)
-        ops.moe_grouped_gemm(gateup, hbuf, layer.w13, pair_token, desc1)
+        ops.moe_grouped_gemm128(gateup, hbuf, layer.w13, pair_token, desc1)
         h = torch.empty(P, I, dtype=torch.bfloat16, device=hbuf.device)
         ops.silu_mul(h, gateup)
-        desc2 = ops.build_moe_tile_desc(pair_expert, H // 64, cfg.num_experts)
+        desc2 = ops.build_moe_tile_desc(pair_expert, H // 64,
+                                        cfg.num_experts, bm=128)
         z = torch.empty(P, H, dtype=torch.bfloat16, device=hbuf.device)
-        ops.moe_grouped_gemm(z, h, layer.w2,
-                             torch.arange(P, device=hbuf.device, dtype=torch.int32),
-                             desc2)
+        ops.moe_grouped_gemm128(z, h, layer.w2,
+                                torch.arange(P, device=hbuf.device, dtype=torch.int32),
+                                desc2)
         ops.moe_combine(out, z, pair_w, pair_token)
         return out

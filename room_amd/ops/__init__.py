@@ -113,6 +113,14 @@ def moe_grouped_gemm(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
     return out
 
 
+def moe_grouped_gemm128(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
+                        pair_token: torch.Tensor,
+                        tile_desc: torch.Tensor) -> torch.Tensor:
+    """BM=128 variant: expert weight panels read once per 128-row m-tile."""
+    _require().moe_grouped_gemm128(out, x, w, pair_token, tile_desc)
+    return out
+
+
 def moe_combine(out: torch.Tensor, z: torch.Tensor, pair_w: torch.Tensor,
                 pair_token: torch.Tensor) -> torch.Tensor:
     _require().moe_combine(out, z, pair_w, pair_token)
@@ -122,7 +130,8 @@ def moe_combine(out: torch.Tensor, z: torch.Tensor, pair_w: torch.Tensor,
 def sample_tokens(logits: torch.Tensor, seeds: torch.Tensor, top_k: int = 40,
                   temperature: float = 0.7, top_p: float = 0.95) -> torch.Tensor:
     out = torch.empty(logits.size(0), dtype=torch.int32, device=logits.device)
-    _require().sample_tokens(out, logits, seeds, top_k, temperature, top_p)
+    # v3: register-resident top-8 scan + tournament (4× the v2 LDS variant)
+    _require().sample_tokens_v3(out, logits, seeds, top_k, temperature, top_p)
     return out
 
 

@@ -24,6 +24,8 @@ void moe_gemv_down(torch::Tensor out, torch::Tensor h, torch::Tensor w2,
                    torch::Tensor pair_expert);
 void moe_grouped_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                       torch::Tensor pair_token, torch::Tensor tile_desc);
+void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                         torch::Tensor pair_token, torch::Tensor tile_desc);
 void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
                  torch::Tensor pair_token);
 void sample_tokens(torch::Tensor out_tokens, torch::Tensor logits,
@@ -54,6 +56,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_gemv_h", &moe_gemv_h, "MoE gate/up GEMV + silu-mul (decode)");
   m.def("moe_gemv_down", &moe_gemv_down, "MoE down GEMV + weighted scatter-add");
   m.def("moe_grouped_gemm", &moe_grouped_gemm, "grouped MFMA GEMM (prefill)");
+  m.def("moe_grouped_gemm128", &moe_grouped_gemm128, "BM=128 grouped MFMA GEMM");
   m.def("moe_combine", &moe_combine, "weighted scatter-add combine");
   m.def("sample_tokens", &sample_tokens, "fused temperature/top-k/top-p sampling");
   m.def("paged_attention_split", &paged_attention_split,

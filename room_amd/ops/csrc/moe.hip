@@ -313,3 +313,105 @@ void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
                      pair_w.data_ptr<float>(), pair_token.data_ptr<int>(), H);
   HIP_CHECK_KERNEL();
 }
+
+// ------------------------------------------------ BM=128 grouped MFMA GEMM
+// The BM=16 tile re-reads each expert's weight panel ceil(rows/16)× per layer
+// (~8× at 2k-token prefill → 6.4 GB/layer). BM=128 reads W once per expert
+// m-tile: 8 waves (512 threads), wave w owns rows [16w, 16w+16) × BN=64.
+// LDS rows padded +8 bf16 so ds_read_b128 across rows spreads banks (G4).
+
+#define G2_BM 128
+#define G2_BN 64
+#define G2_BK 64
+#define G2_PAD 8
+
+__global__ __launch_bounds__(512)
+void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
+                                const short* __restrict__ x,   // [T, H]
+                                const short* __restrict__ w,   // [E, N, H]
+                                const int* __restrict__ pair_token,
+                                const int* __restrict__ tile_desc,  // [G,4]
+                                int H, int N) {
+  const int g = blockIdx.x;
+  const int e = tile_desc[g * 4 + 0];
+  const int row0 = tile_desc[g * 4 + 1];
+  const int msize = tile_desc[g * 4 + 2];
+  const int n0 = tile_desc[g * 4 + 3] * G2_BN;
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;       // 0..7: wave's 16-row m-slice
+  const int lane = tid & 63;
+
+  __shared__ short xs[G2_BM][G2_BK + G2_PAD];
+  __shared__ short ws[G2_BN][G2_BK + G2_PAD];
+
+  const short* wbase = w + (long)e * N * H;
+
+  cfrag_t acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                    {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+
+  for (int k0 = 0; k0 < H; k0 += G2_BK) {
+    // stage X tile: 128 rows × 64 k = 1024 vec8 → 2 per thread
+    #pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = tid + it * 512;
+      const int r = (idx * 8) / G2_BK;
+      const int c = (idx * 8) % G2_BK;
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (r < msize) {
+        const int tok = pair_token[row0 + r];
+        v = *reinterpret_cast<const bf16x8*>(x + (long)tok * H + k0 + c);
+      }
+      *reinterpret_cast<bf16x8*>(&xs[r][c]) = v;
+    }
+    // stage W tile: 64 rows × 64 k = 512 vec8 → 1 per thread
+    {
+      const int r = (tid * 8) / G2_BK;
+      const int c = (tid * 8) % G2_BK;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          wbase + (long)(n0 + r) * H + k0 + c);
+      *reinterpret_cast<bf16x8*>(&ws[r][c]) = v;
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int kk = 0; kk < G2_BK / 32; ++kk) {
+      const int arow = wid * 16 + (lane & 15);
+      const int akoff = kk * 32 + (lane >> 4) * 8;
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(&xs[arow][akoff]);
+      #pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int bcol = nf * 16 + (lane & 15);
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(&ws[bcol][akoff]);
+        acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nf], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  const int crow = wid * 16 + (lane >> 4) * 4;
+  const int ccol_base = lane & 15;
+  #pragma unroll
+  for (int nf = 0; nf < 4; ++nf) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = crow + r;
+      if (m < msize)
+        out[(long)(row0 + m) * N + n0 + nf * 16 + ccol_base] = f2bf(acc[nf][r]);
+    }
+  }
+}
+
+void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                         torch::Tensor pair_token, torch::Tensor tile_desc) {
+  const int H = x.size(-1), N = out.size(-1);
+  const int G = tile_desc.size(0);
+  TORCH_CHECK(H % G2_BK == 0 && N % G2_BN == 0);
+  dim3 grid(G), block(512);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(moe_grouped_gemm128_kernel, grid, block, 0, s,
+                     (short*)out.data_ptr(), (const short*)x.data_ptr(),
+                     (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
+                     tile_desc.data_ptr<int>(), H, N);
+  HIP_CHECK_KERNEL();
+}

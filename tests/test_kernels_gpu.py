@@ -296,3 +296,22 @@ def test_paged_attention_split_matches_ref():
                               part, part_ml, scale)
     expect = ref.paged_attention_ref(q, kcache, vcache, bt, seq_ids, q_pos, scale)
     assert bf16_close(out, expect, atol=3e-2)
+
+
+def test_moe_grouped_gemm128_vs_matmul():
+    torch.manual_seed(15)
+    E, H, N = 4, 2048, 1536
+    T = 300  # several 128-row tiles plus a ragged tail
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV) * 0.3
+    w = torch.randn(E, N, H, dtype=torch.bfloat16, device=DEV) * 0.05
+    pair_expert = torch.randint(0, E, (T,), device=DEV).int()
+    order = torch.argsort(pair_expert)
+    pair_expert = pair_expert[order].contiguous()
+    pair_token = order.int().contiguous()
+    tile_desc = ops.build_moe_tile_desc(pair_expert, N // 64, E, bm=128).to(DEV)
+    out = torch.empty(T, N, dtype=torch.bfloat16, device=DEV)
+    ops.moe_grouped_gemm128(out, x, w, pair_token, tile_desc)
+    for p in [0, 1, 100, 150, 299]:
+        t, e = int(pair_token[p]), int(pair_expert[p])
+        expect = (x[t].float() @ w[e].float().T).to(torch.bfloat16)
+        assert bf16_close(out[p], expect, atol=6e-2, rtol=6e-2), f"pair {p}"
