@@ -303,6 +303,7 @@ class LocalEngine:
         t0 = time.time()
         budget = PREFILL_CHUNK
         tokens, seq_ids, q_pos, last_rows, sampled_reqs = [], [], [], [], []
+        segments = []  # (row0, count) per request — host-side q-tile info
         for r in reqs:
             if budget <= 0:
                 break
@@ -310,6 +311,7 @@ class LocalEngine:
             chunk = r.pending_prefill[:take]
             r.pending_prefill = r.pending_prefill[take:]
             self.cache.ensure_capacity(r.slot, r.pos + take)
+            segments.append((len(tokens), take))
             tokens.extend(chunk)
             seq_ids.extend([r.slot] * take)
             q_pos.extend(range(r.pos, r.pos + take))
@@ -323,9 +325,11 @@ class LocalEngine:
         seq_t = torch.tensor(seq_ids, dtype=torch.int32, device=dev)
         pos_t = torch.tensor(q_pos, dtype=torch.int32, device=dev)
         rows_t = torch.tensor(last_rows, dtype=torch.int64, device=dev)
+        qtiles = (ops.build_qtile_desc(segments, dev)
+                  if len(tokens) > 8 else None)
         logits = self.model.forward(tokens_t, seq_t, pos_t, self.cache.block_table,
                                     self.cache.kcaches, self.cache.vcaches,
-                                    logits_rows=rows_t)
+                                    logits_rows=rows_t, qtile_desc=qtiles)
         self.stats["prefill_tokens"] += len(tokens)
         if sampled_reqs:
             self._sample_and_append(sampled_reqs, logits)

@@ -107,7 +107,8 @@ class Qwen3MoEModel:
     def forward(self, tokens: torch.Tensor, seq_ids: torch.Tensor,
                 q_pos: torch.Tensor, block_table: torch.Tensor,
                 kcaches: list[torch.Tensor], vcaches: list[torch.Tensor],
-                logits_rows: torch.Tensor | None = None) -> torch.Tensor:
+                logits_rows: torch.Tensor | None = None,
+                qtile_desc: torch.Tensor | None = None) -> torch.Tensor:
         """tokens/seq_ids/q_pos: [T] on device. kcaches/vcaches: one pair per
         layer. logits_rows: row indices to compute logits for (default: all).
         Returns fp32 logits [R, vocab]."""
@@ -160,6 +161,10 @@ class Qwen3MoEModel:
                                           part_ml, self.scale)
                 ops.gemv(obuf, attn.reshape(T, qdim), layer.wo)
                 o = obuf
+            elif qtile_desc is not None:
+                ops.flash_prefill(attn, q, kcaches[li], vcaches[li], block_table,
+                                  seq_ids, q_pos, qtile_desc, self.scale)
+                o = F.linear(attn.reshape(T, qdim), layer.wo)
             else:
                 ops.paged_attention(attn, q, kcaches[li], vcaches[li], block_table,
                                     seq_ids, q_pos, self.scale)
@@ -209,12 +214,14 @@ class Qwen3MoEModel:
 
         # prefill path: sort pairs by expert, grouped MFMA GEMMs
         flat_expert = topk_ids.flatten()
-        order = torch.argsort(flat_expert).int()
-        pair_expert = flat_expert[order.long()].contiguous()
+        order = torch.argsort(flat_expert)
+        pair_expert = flat_expert[order].int().contiguous()
         pair_token = (torch.arange(T, device=hbuf.device, dtype=torch.int32)
-                      .repeat_interleave(K))[order.long()].contiguous()
-        pair_w = topk_w.flatten()[order.long()].contiguous()
+                      .repeat_interleave(K))[order].contiguous()
         P = pair_token.numel()
+        inv_order = torch.empty_like(order)
+        inv_order[order] = torch.arange(P, device=hbuf.device)
+        inv_order = inv_order.int().contiguous()
         desc1 = ops.build_moe_tile_desc(pair_expert, (2 * I) // 64,
                                         cfg.num_experts, bm=128)
         gateup = torch.empty(P, 2 * I, dtype=torch.bfloat16, device=hbuf.device)
@@ -227,5 +234,6 @@ class Qwen3MoEModel:
         ops.moe_grouped_gemm128(z, h, layer.w2,
                                 torch.arange(P, device=hbuf.device, dtype=torch.int32),
                                 desc2)
-        ops.moe_combine(out, z, pair_w, pair_token)
-        return out
+        out_bf = torch.empty(T, H, dtype=torch.bfloat16, device=hbuf.device)
+        ops.moe_combine_gather(out_bf, z, topk_w.contiguous(), inv_order)
+        return out_bf

@@ -86,6 +86,32 @@ def gemv(y: torch.Tensor, x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return y
 
 
+def flash_prefill(out: torch.Tensor, q: torch.Tensor, kcache: torch.Tensor,
+                  vcache: torch.Tensor, block_table: torch.Tensor,
+                  seq_ids: torch.Tensor, q_pos: torch.Tensor,
+                  tile_desc: torch.Tensor, scale: float) -> torch.Tensor:
+    """MFMA flash prefill; tile_desc [G,2] = (row0, n_tokens≤16), one sequence
+    per tile (build with build_qtile_desc)."""
+    _require().flash_prefill(out, q, kcache, vcache, block_table, seq_ids,
+                             q_pos, tile_desc, scale)
+    return out
+
+
+def build_qtile_desc(segments: list, device) -> torch.Tensor:
+    """Q-tile descriptors from host-known prefill segments [(row0, count), ...]
+    (one sequence per segment) — no device sync."""
+    desc = []
+    for row0, count in segments:
+        r = row0
+        while r < row0 + count:
+            n = min(16, row0 + count - r)
+            desc.append((r, n))
+            r += n
+    if not desc:
+        return torch.zeros(0, 2, dtype=torch.int32, device=device)
+    return torch.tensor(desc, dtype=torch.int32, device=device)
+
+
 def moe_router(logits: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:
     T = logits.size(0)
     ids = torch.empty(T, k, dtype=torch.int32, device=logits.device)
@@ -124,6 +150,12 @@ def moe_grouped_gemm128(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
 def moe_combine(out: torch.Tensor, z: torch.Tensor, pair_w: torch.Tensor,
                 pair_token: torch.Tensor) -> torch.Tensor:
     _require().moe_combine(out, z, pair_w, pair_token)
+    return out
+
+
+def moe_combine_gather(out: torch.Tensor, z: torch.Tensor, topk_w: torch.Tensor,
+                       inv_order: torch.Tensor) -> torch.Tensor:
+    _require().moe_combine_gather(out, z, topk_w, inv_order)
     return out
 
 

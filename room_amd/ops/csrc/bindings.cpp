@@ -28,6 +28,8 @@ void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                          torch::Tensor pair_token, torch::Tensor tile_desc);
 void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
                  torch::Tensor pair_token);
+void moe_combine_gather(torch::Tensor out, torch::Tensor z, torch::Tensor topk_w,
+                        torch::Tensor inv_order);
 void sample_tokens(torch::Tensor out_tokens, torch::Tensor logits,
                    torch::Tensor seeds, int64_t top_k, double temperature,
                    double top_p);
@@ -36,6 +38,10 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
                            torch::Tensor block_table, torch::Tensor seq_ids,
                            torch::Tensor q_pos, torch::Tensor part,
                            torch::Tensor part_ml, double scale);
+void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
+                   torch::Tensor vcache, torch::Tensor block_table,
+                   torch::Tensor seq_ids, torch::Tensor q_pos,
+                   torch::Tensor tile_desc, double scale);
 void gemv(torch::Tensor y, torch::Tensor x, torch::Tensor w);
 void sample_tokens_v3(torch::Tensor out_tokens, torch::Tensor logits,
                       torch::Tensor seeds, int64_t top_k, double temperature,
@@ -58,9 +64,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_grouped_gemm", &moe_grouped_gemm, "grouped MFMA GEMM (prefill)");
   m.def("moe_grouped_gemm128", &moe_grouped_gemm128, "BM=128 grouped MFMA GEMM");
   m.def("moe_combine", &moe_combine, "weighted scatter-add combine");
+  m.def("moe_combine_gather", &moe_combine_gather, "atomics-free MoE combine");
   m.def("sample_tokens", &sample_tokens, "fused temperature/top-k/top-p sampling");
   m.def("paged_attention_split", &paged_attention_split,
         "split-KV flash-decode paged attention");
+  m.def("flash_prefill", &flash_prefill, "MFMA flash-attention prefill");
   m.def("gemv", &gemv, "dense skinny-batch GEMV (decode projections)");
   m.def("sample_tokens_v3", &sample_tokens_v3, "register top-8 sampler");
   m.def("sample_scan_probe", &sample_scan_probe, "scan-cost probe");

@@ -415,3 +415,45 @@ void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                      tile_desc.data_ptr<int>(), H, N);
   HIP_CHECK_KERNEL();
 }
+
+// gather-combine: out[t] = Σ_k w[t,k] * z[inv_order[t*K+k]] — replaces the
+// atomicAdd combine (33M atomics/layer at 2k-token prefill → 0.9 ms/layer).
+// inv_order maps (token, k) → sorted-pair row of z.
+__global__ void moe_combine_gather_kernel(short* __restrict__ out,  // [T, H] bf16
+                                          const short* __restrict__ z,  // [P, H]
+                                          const float* __restrict__ topk_w,  // [T, K]
+                                          const int* __restrict__ inv_order,  // [T*K]
+                                          int H, int K) {
+  const int t = blockIdx.x;
+  float w[8];
+  int p[8];
+  for (int k = 0; k < K; ++k) {
+    w[k] = topk_w[(long)t * K + k];
+    p[k] = inv_order[(long)t * K + k];
+  }
+  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int k = 0; k < K; ++k) {
+      bf16x8 zv = *reinterpret_cast<const bf16x8*>(z + (long)p[k] * H + i);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += w[k] * bf2f(zv[j]);
+    }
+    bf16x8 o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(acc[j]);
+    *reinterpret_cast<bf16x8*>(out + (long)t * H + i) = o;
+  }
+}
+
+void moe_combine_gather(torch::Tensor out, torch::Tensor z, torch::Tensor topk_w,
+                        torch::Tensor inv_order) {
+  const int T = out.size(0), H = out.size(-1);
+  const int K = topk_w.size(-1);
+  TORCH_CHECK(K <= 8 && out.dtype() == torch::kBFloat16);
+  dim3 grid(T), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(moe_combine_gather_kernel, grid, block, 0, s,
+                     (short*)out.data_ptr(), (const short*)z.data_ptr(),
+                     topk_w.data_ptr<float>(), inv_order.data_ptr<int>(), H, K);
+  HIP_CHECK_KERNEL();
+}

@@ -315,3 +315,28 @@ def test_moe_grouped_gemm128_vs_matmul():
         t, e = int(pair_token[p]), int(pair_expert[p])
         expect = (x[t].float() @ w[e].float().T).to(torch.bfloat16)
         assert bf16_close(out[p], expect, atol=6e-2, rtol=6e-2), f"pair {p}"
+
+
+def test_flash_prefill_vs_ref():
+    torch.manual_seed(16)
+    Hq, Hk, D = 32, 4, 128
+    kcache, vcache, bt = _setup_cache(2, 128)
+    # two sequences: 50 tokens from pos 0, 37 tokens continuing from pos 20
+    segments, seq_ids, q_pos = [], [], []
+    row = 0
+    for s, start, n in [(0, 0, 50), (1, 20, 37)]:
+        segments.append((row, n))
+        for p in range(start, start + n):
+            seq_ids.append(s)
+            q_pos.append(p)
+        row += n
+    T = len(seq_ids)
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=DEV)
+    seq_ids = torch.tensor(seq_ids, dtype=torch.int32, device=DEV)
+    q_pos = torch.tensor(q_pos, dtype=torch.int32, device=DEV)
+    desc = ops.build_qtile_desc(segments, DEV)
+    out = torch.empty_like(q)
+    scale = D ** -0.5
+    ops.flash_prefill(out, q, kcache, vcache, bt, seq_ids, q_pos, desc, scale)
+    expect = ref.paged_attention_ref(q, kcache, vcache, bt, seq_ids, q_pos, scale)
+    assert bf16_close(out, expect, atol=4e-2, rtol=4e-2)
