@@ -222,18 +222,15 @@ class Qwen3MoEModel:
         inv_order = torch.empty_like(order)
         inv_order[order] = torch.arange(P, device=hbuf.device)
         inv_order = inv_order.int().contiguous()
-        desc1 = ops.build_moe_tile_desc(pair_expert, (2 * I) // 64,
-                                        cfg.num_experts, bm=128)
+        desc = ops.moe_build_desc_device(pair_expert, cfg.num_experts, bm=128)
         gateup = torch.empty(P, 2 * I, dtype=torch.bfloat16, device=hbuf.device)
-        ops.moe_grouped_gemm128(gateup, hbuf, layer.w13, pair_token, desc1)
+        ops.moe_grouped_gemm128(gateup, hbuf, layer.w13, pair_token, desc)
         h = torch.empty(P, I, dtype=torch.bfloat16, device=hbuf.device)
         ops.silu_mul(h, gateup)
-        desc2 = ops.build_moe_tile_desc(pair_expert, H // 64,
-                                        cfg.num_experts, bm=128)
         z = torch.empty(P, H, dtype=torch.bfloat16, device=hbuf.device)
         ops.moe_grouped_gemm128(z, h, layer.w2,
                                 torch.arange(P, device=hbuf.device, dtype=torch.int32),
-                                desc2)
+                                desc)
         out_bf = torch.empty(T, H, dtype=torch.bfloat16, device=hbuf.device)
         ops.moe_combine_gather(out_bf, z, topk_w.contiguous(), inv_order)
         return out_bf

@@ -153,6 +153,19 @@ def moe_combine(out: torch.Tensor, z: torch.Tensor, pair_w: torch.Tensor,
     return out
 
 
+def moe_build_desc_device(pair_expert_sorted: torch.Tensor, num_experts: int,
+                          bm: int = 128) -> torch.Tensor:
+    """Sync-free [GMAX,3] (expert,row0,msize) descriptors; GMAX computed from
+    the host-known pair count, empty tiles zero-sized."""
+    P = pair_expert_sorted.numel()
+    gmax = num_experts + (P + bm - 1) // bm
+    counts = torch.bincount(pair_expert_sorted.long(), minlength=num_experts)
+    desc = torch.empty(gmax, 3, dtype=torch.int32,
+                       device=pair_expert_sorted.device)
+    _require().moe_build_desc(desc, counts, bm)
+    return desc
+
+
 def moe_combine_gather(out: torch.Tensor, z: torch.Tensor, topk_w: torch.Tensor,
                        inv_order: torch.Tensor) -> torch.Tensor:
     _require().moe_combine_gather(out, z, topk_w, inv_order)

@@ -330,13 +330,14 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
                                 const short* __restrict__ x,   // [T, H]
                                 const short* __restrict__ w,   // [E, N, H]
                                 const int* __restrict__ pair_token,
-                                const int* __restrict__ tile_desc,  // [G,4]
+                                const int* __restrict__ tile_desc,  // [G,3]
                                 int H, int N) {
   const int g = blockIdx.x;
-  const int e = tile_desc[g * 4 + 0];
-  const int row0 = tile_desc[g * 4 + 1];
-  const int msize = tile_desc[g * 4 + 2];
-  const int n0 = tile_desc[g * 4 + 3] * G2_BN;
+  const int msize = tile_desc[g * 3 + 2];
+  if (msize == 0) return;                 // past the live tile count
+  const int e = tile_desc[g * 3 + 0];
+  const int row0 = tile_desc[g * 3 + 1];
+  const int n0 = blockIdx.y * G2_BN;      // n-tile from the grid
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6;       // 0..7: wave's 16-row m-slice
@@ -407,7 +408,8 @@ void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   const int H = x.size(-1), N = out.size(-1);
   const int G = tile_desc.size(0);
   TORCH_CHECK(H % G2_BK == 0 && N % G2_BN == 0);
-  dim3 grid(G), block(512);
+  TORCH_CHECK(tile_desc.size(1) == 3, "desc is [G,3]; n-tile comes from grid.y");
+  dim3 grid(G, N / G2_BN), block(512);
   hipStream_t s = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(moe_grouped_gemm128_kernel, grid, block, 0, s,
                      (short*)out.data_ptr(), (const short*)x.data_ptr(),
@@ -455,5 +457,40 @@ void moe_combine_gather(torch::Tensor out, torch::Tensor z, torch::Tensor topk_w
   hipLaunchKernelGGL(moe_combine_gather_kernel, grid, block, 0, s,
                      (short*)out.data_ptr(), (const short*)z.data_ptr(),
                      topk_w.data_ptr<float>(), inv_order.data_ptr<int>(), H, K);
+  HIP_CHECK_KERNEL();
+}
+
+// device-side tile-descriptor builder: removes the per-layer host syncs that
+// data-dependent repeat_interleave shapes forced (profiled: ~6 syncs/layer
+// serialized prefill on the CPU). desc[GMAX,3] = (expert, row_start, m_size);
+// unused slots get m_size=0 and the GEMM early-exits. GMAX is host-computed
+// from P alone (≤ E + ceil(P/bm)), so grids stay static.
+__global__ void moe_build_desc_kernel(int* __restrict__ desc,      // [GMAX, 3]
+                                      const long* __restrict__ counts,  // [E]
+                                      int E, int bm, int gmax) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  int g = 0;
+  long row = 0;
+  for (int e = 0; e < E; ++e) {
+    const long c = counts[e];
+    for (long m0 = 0; m0 < c && g < gmax; m0 += bm) {
+      desc[g * 3 + 0] = e;
+      desc[g * 3 + 1] = (int)(row + m0);
+      desc[g * 3 + 2] = (int)min((long)bm, c - m0);
+      ++g;
+    }
+    row += c;
+  }
+  for (; g < gmax; ++g) desc[g * 3 + 2] = 0;
+}
+
+void moe_build_desc(torch::Tensor desc, torch::Tensor counts, int64_t bm) {
+  const int E = counts.numel();
+  const int gmax = desc.size(0);
+  TORCH_CHECK(counts.dtype() == torch::kInt64 && desc.size(1) == 3);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(moe_build_desc_kernel, dim3(1), dim3(64), 0, s,
+                     desc.data_ptr<int>(), counts.data_ptr<long>(), E, (int)bm,
+                     gmax);
   HIP_CHECK_KERNEL();
 }

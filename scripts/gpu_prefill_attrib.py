@@ -23,10 +23,13 @@ tokens = torch.randint(0, cfg.vocab_size, (T,), device=dev)
 seq_ids = torch.full((T,), slot, dtype=torch.int32, device=dev)
 q_pos = torch.arange(T, dtype=torch.int32, device=dev)
 rows = torch.tensor([T - 1], device=dev)
+from room_amd import ops
+qtiles = ops.build_qtile_desc([(0, T)], dev)
 
 def fwd():
     return model.forward(tokens, seq_ids, q_pos, cache.block_table,
-                         cache.kcaches, cache.vcaches, logits_rows=rows)
+                         cache.kcaches, cache.vcaches, logits_rows=rows,
+                         qtile_desc=qtiles)
 
 # warm
 fwd(); torch.cuda.synchronize()

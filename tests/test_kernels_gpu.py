@@ -308,7 +308,7 @@ def test_moe_grouped_gemm128_vs_matmul():
     order = torch.argsort(pair_expert)
     pair_expert = pair_expert[order].contiguous()
     pair_token = order.int().contiguous()
-    tile_desc = ops.build_moe_tile_desc(pair_expert, N // 64, E, bm=128).to(DEV)
+    tile_desc = ops.moe_build_desc_device(pair_expert, E, bm=128)
     out = torch.empty(T, N, dtype=torch.bfloat16, device=DEV)
     ops.moe_grouped_gemm128(out, x, w, pair_token, tile_desc)
     for p in [0, 1, 100, 150, 299]:
