@@ -98,7 +98,13 @@ def main() -> None:
             q.create_entity(db, f"Benchmark note {j}", room_id=room_id,
                             observations=[filler])
 
-    mgr = AgentLoopManager(ldb)
+    from room_amd.memory.vector_store import GpuVectorStore, MemoryService
+    memsvc = MemoryService(ldb, store=GpuVectorStore(
+        capacity=100_000, device="cuda" if use_gpu else "cpu"))
+    with ldb as db:
+        memsvc.store.rebuild_from_db(db)
+        memsvc.index_pending()
+    mgr = AgentLoopManager(ldb, memory=memsvc)
     pad = " ".join(f"obs{i} filler" for i in range(args.prompt_pad // 2))
 
     async def one_step() -> list[float]:

@@ -56,10 +56,12 @@ class AgentLoopManager:
 
     def __init__(self, ldb: LockedDb, bus: EventBus | None = None,
                  embed_fn: Callable[[str], list[float]] | None = None,
+                 memory=None,
                  time_source: Callable[[], float] = time.time):
         self.ldb = ldb
         self.bus = bus or EventBus()
-        self.embed_fn = embed_fn
+        self.memory = memory  # MemoryService (GPU vector store) when available
+        self.embed_fn = embed_fn or (memory.embed if memory is not None else None)
         self.running_loops: dict[int, LoopState] = {}
         self.time = time_source
         agent_tools.register_wake_callbacks(self.trigger_agent, self.wake_room_workers)
@@ -227,7 +229,8 @@ class AgentLoopManager:
         def tool_executor(call: ToolCall) -> str:
             with self.ldb as db:
                 return agent_tools.execute_agent_tool(db, room_id, worker_id, call,
-                                                      embed_fn=self.embed_fn)
+                                                      embed_fn=self.embed_fn,
+                                                      memory=self.memory)
 
         options = AgentExecutionOptions(
             prompt=prompt, model=model,
@@ -331,7 +334,10 @@ class AgentLoopManager:
         # top-5 room memory via hybrid search on the objective context
         query = f"{room.get('goal') or room['name']}"
         vec = self.embed_fn(query) if self.embed_fn else None
-        hits = q.hybrid_search(db, query, vec, limit=MEMORY_TOP_K, room_id=room["id"])
+        semantic = (self.memory.store.search(vec, k=20)
+                    if (self.memory is not None and vec is not None) else None)
+        hits = q.hybrid_search(db, query, vec, limit=MEMORY_TOP_K,
+                               room_id=room["id"], semantic_hits=semantic)
         if hits:
             mem_lines = [f"- {h['name']}: {'; '.join(h['observations'][:2])}"
                          for h in hits]

@@ -111,7 +111,8 @@ def tools_for_role(role: str | None) -> list[ToolDef]:
 
 def execute_agent_tool(db: sqlite3.Connection, room_id: int, worker_id: int,
                        call: ToolCall,
-                       embed_fn: Callable[[str], list[float]] | None = None) -> str:
+                       embed_fn: Callable[[str], list[float]] | None = None,
+                       memory=None) -> str:
     """In-process tool dispatch. Returns a string result fed back to the model."""
     name, args = call.name, call.arguments
     try:
@@ -168,6 +169,20 @@ def execute_agent_tool(db: sqlite3.Connection, room_id: int, worker_id: int,
             return json.dumps({"worker_id": w["id"], "name": w["name"]})
 
         if name == "room_remember":
+            if memory is not None:
+                # embeds + upserts the GPU store and the durable SQLite copy
+                ent = q.get_entity_by_name(db, args["name"], room_id)
+                if ent is None:
+                    ent = q.create_entity(db, args["name"],
+                                          category=args.get("category"),
+                                          room_id=room_id)
+                q.add_observation(db, ent["id"], args["content"],
+                                  source=f"worker:{worker_id}")
+                vec = memory.embed(f"{args['name']} {args['content']}")
+                q.upsert_embedding(db, ent["id"], vec,
+                                   memory.embedder.text_hash(args["content"]))
+                memory.store.upsert(ent["id"], vec)
+                return json.dumps({"entity_id": ent["id"]})
             ent = q.get_entity_by_name(db, args["name"], room_id)
             if ent is None:
                 ent = q.create_entity(db, args["name"], category=args.get("category"),
@@ -181,9 +196,16 @@ def execute_agent_tool(db: sqlite3.Connection, room_id: int, worker_id: int,
             return json.dumps({"entity_id": ent["id"]})
 
         if name == "room_recall":
-            vec = embed_fn(args["query"]) if embed_fn else None
-            hits = q.hybrid_search(db, args["query"], vec,
-                                   limit=args.get("limit", 5), room_id=room_id)
+            if memory is not None:
+                qvec = memory.embed(args["query"])
+                semantic = memory.store.search(qvec, k=20)
+                hits = q.hybrid_search(db, args["query"], qvec,
+                                       limit=args.get("limit", 5), room_id=room_id,
+                                       semantic_hits=semantic)
+            else:
+                vec = embed_fn(args["query"]) if embed_fn else None
+                hits = q.hybrid_search(db, args["query"], vec,
+                                       limit=args.get("limit", 5), room_id=room_id)
             return json.dumps([{"name": h["name"], "score": round(h["score"], 4),
                                 "observations": h["observations"][:3]} for h in hits])
 

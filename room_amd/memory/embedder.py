@@ -1,0 +1,58 @@
+"""384-dim text embedder.
+
+The reference runs Xenova/all-MiniLM-L6-v2 via ONNX on CPU (mean-pool +
+L2-normalize, src/shared/embeddings.ts:48-69). This environment has no
+network to fetch those weights, so the default embedder is a deterministic
+hashed n-gram projection (feature hashing) producing the same 384-dim fp32
+L2-normalized output format and the same blob codec — semantically useful
+(token overlap → cosine similarity), fully self-contained, and identical
+across processes. A learned encoder can be dropped in behind `embed()`
+without touching callers (the GPU store only sees 384-dim vectors).
+"""
+from __future__ import annotations
+
+import hashlib
+import math
+import re
+
+from ..core.constants import EMBEDDING_DIM
+
+_TOKEN_RE = re.compile(r"[a-z0-9]+")
+
+
+def _features(text: str) -> list[str]:
+    toks = _TOKEN_RE.findall(text.lower())
+    feats = list(toks)
+    feats += [f"{a}_{b}" for a, b in zip(toks, toks[1:])]  # bigrams
+    return feats
+
+
+def embed(text: str) -> list[float]:
+    """Deterministic 384-dim L2-normalized embedding."""
+    vec = [0.0] * EMBEDDING_DIM
+    for f in _features(text):
+        h = hashlib.blake2s(f.encode(), digest_size=8).digest()
+        idx = int.from_bytes(h[:4], "little") % EMBEDDING_DIM
+        sign = 1.0 if h[4] & 1 else -1.0
+        vec[idx] += sign
+    norm = math.sqrt(sum(x * x for x in vec))
+    if norm == 0:
+        vec[0] = 1.0
+        return vec
+    return [x / norm for x in vec]
+
+
+def embed_batch(texts: list[str]) -> list[list[float]]:
+    return [embed(t) for t in texts]
+
+
+def cosine_similarity(a: list[float], b: list[float]) -> float:
+    dot = sum(x * y for x, y in zip(a, b))
+    na = math.sqrt(sum(x * x for x in a)) or 1.0
+    nb = math.sqrt(sum(x * x for x in b)) or 1.0
+    return dot / (na * nb)
+
+
+def text_hash(text: str) -> str:
+    """16-hex content hash (same shape as the reference's, embeddings.ts:124-126)."""
+    return hashlib.sha256(text.encode()).hexdigest()[:16]
