@@ -1,0 +1,817 @@
+"""HTTP + WebSocket API server (reference surface: src/server/index.ts 1020 LoC
++ 19 route modules + ws.ts).
+
+FastAPI/uvicorn replaces the reference's raw node:http server. Preserved
+contract: Bearer dual-token auth with member RBAC, /api/auth/handshake
+(localhost-only), the REST resource paths, webhook pass-through before auth
+with a 30/min per-token limit, WS channel subscribe protocol
+{type, channel, data, timestamp}, and the event-bus fan-out.
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+from collections import defaultdict, deque
+from typing import Any, Optional
+
+from fastapi import (Body, Depends, FastAPI, HTTPException, Query, Request,
+                     WebSocket, WebSocketDisconnect)
+from fastapi.responses import HTMLResponse, JSONResponse
+
+from ..core import goals as goals_mod
+from ..core import quorum as quorum_mod
+from ..core import room as room_mod
+from ..core import self_mod, skills as skills_mod, wallet as wallet_mod
+from ..core.constants import WEBHOOK_RATE_LIMIT_PER_MIN
+from ..core.events import EventBus
+from ..core.secret_store import decrypt_secret, encrypt_secret
+from ..db import LockedDb
+from ..db import queries as q
+from .auth import ROLE_MEMBER, AuthManager, member_can_write
+
+
+class ServerState:
+    def __init__(self, ldb: LockedDb, auth: AuthManager, bus: EventBus,
+                 loop_mgr=None, runner=None, memory=None, runtime=None):
+        self.ldb = ldb
+        self.auth = auth
+        self.bus = bus
+        self.loop_mgr = loop_mgr
+        self.runner = runner
+        self.memory = memory
+        self.runtime = runtime
+        self.started_at = time.time()
+        self.webhook_hits: dict[str, deque] = defaultdict(deque)
+
+
+def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
+               auth: AuthManager | None = None,
+               bus: EventBus | None = None) -> FastAPI:
+    bus = bus or (loop_mgr.bus if loop_mgr else EventBus())
+    auth = auth or AuthManager(skip_token_file=True)
+    state = ServerState(ldb, auth, bus, loop_mgr, runner, memory)
+
+    app = FastAPI(title="room_amd", version="0.1.0")
+    app.state.ctx = state
+
+    # ---------------------------------------------------------------- auth
+
+    def get_role(request: Request) -> str:
+        hdr = request.headers.get("authorization", "")
+        token = hdr[7:] if hdr.lower().startswith("bearer ") else None
+        role = auth.role_for(token)
+        if role is None:
+            raise HTTPException(401, "invalid or missing token")
+        return role
+
+    def require_write(request: Request, role: str) -> None:
+        if role == ROLE_MEMBER:
+            template = request.scope.get("route").path if request.scope.get("route") else ""
+            template = (template.replace("{room_id}", "{room_id}")
+                        .replace("{decision_id}", "{decision_id}"))
+            if not member_can_write(request.method, template):
+                raise HTTPException(403, "member tokens are read-only here")
+
+    def rw(request: Request) -> str:
+        role = get_role(request)
+        if request.method not in ("GET", "HEAD", "OPTIONS"):
+            require_write(request, role)
+        return role
+
+    # --------------------------------------------------------- auth routes
+
+    @app.post("/api/auth/handshake")
+    async def handshake(request: Request):
+        # localhost-only (reference index.ts:502-522)
+        host = request.client.host if request.client else ""
+        if host not in ("127.0.0.1", "::1", "localhost", "testclient"):
+            raise HTTPException(403, "handshake is localhost-only")
+        token = auth.issue_user_token()
+        return {"token": token, "role": "user"}
+
+    @app.get("/api/auth/verify")
+    async def verify(role: str = Depends(get_role)):
+        return {"ok": True, "role": role}
+
+    # ------------------------------------------------------------ webhooks
+    # pass-through BEFORE auth (index.ts:601-608), 30/min per token
+
+    def _webhook_limit(token: str) -> None:
+        dq = state.webhook_hits[token]
+        now = time.time()
+        while dq and dq[0] < now - 60:
+            dq.popleft()
+        if len(dq) >= WEBHOOK_RATE_LIMIT_PER_MIN:
+            raise HTTPException(429, "webhook rate limit")
+        dq.append(now)
+
+    @app.post("/api/hooks/task/{token}")
+    async def hook_task(token: str, payload: dict = Body(default={})):
+        _webhook_limit(token)
+        with ldb as db:
+            task = q.get_task_by_webhook_token(db, token)
+        if task is None:
+            raise HTTPException(404, "unknown webhook token")
+        if runner is not None:
+            asyncio.create_task(runner.execute_task(task["id"]))
+        return {"queued": True, "task_id": task["id"]}
+
+    @app.post("/api/hooks/queen/{token}")
+    async def hook_queen(token: str, payload: dict = Body(default={})):
+        _webhook_limit(token)
+        with ldb as db:
+            room = q.get_room_by_webhook_token(db, token)
+            if room is None:
+                raise HTTPException(404, "unknown webhook token")
+            if room["status"] != "active":
+                raise HTTPException(409, f"room is {room['status']}")
+            message = payload.get("message") or json.dumps(payload)[:2000]
+            esc = q.create_escalation(db, room["id"], message)
+        if loop_mgr is not None and room["queen_worker_id"]:
+            loop_mgr.trigger_agent(room["queen_worker_id"])
+        bus.emit(f"room:{room['id']}", "escalation", {"id": esc["id"]})
+        return {"escalation_id": esc["id"]}
+
+    # --------------------------------------------------------------- rooms
+
+    @app.get("/api/rooms")
+    async def list_rooms(role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_rooms(db)
+
+    @app.post("/api/rooms")
+    async def create_room(payload: dict = Body(...), role: str = Depends(rw)):
+        with ldb as db:
+            room = room_mod.create_room(
+                db, payload["name"], goal=payload.get("goal"),
+                worker_model=payload.get("worker_model", "qwen3-coder-30b"),
+                queen_cycle_gap_ms=payload.get("queen_cycle_gap_ms"))
+        bus.emit("rooms", "room_created", {"id": room["id"]})
+        return room
+
+    @app.get("/api/rooms/{room_id}")
+    async def get_room(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            room = q.get_room(db, room_id)
+        if room is None:
+            raise HTTPException(404)
+        return room
+
+    @app.patch("/api/rooms/{room_id}")
+    async def update_room(room_id: int, payload: dict = Body(...),
+                          role: str = Depends(rw)):
+        with ldb as db:
+            room = q.update_room(db, room_id, **{
+                k: v for k, v in payload.items()
+                if k in ("name", "goal", "autonomy_mode", "max_concurrent_tasks",
+                         "worker_model", "queen_cycle_gap_ms", "queen_max_turns",
+                         "queen_quiet_from", "queen_quiet_until", "visibility",
+                         "queen_nickname")})
+        if room is None:
+            raise HTTPException(404)
+        return room
+
+    @app.delete("/api/rooms/{room_id}")
+    async def delete_room(room_id: int, role: str = Depends(rw)):
+        if loop_mgr is not None:
+            with ldb as db:
+                for w in q.list_room_workers(db, room_id):
+                    loop_mgr.stop_agent(w["id"])
+        with ldb as db:
+            room_mod.delete_room(db, room_id)
+        return {"deleted": True}
+
+    @app.get("/api/rooms/{room_id}/status")
+    async def room_status(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            try:
+                return room_mod.get_room_status(db, room_id)
+            except ValueError:
+                raise HTTPException(404)
+
+    @app.post("/api/rooms/{room_id}/start")
+    async def start_room(room_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            room = room_mod.resume_room(db, room_id)
+        if loop_mgr is not None and room["queen_worker_id"]:
+            await loop_mgr.start_agent_loop(room_id, room["queen_worker_id"])
+        bus.emit(f"room:{room_id}", "room_started", {})
+        return room
+
+    @app.post("/api/rooms/{room_id}/pause")
+    async def pause_room(room_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            room = room_mod.pause_room(db, room_id)
+        return room
+
+    @app.post("/api/rooms/{room_id}/stop")
+    async def stop_room(room_id: int, role: str = Depends(rw)):
+        if loop_mgr is not None:
+            with ldb as db:
+                workers = q.list_room_workers(db, room_id)
+            for w in workers:
+                loop_mgr.stop_agent(w["id"])
+        with ldb as db:
+            return room_mod.stop_room(db, room_id)
+
+    @app.post("/api/rooms/{room_id}/restart")
+    async def restart_room(room_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            return room_mod.restart_room(db, room_id)
+
+    @app.get("/api/rooms/{room_id}/activity")
+    async def room_activity(room_id: int, limit: int = 50,
+                            role: str = Depends(get_role)):
+        with ldb as db:
+            return q.get_room_activity(db, room_id, limit=limit)
+
+    @app.get("/api/rooms/{room_id}/cycles")
+    async def room_cycles(room_id: int, limit: int = 50,
+                          role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_room_cycles(db, room_id, limit=limit)
+
+    @app.get("/api/cycles/{cycle_id}/logs")
+    async def cycle_logs(cycle_id: int, after_seq: int = -1,
+                         role: str = Depends(get_role)):
+        with ldb as db:
+            return q.get_cycle_logs(db, cycle_id, after_seq=after_seq)
+
+    # ------------------------------------------------------------- workers
+
+    @app.get("/api/rooms/{room_id}/workers")
+    async def list_workers(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_room_workers(db, room_id)
+
+    @app.post("/api/rooms/{room_id}/workers")
+    async def create_worker(room_id: int, payload: dict = Body(...),
+                            role: str = Depends(rw)):
+        from ..core.constants import WORKER_ROLE_PRESETS
+        preset = WORKER_ROLE_PRESETS.get(payload.get("role", ""), {})
+        with ldb as db:
+            w = q.create_worker(
+                db, payload["name"],
+                payload.get("system_prompt") or preset.get("systemPromptPrefix", ""),
+                role=payload.get("role"), room_id=room_id,
+                model=payload.get("model"),
+                cycle_gap_ms=payload.get("cycle_gap_ms") or preset.get("cycleGapMs"),
+                max_turns=payload.get("max_turns") or preset.get("maxTurns"))
+        bus.emit(f"room:{room_id}", "worker_created", {"id": w["id"]})
+        return w
+
+    @app.get("/api/workers/{worker_id}")
+    async def get_worker(worker_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            w = q.get_worker(db, worker_id)
+        if w is None:
+            raise HTTPException(404)
+        return w
+
+    @app.patch("/api/workers/{worker_id}")
+    async def update_worker(worker_id: int, payload: dict = Body(...),
+                            role: str = Depends(rw)):
+        with ldb as db:
+            w = q.update_worker(db, worker_id, **{
+                k: v for k, v in payload.items()
+                if k in ("name", "role", "system_prompt", "description", "model",
+                         "cycle_gap_ms", "max_turns", "wip")})
+        if w is None:
+            raise HTTPException(404)
+        return w
+
+    @app.delete("/api/workers/{worker_id}")
+    async def delete_worker(worker_id: int, role: str = Depends(rw)):
+        if loop_mgr is not None:
+            loop_mgr.stop_agent(worker_id)
+        with ldb as db:
+            q.delete_worker(db, worker_id)
+        return {"deleted": True}
+
+    @app.post("/api/workers/{worker_id}/start")
+    async def start_worker(worker_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            w = q.get_worker(db, worker_id)
+        if w is None:
+            raise HTTPException(404)
+        if loop_mgr is not None and w["room_id"]:
+            await loop_mgr.start_agent_loop(w["room_id"], worker_id)
+            loop_mgr.trigger_agent(worker_id)  # the MCP nudge semantics
+        return {"started": True}
+
+    @app.post("/api/workers/{worker_id}/stop")
+    async def stop_worker(worker_id: int, role: str = Depends(rw)):
+        if loop_mgr is not None:
+            loop_mgr.stop_agent(worker_id)
+        return {"stopped": True}
+
+    # --------------------------------------------------------------- goals
+
+    @app.get("/api/rooms/{room_id}/goals")
+    async def list_goals(room_id: int, status: Optional[str] = None,
+                         tree: bool = False, role: str = Depends(get_role)):
+        with ldb as db:
+            if tree:
+                return goals_mod.get_goal_tree(db, room_id)
+            return q.list_room_goals(db, room_id, status=status)
+
+    @app.post("/api/rooms/{room_id}/goals")
+    async def create_goal(room_id: int, payload: dict = Body(...),
+                          role: str = Depends(rw)):
+        with ldb as db:
+            return q.create_goal(db, room_id, payload["description"],
+                                 parent_goal_id=payload.get("parent_goal_id"),
+                                 assigned_worker_id=payload.get("assigned_worker_id"))
+
+    @app.patch("/api/goals/{goal_id}")
+    async def update_goal(goal_id: int, payload: dict = Body(...),
+                          role: str = Depends(rw)):
+        with ldb as db:
+            if payload.get("status") == "completed":
+                return goals_mod.complete_goal(db, goal_id,
+                                               observation=payload.get("observation"))
+            if payload.get("status") == "abandoned":
+                return goals_mod.abandon_goal(db, goal_id,
+                                              reason=payload.get("reason"))
+            if "progress" in payload:
+                return goals_mod.update_goal_progress(
+                    db, goal_id, payload["progress"],
+                    observation=payload.get("observation"))
+            g = q.update_goal(db, goal_id, **{
+                k: v for k, v in payload.items()
+                if k in ("description", "assigned_worker_id", "status")})
+        if g is None:
+            raise HTTPException(404)
+        return g
+
+    # ----------------------------------------------------------- decisions
+
+    @app.get("/api/rooms/{room_id}/decisions")
+    async def list_decisions(room_id: int, status: Optional[str] = None,
+                             role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_room_decisions(db, room_id, status=status)
+
+    @app.post("/api/rooms/{room_id}/decisions")
+    async def announce(room_id: int, payload: dict = Body(...),
+                       role: str = Depends(rw)):
+        with ldb as db:
+            d = quorum_mod.announce(db, room_id, payload.get("proposer_id"),
+                                    payload["proposal"],
+                                    payload.get("decision_type", "low_impact"),
+                                    delay_minutes=payload.get("delay_minutes"))
+        bus.emit(f"room:{room_id}", "decision", {"id": d["id"]})
+        return d
+
+    @app.post("/api/decisions/{decision_id}/object")
+    async def object_decision(decision_id: int, payload: dict = Body(...),
+                              role: str = Depends(rw)):
+        with ldb as db:
+            try:
+                return quorum_mod.object_to(db, decision_id,
+                                            payload.get("worker_id", 0),
+                                            payload.get("reason", ""))
+            except ValueError as e:
+                raise HTTPException(409, str(e))
+
+    @app.post("/api/decisions/{decision_id}/vote")
+    async def vote(decision_id: int, payload: dict = Body(...),
+                   role: str = Depends(rw)):
+        with ldb as db:
+            try:
+                return quorum_mod.vote(db, decision_id, payload["worker_id"],
+                                       payload["vote"], payload.get("reasoning"))
+            except ValueError as e:
+                raise HTTPException(409, str(e))
+
+    @app.post("/api/decisions/{decision_id}/keeper-vote")
+    async def keeper_vote(decision_id: int, payload: dict = Body(...),
+                          role: str = Depends(rw)):
+        with ldb as db:
+            try:
+                return quorum_mod.keeper_vote(db, decision_id, payload["vote"])
+            except ValueError as e:
+                raise HTTPException(409, str(e))
+
+    @app.get("/api/decisions/{decision_id}/votes")
+    async def decision_votes(decision_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            return q.get_votes(db, decision_id)
+
+    # ---------------------------------------------------------------- tasks
+
+    @app.get("/api/tasks")
+    async def list_tasks(room_id: Optional[int] = None,
+                         status: Optional[str] = None,
+                         role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_tasks(db, room_id=room_id, status=status)
+
+    @app.post("/api/tasks")
+    async def create_task(payload: dict = Body(...), role: str = Depends(rw)):
+        from ..core.cron import validate_cron
+        import secrets as _secrets
+        if payload.get("trigger_type", "cron") == "cron" \
+                and payload.get("cron_expression") \
+                and not validate_cron(payload["cron_expression"]):
+            raise HTTPException(422, "invalid cron expression")
+        webhook_token = (_secrets.token_hex(16)
+                         if payload.get("trigger_type") == "webhook" else None)
+        with ldb as db:
+            return q.create_task(
+                db, payload["name"], payload["prompt"],
+                trigger_type=payload.get("trigger_type", "cron"),
+                cron_expression=payload.get("cron_expression"),
+                scheduled_at=payload.get("scheduled_at"),
+                room_id=payload.get("room_id"), worker_id=payload.get("worker_id"),
+                session_continuity=payload.get("session_continuity", False),
+                max_runs=payload.get("max_runs"),
+                description=payload.get("description"),
+                webhook_token=webhook_token,
+                timeout_minutes=payload.get("timeout_minutes"),
+                max_turns=payload.get("max_turns"))
+
+    @app.get("/api/tasks/{task_id}")
+    async def get_task(task_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            t = q.get_task(db, task_id)
+        if t is None:
+            raise HTTPException(404)
+        return t
+
+    @app.patch("/api/tasks/{task_id}")
+    async def update_task(task_id: int, payload: dict = Body(...),
+                          role: str = Depends(rw)):
+        with ldb as db:
+            t = q.update_task(db, task_id, **{
+                k: v for k, v in payload.items()
+                if k in ("name", "description", "prompt", "cron_expression",
+                         "status", "max_runs", "worker_id", "session_continuity",
+                         "timeout_minutes", "max_turns", "scheduled_at")})
+        if t is None:
+            raise HTTPException(404)
+        return t
+
+    @app.delete("/api/tasks/{task_id}")
+    async def delete_task(task_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            q.delete_task(db, task_id)
+        return {"deleted": True}
+
+    @app.post("/api/tasks/{task_id}/run")
+    async def run_task(task_id: int, role: str = Depends(rw)):
+        if runner is None:
+            raise HTTPException(503, "task runner not available")
+        asyncio.create_task(runner.execute_task(task_id))
+        return {"queued": True}
+
+    @app.get("/api/tasks/{task_id}/runs")
+    async def task_runs(task_id: int, limit: int = 20,
+                        role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_task_runs(db, task_id, limit=limit)
+
+    @app.get("/api/runs/{run_id}")
+    async def get_run(run_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            r = q.get_task_run(db, run_id)
+        if r is None:
+            raise HTTPException(404)
+        return r
+
+    @app.get("/api/runs/{run_id}/logs")
+    async def run_logs(run_id: int, after_seq: int = -1,
+                       role: str = Depends(get_role)):
+        with ldb as db:
+            return q.get_console_logs(db, run_id, after_seq=after_seq)
+
+    # --------------------------------------------------------------- memory
+
+    @app.get("/api/memory/search")
+    async def memory_search(query: str, room_id: Optional[int] = None,
+                            limit: int = 5, role: str = Depends(get_role)):
+        if memory is not None:
+            return memory.recall(room_id, query, limit=limit)
+        with ldb as db:
+            return q.hybrid_search(db, query, None, limit=limit, room_id=room_id)
+
+    @app.post("/api/memory/entities")
+    async def remember(payload: dict = Body(...), role: str = Depends(rw)):
+        if memory is not None:
+            eid = memory.remember(payload.get("room_id"), payload["name"],
+                                  payload["content"],
+                                  category=payload.get("category"))
+            return {"entity_id": eid}
+        with ldb as db:
+            ent = q.create_entity(db, payload["name"],
+                                  category=payload.get("category"),
+                                  room_id=payload.get("room_id"),
+                                  observations=[payload["content"]])
+        return {"entity_id": ent["id"]}
+
+    @app.get("/api/memory/entities/{entity_id}")
+    async def get_entity(entity_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            e = q.get_entity(db, entity_id)
+            if e is None:
+                raise HTTPException(404)
+            e = dict(e)
+            e["observations"] = q.get_observations(db, entity_id)
+        return e
+
+    @app.delete("/api/memory/entities/{entity_id}")
+    async def delete_entity(entity_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            q.delete_entity(db, entity_id)
+        if memory is not None:
+            memory.store.remove(entity_id)
+        return {"deleted": True}
+
+    # --------------------------------------------------------------- skills
+
+    @app.get("/api/rooms/{room_id}/skills")
+    async def list_skills(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_room_skills(db, room_id)
+
+    @app.post("/api/rooms/{room_id}/skills")
+    async def create_skill(room_id: int, payload: dict = Body(...),
+                           role: str = Depends(rw)):
+        with ldb as db:
+            return q.create_skill(db, room_id, payload["name"], payload["content"],
+                                  activation_context=payload.get("activation_context"),
+                                  auto_activate=payload.get("auto_activate", False))
+
+    @app.patch("/api/skills/{skill_id}")
+    async def update_skill(skill_id: int, payload: dict = Body(...),
+                           role: str = Depends(rw)):
+        with ldb as db:
+            s = q.update_skill(db, skill_id, content=payload.get("content"),
+                               activation_context=payload.get("activation_context"),
+                               auto_activate=payload.get("auto_activate"))
+        if s is None:
+            raise HTTPException(404)
+        return s
+
+    @app.delete("/api/skills/{skill_id}")
+    async def delete_skill(skill_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            q.delete_skill(db, skill_id)
+        return {"deleted": True}
+
+    # ---------------------------------------------------------- escalations
+
+    @app.get("/api/rooms/{room_id}/escalations")
+    async def list_escalations(room_id: int, status: Optional[str] = None,
+                               role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_escalations(db, room_id, status=status)
+
+    @app.post("/api/rooms/{room_id}/escalations")
+    async def create_escalation(room_id: int, payload: dict = Body(...),
+                                role: str = Depends(rw)):
+        with ldb as db:
+            e = q.create_escalation(db, room_id, payload["question"],
+                                    from_agent_id=payload.get("from_agent_id"))
+        bus.emit(f"room:{room_id}", "escalation", {"id": e["id"]})
+        return e
+
+    @app.post("/api/escalations/{escalation_id}/answer")
+    async def answer_escalation(escalation_id: int, payload: dict = Body(...),
+                                role: str = Depends(rw)):
+        with ldb as db:
+            q.answer_escalation(db, escalation_id, payload["answer"])
+            row = db.execute("SELECT * FROM escalations WHERE id = ?",
+                             (escalation_id,)).fetchone()
+        if row and loop_mgr is not None:
+            with ldb as db:
+                room = q.get_room(db, row["room_id"])
+            if room and room.get("queen_worker_id"):
+                loop_mgr.trigger_agent(room["queen_worker_id"])
+        return row or {}
+
+    # ------------------------------------------------------------- self-mod
+
+    @app.get("/api/rooms/{room_id}/self-mod")
+    async def self_mod_audit(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_self_mod_audit(db, room_id)
+
+    @app.post("/api/self-mod/{audit_id}/revert")
+    async def self_mod_revert(audit_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            try:
+                return self_mod.revert_modification(db, audit_id)
+            except ValueError as e:
+                raise HTTPException(409, str(e))
+
+    # --------------------------------------------------------------- wallet
+
+    @app.get("/api/rooms/{room_id}/wallet")
+    async def get_wallet(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            w = q.get_room_wallet(db, room_id)
+        if w is None:
+            raise HTTPException(404)
+        return {k: v for k, v in w.items() if k != "private_key_encrypted"}
+
+    @app.get("/api/rooms/{room_id}/wallet/transactions")
+    async def wallet_txs(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            w = q.get_room_wallet(db, room_id)
+            if w is None:
+                raise HTTPException(404)
+            return q.list_wallet_txs(db, w["id"])
+
+    @app.post("/api/rooms/{room_id}/wallet/send")
+    async def wallet_send(room_id: int, payload: dict = Body(...),
+                          role: str = Depends(rw)):
+        with ldb as db:
+            try:
+                return wallet_mod.send_token(
+                    db, room_id, payload["to_address"], payload["amount"],
+                    chain=payload.get("chain", "base"),
+                    token=payload.get("token", "usdc"),
+                    description=payload.get("description"))
+            except ValueError as e:
+                raise HTTPException(422, str(e))
+
+    # ---------------------------------------------------------- credentials
+
+    @app.get("/api/rooms/{room_id}/credentials")
+    async def list_credentials(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_credentials(db, room_id)  # values never exposed
+
+    @app.post("/api/rooms/{room_id}/credentials")
+    async def set_credential(room_id: int, payload: dict = Body(...),
+                             role: str = Depends(rw)):
+        with ldb as db:
+            row = q.set_credential(db, room_id, payload["name"],
+                                   encrypt_secret(payload["value"]),
+                                   cred_type=payload.get("type", "other"))
+        return {k: v for k, v in row.items() if k != "value_encrypted"}
+
+    @app.delete("/api/rooms/{room_id}/credentials/{name}")
+    async def delete_credential(room_id: int, name: str, role: str = Depends(rw)):
+        with ldb as db:
+            q.delete_credential(db, room_id, name)
+        return {"deleted": True}
+
+    # ------------------------------------------------------- room messages
+
+    @app.get("/api/rooms/{room_id}/messages")
+    async def list_messages(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_room_messages(db, room_id)
+
+    @app.post("/api/rooms/{room_id}/messages")
+    async def send_message(room_id: int, payload: dict = Body(...),
+                           role: str = Depends(rw)):
+        with ldb as db:
+            m = q.create_room_message(db, room_id,
+                                      payload.get("direction", "inbound"),
+                                      payload.get("subject", ""),
+                                      payload["body"],
+                                      from_room_id=payload.get("from_room_id"))
+        bus.emit(f"room:{room_id}", "message", {"id": m["id"]})
+        return m
+
+    # ----------------------------------------------------------- chat/clerk
+
+    @app.get("/api/rooms/{room_id}/chat")
+    async def room_chat_history(room_id: int, limit: int = 50,
+                                role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_chat_messages(db, room_id, limit=limit)
+
+    @app.post("/api/rooms/{room_id}/chat")
+    async def room_chat(room_id: int, payload: dict = Body(...),
+                        role: str = Depends(rw)):
+        with ldb as db:
+            q.add_chat_message(db, room_id, "user", payload["content"])
+            e = q.create_escalation(db, room_id, payload["content"])
+        if loop_mgr is not None:
+            with ldb as db:
+                room = q.get_room(db, room_id)
+            if room and room.get("queen_worker_id"):
+                loop_mgr.trigger_agent(room["queen_worker_id"])
+        return {"escalation_id": e["id"]}
+
+    @app.get("/api/clerk/messages")
+    async def clerk_messages(limit: int = 50, role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_clerk_messages(db, limit=limit)
+
+    @app.post("/api/clerk/chat")
+    async def clerk_chat(payload: dict = Body(...), role: str = Depends(rw)):
+        from ..core.clerk import clerk_chat as do_chat
+        reply = await asyncio.to_thread(do_chat, ldb, payload["content"],
+                                        memory)
+        bus.emit("clerk", "message", {"content": reply})
+        return {"reply": reply}
+
+    # ------------------------------------------------------------- settings
+
+    @app.get("/api/settings")
+    async def list_settings(role: str = Depends(get_role)):
+        with ldb as db:
+            return q.list_settings(db)
+
+    @app.put("/api/settings/{key}")
+    async def set_setting(key: str, payload: dict = Body(...),
+                          role: str = Depends(rw)):
+        with ldb as db:
+            q.set_setting(db, key, str(payload.get("value", "")))
+        return {"key": key, "value": payload.get("value")}
+
+    # --------------------------------------------------------------- status
+
+    @app.get("/api/status")
+    async def status(role: str = Depends(get_role)):
+        with ldb as db:
+            rooms = q.list_rooms(db)
+        out = {
+            "version": "0.1.0",
+            "uptime_s": int(time.time() - state.started_at),
+            "rooms": len(rooms),
+            "active_rooms": sum(1 for r in rooms if r["status"] == "active"),
+            "running_loops": len(loop_mgr.running_loops) if loop_mgr else 0,
+        }
+        try:
+            import torch
+            if torch.cuda.is_available():
+                free_b, total_b = torch.cuda.mem_get_info()
+                out["gpu"] = {"name": torch.cuda.get_device_name(0),
+                              "hbm_free_gb": round(free_b / 2**30, 1),
+                              "hbm_total_gb": round(total_b / 2**30, 1)}
+        except Exception:
+            pass
+        return out
+
+    @app.get("/api/feed")
+    async def public_feed(limit: int = 50, role: str = Depends(get_role)):
+        from ..core.public_feed import get_public_feed
+        with ldb as db:
+            return get_public_feed(db, limit=limit)
+
+    # -------------------------------------------------------------- UI stub
+
+    @app.get("/", response_class=HTMLResponse)
+    async def index():
+        return ("<html><head><title>room_amd</title></head><body>"
+                "<h1>room_amd — MI355X-native agent swarm runtime</h1>"
+                "<p>API at /api/* (Bearer auth), WebSocket at /ws?token=...</p>"
+                "</body></html>")
+
+    # ------------------------------------------------------------ websocket
+
+    @app.websocket("/ws")
+    async def ws_endpoint(ws: WebSocket, token: str = Query(default="")):
+        if auth.role_for(token) is None:
+            await ws.close(code=4401)
+            return
+        await ws.accept()
+        subscribed: set[str] = set()
+        queue: asyncio.Queue = asyncio.Queue(maxsize=1000)
+        loop = asyncio.get_event_loop()
+
+        def on_event(channel: str, event: dict) -> None:
+            if channel in subscribed or "*" in subscribed:
+                try:
+                    loop.call_soon_threadsafe(queue.put_nowait, event)
+                except RuntimeError:
+                    pass
+
+        unsub = bus.on("*", lambda ch, ev: on_event(ch, ev))
+
+        async def sender():
+            while True:
+                try:
+                    event = await asyncio.wait_for(queue.get(), timeout=30)
+                    await ws.send_json(event)
+                except asyncio.TimeoutError:
+                    await ws.send_json({"type": "ping", "channel": "system",
+                                        "data": None,
+                                        "timestamp": int(time.time() * 1000)})
+
+        send_task = asyncio.create_task(sender())
+        try:
+            while True:
+                msg = await ws.receive_json()
+                if msg.get("type") == "subscribe":
+                    subscribed.add(msg.get("channel", ""))
+                elif msg.get("type") == "unsubscribe":
+                    subscribed.discard(msg.get("channel", ""))
+                elif msg.get("type") == "ping":
+                    await ws.send_json({"type": "pong", "channel": "system",
+                                        "data": None,
+                                        "timestamp": int(time.time() * 1000)})
+        except WebSocketDisconnect:
+            pass
+        finally:
+            send_task.cancel()
+            unsub()
+
+    return app

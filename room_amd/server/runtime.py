@@ -1,0 +1,131 @@
+"""Server runtime loops (reference: src/server/runtime.ts).
+
+asyncio tasks: cron-triggered task execution (15s refresh), due-once tasks,
+maintenance (stale-run cleanup, 60s), queen inbox poll (2.5s) that wakes a
+queen when pending escalations appear, and background embedding indexing.
+"""
+from __future__ import annotations
+
+import asyncio
+from datetime import datetime
+from typing import Optional
+
+from ..core.cron import CronExpression
+from ..core.tasks import TaskRunner
+from ..db import LockedDb
+from ..db import queries as q
+
+CRON_REFRESH_S = float(__import__("os").environ.get("ROOMAMD_CRON_REFRESH_S", "15"))
+MAINTENANCE_S = 60.0
+INBOX_POLL_S = 2.5
+INDEXER_S = 10.0
+
+
+class ServerRuntime:
+    def __init__(self, ldb: LockedDb, runner: TaskRunner, loop_mgr=None,
+                 memory=None, bus=None):
+        self.ldb = ldb
+        self.runner = runner
+        self.loop_mgr = loop_mgr
+        self.memory = memory
+        self.bus = bus
+        self._tasks: list[asyncio.Task] = []
+        self._stop = asyncio.Event()
+        self._last_cron_minute: dict[int, str] = {}
+
+    async def start(self) -> None:
+        with self.ldb as db:
+            n = q.cleanup_stale_cycles(db)
+            n += q.cleanup_all_running_runs(db)
+        self._tasks = [
+            asyncio.create_task(self._cron_loop()),
+            asyncio.create_task(self._maintenance_loop()),
+            asyncio.create_task(self._inbox_loop()),
+            asyncio.create_task(self._indexer_loop()),
+        ]
+
+    async def stop(self) -> None:
+        self._stop.set()
+        for t in self._tasks:
+            t.cancel()
+        for t in self._tasks:
+            try:
+                await t
+            except (asyncio.CancelledError, Exception):
+                pass
+        self._tasks = []
+
+    async def _sleep(self, s: float) -> bool:
+        try:
+            await asyncio.wait_for(self._stop.wait(), timeout=s)
+            return True
+        except asyncio.TimeoutError:
+            return False
+
+    # --- cron + due-once tasks
+    async def _cron_loop(self) -> None:
+        while not self._stop.is_set():
+            try:
+                now = datetime.now()
+                minute_key = now.strftime("%Y-%m-%d %H:%M")
+                with self.ldb as db:
+                    crons = [t for t in q.list_tasks(db, status="active")
+                             if t["trigger_type"] == "cron" and t["cron_expression"]]
+                    due_once = q.get_due_once_tasks(db)
+                for t in crons:
+                    if self._last_cron_minute.get(t["id"]) == minute_key:
+                        continue
+                    try:
+                        if CronExpression(t["cron_expression"]).matches(now):
+                            self._last_cron_minute[t["id"]] = minute_key
+                            asyncio.create_task(self.runner.execute_task(t["id"]))
+                    except ValueError:
+                        pass
+                for t in due_once:
+                    if not self.runner.is_task_running(t["id"]):
+                        asyncio.create_task(self.runner.execute_task(t["id"]))
+            except Exception:
+                pass
+            if await self._sleep(CRON_REFRESH_S):
+                return
+
+    # --- maintenance
+    async def _maintenance_loop(self) -> None:
+        while not self._stop.is_set():
+            try:
+                with self.ldb as db:
+                    q.cleanup_stale_runs(db)
+            except Exception:
+                pass
+            if await self._sleep(MAINTENANCE_S):
+                return
+
+    # --- queen inbox poll: wake queens with pending escalations/messages
+    async def _inbox_loop(self) -> None:
+        while not self._stop.is_set():
+            try:
+                if self.loop_mgr is not None:
+                    with self.ldb as db:
+                        rooms = q.list_rooms(db)
+                        for room in rooms:
+                            if room["status"] != "active" or not room["queen_worker_id"]:
+                                continue
+                            pend = q.list_escalations(db, room["id"], status="answered")
+                            unread = q.get_unread_room_messages(db, room["id"])
+                            if pend or unread:
+                                self.loop_mgr.trigger_agent(room["queen_worker_id"])
+            except Exception:
+                pass
+            if await self._sleep(INBOX_POLL_S):
+                return
+
+    # --- background embedding indexer
+    async def _indexer_loop(self) -> None:
+        while not self._stop.is_set():
+            try:
+                if self.memory is not None:
+                    await asyncio.to_thread(self.memory.index_pending)
+            except Exception:
+                pass
+            if await self._sleep(INDEXER_S):
+                return
