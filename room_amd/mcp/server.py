@@ -1,0 +1,516 @@
+"""MCP stdio server (reference: src/mcp/server.ts + 17 tool modules, 76 tools).
+
+Implements the Model Context Protocol over stdio as JSON-RPC 2.0
+(initialize / tools/list / tools/call). Tools share the same SQLite file as
+the HTTP server (ROOMAMD_DB_PATH env, reference QUOROOM_DB_PATH); agent wakes
+cross the process boundary via an HTTP nudge using the token in
+~/.roomamd/api.token (reference mcp/nudge.ts:14-43).
+"""
+from __future__ import annotations
+
+import json
+import sys
+from typing import Any, Callable
+
+from ..core import goals as goals_mod
+from ..core import quorum as quorum_mod
+from ..core import room as room_mod
+from ..core import self_mod, skills as skills_mod, wallet as wallet_mod
+from ..db import LockedDb, connect
+from ..db import queries as q
+
+PROTOCOL_VERSION = "2024-11-05"
+SERVER_INFO = {"name": "room-amd", "version": "0.1.0"}
+
+
+def nudge_worker(worker_id: int) -> bool:
+    """Cross-process wake: POST /api/workers/{id}/start on the local server."""
+    import urllib.request
+
+    from ..server.auth import data_dir
+    try:
+        port = int((data_dir() / "api.port").read_text().strip())
+        token = (data_dir() / "api.token").read_text().strip()
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{port}/api/workers/{worker_id}/start",
+            method="POST", headers={"Authorization": f"Bearer {token}"},
+            data=b"{}")
+        urllib.request.urlopen(req, timeout=3)
+        return True
+    except Exception:
+        return False  # fail-silent like the reference nudge
+
+
+class McpServer:
+    def __init__(self, ldb: LockedDb, memory=None, nudge=nudge_worker):
+        self.ldb = ldb
+        self.memory = memory
+        self.nudge = nudge
+        self.tools: dict[str, tuple[dict, Callable[[dict], Any]]] = {}
+        self._register_all()
+
+    # ------------------------------------------------------------ registry
+
+    def tool(self, name: str, description: str, schema: dict):
+        def deco(fn):
+            self.tools[name] = (
+                {"name": name, "description": description,
+                 "inputSchema": {"type": "object", "properties": schema.get("p", {}),
+                                 "required": schema.get("r", [])}}, fn)
+            return fn
+        return deco
+
+    def _register_all(self) -> None:
+        S = {"type": "string"}
+        I = {"type": "integer"}
+        B = {"type": "boolean"}
+        N = {"type": "number"}
+        t = self.tool
+        ldb = self.ldb
+
+        # ---- rooms
+        @t("room_list_rooms", "List all rooms.", {})
+        def _(args):
+            with ldb as db:
+                return [{"id": r["id"], "name": r["name"], "status": r["status"],
+                         "goal": r["goal"]} for r in q.list_rooms(db)]
+
+        @t("room_create_room", "Create a room with a queen worker.",
+           {"p": {"name": S, "goal": S, "worker_model": S}, "r": ["name"]})
+        def _(args):
+            with ldb as db:
+                r = room_mod.create_room(db, args["name"], goal=args.get("goal"),
+                                         worker_model=args.get("worker_model",
+                                                               "qwen3-coder-30b"))
+            return {"room_id": r["id"], "queen_worker_id": r["queen_worker_id"]}
+
+        @t("room_get_status", "Room status summary.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                s = room_mod.get_room_status(db, args["room_id"])
+            return {"room": s["room"]["name"], "status": s["room"]["status"],
+                    "workers": [{"id": w["id"], "name": w["name"],
+                                 "state": w["agent_state"]} for w in s["workers"]],
+                    "active_goals": len(s["active_goals"]),
+                    "pending_decisions": len(s["pending_decisions"])}
+
+        @t("room_pause_room", "Pause a room.", {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                room_mod.pause_room(db, args["room_id"])
+            return {"paused": True}
+
+        @t("room_resume_room", "Resume a room.", {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                room_mod.resume_room(db, args["room_id"])
+            return {"resumed": True}
+
+        @t("room_restart_room", "Restart a room (clears goals/decisions).",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                room_mod.restart_room(db, args["room_id"])
+            return {"restarted": True}
+
+        # ---- workers
+        @t("room_list_workers", "List a room's workers.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return q.list_room_workers(db, args["room_id"])
+
+        @t("room_create_worker", "Create a worker in a room.",
+           {"p": {"room_id": I, "name": S, "role": S, "system_prompt": S},
+            "r": ["room_id", "name"]})
+        def _(args):
+            from ..core.constants import WORKER_ROLE_PRESETS
+            preset = WORKER_ROLE_PRESETS.get(args.get("role", ""), {})
+            with ldb as db:
+                w = q.create_worker(db, args["name"],
+                                    args.get("system_prompt")
+                                    or preset.get("systemPromptPrefix", ""),
+                                    role=args.get("role"), room_id=args["room_id"],
+                                    cycle_gap_ms=preset.get("cycleGapMs"),
+                                    max_turns=preset.get("maxTurns"))
+            return {"worker_id": w["id"]}
+
+        @t("room_start_worker", "Wake a worker (cross-process nudge).",
+           {"p": {"worker_id": I}, "r": ["worker_id"]})
+        def _(args):
+            ok = self.nudge(args["worker_id"])
+            return {"nudged": ok}
+
+        @t("room_save_wip", "Save a worker's work-in-progress.",
+           {"p": {"worker_id": I, "wip": S}, "r": ["worker_id", "wip"]})
+        def _(args):
+            with ldb as db:
+                q.set_worker_wip(db, args["worker_id"], args["wip"])
+            return {"saved": True}
+
+        # ---- goals
+        @t("room_set_goal", "Create a goal in a room.",
+           {"p": {"room_id": I, "description": S, "parent_goal_id": I,
+                  "assigned_worker_id": I}, "r": ["room_id", "description"]})
+        def _(args):
+            with ldb as db:
+                g = q.create_goal(db, args["room_id"], args["description"],
+                                  parent_goal_id=args.get("parent_goal_id"),
+                                  assigned_worker_id=args.get("assigned_worker_id"))
+            return {"goal_id": g["id"]}
+
+        @t("room_delegate_task", "Assign a goal to a worker and wake them.",
+           {"p": {"room_id": I, "description": S, "worker_id": I},
+            "r": ["room_id", "description", "worker_id"]})
+        def _(args):
+            with ldb as db:
+                g = q.create_goal(db, args["room_id"], args["description"],
+                                  assigned_worker_id=args["worker_id"])
+                q.update_goal(db, g["id"], status="in_progress")
+            self.nudge(args["worker_id"])
+            return {"goal_id": g["id"]}
+
+        @t("room_goal_tree", "The room's hierarchical goal tree.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return goals_mod.get_goal_tree(db, args["room_id"])
+
+        @t("room_complete_goal", "Mark a goal completed.",
+           {"p": {"goal_id": I, "observation": S}, "r": ["goal_id"]})
+        def _(args):
+            with ldb as db:
+                g = goals_mod.complete_goal(db, args["goal_id"],
+                                            observation=args.get("observation"))
+            return {"goal_id": g["id"], "status": g["status"]}
+
+        @t("room_update_goal_progress", "Report goal progress 0..1.",
+           {"p": {"goal_id": I, "progress": N}, "r": ["goal_id", "progress"]})
+        def _(args):
+            with ldb as db:
+                g = goals_mod.update_goal_progress(db, args["goal_id"],
+                                                   args["progress"])
+            return {"goal_id": g["id"], "progress": g["progress"]}
+
+        # ---- quorum
+        @t("room_announce", "Announce a decision (auto-effective unless objected).",
+           {"p": {"room_id": I, "proposer_id": I, "proposal": S,
+                  "decision_type": S}, "r": ["room_id", "proposal"]})
+        def _(args):
+            with ldb as db:
+                d = quorum_mod.announce(db, args["room_id"],
+                                        args.get("proposer_id"), args["proposal"],
+                                        args.get("decision_type", "low_impact"))
+            return {"decision_id": d["id"], "status": d["status"]}
+
+        @t("room_object", "Object to an announced decision.",
+           {"p": {"decision_id": I, "worker_id": I, "reason": S},
+            "r": ["decision_id", "worker_id", "reason"]})
+        def _(args):
+            with ldb as db:
+                d = quorum_mod.object_to(db, args["decision_id"],
+                                         args["worker_id"], args["reason"])
+            return {"status": d["status"]}
+
+        @t("room_vote", "Vote on an open decision.",
+           {"p": {"decision_id": I, "worker_id": I, "vote": S},
+            "r": ["decision_id", "worker_id", "vote"]})
+        def _(args):
+            with ldb as db:
+                v = quorum_mod.vote(db, args["decision_id"], args["worker_id"],
+                                    args["vote"])
+            return {"vote_id": v["id"]}
+
+        @t("room_keeper_vote", "Keeper override vote on a decision.",
+           {"p": {"decision_id": I, "vote": S}, "r": ["decision_id", "vote"]})
+        def _(args):
+            with ldb as db:
+                d = quorum_mod.keeper_vote(db, args["decision_id"], args["vote"])
+            return {"status": d["status"]}
+
+        @t("room_list_decisions", "List a room's decisions.",
+           {"p": {"room_id": I, "status": S}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return q.list_room_decisions(db, args["room_id"],
+                                             status=args.get("status"))
+
+        # ---- memory
+        @t("room_remember", "Store a fact in room memory.",
+           {"p": {"room_id": I, "name": S, "content": S, "category": S},
+            "r": ["name", "content"]})
+        def _(args):
+            if self.memory is not None:
+                eid = self.memory.remember(args.get("room_id"), args["name"],
+                                           args["content"],
+                                           category=args.get("category"))
+                return {"entity_id": eid}
+            with ldb as db:
+                ent = q.create_entity(db, args["name"],
+                                      category=args.get("category"),
+                                      room_id=args.get("room_id"),
+                                      observations=[args["content"]])
+            return {"entity_id": ent["id"]}
+
+        @t("room_recall", "Hybrid search over room memory.",
+           {"p": {"room_id": I, "query": S, "limit": I}, "r": ["query"]})
+        def _(args):
+            if self.memory is not None:
+                hits = self.memory.recall(args.get("room_id"), args["query"],
+                                          limit=args.get("limit", 5))
+            else:
+                with ldb as db:
+                    hits = q.hybrid_search(db, args["query"], None,
+                                           limit=args.get("limit", 5),
+                                           room_id=args.get("room_id"))
+            return [{"name": h["name"], "score": round(h["score"], 4),
+                     "observations": h["observations"][:3]} for h in hits]
+
+        # ---- skills
+        @t("room_create_skill", "Save a reusable skill.",
+           {"p": {"room_id": I, "name": S, "content": S, "activation_context": S,
+                  "auto_activate": B}, "r": ["name", "content"]})
+        def _(args):
+            with ldb as db:
+                s = skills_mod.create_agent_skill(
+                    db, args.get("room_id"), args["name"], args["content"],
+                    activation_context=args.get("activation_context"),
+                    auto_activate=args.get("auto_activate", False))
+            return {"skill_id": s["id"]}
+
+        @t("room_list_skills", "List a room's skills.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return [{"id": s["id"], "name": s["name"], "version": s["version"]}
+                        for s in q.list_room_skills(db, args["room_id"])]
+
+        @t("room_modify_skill", "Audited self-modification of a skill.",
+           {"p": {"room_id": I, "worker_id": I, "skill_id": I, "content": S,
+                  "reason": S}, "r": ["worker_id", "skill_id", "content"]})
+        def _(args):
+            with ldb as db:
+                out = self_mod.perform_skill_modification(
+                    db, args.get("room_id"), args["worker_id"], args["skill_id"],
+                    args["content"], reason=args.get("reason"))
+            return {"audit_id": out["audit_id"]}
+
+        @t("room_revert_modification", "Revert an audited self-modification.",
+           {"p": {"audit_id": I}, "r": ["audit_id"]})
+        def _(args):
+            with ldb as db:
+                return self_mod.revert_modification(db, args["audit_id"])
+
+        # ---- tasks / scheduler
+        @t("room_create_task", "Create a scheduled task (cron/once/manual/webhook).",
+           {"p": {"name": S, "prompt": S, "trigger_type": S, "cron_expression": S,
+                  "scheduled_at": S, "room_id": I, "max_runs": I},
+            "r": ["name", "prompt"]})
+        def _(args):
+            with ldb as db:
+                t_ = q.create_task(db, args["name"], args["prompt"],
+                                   trigger_type=args.get("trigger_type", "cron"),
+                                   cron_expression=args.get("cron_expression"),
+                                   scheduled_at=args.get("scheduled_at"),
+                                   room_id=args.get("room_id"),
+                                   max_runs=args.get("max_runs"))
+            return {"task_id": t_["id"]}
+
+        @t("room_list_tasks", "List tasks.", {"p": {"room_id": I}})
+        def _(args):
+            with ldb as db:
+                return [{"id": t_["id"], "name": t_["name"], "status": t_["status"],
+                         "trigger_type": t_["trigger_type"]}
+                        for t_ in q.list_tasks(db, room_id=args.get("room_id"))]
+
+        @t("room_task_runs", "Recent runs of a task.",
+           {"p": {"task_id": I}, "r": ["task_id"]})
+        def _(args):
+            with ldb as db:
+                return q.list_task_runs(db, args["task_id"])
+
+        # ---- inbox / escalations
+        @t("room_escalate", "Escalate a question to the keeper.",
+           {"p": {"room_id": I, "question": S, "from_agent_id": I},
+            "r": ["room_id", "question"]})
+        def _(args):
+            with ldb as db:
+                e = q.create_escalation(db, args["room_id"], args["question"],
+                                        from_agent_id=args.get("from_agent_id"))
+            return {"escalation_id": e["id"]}
+
+        @t("room_answer_escalation", "Answer a pending escalation.",
+           {"p": {"escalation_id": I, "answer": S},
+            "r": ["escalation_id", "answer"]})
+        def _(args):
+            with ldb as db:
+                q.answer_escalation(db, args["escalation_id"], args["answer"])
+            return {"answered": True}
+
+        @t("room_list_escalations", "List escalations for a room.",
+           {"p": {"room_id": I, "status": S}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return q.list_escalations(db, args["room_id"],
+                                          status=args.get("status"))
+
+        @t("room_send_room_message", "Send an inter-room message.",
+           {"p": {"room_id": I, "to_room_id": S, "subject": S, "body": S},
+            "r": ["room_id", "body"]})
+        def _(args):
+            with ldb as db:
+                m = q.create_room_message(db, args["room_id"], "outbound",
+                                          args.get("subject", ""), args["body"],
+                                          to_room_id=args.get("to_room_id"))
+            return {"message_id": m["id"]}
+
+        # ---- wallet / identity
+        @t("room_wallet_address", "The room wallet's address.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                addr = wallet_mod.get_wallet_address(db, args["room_id"])
+            return {"address": addr}
+
+        @t("room_wallet_send", "Send tokens from the room wallet.",
+           {"p": {"room_id": I, "to_address": S, "amount": S, "chain": S,
+                  "token": S}, "r": ["room_id", "to_address", "amount"]})
+        def _(args):
+            with ldb as db:
+                return wallet_mod.send_token(db, args["room_id"],
+                                             args["to_address"], args["amount"],
+                                             chain=args.get("chain", "base"),
+                                             token=args.get("token", "usdc"))
+
+        @t("room_payment_audit", "Wallet transaction history.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                w = q.get_room_wallet(db, args["room_id"])
+                if w is None:
+                    return []
+                return q.list_wallet_txs(db, w["id"])
+
+        @t("room_register_identity", "Register the room's on-chain identity "
+           "(ERC-8004 metadata prepared; broadcast requires RPC).",
+           {"p": {"room_id": I, "chain": S}, "r": ["room_id"]})
+        def _(args):
+            from ..core.identity import register_identity
+            with ldb as db:
+                return register_identity(db, args["room_id"],
+                                         chain=args.get("chain", "base"))
+
+        # ---- settings / resources
+        @t("room_get_setting", "Read a settings key.", {"p": {"key": S}, "r": ["key"]})
+        def _(args):
+            with ldb as db:
+                return {"key": args["key"], "value": q.get_setting(db, args["key"])}
+
+        @t("room_set_setting", "Write a settings key.",
+           {"p": {"key": S, "value": S}, "r": ["key", "value"]})
+        def _(args):
+            with ldb as db:
+                q.set_setting(db, args["key"], args["value"])
+            return {"ok": True}
+
+        @t("room_activity_feed", "Recent room activity.",
+           {"p": {"room_id": I, "limit": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return q.get_room_activity(db, args["room_id"],
+                                           limit=args.get("limit", 20))
+
+        # ---- watches
+        @t("room_watch_path", "Watch a filesystem path (validated allowlist).",
+           {"p": {"path": S, "action_prompt": S, "room_id": I}, "r": ["path"]})
+        def _(args):
+            from ..core.watch_path import validate_watch_path
+            ok, why = validate_watch_path(args["path"])
+            if not ok:
+                return {"error": why}
+            with ldb as db:
+                w = q.create_watch(db, args["path"],
+                                   action_prompt=args.get("action_prompt"),
+                                   room_id=args.get("room_id"))
+            return {"watch_id": w["id"]}
+
+        @t("room_list_watches", "List watches.", {"p": {"room_id": I}})
+        def _(args):
+            with ldb as db:
+                return q.list_watches(db, room_id=args.get("room_id"))
+
+    # ------------------------------------------------------------ JSON-RPC
+
+    def handle(self, msg: dict) -> dict | None:
+        mid = msg.get("id")
+        method = msg.get("method", "")
+        if method == "initialize":
+            return self._result(mid, {
+                "protocolVersion": PROTOCOL_VERSION,
+                "capabilities": {"tools": {}},
+                "serverInfo": SERVER_INFO})
+        if method == "notifications/initialized":
+            return None
+        if method == "tools/list":
+            return self._result(mid, {"tools": [t[0] for t in self.tools.values()]})
+        if method == "tools/call":
+            params = msg.get("params", {})
+            name = params.get("name", "")
+            args = params.get("arguments", {}) or {}
+            if name not in self.tools:
+                return self._error(mid, -32602, f"unknown tool: {name}")
+            try:
+                out = self.tools[name][1](args)
+                return self._result(mid, {"content": [
+                    {"type": "text", "text": json.dumps(out, default=str)}]})
+            except Exception as e:
+                return self._result(mid, {"content": [
+                    {"type": "text", "text": json.dumps({"error": str(e)})}],
+                    "isError": True})
+        if method == "ping":
+            return self._result(mid, {})
+        if mid is None:
+            return None
+        return self._error(mid, -32601, f"unknown method: {method}")
+
+    @staticmethod
+    def _result(mid, result) -> dict:
+        return {"jsonrpc": "2.0", "id": mid, "result": result}
+
+    @staticmethod
+    def _error(mid, code, message) -> dict:
+        return {"jsonrpc": "2.0", "id": mid,
+                "error": {"code": code, "message": message}}
+
+    def run_stdio(self) -> None:
+        """Blocking stdio loop: newline-delimited JSON-RPC."""
+        for line in sys.stdin:
+            line = line.strip()
+            if not line:
+                continue
+            try:
+                msg = json.loads(line)
+            except ValueError:
+                continue
+            resp = self.handle(msg)
+            if resp is not None:
+                sys.stdout.write(json.dumps(resp) + "\n")
+                sys.stdout.flush()
+
+
+def main() -> None:
+    import os
+
+    from ..memory.vector_store import GpuVectorStore, MemoryService
+    from ..server.auth import data_dir
+
+    db_path = os.environ.get("ROOMAMD_DB_PATH", str(data_dir() / "data.db"))
+    ldb = LockedDb(connect(db_path))
+    memory = MemoryService(ldb, store=GpuVectorStore(capacity=100_000,
+                                                     device="cpu"))
+    McpServer(ldb, memory=memory).run_stdio()
+
+
+if __name__ == "__main__":
+    main()
