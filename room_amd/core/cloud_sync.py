@@ -1,0 +1,131 @@
+"""Cloud sync (reference: src/shared/cloud-sync.ts — optional registration with
+a cloud relay, per-room token store file, 5-min heartbeats, inter-room message
+relay; everything fail-silent).
+
+Offline default: disabled unless ROOMAMD_CLOUD_API is set. Token store format
+(~/.roomamd/cloud-room-tokens.json) preserved.
+"""
+from __future__ import annotations
+
+import json
+import os
+import time
+import urllib.request
+from pathlib import Path
+
+from ..db import LockedDb
+from ..db import queries as q
+from .telemetry import get_machine_id
+
+HEARTBEAT_INTERVAL_S = 300
+
+
+def cloud_api() -> str | None:
+    return os.environ.get("ROOMAMD_CLOUD_API") or None
+
+
+def _tokens_path() -> Path:
+    return Path(os.environ.get("ROOMAMD_DATA_DIR",
+                               str(Path.home() / ".roomamd"))) / "cloud-room-tokens.json"
+
+
+def load_room_tokens() -> dict:
+    p = _tokens_path()
+    if p.exists():
+        try:
+            return json.loads(p.read_text())
+        except (ValueError, OSError):
+            return {}
+    return {}
+
+
+def save_room_token(room_id: int, token: str) -> None:
+    tokens = load_room_tokens()
+    tokens[str(room_id)] = token
+    p = _tokens_path()
+    try:
+        p.parent.mkdir(parents=True, exist_ok=True)
+        p.write_text(json.dumps(tokens))
+        p.chmod(0o600)
+    except OSError:
+        pass
+
+
+def _post(path: str, payload: dict, token: str | None = None) -> dict | None:
+    api = cloud_api()
+    if not api:
+        return None
+    try:
+        headers = {"Content-Type": "application/json"}
+        if token:
+            headers["Authorization"] = f"Bearer {token}"
+        req = urllib.request.Request(f"{api}{path}",
+                                     data=json.dumps(payload).encode(),
+                                     headers=headers)
+        with urllib.request.urlopen(req, timeout=5) as resp:
+            return json.load(resp)
+    except Exception:
+        return None  # fail-silent by design
+
+
+def register_with_cloud(ldb: LockedDb, room_id: int) -> str | None:
+    with ldb as db:
+        room = q.get_room(db, room_id)
+    if room is None:
+        return None
+    out = _post("/api/runtime/register",
+                {"machine_id": get_machine_id(), "room_id": room_id,
+                 "name": room["name"]})
+    if out and out.get("token"):
+        save_room_token(room_id, out["token"])
+        return out["token"]
+    return None
+
+
+def send_heartbeat(ldb: LockedDb, room_id: int) -> bool:
+    token = load_room_tokens().get(str(room_id))
+    if not token:
+        return False
+    with ldb as db:
+        usage = q.get_room_token_usage(db, room_id)
+        workers = q.list_room_workers(db, room_id)
+    out = _post("/api/runtime/heartbeat",
+                {"room_id": room_id, "ts": int(time.time()),
+                 "cycles": usage["cycles"], "workers": len(workers)},
+                token=token)
+    return out is not None
+
+
+def send_cloud_room_message(ldb: LockedDb, room_id: int, to_room: str,
+                            subject: str, body: str) -> bool:
+    token = load_room_tokens().get(str(room_id))
+    out = _post("/api/runtime/messages",
+                {"to": to_room, "subject": subject, "body": body}, token=token)
+    with ldb as db:  # outbound row recorded regardless (durable intent)
+        q.create_room_message(db, room_id, "outbound", subject, body,
+                              to_room_id=to_room)
+    return out is not None
+
+
+def fetch_cloud_room_messages(ldb: LockedDb, room_id: int) -> int:
+    """Pull relayed inbound messages into room_messages (runtime 60s loop)."""
+    api = cloud_api()
+    token = load_room_tokens().get(str(room_id))
+    if not api or not token:
+        return 0
+    try:
+        req = urllib.request.Request(
+            f"{api}/api/runtime/messages?room_id={room_id}",
+            headers={"Authorization": f"Bearer {token}"})
+        with urllib.request.urlopen(req, timeout=5) as resp:
+            msgs = json.load(resp)
+    except Exception:
+        return 0
+    n = 0
+    with ldb as db:
+        for m in msgs if isinstance(msgs, list) else []:
+            q.create_room_message(db, room_id, "inbound",
+                                  m.get("subject", ""), m.get("body", ""),
+                                  from_room_id=m.get("from"))
+            n += 1
+    return n

@@ -1,0 +1,150 @@
+"""Aux subsystem tests: prompt sync, templates, telemetry, cloud token store,
+watch paths, identity, public feed, clerk."""
+import json
+import os
+import time
+
+import pytest
+
+from room_amd.core import prompt_sync, room, templates
+from room_amd.db import LockedDb
+from room_amd.db import queries as q
+
+
+def test_worker_prompt_export_import(db, tmp_path):
+    r = room.create_room(db, "ps", worker_model="stub")
+    wid = r["queen_worker_id"]
+    files = prompt_sync.export_worker_prompts(db, r["id"], base=str(tmp_path))
+    assert len(files) == 1
+    text = open(files[0]).read()
+    assert text.startswith("---\n")
+    meta, prompt = prompt_sync.parse_worker_md(text)
+    assert int(meta["worker_id"]) == wid
+    assert "control plane" in prompt
+
+    # edit the file with a NEWER mtime → import overwrites the DB
+    edited = text.replace("control plane", "REWRITTEN PROMPT")
+    open(files[0], "w").write(edited)
+    future = time.time() + 5
+    os.utime(files[0], (future, future))
+    out = prompt_sync.import_worker_prompts(db, r["id"], base=str(tmp_path))
+    assert out[0]["action"] == "imported"
+    assert "REWRITTEN PROMPT" in q.get_worker(db, wid)["system_prompt"]
+
+    # older file → DB wins
+    past = time.time() - 9999
+    os.utime(files[0], (past, past))
+    q.update_worker(db, wid, system_prompt="db version")
+    out = prompt_sync.import_worker_prompts(db, r["id"], base=str(tmp_path))
+    assert out[0]["action"] == "kept-db"
+    # force overrides
+    out = prompt_sync.import_worker_prompts(db, r["id"], base=str(tmp_path),
+                                            force=True)
+    assert out[0]["action"] == "imported"
+
+
+def test_room_template_instantiation(db):
+    t = templates.list_templates()
+    assert "saas-builder" in t["rooms"]
+    r = templates.instantiate_room_template(db, "saas-builder", "my-startup",
+                                            worker_model="stub")
+    workers = q.list_room_workers(db, r["id"])
+    roles = {w["role"] for w in workers}
+    assert "queen" in roles and "executor" in roles and "researcher" in roles
+    assert len(workers) == 5  # queen + 4 template workers
+    with pytest.raises(ValueError):
+        templates.instantiate_room_template(db, "nope", "x")
+
+
+def test_telemetry_local_records(tmp_path, monkeypatch):
+    monkeypatch.setenv("ROOMAMD_DATA_DIR", str(tmp_path))
+    from room_amd.core import telemetry
+    mid = telemetry.get_machine_id()
+    assert len(mid) == 12
+    telemetry.submit_crash_report("boom", context="test")
+    telemetry.submit_heartbeat({"rooms": 1})
+    crash = (tmp_path / "telemetry" / "crash.jsonl").read_text()
+    assert json.loads(crash)["error"] == "boom"
+    hb = (tmp_path / "telemetry" / "heartbeat.jsonl").read_text()
+    assert json.loads(hb)["stats"] == {"rooms": 1}
+
+
+def test_cloud_token_store(tmp_path, monkeypatch, db):
+    monkeypatch.setenv("ROOMAMD_DATA_DIR", str(tmp_path))
+    from room_amd.core import cloud_sync
+    assert cloud_sync.cloud_api() is None
+    cloud_sync.save_room_token(7, "tok-abc")
+    assert cloud_sync.load_room_tokens() == {"7": "tok-abc"}
+    # offline: registration and heartbeats fail silently
+    r = room.create_room(db, "cs", worker_model="stub")
+    assert cloud_sync.register_with_cloud(LockedDb(db), r["id"]) is None
+    assert cloud_sync.send_heartbeat(LockedDb(db), r["id"]) is False
+
+
+def test_web_tools_offline_degrade():
+    from room_amd.core import web_tools
+    out = web_tools.web_fetch("ftp://bad")
+    assert not out["ok"]
+    act = web_tools.browser_action("s1", "click", selector="#x")
+    assert not act["ok"] and act["session"] == "s1"
+    assert web_tools.close_browser("s1")
+    assert not web_tools.close_browser("s1")
+
+
+def test_identity_metadata(db):
+    from room_amd.core import identity
+    r = room.create_room(db, "idroom", goal="g", worker_model="stub")
+    out = identity.register_identity(db, r["id"])
+    assert out["registry"].startswith("0x8004A169")
+    assert out["agent_uri"].startswith("data:application/json;base64,")
+    import base64
+    meta = json.loads(base64.b64decode(out["agent_uri"].split(",", 1)[1]))
+    assert meta["name"] == "idroom"
+    w = q.get_room_wallet(db, r["id"])
+    assert w["erc8004_agent_id"].startswith("pending:base:0x")
+
+
+def test_public_feed_visibility(db):
+    from room_amd.core.public_feed import get_public_feed, get_public_room_profile
+    r = room.create_room(db, "pub", goal="g", worker_model="stub")
+    assert get_public_feed(db) == []  # private by default
+    q.update_room(db, r["id"], visibility="public")
+    feed = get_public_feed(db)
+    assert feed and feed[0]["room_name"] == "pub"
+    assert "details" not in feed[0]
+    prof = get_public_room_profile(db, r["id"])
+    assert prof["worker_count"] == 1
+
+
+def test_clerk_chat_and_tools(db):
+    from room_amd.core.clerk import clerk_chat, execute_clerk_tool
+    from room_amd.engine.types import ToolCall
+    ldb = LockedDb(db)
+    out = execute_clerk_tool(ldb, ToolCall("clerk_create_room",
+                                           {"name": "clerked"}))
+    rid = json.loads(out)["room_id"]
+    status = json.loads(execute_clerk_tool(
+        ldb, ToolCall("clerk_room_status", {"room_id": rid})))
+    assert status["room"] == "clerked"
+    reply = clerk_chat(ldb, "list the rooms please", model="stub")
+    assert isinstance(reply, str) and reply
+    with ldb as conn:
+        msgs = q.list_clerk_messages(conn)
+        assert any(m["role"] == "assistant" for m in msgs)
+        usage = conn.execute("SELECT * FROM clerk_usage").fetchall()
+        assert usage and usage[0]["source"] == "chat"
+
+
+def test_commentary_engine(db):
+    from room_amd.core.clerk import CommentaryEngine
+    from room_amd.core.events import EventBus
+    ldb = LockedDb(db)
+    bus = EventBus()
+    eng = CommentaryEngine(ldb, bus, model="stub", pace_s=0.0)
+    bus.emit("room:1", "cycle_finished", {"cycle_id": 1})
+    line = eng.tick()
+    assert line is not None
+    with ldb as conn:
+        msgs = q.list_clerk_messages(conn)
+        assert any(m["role"] == "commentary" for m in msgs)
+    eng.stop()
