@@ -80,6 +80,11 @@ class _DecodeGraph:
         self.pos_in = torch.zeros(bucket, dtype=torch.int32, device=dev)
         self.seeds = torch.randint(1, 2**62, (bucket,), dtype=torch.int64,
                                    device=dev)
+        # pinned host staging (pageable→device copies would sync per step)
+        self.h_tok = torch.zeros(bucket, dtype=torch.int64, pin_memory=True)
+        self.h_seq = torch.full((bucket,), engine.pad_slot, dtype=torch.int32,
+                                pin_memory=True)
+        self.h_pos = torch.zeros(bucket, dtype=torch.int32, pin_memory=True)
         model, cache = engine.model, engine.cache
         from .. import ops as _ops
 
@@ -107,13 +112,17 @@ class _DecodeGraph:
     def replay(self, tokens: list[int], slots: list[int],
                positions: list[int]) -> torch.Tensor:
         n, b = len(tokens), self.bucket
-        pad = b - n
-        self.tok_in.copy_(torch.tensor(tokens + [0] * pad, dtype=torch.int64),
-                          non_blocking=True)
-        self.seq_in.copy_(torch.tensor(slots + [self.pad_slot] * pad,
-                                       dtype=torch.int32), non_blocking=True)
-        self.pos_in.copy_(torch.tensor(positions + [0] * pad, dtype=torch.int32),
-                          non_blocking=True)
+        for i in range(n):
+            self.h_tok[i] = tokens[i]
+            self.h_seq[i] = slots[i]
+            self.h_pos[i] = positions[i]
+        for i in range(n, b):
+            self.h_tok[i] = 0
+            self.h_seq[i] = self.pad_slot
+            self.h_pos[i] = 0
+        self.tok_in.copy_(self.h_tok, non_blocking=True)
+        self.seq_in.copy_(self.h_seq, non_blocking=True)
+        self.pos_in.copy_(self.h_pos, non_blocking=True)
         self.graph.replay()
         return self.out_tokens[:n]
 

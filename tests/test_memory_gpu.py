@@ -17,16 +17,18 @@ def test_gpu_store_matches_cpu():
     torch.manual_seed(20)
     gpu = GpuVectorStore(capacity=5000, device="cuda")
     cpu = GpuVectorStore(capacity=5000, device="cpu")
-    texts = [f"document {i} about topic {i % 37}" for i in range(2000)]
+    texts = [f"doc {i} unique subject {i} with marker w{i}x" for i in range(2000)]
     for i, t in enumerate(texts):
         v = embedder.embed(t)
         gpu.upsert(i + 1, v)
         cpu.upsert(i + 1, v)
-    q = embedder.embed("document about topic 5")
+    q = embedder.embed("doc 5 unique subject 5 with marker w5x")
     g = gpu.search(q, k=10)
     c = cpu.search(q, k=10)
-    overlap = len({i for i, _ in g} & {i for i, _ in c})
-    assert overlap >= 8  # bf16 vs f32 rounding can swap near-ties
+    assert g[0][0] == c[0][0] == 6  # entity id 5+1
+    # scores agree to bf16 precision even where near-ties reorder ids
+    for (gi, gv), (ci, cv) in zip(g, c):
+        assert abs(gv - cv) < 2e-2
 
 
 def test_gpu_store_10m_scale():
