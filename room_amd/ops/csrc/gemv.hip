@@ -1,16 +1,14 @@
 // Dense skinny-batch GEMV: y[B, N] = x[B, H] @ W[N, H]^T  (decode projections).
 //
 // hipBLASLt runs these M≤8 shapes at ~1.1 TB/s (measured, scripts/
-// gpu_op_microbench.py). v1 here (wave-per-output) reached ~2.2 TB/s but its
-// waves lived only 4 iterations — launch/drain-bound. v2: each wave owns FOUR
-// consecutive output rows, streaming their W rows together; x rows (L2-hot)
-// are unpacked once per k-chunk and reused across the 4 rows.
+// gpu_op_microbench.py); this kernel streams W once for ALL batch rows
+// (x rows are L2-resident) with 16 B/lane coalesced loads. Used for the
+// QKV/O/router/lm_head projections at decode; prefill keeps hipBLASLt.
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include "common.h"
 
 #define GEMV_MAXB 8
-#define GEMV_ROWS 4   // output rows per wave
 
 template <bool F32OUT>
 __global__ __launch_bounds__(256)
@@ -20,53 +18,37 @@ void gemv_kernel(void* __restrict__ y,           // [B, N] bf16 or f32
                  int B, int H, int N) {
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int n0 = (blockIdx.x * 4 + wid) * GEMV_ROWS;
-  if (n0 >= N) return;
+  const int n = blockIdx.x * 4 + wid;
+  if (n >= N) return;
 
-  float acc[GEMV_ROWS][GEMV_MAXB];
+  float acc[GEMV_MAXB];
   #pragma unroll
-  for (int r = 0; r < GEMV_ROWS; ++r)
-    #pragma unroll
-    for (int b = 0; b < GEMV_MAXB; ++b) acc[r][b] = 0.f;
+  for (int b = 0; b < GEMV_MAXB; ++b) acc[b] = 0.f;
 
+  const short* wrow = w + (long)n * H;
   for (int base = lane * 8; base < H; base += WAVE * 8) {
-    float xf[GEMV_MAXB][8];
+    bf16x8 wv = *reinterpret_cast<const bf16x8*>(wrow + base);
+    float wf[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) wf[j] = bf2f(wv[j]);
     #pragma unroll
     for (int b = 0; b < GEMV_MAXB; ++b) {
       if (b < B) {
         bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
         #pragma unroll
-        for (int j = 0; j < 8; ++j) xf[b][j] = bf2f(xv[j]);
-      }
-    }
-    #pragma unroll
-    for (int r = 0; r < GEMV_ROWS; ++r) {
-      const int n = n0 + r;
-      if (n >= N) break;
-      bf16x8 wv = *reinterpret_cast<const bf16x8*>(w + (long)n * H + base);
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float wf = bf2f(wv[j]);
-        #pragma unroll
-        for (int b = 0; b < GEMV_MAXB; ++b)
-          if (b < B) acc[r][b] += wf * xf[b][j];
+        for (int j = 0; j < 8; ++j) acc[b] += wf[j] * bf2f(xv[j]);
       }
     }
   }
   #pragma unroll
-  for (int r = 0; r < GEMV_ROWS; ++r) {
-    const int n = n0 + r;
-    if (n >= N) break;
-    #pragma unroll
-    for (int b = 0; b < GEMV_MAXB; ++b) {
-      if (b < B) {
-        float v = wave_reduce_sum(acc[r][b]);
-        if (lane == 0) {
-          if (F32OUT)
-            ((float*)y)[(long)b * N + n] = v;
-          else
-            ((short*)y)[(long)b * N + n] = f2bf(v);
-        }
+  for (int b = 0; b < GEMV_MAXB; ++b) {
+    if (b < B) {
+      float r = wave_reduce_sum(acc[b]);
+      if (lane == 0) {
+        if (F32OUT)
+          ((float*)y)[(long)b * N + n] = r;
+        else
+          ((short*)y)[(long)b * N + n] = f2bf(r);
       }
     }
   }
@@ -77,7 +59,7 @@ void gemv(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
   TORCH_CHECK(B <= GEMV_MAXB, "gemv handles B<=8 (decode); use hipBLASLt above");
   TORCH_CHECK(x.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16);
   TORCH_CHECK(H % (WAVE * 8) == 0, "H must be a multiple of 512");
-  dim3 grid((N + 4 * GEMV_ROWS - 1) / (4 * GEMV_ROWS)), block(256);
+  dim3 grid((N + 3) / 4), block(256);
   hipStream_t s = c10::hip::getCurrentHIPStream();
   if (y.dtype() == torch::kFloat32) {
     hipLaunchKernelGGL(gemv_kernel<true>, grid, block, 0, s,
