@@ -270,14 +270,14 @@ def test_agent_loop_trigger_and_stop(ldb):
     async def go():
         state = await mgr.start_agent_loop(r["id"], r["queen_worker_id"])
         # wait for first cycle to complete
-        for _ in range(200):
+        for _ in range(600):
             await asyncio.sleep(0.02)
             if state.cycle_count >= 1:
                 break
         assert state.cycle_count >= 1
         # loop is now sleeping 1h; trigger wakes it immediately
         mgr.trigger_agent(r["queen_worker_id"])
-        for _ in range(200):
+        for _ in range(600):
             await asyncio.sleep(0.02)
             if state.cycle_count >= 2:
                 break

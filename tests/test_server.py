@@ -134,7 +134,7 @@ def test_tasks_and_runs(server):
                     headers=h).json()
     assert client.post(f"/api/tasks/{t['id']}/run", headers=h).json()["queued"]
     import time
-    for _ in range(100):
+    for _ in range(300):
         runs = client.get(f"/api/tasks/{t['id']}/runs", headers=h).json()
         if runs and runs[0]["status"] != "running":
             break
