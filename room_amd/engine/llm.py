@@ -61,13 +61,17 @@ class _Session:
 
 
 class _DecodeGraph:
-    """One captured hipGraph per (batch bucket, sampling params).
+    """One captured hipGraph per (batch bucket, sampling params), replayed K
+    steps per host sync (multi-step decode).
 
-    The decode step is ~1500 tiny kernel launches across 48 layers; capturing
-    it collapses launch overhead to a single hipGraphLaunch. Inputs are
-    persistent device buffers; the sampler RNG state lives on-device so
-    replays draw fresh randomness with no host work. Padded lanes point at a
-    reserved pad sequence slot (their KV writes land in a scrap block)."""
+    Inside the capture: forward → fused sampler (on-device RNG state) →
+    token-history append (device step counter) → sampled tokens copied into
+    the next step's input → positions += 1. A block of K replays therefore
+    runs K full decode steps with ZERO host work; the host reads the K×B
+    token history once at the end. Padded lanes use a reserved pad slot whose
+    KV writes land in scrap block 0."""
+
+    KMAX = 32
 
     def __init__(self, engine: "LocalEngine", bucket: int,
                  temperature: float, top_p: float, top_k: int):
@@ -80,38 +84,47 @@ class _DecodeGraph:
         self.pos_in = torch.zeros(bucket, dtype=torch.int32, device=dev)
         self.seeds = torch.randint(1, 2**62, (bucket,), dtype=torch.int64,
                                    device=dev)
-        # pinned host staging (pageable→device copies would sync per step)
+        self.hist = torch.zeros(self.KMAX * bucket, dtype=torch.int32, device=dev)
+        self.ctr = torch.zeros(1, dtype=torch.int32, device=dev)
+        self.hist_host = torch.zeros(self.KMAX * bucket, dtype=torch.int32,
+                                     pin_memory=True)
+        # pinned host staging for the block entry
         self.h_tok = torch.zeros(bucket, dtype=torch.int64, pin_memory=True)
         self.h_seq = torch.full((bucket,), engine.pad_slot, dtype=torch.int32,
                                 pin_memory=True)
         self.h_pos = torch.zeros(bucket, dtype=torch.int32, pin_memory=True)
         model, cache = engine.model, engine.cache
         from .. import ops as _ops
+        C = _ops._require()
 
-        def run():
+        def run_step():
             logits = model.forward(self.tok_in, self.seq_in, self.pos_in,
                                    cache.block_table, cache.kcaches,
                                    cache.vcaches)
             toks = torch.empty(bucket, dtype=torch.int32, device=dev)
-            _ops._require().sample_tokens_v3(toks, logits, self.seeds, top_k,
-                                          temperature, top_p)
-            return toks
+            C.sample_tokens_v3(toks, logits, self.seeds, top_k, temperature,
+                               top_p)
+            C.hist_append(self.hist, self.ctr, toks, self.KMAX)
+            self.tok_in.copy_(toks)      # feed the next replay
+            self.pos_in.add_(1)
 
         # warmup on a side stream (required before capture)
-        s = torch.cuda.Stream(dev)
-        s.wait_stream(torch.cuda.current_stream(dev))
-        with torch.cuda.stream(s):
+        strm = torch.cuda.Stream(dev)
+        strm.wait_stream(torch.cuda.current_stream(dev))
+        with torch.cuda.stream(strm):
             for _ in range(2):
-                run()
-        torch.cuda.current_stream(dev).wait_stream(s)
+                run_step()
+        torch.cuda.current_stream(dev).wait_stream(strm)
         torch.cuda.synchronize(dev)
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
-            self.out_tokens = run()
+            run_step()
 
-    def replay(self, tokens: list[int], slots: list[int],
-               positions: list[int]) -> torch.Tensor:
+    def run_block(self, tokens: list[int], slots: list[int],
+                  positions: list[int], k: int) -> list[int]:
+        """Run k decode steps; returns the flat K×B token history (host)."""
         n, b = len(tokens), self.bucket
+        k = min(k, self.KMAX)
         for i in range(n):
             self.h_tok[i] = tokens[i]
             self.h_seq[i] = slots[i]
@@ -123,8 +136,11 @@ class _DecodeGraph:
         self.tok_in.copy_(self.h_tok, non_blocking=True)
         self.seq_in.copy_(self.h_seq, non_blocking=True)
         self.pos_in.copy_(self.h_pos, non_blocking=True)
-        self.graph.replay()
-        return self.out_tokens[:n]
+        self.ctr.zero_()
+        for _ in range(k):
+            self.graph.replay()
+        self.hist_host[:k * b].copy_(self.hist[:k * b])  # syncs
+        return self.hist_host[:k * b].tolist()
 
 
 class LocalEngine:
@@ -349,49 +365,76 @@ class LocalEngine:
             return
         t0 = time.time()
         dev = self.device
-        for r in reqs:
-            self.cache.ensure_capacity(r.slot, r.pos + 1)
         tokens = [r.last_token for r in reqs]
         slots = [r.slot for r in reqs]
         positions = [r.pos for r in reqs]
-        for r in reqs:
-            r.pos += 1
 
-        toks_t = None
-        if self.graphs_enabled and not self._graphs_broken and len(reqs) <= 64:
-            r0 = reqs[0]
+        # multi-step graph block: K bounded by the smallest remaining budget so
+        # every request stops exactly at max_new (admission latency is K steps)
+        steps = 1
+        graph_ok = (self.graphs_enabled and not self._graphs_broken
+                    and len(reqs) <= 8)
+        r0 = reqs[0]
+        uniform = all((r.temperature, r.top_p, r.top_k)
+                      == (r0.temperature, r0.top_p, r0.top_k) for r in reqs)
+        if graph_ok and uniform:
+            steps = max(1, min(min(r.max_new_tokens - len(r.out_tokens)
+                                   for r in reqs), _DecodeGraph.KMAX))
+            for r in reqs:
+                self.cache.ensure_capacity(r.slot, r.pos + steps)
             bucket = next(b for b in DECODE_BUCKETS if b >= len(reqs))
             key = (bucket, round(r0.temperature, 3), round(r0.top_p, 3), r0.top_k)
-            uniform = all((r.temperature, r.top_p, r.top_k)
-                          == (r0.temperature, r0.top_p, r0.top_k) for r in reqs)
-            if uniform:
-                try:
-                    g = self._graphs.get(key)
-                    if g is None:
-                        g = _DecodeGraph(self, bucket, r0.temperature, r0.top_p,
-                                         r0.top_k)
-                        self._graphs[key] = g
-                        self.stats["graphs_captured"] = len(self._graphs)
-                    toks_t = g.replay(tokens, slots, positions)
-                except Exception as e:
-                    import sys
-                    print(f"[room_amd] hipGraph decode disabled: {e}",
-                          file=sys.stderr)
-                    self._graphs_broken = True
-                    toks_t = None
+            try:
+                g = self._graphs.get(key)
+                if g is None:
+                    g = _DecodeGraph(self, bucket, r0.temperature, r0.top_p,
+                                     r0.top_k)
+                    self._graphs[key] = g
+                    self.stats["graphs_captured"] = len(self._graphs)
+                hist = g.run_block(tokens, slots, positions, steps)
+                finished = []
+                for si in range(steps):
+                    row = hist[si * g.bucket: si * g.bucket + len(reqs)]
+                    for r, t in zip(reqs, row):
+                        if r in finished:
+                            continue
+                        r.last_token = int(t)
+                        r.out_tokens.append(int(t))
+                        r.pos += 1
+                        if (len(r.out_tokens) >= r.max_new_tokens
+                                or t in (tok.EOS, tok.IM_END)):
+                            finished.append(r)
+                for r in finished:
+                    if r.session_key and r.session_key in self.sessions:
+                        self.sessions[r.session_key].tokens = (
+                            r.prompt_tokens + r.out_tokens[:-1])
+                    self._active.remove(r)
+                    r.done.set()
+                self.stats["decode_steps"] += steps
+                self.stats["decode_tokens"] += steps * len(reqs)
+                self.stats["decode_time"] += time.time() - t0
+                return
+            except Exception as e:
+                import sys
+                print(f"[room_amd] hipGraph decode disabled: {e}",
+                      file=sys.stderr)
+                self._graphs_broken = True
 
-        if toks_t is None:  # eager fallback
-            tokens_t = torch.tensor(tokens, dtype=torch.int64, device=dev)
-            seq_t = torch.tensor(slots, dtype=torch.int32, device=dev)
-            pos_t = torch.tensor(positions, dtype=torch.int32, device=dev)
-            logits = self.model.forward(tokens_t, seq_t, pos_t,
-                                        self.cache.block_table,
-                                        self.cache.kcaches, self.cache.vcaches)
-            seeds = torch.randint(1, 2**62, (logits.size(0),), dtype=torch.int64,
-                                  device=dev)
-            r0 = reqs[0]
-            toks_t = ops.sample_tokens(logits, seeds, top_k=r0.top_k,
-                                       temperature=r0.temperature, top_p=r0.top_p)
+        # eager fallback: one step
+        for r in reqs:
+            self.cache.ensure_capacity(r.slot, r.pos + 1)
+        for r in reqs:
+            r.pos += 1
+        tokens_t = torch.tensor(tokens, dtype=torch.int64, device=dev)
+        seq_t = torch.tensor(slots, dtype=torch.int32, device=dev)
+        pos_t = torch.tensor(positions, dtype=torch.int32, device=dev)
+        logits = self.model.forward(tokens_t, seq_t, pos_t,
+                                    self.cache.block_table,
+                                    self.cache.kcaches, self.cache.vcaches)
+        seeds = torch.randint(1, 2**62, (logits.size(0),), dtype=torch.int64,
+                              device=dev)
+        toks_t = ops.sample_tokens(logits, seeds, top_k=r0.top_k,
+                                   temperature=r0.temperature, top_p=r0.top_p)
         self._finish_tokens(reqs, toks_t)
         self.stats["decode_steps"] += 1
         self.stats["decode_tokens"] += len(reqs)

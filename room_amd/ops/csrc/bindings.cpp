@@ -48,6 +48,8 @@ void sample_tokens_v3(torch::Tensor out_tokens, torch::Tensor logits,
                       torch::Tensor seeds, int64_t top_k, double temperature,
                       double top_p);
 void sample_scan_probe(torch::Tensor out, torch::Tensor logits);
+void hist_append(torch::Tensor hist, torch::Tensor ctr, torch::Tensor toks,
+                 int64_t kmax);
 void vs_topk(torch::Tensor out_v, torch::Tensor out_i, torch::Tensor cand_v,
              torch::Tensor cand_i, torch::Tensor mat, torch::Tensor query,
              int64_t K);
@@ -74,5 +76,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemv", &gemv, "dense skinny-batch GEMV (decode projections)");
   m.def("sample_tokens_v3", &sample_tokens_v3, "register top-8 sampler");
   m.def("sample_scan_probe", &sample_scan_probe, "scan-cost probe");
+  m.def("hist_append", &hist_append, "token-history append (multi-step decode)");
   m.def("vs_topk", &vs_topk, "vector-store cosine top-k over bf16 matrix");
 }

@@ -313,3 +313,27 @@ void sample_scan_probe(torch::Tensor out, torch::Tensor logits) {
                      out.data_ptr<int>(), logits.data_ptr<float>(), V);
   HIP_CHECK_KERNEL();
 }
+
+// token-history append for multi-step decode graphs: each graph replay
+// appends the step's sampled tokens into a ring and bumps the device counter,
+// so the host syncs once per K-step block instead of every step.
+__global__ void hist_append_kernel(int* __restrict__ hist,   // [KMAX * B]
+                                   int* __restrict__ ctr,    // [1]
+                                   const int* __restrict__ toks,  // [B]
+                                   int B, int kmax) {
+  const int i = threadIdx.x;
+  const int step = ctr[0];
+  if (step < kmax && i < B) hist[step * B + i] = toks[i];
+  __syncthreads();
+  if (i == 0) ctr[0] = step + 1;
+}
+
+void hist_append(torch::Tensor hist, torch::Tensor ctr, torch::Tensor toks,
+                 int64_t kmax) {
+  const int B = toks.numel();
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(hist_append_kernel, dim3(1), dim3(std::max(B, 64)), 0, s,
+                     hist.data_ptr<int>(), ctr.data_ptr<int>(),
+                     toks.data_ptr<int>(), B, (int)kmax);
+  HIP_CHECK_KERNEL();
+}
