@@ -1,0 +1,183 @@
+"""Within-probe A/B of dense-GEMV variants with COLD weight reads.
+
+The op microbench re-reads one weight tensor 100× (L2/L3-warm, flattering).
+Here each iteration rotates through 24 weight copies (~500 MB > 256 MB L3) so
+reads hit HBM like a real decode step. Variants:
+  V0 wave-per-output (shipped)        V1 wave-per-2-outputs
+  V2 512-thread blocks, wave/output   V3 thread-per-output (serial row read)
+"""
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import torch
+from torch.utils.cpp_extension import load_inline
+
+SRC = r"""
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+#define WAVE 64
+__device__ __forceinline__ float bf2f(short u) {
+  union { float f; unsigned u32; } c; c.u32 = (unsigned)(unsigned short)u << 16;
+  return c.f;
+}
+__device__ __forceinline__ short f2bf(float f) {
+  union { float f; unsigned u32; } c; c.f = f;
+  unsigned lsb = (c.u32 >> 16) & 1; c.u32 += 0x7fff + lsb;
+  return (short)(c.u32 >> 16);
+}
+__device__ __forceinline__ float wsum(float v) {
+  #pragma unroll
+  for (int o = 32; o > 0; o >>= 1) v += __shfl_xor(v, o, WAVE);
+  return v;
+}
+
+template <int NW>  // waves per block
+__global__ __launch_bounds__(NW * 64)
+void v0(short* y, const short* x, const short* w, int B, int H, int N) {
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int n = blockIdx.x * NW + wid;
+  if (n >= N) return;
+  float acc[8];
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) acc[b] = 0.f;
+  const short* wr = w + (long)n * H;
+  for (int base = lane * 8; base < H; base += WAVE * 8) {
+    bf16x8 wv = *(const bf16x8*)(wr + base);
+    float wf[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) wf[j] = bf2f(wv[j]);
+    #pragma unroll
+    for (int b = 0; b < 8; ++b) if (b < B) {
+      bf16x8 xv = *(const bf16x8*)(x + (long)b * H + base);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) acc[b] += wf[j] * bf2f(xv[j]);
+    }
+  }
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) if (b < B) {
+    float r = wsum(acc[b]);
+    if (lane == 0) y[(long)b * N + n] = f2bf(r);
+  }
+}
+
+__global__ __launch_bounds__(256)
+void v1(short* y, const short* x, const short* w, int B, int H, int N) {
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int n0 = (blockIdx.x * 4 + wid) * 2;
+  if (n0 >= N) return;
+  float a0[8], a1[8];
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) { a0[b] = 0.f; a1[b] = 0.f; }
+  const short* w0 = w + (long)n0 * H;
+  const short* w1 = w + (long)(n0 + 1) * H;
+  for (int base = lane * 8; base < H; base += WAVE * 8) {
+    bf16x8 wv0 = *(const bf16x8*)(w0 + base);
+    bf16x8 wv1 = *(const bf16x8*)(w1 + base);
+    #pragma unroll
+    for (int b = 0; b < 8; ++b) if (b < B) {
+      bf16x8 xv = *(const bf16x8*)(x + (long)b * H + base);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xf = bf2f(xv[j]);
+        a0[b] += bf2f(wv0[j]) * xf;
+        a1[b] += bf2f(wv1[j]) * xf;
+      }
+    }
+  }
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) if (b < B) {
+    float r0 = wsum(a0[b]), r1 = wsum(a1[b]);
+    if (lane == 0) { y[(long)b * N + n0] = f2bf(r0);
+                     y[(long)b * N + n0 + 1] = f2bf(r1); }
+  }
+}
+
+__global__ __launch_bounds__(256)
+void v3(short* y, const short* x, const short* w, int B, int H, int N) {
+  // thread-per-output: serial 16B-chunk row read, no cross-lane reduce
+  const int n = blockIdx.x * 256 + threadIdx.x;
+  if (n >= N) return;
+  float acc[8];
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) acc[b] = 0.f;
+  const short* wr = w + (long)n * H;
+  for (int k = 0; k < H; k += 8) {
+    bf16x8 wv = *(const bf16x8*)(wr + k);
+    #pragma unroll
+    for (int b = 0; b < 8; ++b) if (b < B) {
+      bf16x8 xv = *(const bf16x8*)(x + (long)b * H + k);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) acc[b] += bf2f(wv[j]) * bf2f(xv[j]);
+    }
+  }
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) if (b < B)
+    y[(long)b * N + n] = f2bf(acc[b]);
+}
+
+void run(int variant, torch::Tensor y, torch::Tensor x, torch::Tensor w) {
+  int B = x.size(0), H = x.size(1), N = w.size(0);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  if (variant == 0)
+    hipLaunchKernelGGL((v0<4>), dim3((N + 3) / 4), dim3(256), 0, s,
+      (short*)y.data_ptr(), (const short*)x.data_ptr(), (const short*)w.data_ptr(), B, H, N);
+  else if (variant == 1)
+    hipLaunchKernelGGL(v1, dim3((N + 7) / 8), dim3(256), 0, s,
+      (short*)y.data_ptr(), (const short*)x.data_ptr(), (const short*)w.data_ptr(), B, H, N);
+  else if (variant == 2)
+    hipLaunchKernelGGL((v0<8>), dim3((N + 7) / 8), dim3(512), 0, s,
+      (short*)y.data_ptr(), (const short*)x.data_ptr(), (const short*)w.data_ptr(), B, H, N);
+  else
+    hipLaunchKernelGGL(v3, dim3((N + 255) / 256), dim3(256), 0, s,
+      (short*)y.data_ptr(), (const short*)x.data_ptr(), (const short*)w.data_ptr(), B, H, N);
+}
+"""
+
+mod = load_inline(name="gemv_ab", cpp_sources="", cuda_sources=SRC,
+                  functions=["run"], with_cuda=True, verbose=False,
+                  extra_cuda_cflags=["-O3", "--offload-arch=gfx950"])
+
+DEV = "cuda"
+B, H, N = 5, 2048, 5120
+COPIES = 24
+torch.manual_seed(0)
+x = torch.randn(B, H, dtype=torch.bfloat16, device=DEV)
+ws = [torch.randn(N, H, dtype=torch.bfloat16, device=DEV) * 0.02
+      for _ in range(COPIES)]
+y = torch.empty(B, N, dtype=torch.bfloat16, device=DEV)
+
+# correctness vs torch
+ref = (x.float() @ ws[0].float().T).to(torch.bfloat16)
+for v in range(4):
+    mod.run(v, y, x, ws[0])
+    torch.cuda.synchronize()
+    assert torch.allclose(y.float(), ref.float(), atol=6e-2, rtol=6e-2), f"V{v}"
+print("all variants correct")
+
+ITER = 240
+for v in range(4):
+    for i in range(COPIES):
+        mod.run(v, y, x, ws[i])
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for i in range(ITER):
+        mod.run(v, y, x, ws[i % COPIES])
+    torch.cuda.synchronize()
+    us = (time.time() - t0) / ITER * 1e6
+    tb = N * H * 2 / us / 1e6
+    print(f"V{v}: {us:7.1f} µs  {tb:5.2f} TB/s (cold rotation)")
+# warm single-buffer comparison for V0
+for i in range(20):
+    mod.run(0, y, x, ws[0])
+torch.cuda.synchronize()
+t0 = time.time()
+for i in range(ITER):
+    mod.run(0, y, x, ws[0])
+torch.cuda.synchronize()
+us = (time.time() - t0) / ITER * 1e6
+print(f"V0 warm single-buffer: {us:.1f} µs ({N*H*2/us/1e6:.2f} TB/s)")
