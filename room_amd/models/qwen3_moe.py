@@ -124,6 +124,12 @@ class Qwen3MoEModel:
         x = self.embed[tokens.long()]              # [T, H] bf16 (residual stream)
         hbuf = torch.empty_like(x)                 # normed activations
         moe_out = None                             # pending delta for fused add
+        # layer-invariant MoE pair/token indices (was re-built 48×/step: an
+        # arange+repeat_interleave kernel pair per layer in the decode graph)
+        K = cfg.num_experts_per_tok
+        self._pair_token_flat = torch.arange(
+            T, device=dev, dtype=torch.int32).repeat_interleave(K)
+        self._arangeP = torch.arange(T * K, device=dev, dtype=torch.int32)
         if decode:
             qkv = torch.empty(T, qdim + 2 * kvdim, dtype=torch.bfloat16, device=dev)
             obuf = torch.empty(T, cfg.hidden_size, dtype=torch.bfloat16, device=dev)
@@ -202,8 +208,7 @@ class Qwen3MoEModel:
 
         if T < cfg.moe_grouped_threshold:
             # decode path: GEMV per (token, expert) pair
-            pair_token = torch.arange(T, device=hbuf.device,
-                                      dtype=torch.int32).repeat_interleave(K)
+            pair_token = self._pair_token_flat
             pair_expert = topk_ids.flatten().contiguous()
             pair_w = topk_w.flatten().contiguous()
             P = pair_token.numel()
@@ -216,8 +221,7 @@ class Qwen3MoEModel:
         flat_expert = topk_ids.flatten()
         order = torch.argsort(flat_expert)
         pair_expert = flat_expert[order].int().contiguous()
-        pair_token = (torch.arange(T, device=hbuf.device, dtype=torch.int32)
-                      .repeat_interleave(K))[order].contiguous()
+        pair_token = self._pair_token_flat[order].contiguous()
         P = pair_token.numel()
         inv_order = torch.empty_like(order)
         inv_order[order] = torch.arange(P, device=hbuf.device)
@@ -228,9 +232,7 @@ class Qwen3MoEModel:
         h = torch.empty(P, I, dtype=torch.bfloat16, device=hbuf.device)
         ops.silu_mul(h, gateup)
         z = torch.empty(P, H, dtype=torch.bfloat16, device=hbuf.device)
-        ops.moe_grouped_gemm128(z, h, layer.w2,
-                                torch.arange(P, device=hbuf.device, dtype=torch.int32),
-                                desc)
+        ops.moe_grouped_gemm128(z, h, layer.w2, self._arangeP, desc)
         out_bf = torch.empty(T, H, dtype=torch.bfloat16, device=hbuf.device)
         ops.moe_combine_gather(out_bf, z, topk_w.contiguous(), inv_order)
         return out_bf

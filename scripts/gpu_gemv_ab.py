@@ -138,7 +138,8 @@ void run(int variant, torch::Tensor y, torch::Tensor x, torch::Tensor w) {
 }
 """
 
-mod = load_inline(name="gemv_ab", cpp_sources="", cuda_sources=SRC,
+CPP = "void run(int variant, torch::Tensor y, torch::Tensor x, torch::Tensor w);"
+mod = load_inline(name="gemv_ab", cpp_sources=CPP, cuda_sources=SRC,
                   functions=["run"], with_cuda=True, verbose=False,
                   extra_cuda_cflags=["-O3", "--offload-arch=gfx950"])
 
