@@ -135,9 +135,9 @@ class Qwen3MoEModel:
             obuf = torch.empty(T, cfg.hidden_size, dtype=torch.bfloat16, device=dev)
             router_logits = torch.empty(T, cfg.num_experts, dtype=torch.float32,
                                         device=dev)
-            part = torch.empty(T, cfg.num_q_heads, 16, cfg.head_dim,
+            part = torch.empty(T, cfg.num_q_heads, 32, cfg.head_dim,
                                dtype=torch.float32, device=dev)
-            part_ml = torch.empty(T, cfg.num_q_heads, 16, 2, dtype=torch.float32,
+            part_ml = torch.empty(T, cfg.num_q_heads, 32, 2, dtype=torch.float32,
                                   device=dev)
 
         for li, layer in enumerate(self.layers):

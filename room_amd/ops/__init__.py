@@ -74,7 +74,7 @@ def paged_attention_split(out: torch.Tensor, q: torch.Tensor, kcache: torch.Tens
                           seq_ids: torch.Tensor, q_pos: torch.Tensor,
                           part: torch.Tensor, part_ml: torch.Tensor,
                           scale: float) -> torch.Tensor:
-    """Split-KV flash-decode: part [T, Hq, 16, 128] f32, part_ml [T, Hq, 16, 2]."""
+    """Split-KV flash-decode: part [T, Hq, 32, 128] f32, part_ml [T, Hq, 32, 2]."""
     _require().paged_attention_split(out, q, kcache, vcache, block_table,
                                      seq_ids, q_pos, part, part_ml, scale)
     return out

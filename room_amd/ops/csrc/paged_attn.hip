@@ -198,7 +198,7 @@ void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
 // head); each writes an unnormalized partial (acc, m, l); a merge kernel
 // combines. NSPLITS is static so the decode step stays hipGraph-capturable.
 
-#define NSPLITS 16
+#define NSPLITS 32
 
 __global__ __launch_bounds__(256)
 void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]

@@ -80,8 +80,8 @@ qpos = torch.full((B,), 512, dtype=torch.int32, device=DEV)
 total += bench("write_kv",
                lambda: ops.write_kv(kcache, vcache, k, k, bt, seq_ids, qpos))
 attn = torch.empty_like(q)
-part = torch.empty(B, 32, 16, 128, dtype=torch.float32, device=DEV)
-part_ml = torch.empty(B, 32, 16, 2, dtype=torch.float32, device=DEV)
+part = torch.empty(B, 32, 32, 128, dtype=torch.float32, device=DEV)
+part_ml = torch.empty(B, 32, 32, 2, dtype=torch.float32, device=DEV)
 total += bench("paged_attn_split seq=512",
                lambda: ops.paged_attention_split(attn, q, kcache, vcache, bt,
                                                  seq_ids, qpos, part, part_ml,
