@@ -113,9 +113,6 @@ def main() -> None:
             for wid in agent_ids])
         return durations
 
-    async def _noop():
-        return None
-
     queen_latencies: list[float] = []
 
     def run_step() -> None:

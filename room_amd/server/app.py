@@ -750,6 +750,49 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
             pass
         return out
 
+    @app.get("/api/templates")
+    async def list_templates(role: str = Depends(get_role)):
+        from ..core.templates import list_templates as lt
+        return lt()
+
+    @app.post("/api/rooms/from-template")
+    async def room_from_template(payload: dict = Body(...),
+                                 role: str = Depends(rw)):
+        from ..core.templates import instantiate_room_template
+        with ldb as db:
+            try:
+                room = instantiate_room_template(
+                    db, payload["template"], payload["name"],
+                    worker_model=payload.get("worker_model", "qwen3-coder-30b"))
+            except ValueError as e:
+                raise HTTPException(422, str(e))
+        bus.emit("rooms", "room_created", {"id": room["id"]})
+        return room
+
+    @app.post("/api/rooms/{room_id}/prompts/export")
+    async def export_prompts(room_id: int, role: str = Depends(rw)):
+        from ..core.prompt_sync import export_worker_prompts
+        with ldb as db:
+            files = export_worker_prompts(db, room_id)
+        return {"files": files}
+
+    @app.post("/api/rooms/{room_id}/prompts/import")
+    async def import_prompts(room_id: int, payload: dict = Body(default={}),
+                             role: str = Depends(rw)):
+        from ..core.prompt_sync import import_worker_prompts
+        with ldb as db:
+            return import_worker_prompts(db, room_id,
+                                         force=payload.get("force", False))
+
+    @app.get("/api/rooms/{room_id}/profile")
+    async def room_profile(room_id: int, role: str = Depends(get_role)):
+        from ..core.public_feed import get_public_room_profile
+        with ldb as db:
+            prof = get_public_room_profile(db, room_id)
+        if prof is None:
+            raise HTTPException(404, "room not public")
+        return prof
+
     @app.get("/api/feed")
     async def public_feed(limit: int = 50, role: str = Depends(get_role)):
         from ..core.public_feed import get_public_feed

@@ -23,12 +23,16 @@ INDEXER_S = 10.0
 
 class ServerRuntime:
     def __init__(self, ldb: LockedDb, runner: TaskRunner, loop_mgr=None,
-                 memory=None, bus=None):
+                 memory=None, bus=None, commentary_model: str | None = None):
         self.ldb = ldb
         self.runner = runner
         self.loop_mgr = loop_mgr
         self.memory = memory
         self.bus = bus
+        self.commentary = None
+        if bus is not None and commentary_model:
+            from ..core.clerk import CommentaryEngine
+            self.commentary = CommentaryEngine(ldb, bus, model=commentary_model)
         self._tasks: list[asyncio.Task] = []
         self._stop = asyncio.Event()
         self._last_cron_minute: dict[int, str] = {}
@@ -43,6 +47,8 @@ class ServerRuntime:
             asyncio.create_task(self._inbox_loop()),
             asyncio.create_task(self._indexer_loop()),
         ]
+        if self.commentary is not None:
+            self._tasks.append(asyncio.create_task(self._commentary_loop()))
 
     async def stop(self) -> None:
         self._stop.set()
@@ -117,6 +123,16 @@ class ServerRuntime:
             except Exception:
                 pass
             if await self._sleep(INBOX_POLL_S):
+                return
+
+    # --- clerk commentary narration
+    async def _commentary_loop(self) -> None:
+        while not self._stop.is_set():
+            try:
+                await asyncio.to_thread(self.commentary.tick)
+            except Exception:
+                pass
+            if await self._sleep(self.commentary.pace_s):
                 return
 
     # --- background embedding indexer

@@ -236,3 +236,20 @@ def test_websocket_subscribe_and_events(server):
         msg = ws.receive_json()
         assert msg["type"] == "room_created"
         assert msg["channel"] == "rooms"
+
+
+def test_templates_and_prompt_sync_api(server, tmp_path, monkeypatch):
+    client, h, *_ = server
+    monkeypatch.setenv("ROOMAMD_DATA_DIR", str(tmp_path))
+    t = client.get("/api/templates", headers=h).json()
+    assert "saas-builder" in t["rooms"]
+    room = client.post("/api/rooms/from-template",
+                       json={"template": "research-lab", "name": "lab",
+                             "worker_model": "stub"}, headers=h).json()
+    workers = client.get(f"/api/rooms/{room['id']}/workers", headers=h).json()
+    assert len(workers) == 4  # queen + 3
+    out = client.post(f"/api/rooms/{room['id']}/prompts/export", headers=h).json()
+    assert len(out["files"]) == 4
+    imp = client.post(f"/api/rooms/{room['id']}/prompts/import", json={},
+                      headers=h).json()
+    assert len(imp) == 4
