@@ -4,7 +4,6 @@ system entries, callback for WS fan-out)."""
 from __future__ import annotations
 
 import time
-from typing import Optional
 
 from ..db import LockedDb
 from ..db import queries as q

@@ -10,8 +10,7 @@ import re
 import time
 from dataclasses import dataclass
 
-from .constants import (RATE_LIMIT_MAX_RETRIES, RATE_LIMIT_MAX_WAIT_MS,
-                        RATE_LIMIT_MIN_WAIT_MS)
+from .constants import RATE_LIMIT_MAX_WAIT_MS, RATE_LIMIT_MIN_WAIT_MS
 
 RATE_LIMIT_PATTERNS = [
     re.compile(r"rate.?limit", re.I),

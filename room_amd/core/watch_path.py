@@ -2,7 +2,6 @@
 only, denies .ssh/.aws/etc., resolves symlinks)."""
 from __future__ import annotations
 
-import os
 import tempfile
 from pathlib import Path
 

@@ -9,7 +9,6 @@ deployments.
 """
 from __future__ import annotations
 
-import json
 import re
 import urllib.error
 import urllib.parse

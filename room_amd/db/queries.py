@@ -14,13 +14,8 @@ import time
 from datetime import datetime, timedelta
 from typing import Any, Iterable, Optional
 
-from ..core.constants import (
-    DEFAULT_ROOM_CONFIG,
-    HYBRID_FTS_WEIGHT,
-    HYBRID_RRF_K,
-    HYBRID_SEMANTIC_WEIGHT,
-    EMBEDDING_DIM,
-)
+from ..core.constants import (DEFAULT_ROOM_CONFIG, HYBRID_FTS_WEIGHT,
+                              HYBRID_RRF_K, HYBRID_SEMANTIC_WEIGHT)
 
 # ---------------------------------------------------------------- helpers
 

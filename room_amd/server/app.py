@@ -13,19 +13,19 @@ import asyncio
 import json
 import time
 from collections import defaultdict, deque
-from typing import Any, Optional
+from typing import Optional
 
 from fastapi import (Body, Depends, FastAPI, HTTPException, Query, Request,
                      WebSocket, WebSocketDisconnect)
-from fastapi.responses import HTMLResponse, JSONResponse
+from fastapi.responses import HTMLResponse
 
 from ..core import goals as goals_mod
 from ..core import quorum as quorum_mod
 from ..core import room as room_mod
-from ..core import self_mod, skills as skills_mod, wallet as wallet_mod
+from ..core import self_mod, wallet as wallet_mod
 from ..core.constants import WEBHOOK_RATE_LIMIT_PER_MIN
 from ..core.events import EventBus
-from ..core.secret_store import decrypt_secret, encrypt_secret
+from ..core.secret_store import encrypt_secret
 from ..db import LockedDb
 from ..db import queries as q
 from .auth import ROLE_MEMBER, AuthManager, member_can_write

@@ -7,9 +7,7 @@ via POST /api/rooms/{id}/start.
 """
 from __future__ import annotations
 
-import asyncio
 import os
-from pathlib import Path
 
 from ..core.agent_loop import AgentLoopManager
 from ..core.events import EventBus

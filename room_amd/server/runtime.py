@@ -8,7 +8,6 @@ from __future__ import annotations
 
 import asyncio
 from datetime import datetime
-from typing import Optional
 
 from ..core.cron import CronExpression
 from ..core.tasks import TaskRunner
