@@ -153,6 +153,29 @@ def moe_combine(out: torch.Tensor, z: torch.Tensor, pair_w: torch.Tensor,
     return out
 
 
+def moe_gemv_dedup(out: torch.Tensor, x: torch.Tensor, w13: torch.Tensor,
+                   w2: torch.Tensor, topk_ids: torch.Tensor,
+                   topk_w: torch.Tensor, scratch: dict) -> torch.Tensor:
+    """Expert-deduped decode MoE: weight rows read once per active expert.
+    scratch holds persistent buffers keyed by (E, I): counts/tok_list/w_list/h."""
+    E = w13.size(0)
+    I = w2.size(2)
+    key = (E, I)
+    if key not in scratch:
+        dev = x.device
+        scratch[key] = {
+            "counts": torch.zeros(E, dtype=torch.int32, device=dev),
+            "tok_list": torch.zeros(E, 64, dtype=torch.int32, device=dev),
+            "w_list": torch.zeros(E, 64, dtype=torch.float32, device=dev),
+            "h": torch.zeros(E * 64, I, dtype=torch.bfloat16, device=dev),
+        }
+    sc = scratch[key]
+    _require().moe_gemv_dedup(out, x, w13, w2, topk_ids.contiguous(),
+                              topk_w.contiguous(), sc["counts"],
+                              sc["tok_list"], sc["w_list"], sc["h"])
+    return out
+
+
 def moe_build_desc_device(pair_expert_sorted: torch.Tensor, num_experts: int,
                           bm: int = 128) -> torch.Tensor:
     """Sync-free [GMAX,3] (expert,row0,msize) descriptors; GMAX computed from

@@ -26,6 +26,11 @@ void moe_grouped_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                       torch::Tensor pair_token, torch::Tensor tile_desc);
 void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                          torch::Tensor pair_token, torch::Tensor tile_desc);
+void moe_gemv_dedup(torch::Tensor out, torch::Tensor x, torch::Tensor w13,
+                    torch::Tensor w2, torch::Tensor topk_ids,
+                    torch::Tensor topk_w, torch::Tensor counts,
+                    torch::Tensor tok_list, torch::Tensor w_list,
+                    torch::Tensor h);
 void moe_build_desc(torch::Tensor desc, torch::Tensor counts, int64_t bm);
 void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
                  torch::Tensor pair_token);
@@ -66,6 +71,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_gemv_down", &moe_gemv_down, "MoE down GEMV + weighted scatter-add");
   m.def("moe_grouped_gemm", &moe_grouped_gemm, "grouped MFMA GEMM (prefill)");
   m.def("moe_grouped_gemm128", &moe_grouped_gemm128, "BM=128 grouped MFMA GEMM");
+  m.def("moe_gemv_dedup", &moe_gemv_dedup, "expert-deduped decode MoE");
   m.def("moe_build_desc", &moe_build_desc, "device-side tile desc builder");
   m.def("moe_combine", &moe_combine, "weighted scatter-add combine");
   m.def("moe_combine_gather", &moe_combine_gather, "atomics-free MoE combine");

@@ -494,3 +494,171 @@ void moe_build_desc(torch::Tensor desc, torch::Tensor counts, int64_t bm) {
                      gmax);
   HIP_CHECK_KERNEL();
 }
+
+// ------------------------------------------------ deduped decode GEMV (v2)
+// The pair-GEMV reads an expert's gate/up/down rows once per (token,expert)
+// pair; with B×K pairs over 128 experts many experts serve several tokens
+// (B=5: ~40 pairs → ~35 unique experts). v2 groups pairs by expert with a
+// tiny list-building kernel (static [E, CAP] shapes → hipGraph-safe), then
+// each weight row is read ONCE and applied to all of that expert's tokens
+// (x/h rows are L2-resident at decode sizes).
+
+#define MD_CAP 64   // max pairs per expert per step (B≤8 × K=8)
+
+__global__ void moe_build_lists_kernel(int* __restrict__ counts,   // [E]
+                                       int* __restrict__ tok_list,  // [E, CAP]
+                                       float* __restrict__ w_list,  // [E, CAP]
+                                       const int* __restrict__ topk_ids,  // [T,K]
+                                       const float* __restrict__ topk_w,
+                                       int T, int K, int E) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i == 0) {
+    // zero counts first (single wavefront; E ≤ 128)
+  }
+  // two-phase in one launch is racy across blocks; use one block
+  if (blockIdx.x != 0) return;
+  for (int e = threadIdx.x; e < E; e += blockDim.x) counts[e] = 0;
+  __syncthreads();
+  for (int p = threadIdx.x; p < T * K; p += blockDim.x) {
+    const int e = topk_ids[p];
+    const int slot = atomicAdd(&counts[e], 1);
+    if (slot < MD_CAP) {
+      tok_list[e * MD_CAP + slot] = p / K;   // token index
+      w_list[e * MD_CAP + slot] = topk_w[p];
+    }
+  }
+}
+
+// h[e*CAP + i][I] = silu(x_t · g_j) * (x_t · u_j); grid (E, I/4), 4 waves.
+// CHUNKS = H / 512 (bf16x8 row-cache registers per lane); templated so the
+// register file stays statically indexed (guide rule #20).
+template <int CHUNKS>
+__global__ __launch_bounds__(256)
+void moe_gemv_h2_kernel(short* __restrict__ h,           // [E*CAP, I]
+                        const short* __restrict__ x,      // [T, H]
+                        const short* __restrict__ w13,    // [E, 2I, H]
+                        const int* __restrict__ counts,
+                        const int* __restrict__ tok_list,
+                        int H, int I) {
+  const int e = blockIdx.x;
+  const int n = counts[e];
+  if (n == 0) return;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int j = blockIdx.y * 4 + wid;
+  if (j >= I) return;
+  const short* grow = w13 + ((long)e * 2 * I + j) * H;
+  const short* urow = w13 + ((long)e * 2 * I + I + j) * H;
+
+  bf16x8 greg[CHUNKS], ureg[CHUNKS];
+  #pragma unroll
+  for (int c = 0; c < CHUNKS; ++c) {
+    greg[c] = *reinterpret_cast<const bf16x8*>(grow + (lane * CHUNKS + c) * 8);
+    ureg[c] = *reinterpret_cast<const bf16x8*>(urow + (lane * CHUNKS + c) * 8);
+  }
+  const int lim = min(n, MD_CAP);
+  for (int i = 0; i < lim; ++i) {
+    const int t = tok_list[e * MD_CAP + i];
+    const short* xrow = x + (long)t * H;
+    float dg = 0.f, du = 0.f;
+    #pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      bf16x8 xv = *reinterpret_cast<const bf16x8*>(xrow + (lane * CHUNKS + c) * 8);
+      #pragma unroll
+      for (int q_ = 0; q_ < 8; ++q_) {
+        const float xf = bf2f(xv[q_]);
+        dg += xf * bf2f(greg[c][q_]);
+        du += xf * bf2f(ureg[c][q_]);
+      }
+    }
+    dg = wave_reduce_sum(dg);
+    du = wave_reduce_sum(du);
+    if (lane == 0) {
+      const float s = dg / (1.0f + __expf(-dg));
+      h[((long)e * MD_CAP + i) * I + j] = f2bf(s * du);
+    }
+  }
+}
+
+// out[t] += w_p * (h_row · w2_row_o); grid (E, H/4), 4 waves.
+__global__ __launch_bounds__(256)
+void moe_gemv_down2_kernel(float* __restrict__ out,       // [T, H] f32
+                           const short* __restrict__ h,    // [E*CAP, I]
+                           const short* __restrict__ w2,   // [E, H, I]
+                           const int* __restrict__ counts,
+                           const int* __restrict__ tok_list,
+                           const float* __restrict__ w_list,
+                           int H, int I) {
+  const int e = blockIdx.x;
+  const int n = counts[e];
+  if (n == 0) return;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int o = blockIdx.y * 4 + wid;
+  if (o >= H) return;
+  const short* wrow = w2 + ((long)e * H + o) * I;
+  // cache the down row: I/64 bf16 per lane (12 for I=768, 4 for I=256);
+  // fully unrolled with a static guard so wreg stays in registers (rule #20)
+  const int per_lane = I / WAVE;
+  float wreg[16];
+  #pragma unroll
+  for (int c = 0; c < 16; ++c)
+    wreg[c] = (c < per_lane) ? bf2f(wrow[lane * per_lane + c]) : 0.f;
+  const int lim = min(n, MD_CAP);
+  for (int i = 0; i < lim; ++i) {
+    const short* hrow = h + ((long)e * MD_CAP + i) * I + lane * per_lane;
+    float d = 0.f;
+    #pragma unroll
+    for (int c = 0; c < 16; ++c)
+      if (c < per_lane) d += wreg[c] * bf2f(hrow[c]);
+    d = wave_reduce_sum(d);
+    if (lane == 0) {
+      const int t = tok_list[e * MD_CAP + i];
+      atomicAdd(out + (long)t * H + o, d * w_list[e * MD_CAP + i]);
+    }
+  }
+}
+
+void moe_gemv_dedup(torch::Tensor out, torch::Tensor x, torch::Tensor w13,
+                    torch::Tensor w2, torch::Tensor topk_ids,
+                    torch::Tensor topk_w, torch::Tensor counts,
+                    torch::Tensor tok_list, torch::Tensor w_list,
+                    torch::Tensor h) {
+  const int T = x.size(0), H = x.size(1);
+  const int K = topk_ids.size(1), E = w13.size(0);
+  const int I = w2.size(2);
+  TORCH_CHECK(T * K <= 128 * MD_CAP && out.dtype() == torch::kFloat32);
+  TORCH_CHECK(tok_list.size(0) == E && tok_list.size(1) == MD_CAP);
+  TORCH_CHECK(h.size(0) == (long)E * MD_CAP && h.size(1) == I);
+  TORCH_CHECK(I % WAVE == 0 && I / WAVE <= 16 && H % (WAVE * 8) == 0);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(moe_build_lists_kernel, dim3(1), dim3(256), 0, s,
+                     counts.data_ptr<int>(), tok_list.data_ptr<int>(),
+                     w_list.data_ptr<float>(), topk_ids.data_ptr<int>(),
+                     topk_w.data_ptr<float>(), T, K, E);
+  HIP_CHECK_KERNEL();
+  const int chunks = H / (WAVE * 8);
+  TORCH_CHECK(chunks == 1 || chunks == 2 || chunks == 4,
+              "moe_gemv_dedup supports H in {512, 1024, 2048}");
+  if (chunks == 1)
+    hipLaunchKernelGGL(moe_gemv_h2_kernel<1>, dim3(E, (I + 3) / 4), dim3(256),
+                       0, s, (short*)h.data_ptr(), (const short*)x.data_ptr(),
+                       (const short*)w13.data_ptr(), counts.data_ptr<int>(),
+                       tok_list.data_ptr<int>(), H, I);
+  else if (chunks == 2)
+    hipLaunchKernelGGL(moe_gemv_h2_kernel<2>, dim3(E, (I + 3) / 4), dim3(256),
+                       0, s, (short*)h.data_ptr(), (const short*)x.data_ptr(),
+                       (const short*)w13.data_ptr(), counts.data_ptr<int>(),
+                       tok_list.data_ptr<int>(), H, I);
+  else
+    hipLaunchKernelGGL(moe_gemv_h2_kernel<4>, dim3(E, (I + 3) / 4), dim3(256),
+                       0, s, (short*)h.data_ptr(), (const short*)x.data_ptr(),
+                       (const short*)w13.data_ptr(), counts.data_ptr<int>(),
+                       tok_list.data_ptr<int>(), H, I);
+  HIP_CHECK_KERNEL();
+  hipLaunchKernelGGL(moe_gemv_down2_kernel, dim3(E, (H + 3) / 4), dim3(256), 0, s,
+                     out.data_ptr<float>(), (const short*)h.data_ptr(),
+                     (const short*)w2.data_ptr(), counts.data_ptr<int>(),
+                     tok_list.data_ptr<int>(), w_list.data_ptr<float>(), H, I);
+  HIP_CHECK_KERNEL();
+}

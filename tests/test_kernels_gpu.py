@@ -340,3 +340,18 @@ def test_flash_prefill_vs_ref():
     ops.flash_prefill(out, q, kcache, vcache, bt, seq_ids, q_pos, desc, scale)
     expect = ref.paged_attention_ref(q, kcache, vcache, bt, seq_ids, q_pos, scale)
     assert bf16_close(out, expect, atol=4e-2, rtol=4e-2)
+
+
+def test_moe_gemv_dedup_matches_ref():
+    torch.manual_seed(17)
+    x, w13, w2, ids, w = _moe_setup(T=6)
+    T, H = x.shape
+    scratch = {}
+    out = torch.zeros(T, H, dtype=torch.float32, device=DEV)
+    ops.moe_gemv_dedup(out, x, w13, w2, ids, w, scratch)
+    expect = ref.moe_ref(x, w13, w2, ids, w)
+    assert bf16_close(out, expect, atol=5e-2, rtol=5e-2)
+    # second call must reset counts correctly (persistent scratch)
+    out2 = torch.zeros(T, H, dtype=torch.float32, device=DEV)
+    ops.moe_gemv_dedup(out2, x, w13, w2, ids, w, scratch)
+    assert torch.allclose(out, out2, atol=1e-3)
