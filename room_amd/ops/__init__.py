@@ -168,11 +168,13 @@ def moe_gemv_dedup(out: torch.Tensor, x: torch.Tensor, w13: torch.Tensor,
             "tok_list": torch.zeros(E, 64, dtype=torch.int32, device=dev),
             "w_list": torch.zeros(E, 64, dtype=torch.float32, device=dev),
             "h": torch.zeros(E * 64, I, dtype=torch.bfloat16, device=dev),
+            "active": torch.full((512,), -1, dtype=torch.int32, device=dev),
         }
     sc = scratch[key]
     _require().moe_gemv_dedup(out, x, w13, w2, topk_ids.contiguous(),
                               topk_w.contiguous(), sc["counts"],
-                              sc["tok_list"], sc["w_list"], sc["h"])
+                              sc["tok_list"], sc["w_list"], sc["h"],
+                              sc["active"])
     return out
 
 

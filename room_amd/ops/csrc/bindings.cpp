@@ -30,7 +30,7 @@ void moe_gemv_dedup(torch::Tensor out, torch::Tensor x, torch::Tensor w13,
                     torch::Tensor w2, torch::Tensor topk_ids,
                     torch::Tensor topk_w, torch::Tensor counts,
                     torch::Tensor tok_list, torch::Tensor w_list,
-                    torch::Tensor h);
+                    torch::Tensor h, torch::Tensor active);
 void moe_build_desc(torch::Tensor desc, torch::Tensor counts, int64_t bm);
 void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
                  torch::Tensor pair_token);
