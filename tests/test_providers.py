@@ -1,0 +1,110 @@
+"""Executor-seam tests: tool-call parsing, the multi-turn loop, session
+compression, crash-resilience persistence."""
+import json
+
+from room_amd.core import room
+from room_amd.db import queries as q
+from room_amd.engine.providers import (StubEngine, compress_session,
+                                       execute_agent, parse_tool_calls,
+                                       register_engine, render_tool_call)
+from room_amd.engine.types import AgentExecutionOptions, ToolDef
+
+
+def test_parse_tool_calls_variants():
+    text = ('thinking...\n<tool_call>{"name": "a", "arguments": {"x": 1}}'
+            '</tool_call>\nand\n<tool_call>{"name": "b"}</tool_call>')
+    calls = parse_tool_calls(text)
+    assert [c.name for c in calls] == ["a", "b"]
+    assert calls[0].arguments == {"x": 1}
+    assert calls[1].arguments == {}
+    assert parse_tool_calls("no calls here") == []
+    assert parse_tool_calls("<tool_call>not json</tool_call>") == []
+
+
+def test_render_roundtrip():
+    text = render_tool_call("room_set_goal", {"description": "d"})
+    calls = parse_tool_calls(text)
+    assert calls[0].name == "room_set_goal"
+    assert calls[0].arguments == {"description": "d"}
+
+
+def test_execute_agent_multi_turn_loop():
+    """Scripted engine: turn 1 emits two tool calls, turn 2 finishes."""
+    eng = StubEngine(scripted=[
+        "working\n" + render_tool_call("t1", {"a": 1}) + render_tool_call("t2", {}),
+        "done, no more calls",
+    ])
+    register_engine("scripted-1", eng)
+    executed = []
+
+    def tool_exec(call):
+        executed.append(call.name)
+        return json.dumps({"ok": call.name})
+
+    res = execute_agent(AgentExecutionOptions(
+        prompt="go", model="scripted-1", max_turns=5,
+        tools=[ToolDef("t1", "", {}), ToolDef("t2", "", {})],
+        tool_executor=tool_exec))
+    assert res.success
+    assert executed == ["t1", "t2"]
+    assert res.turns_used == 2
+    assert res.tool_calls_executed == 2
+    # tool results were appended to the conversation
+    roles = [m["role"] for m in res.messages]
+    assert roles.count("tool") == 2
+    assert res.text == "done, no more calls"
+
+
+def test_execute_agent_max_turns_cap():
+    looping = render_tool_call("t", {})
+    eng = StubEngine(scripted=[looping] * 10)
+    register_engine("scripted-2", eng)
+    res = execute_agent(AgentExecutionOptions(
+        prompt="go", model="scripted-2", max_turns=3,
+        tools=[ToolDef("t", "", {})], tool_executor=lambda c: "{}"))
+    assert res.turns_used == 3
+    assert res.tool_calls_executed == 3
+
+
+def test_compress_session_trims_and_summarizes():
+    msgs = [{"role": "system", "content": "sys"}]
+    for i in range(40):
+        msgs.append({"role": "user", "content": f"msg {i}"})
+        msgs.append({"role": "assistant", "content": f"reply {i}"})
+    out = compress_session(msgs, model="stub")
+    assert len(out) <= 40
+    assert out[0]["role"] == "system"
+    assert any("[session summary]" in str(m.get("content")) for m in out)
+    # recent tail preserved
+    assert out[-1]["content"] == "reply 39"
+
+
+def test_compress_session_noop_below_threshold():
+    msgs = [{"role": "user", "content": "a"}] * 10
+    assert compress_session(msgs, model="stub") == msgs
+
+
+def test_crash_resilience_boot_cleanup(db):
+    """Running cycles/runs are marked failed on boot (reference
+    db-queries.ts:2388-2393)."""
+    r = room.create_room(db, "crash", worker_model="stub")
+    cid = q.create_worker_cycle(db, r["queen_worker_id"], r["id"])
+    t = q.create_task(db, "t", "p", room_id=r["id"])
+    rid = q.create_task_run(db, t["id"])
+    assert q.get_worker_cycle(db, cid)["status"] == "running"
+    n1 = q.cleanup_stale_cycles(db)
+    n2 = q.cleanup_all_running_runs(db)
+    assert n1 == 1 and n2 == 1
+    assert q.get_worker_cycle(db, cid)["error_message"] == "Server restarted"
+    assert q.get_task_run(db, rid)["status"] == "failed"
+
+
+def test_engine_error_propagates():
+    class Boom:
+        def chat(self, *a, **k):
+            raise RuntimeError("engine exploded")
+
+    register_engine("boom", Boom())
+    res = execute_agent(AgentExecutionOptions(prompt="x", model="boom"))
+    assert not res.success
+    assert "engine exploded" in res.error
