@@ -558,11 +558,13 @@ void moe_gemv_h2_kernel(short* __restrict__ h,           // [E*CAP, I]
   const short* grow = w13 + ((long)e * 2 * I + j) * H;
   const short* urow = w13 + ((long)e * 2 * I + I + j) * H;
 
+  // lane-interleaved layout: instruction c reads a contiguous 1 KiB
+  // (64 lanes × 16 B) — lane-consecutive chunks were 4× uncoalesced
   bf16x8 greg[CHUNKS], ureg[CHUNKS];
   #pragma unroll
   for (int c = 0; c < CHUNKS; ++c) {
-    greg[c] = *reinterpret_cast<const bf16x8*>(grow + (lane * CHUNKS + c) * 8);
-    ureg[c] = *reinterpret_cast<const bf16x8*>(urow + (lane * CHUNKS + c) * 8);
+    greg[c] = *reinterpret_cast<const bf16x8*>(grow + (c * WAVE + lane) * 8);
+    ureg[c] = *reinterpret_cast<const bf16x8*>(urow + (c * WAVE + lane) * 8);
   }
   const int lim = min(n, MD_CAP);
   for (int i = 0; i < lim; ++i) {
@@ -571,7 +573,7 @@ void moe_gemv_h2_kernel(short* __restrict__ h,           // [E*CAP, I]
     float dg = 0.f, du = 0.f;
     #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
-      bf16x8 xv = *reinterpret_cast<const bf16x8*>(xrow + (lane * CHUNKS + c) * 8);
+      bf16x8 xv = *reinterpret_cast<const bf16x8*>(xrow + (c * WAVE + lane) * 8);
       #pragma unroll
       for (int q_ = 0; q_ < 8; ++q_) {
         const float xf = bf2f(xv[q_]);
@@ -612,15 +614,15 @@ void moe_gemv_down2_kernel(float* __restrict__ out,       // [T, H] f32
   const int per_lane = I / WAVE;
   float wreg[16];
   #pragma unroll
-  for (int c = 0; c < 16; ++c)
-    wreg[c] = (c < per_lane) ? bf2f(wrow[lane * per_lane + c]) : 0.f;
+  for (int c = 0; c < 16; ++c)   // lane-interleaved: coalesced per instruction
+    wreg[c] = (c < per_lane) ? bf2f(wrow[c * WAVE + lane]) : 0.f;
   const int lim = min(n, MD_CAP);
   for (int i = 0; i < lim; ++i) {
-    const short* hrow = h + ((long)e * MD_CAP + i) * I + lane * per_lane;
+    const short* hrow = h + ((long)e * MD_CAP + i) * I;
     float d = 0.f;
     #pragma unroll
     for (int c = 0; c < 16; ++c)
-      if (c < per_lane) d += wreg[c] * bf2f(hrow[c]);
+      if (c < per_lane) d += wreg[c] * bf2f(hrow[c * WAVE + lane]);
     d = wave_reduce_sum(d);
     if (lane == 0) {
       const int t = tok_list[e * MD_CAP + i];
