@@ -91,7 +91,9 @@ class Qwen3MoEModel:
         self.final_norm_w = torch.ones(h, dtype=torch.bfloat16, device=self.device)
         self.lm_head = torch.empty(cfg.vocab_size, h, dtype=torch.bfloat16,
                                    device=self.device).normal_(0, 0.02, generator=gen)
+        import os as _os
         self._moe_scratch: dict = {}   # dedup-GEMV persistent buffers
+        self.moe_dedup = _os.environ.get('ROOMAMD_MOE_DEDUP') == '1'
         cos_t, sin_t = rope_tables(cfg.max_position, cfg.head_dim, cfg.rope_theta)
         self.cos_t = cos_t.to(self.device)
         self.sin_t = sin_t.to(self.device)
@@ -208,10 +210,21 @@ class Qwen3MoEModel:
         out = torch.zeros(T, H, dtype=torch.float32, device=hbuf.device)
 
         if T < cfg.moe_grouped_threshold:
-            # decode path: expert-deduped GEMV (weight rows read once per
-            # active expert, not once per (token, expert) pair)
-            ops.moe_gemv_dedup(out, hbuf, layer.w13, layer.w2, topk_ids,
-                               topk_w, self._moe_scratch)
+            if self.moe_dedup:
+                # expert-deduped GEMV (weight rows once per active expert);
+                # measured slower than the pair path at B<=8 on random routing
+                # -- kept behind ROOMAMD_MOE_DEDUP=1 pending a win
+                ops.moe_gemv_dedup(out, hbuf, layer.w13, layer.w2, topk_ids,
+                                   topk_w, self._moe_scratch)
+                return out
+            # decode path: GEMV per (token, expert) pair
+            pair_token = self._pair_token_flat
+            pair_expert = topk_ids.flatten().contiguous()
+            pair_w = topk_w.flatten().contiguous()
+            P = pair_token.numel()
+            h = torch.empty(P, I, dtype=torch.bfloat16, device=hbuf.device)
+            ops.moe_gemv_h(h, hbuf, layer.w13, pair_token, pair_expert)
+            ops.moe_gemv_down(out, h, layer.w2, pair_w, pair_token, pair_expert)
             return out
 
         # prefill path: sort pairs by expert, grouped MFMA GEMMs
