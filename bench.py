@@ -118,10 +118,16 @@ def main() -> None:
     def run_step() -> None:
         durs = asyncio.run(one_step())
         queen_latencies.append(durs[0])
-        # quorum vote all-gather across GPU shards (xGMI) each swarm step
+        # cross-GPU swarm collectives each step (BASELINE config 3): quorum
+        # vote all-gather + goal/skill context broadcast from the queen rank
         votes = torch.ones(len(agent_ids), dtype=torch.int8,
                            device=ctx.device if use_gpu else "cpu")
         ctx.quorum_tally(votes)
+        if ctx.is_distributed:
+            with ldb as db:
+                digest = {"goal": q.get_room(db, room_id)["goal"],
+                          "step": len(queen_latencies)} if ctx.rank == 0 else None
+            ctx.broadcast_blob(digest, src=0)
 
     for _ in range(args.warmup):
         run_step()

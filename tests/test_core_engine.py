@@ -356,3 +356,24 @@ def test_send_token_validation(db):
     w = q.get_room_wallet(db, r["id"])
     txs = q.list_wallet_txs(db, w["id"])
     assert any(t["type"] == "send" and t["amount"] == "5.0" for t in txs)
+
+
+def test_session_survives_manager_restart(ldb):
+    """Session durability: a NEW AgentLoopManager (fresh process semantics)
+    resumes the conversation from the agent_sessions row (SURVEY §5
+    checkpoint/resume: everything is DB state)."""
+    with ldb as db:
+        r = room.create_room(db, "restart", goal="persist me", worker_model="stub")
+    mgr1 = AgentLoopManager(ldb)
+    asyncio.run(mgr1.run_cycle(r["id"], r["queen_worker_id"]))
+    with ldb as db:
+        sess1 = q.get_agent_session(db, r["queen_worker_id"])
+    n_msgs_1 = len(json.loads(sess1["messages_json"]))
+
+    mgr2 = AgentLoopManager(ldb)  # "restarted server"
+    asyncio.run(mgr2.run_cycle(r["id"], r["queen_worker_id"]))
+    with ldb as db:
+        sess2 = q.get_agent_session(db, r["queen_worker_id"])
+    assert sess2["turn_count"] == 2
+    # the second cycle continued the persisted history
+    assert len(json.loads(sess2["messages_json"])) > n_msgs_1
