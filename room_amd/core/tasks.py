@@ -98,6 +98,11 @@ class TaskRunner:
         task_id = task["id"]
         start = time.time()
 
+        # built-in executors (reference task-runner.ts:256-329): these run
+        # without a model call
+        if task["executor"] in ("keeper_contact_check", "keeper_reminder"):
+            return self._run_builtin(task)
+
         # session continuity: rotate after 20 runs
         session_id = task["session_id"]
         if task["session_continuity"] and task["run_count"] \
@@ -221,6 +226,35 @@ class TaskRunner:
                           {"run_id": run_id, "task_id": task_id, "status": status})
             self.bus.emit(f"run:{run_id}", "finished", {"status": status})
         return {"run_id": run_id, "status": status, "result": result}
+
+    def _run_builtin(self, task: dict) -> dict:
+        """keeper_contact_check: notify the keeper about pending escalations;
+        keeper_reminder: deliver the task prompt as a keeper notification."""
+        from .notifications import notify_keeper
+        t0 = time.time()
+        with self.ldb as db:
+            run_id = q.create_task_run(db, task["id"])
+        result = ""
+        if task["executor"] == "keeper_contact_check":
+            with self.ldb as db:
+                pending = (q.list_escalations(db, task["room_id"], status="pending")
+                           if task["room_id"] else [])
+            if pending:
+                notify_keeper(f"{len(pending)} pending escalation(s)",
+                              "\n".join(e["question"][:200] for e in pending[:10]),
+                              room_id=task["room_id"], bus=self.bus)
+            result = f"checked: {len(pending)} pending"
+        else:  # keeper_reminder
+            notify_keeper(f"Reminder: {task['name']}", task["prompt"],
+                          room_id=task["room_id"], bus=self.bus)
+            result = "reminder delivered"
+        dur = int((time.time() - t0) * 1000)
+        with self.ldb as db:
+            q.finish_task_run(db, run_id, "completed", result=result,
+                              duration_ms=dur)
+            q.update_task(db, task["id"], last_run=q.now_iso(),
+                          last_result=result, run_count=task["run_count"] + 1)
+        return {"run_id": run_id, "status": "completed", "result": result}
 
     def cancel_running_tasks_for_room(self, room_id: int) -> int:
         n = 0

@@ -18,6 +18,7 @@ CRON_REFRESH_S = float(__import__("os").environ.get("ROOMAMD_CRON_REFRESH_S", "1
 MAINTENANCE_S = 60.0
 INBOX_POLL_S = 2.5
 INDEXER_S = 10.0
+WATCHER_S = 5.0
 
 
 class ServerRuntime:
@@ -48,6 +49,8 @@ class ServerRuntime:
         ]
         if self.commentary is not None:
             self._tasks.append(asyncio.create_task(self._commentary_loop()))
+        self._tasks.append(asyncio.create_task(self._watcher_loop()))
+        self._watch_mtimes: dict[int, float] = {}
 
     async def stop(self) -> None:
         self._stop.set()
@@ -132,6 +135,43 @@ class ServerRuntime:
             except Exception:
                 pass
             if await self._sleep(self.commentary.pace_s):
+                return
+
+    # --- file watchers: mtime changes trigger a queen escalation (reference
+    # watcher tool semantics: watched path + action prompt)
+    async def _watcher_loop(self) -> None:
+        import os as _os
+        self._watch_mtimes = getattr(self, "_watch_mtimes", {})
+        while not self._stop.is_set():
+            try:
+                with self.ldb as db:
+                    watches = [w for w in q.list_watches(db)
+                               if w["status"] == "active"]
+                for w in watches:
+                    try:
+                        mtime = _os.path.getmtime(_os.path.expanduser(w["path"]))
+                    except OSError:
+                        continue
+                    prev = self._watch_mtimes.get(w["id"])
+                    self._watch_mtimes[w["id"]] = mtime
+                    if prev is not None and mtime > prev and w["room_id"]:
+                        with self.ldb as db:
+                            q.create_escalation(
+                                db, w["room_id"],
+                                (w["action_prompt"] or "Watched path changed")
+                                + f" (path: {w['path']})")
+                            db.execute(
+                                "UPDATE watches SET last_triggered = ?,"
+                                " trigger_count = trigger_count + 1 WHERE id = ?",
+                                (q.now_iso(), w["id"]))
+                        if self.loop_mgr is not None:
+                            with self.ldb as db:
+                                room = q.get_room(db, w["room_id"])
+                            if room and room.get("queen_worker_id"):
+                                self.loop_mgr.trigger_agent(room["queen_worker_id"])
+            except Exception:
+                pass
+            if await self._sleep(WATCHER_S):
                 return
 
     # --- background embedding indexer

@@ -89,3 +89,24 @@ def test_learned_context_distills(db):
         q.finish_task_run(db, rid, "completed", result=f"result {i}")
     memo = distill_learned_context(db, t["id"], model="stub")
     assert memo
+
+
+def test_builtin_keeper_executors(db, tmp_path, monkeypatch):
+    monkeypatch.setenv("ROOMAMD_DATA_DIR", str(tmp_path))
+    r = room.create_room(db, "builtin", worker_model="stub")
+    q.create_escalation(db, r["id"], "need approval for X")
+    ldb = LockedDb(db)
+    runner = TaskRunner(ldb, default_model="stub")
+    t1 = q.create_task(db, "contact check", "-", trigger_type="manual",
+                       room_id=r["id"], executor="keeper_contact_check")
+    out = asyncio.run(runner.execute_task(t1["id"]))
+    assert out["status"] == "completed"
+    assert "1 pending" in out["result"]
+    t2 = q.create_task(db, "standup", "time for standup", trigger_type="manual",
+                       room_id=r["id"], executor="keeper_reminder")
+    out2 = asyncio.run(runner.execute_task(t2["id"]))
+    assert out2["result"] == "reminder delivered"
+    from room_amd.core.notifications import read_outbox
+    box = read_outbox()
+    assert len(box) == 2
+    assert box[1]["subject"] == "Reminder: standup"
