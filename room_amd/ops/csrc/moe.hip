@@ -325,6 +325,9 @@ void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
 #define G2_BK 64
 #define G2_PAD 8
 
+// BN templated: wider N-tiles cut the X-tile re-read factor (X is re-read
+// once per N-tile; gateup N=1536 at BN=64 → 24×, at BN=128 → 12×).
+template <int BN>
 __global__ __launch_bounds__(512)
 void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
                                 const short* __restrict__ x,   // [T, H]
@@ -337,19 +340,21 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
   if (msize == 0) return;                 // past the live tile count
   const int e = tile_desc[g * 3 + 0];
   const int row0 = tile_desc[g * 3 + 1];
-  const int n0 = blockIdx.y * G2_BN;      // n-tile from the grid
+  const int n0 = blockIdx.y * BN;         // n-tile from the grid
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6;       // 0..7: wave's 16-row m-slice
   const int lane = tid & 63;
 
   __shared__ short xs[G2_BM][G2_BK + G2_PAD];
-  __shared__ short ws[G2_BN][G2_BK + G2_PAD];
+  __shared__ short ws[BN][G2_BK + G2_PAD];
 
   const short* wbase = w + (long)e * N * H;
+  constexpr int NF = BN / 16;             // n-fragments per wave
 
-  cfrag_t acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
-                    {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  cfrag_t acc[NF];
+  #pragma unroll
+  for (int nf = 0; nf < NF; ++nf) acc[nf] = cfrag_t{0.f, 0.f, 0.f, 0.f};
 
   for (int k0 = 0; k0 < H; k0 += G2_BK) {
     // stage X tile: 128 rows × 64 k = 1024 vec8 → 2 per thread
@@ -365,10 +370,12 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
       }
       *reinterpret_cast<bf16x8*>(&xs[r][c]) = v;
     }
-    // stage W tile: 64 rows × 64 k = 512 vec8 → 1 per thread
-    {
-      const int r = (tid * 8) / G2_BK;
-      const int c = (tid * 8) % G2_BK;
+    // stage W tile: BN rows × 64 k → BN/64 vec8 per thread
+    #pragma unroll
+    for (int it = 0; it < BN / 64; ++it) {
+      const int idx = tid + it * 512;
+      const int r = (idx * 8) / G2_BK;
+      const int c = (idx * 8) % G2_BK;
       bf16x8 v = *reinterpret_cast<const bf16x8*>(
           wbase + (long)(n0 + r) * H + k0 + c);
       *reinterpret_cast<bf16x8*>(&ws[r][c]) = v;
@@ -381,7 +388,7 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
       const int akoff = kk * 32 + (lane >> 4) * 8;
       bf16x8 a = *reinterpret_cast<const bf16x8*>(&xs[arow][akoff]);
       #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NF; ++nf) {
         const int bcol = nf * 16 + (lane & 15);
         bf16x8 b = *reinterpret_cast<const bf16x8*>(&ws[bcol][akoff]);
         acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nf], 0, 0, 0);
@@ -393,7 +400,7 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
   const int crow = wid * 16 + (lane >> 4) * 4;
   const int ccol_base = lane & 15;
   #pragma unroll
-  for (int nf = 0; nf < 4; ++nf) {
+  for (int nf = 0; nf < NF; ++nf) {
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int m = crow + r;
@@ -409,12 +416,24 @@ void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   const int G = tile_desc.size(0);
   TORCH_CHECK(H % G2_BK == 0 && N % G2_BN == 0);
   TORCH_CHECK(tile_desc.size(1) == 3, "desc is [G,3]; n-tile comes from grid.y");
-  dim3 grid(G, N / G2_BN), block(512);
   hipStream_t s = c10::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(moe_grouped_gemm128_kernel, grid, block, 0, s,
-                     (short*)out.data_ptr(), (const short*)x.data_ptr(),
-                     (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
-                     tile_desc.data_ptr<int>(), H, N);
+  static const bool wide = []() {
+    const char* v = getenv("ROOMAMD_MOE_BN64");
+    return !(v && v[0] == '1');   // BN=128 default; BN64 opt-out for A/B
+  }();
+  if (wide && N % 128 == 0) {
+    dim3 grid(G, N / 128), block(512);
+    hipLaunchKernelGGL(moe_grouped_gemm128_kernel<128>, grid, block, 0, s,
+                       (short*)out.data_ptr(), (const short*)x.data_ptr(),
+                       (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
+                       tile_desc.data_ptr<int>(), H, N);
+  } else {
+    dim3 grid(G, N / G2_BN), block(512);
+    hipLaunchKernelGGL(moe_grouped_gemm128_kernel<G2_BN>, grid, block, 0, s,
+                       (short*)out.data_ptr(), (const short*)x.data_ptr(),
+                       (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
+                       tile_desc.data_ptr<int>(), H, N);
+  }
   HIP_CHECK_KERNEL();
 }
 
