@@ -159,10 +159,10 @@ class Qwen3MoEModel:
                                                   cfg.head_dim).contiguous()
             v = qkv[:, qdim + kvdim:].reshape(T, cfg.num_kv_heads,
                                               cfg.head_dim).contiguous()
-            ops.qk_norm_rope(q, k, layer.q_norm_w, layer.k_norm_w, self.cos_t,
-                             self.sin_t, q_pos, cfg.num_q_heads, cfg.num_kv_heads,
-                             cfg.head_dim, cfg.rms_eps)
-            ops.write_kv(kcaches[li], vcaches[li], k, v, block_table, seq_ids, q_pos)
+            ops.qk_rope_write_kv(q, k, v, kcaches[li], vcaches[li],
+                                 layer.q_norm_w, layer.k_norm_w, self.cos_t,
+                                 self.sin_t, block_table, seq_ids, q_pos,
+                                 cfg.rms_eps)
             attn = torch.empty_like(q)
             if decode:
                 ops.paged_attention_split(attn, q, kcaches[li], vcaches[li],

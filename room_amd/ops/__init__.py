@@ -49,6 +49,17 @@ def qk_norm_rope(q: torch.Tensor, k: torch.Tensor, q_w: torch.Tensor,
                             n_qheads, n_kvheads, head_dim, eps)
 
 
+def qk_rope_write_kv(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     kcache: torch.Tensor, vcache: torch.Tensor,
+                     q_w: torch.Tensor, k_w: torch.Tensor,
+                     cos_t: torch.Tensor, sin_t: torch.Tensor,
+                     block_table: torch.Tensor, seq_ids: torch.Tensor,
+                     positions: torch.Tensor, eps: float = 1e-6) -> None:
+    """Fused per-head QK RMSNorm + RoPE + paged KV scatter (one launch)."""
+    _require().qk_rope_write_kv(q, k, v, kcache, vcache, q_w, k_w, cos_t,
+                                sin_t, block_table, seq_ids, positions, eps)
+
+
 def silu_mul(out: torch.Tensor, gateup: torch.Tensor) -> torch.Tensor:
     _require().silu_mul(out, gateup)
     return out

@@ -8,6 +8,12 @@ void qk_norm_rope(torch::Tensor q, torch::Tensor k, torch::Tensor q_w,
                   torch::Tensor k_w, torch::Tensor cos_t, torch::Tensor sin_t,
                   torch::Tensor positions, int64_t n_qheads, int64_t n_kvheads,
                   int64_t head_dim, double eps);
+void qk_rope_write_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                      torch::Tensor kcache, torch::Tensor vcache,
+                      torch::Tensor q_w, torch::Tensor k_w,
+                      torch::Tensor cos_t, torch::Tensor sin_t,
+                      torch::Tensor block_table, torch::Tensor seq_ids,
+                      torch::Tensor positions, double eps);
 void silu_mul(torch::Tensor out, torch::Tensor gateup);
 void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                      torch::Tensor vcache, torch::Tensor block_table,
@@ -63,6 +69,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "residual += x; rmsnorm(residual)");
   m.def("qk_norm_rope", &qk_norm_rope, "fused per-head QK RMSNorm + RoPE");
+  m.def("qk_rope_write_kv", &qk_rope_write_kv,
+        "fused QK-norm + RoPE + paged KV write");
   m.def("silu_mul", &silu_mul, "silu(gate) * up");
   m.def("paged_attention", &paged_attention, "paged causal attention (GQA 8:1)");
   m.def("write_kv", &write_kv, "scatter K/V rows into paged cache");

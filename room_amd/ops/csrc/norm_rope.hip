@@ -218,3 +218,91 @@ void silu_mul(torch::Tensor out, torch::Tensor gateup) {
                      (short*)out.data_ptr(), (const short*)gateup.data_ptr(), inter);
   HIP_CHECK_KERNEL();
 }
+
+// ------------------------------------ fused QK-norm + RoPE + KV-cache write
+// One launch replaces qk_norm_rope + write_kv: Q is normed+rotated in place;
+// K is normed+rotated and scattered straight into the paged cache; V is
+// scattered without a separate pass. grid (T, Hq + 2*Hk), 128 threads.
+__global__ void qk_rope_write_kv_kernel(short* __restrict__ q,     // [T, Hq*D]
+                                        const short* __restrict__ k,  // [T, Hk*D]
+                                        const short* __restrict__ v,
+                                        short* __restrict__ kcache,  // [NB,Hk,16,D]
+                                        short* __restrict__ vcache,
+                                        const short* __restrict__ q_w,
+                                        const short* __restrict__ k_w,
+                                        const float* __restrict__ cos_t,
+                                        const float* __restrict__ sin_t,
+                                        const int* __restrict__ block_table,
+                                        const int* __restrict__ seq_ids,
+                                        const int* __restrict__ pos,
+                                        int n_qheads, int n_kvheads,
+                                        int head_dim, int max_blocks, float eps) {
+  const int t = blockIdx.x;
+  const int h = blockIdx.y;
+  const int i = threadIdx.x;
+  const int p = pos[t];
+  const int blk = block_table[(long)seq_ids[t] * max_blocks + p / 16];
+  const long cache_off = ((long)blk * n_kvheads) * 16 * head_dim
+                         + (long)(p % 16) * head_dim;
+
+  if (h >= n_qheads + n_kvheads) {        // V scatter (no rope)
+    const int hv = h - n_qheads - n_kvheads;
+    vcache[cache_off + (long)hv * 16 * head_dim + i] =
+        v[((long)t * n_kvheads + hv) * head_dim + i];
+    return;
+  }
+
+  const bool is_q = h < n_qheads;
+  const short* src = is_q ? (q + ((long)t * n_qheads + h) * head_dim)
+                          : (k + ((long)t * n_kvheads + (h - n_qheads)) * head_dim);
+  const short* w = is_q ? q_w : k_w;
+
+  float x = bf2f(src[i]);
+  __shared__ float red[16];
+  float total = block_reduce_sum(x * x, red);
+  float scale = rsqrtf(total / head_dim + eps);
+  float xn = x * scale * bf2f(w[i]);
+
+  __shared__ float sh[128];
+  sh[i] = xn;
+  __syncthreads();
+  const int half = head_dim / 2;
+  float out;
+  if (i < half) {
+    out = xn * cos_t[(long)p * half + i] - sh[i + half] * sin_t[(long)p * half + i];
+  } else {
+    out = xn * cos_t[(long)p * half + (i - half)]
+          + sh[i - half] * sin_t[(long)p * half + (i - half)];
+  }
+  if (is_q) {
+    q[((long)t * n_qheads + h) * head_dim + i] = f2bf(out);
+  } else {
+    const int hk = h - n_qheads;
+    kcache[cache_off + (long)hk * 16 * head_dim + i] = f2bf(out);
+  }
+}
+
+void qk_rope_write_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                      torch::Tensor kcache, torch::Tensor vcache,
+                      torch::Tensor q_w, torch::Tensor k_w,
+                      torch::Tensor cos_t, torch::Tensor sin_t,
+                      torch::Tensor block_table, torch::Tensor seq_ids,
+                      torch::Tensor positions, double eps) {
+  const int T = positions.size(0);
+  const int n_kvheads = kcache.size(1);
+  const int n_qheads = q.size(1);
+  const int head_dim = q.size(2);
+  TORCH_CHECK(head_dim == 128 && kcache.size(2) == 16);
+  const int max_blocks = block_table.size(1);
+  dim3 grid(T, n_qheads + 2 * n_kvheads), block(head_dim);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(qk_rope_write_kv_kernel, grid, block, 0, s,
+                     (short*)q.data_ptr(), (const short*)k.data_ptr(),
+                     (const short*)v.data_ptr(), (short*)kcache.data_ptr(),
+                     (short*)vcache.data_ptr(), (const short*)q_w.data_ptr(),
+                     (const short*)k_w.data_ptr(), cos_t.data_ptr<float>(),
+                     sin_t.data_ptr<float>(), block_table.data_ptr<int>(),
+                     seq_ids.data_ptr<int>(), positions.data_ptr<int>(),
+                     n_qheads, n_kvheads, head_dim, max_blocks, (float)eps);
+  HIP_CHECK_KERNEL();
+}

@@ -355,3 +355,29 @@ def test_moe_gemv_dedup_matches_ref():
     out2 = torch.zeros(T, H, dtype=torch.float32, device=DEV)
     ops.moe_gemv_dedup(out2, x, w13, w2, ids, w, scratch)
     assert torch.allclose(out, out2, atol=1e-3)
+
+
+def test_qk_rope_write_kv_fused_matches_unfused():
+    torch.manual_seed(18)
+    T, Hq, Hk, D = 9, 32, 4, 128
+    kcache, vcache, bt = _setup_cache(1, 160)
+    kcache2, vcache2 = kcache.clone(), vcache.clone()
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, Hk, D, dtype=torch.bfloat16, device=DEV)
+    q2, k2 = q.clone(), k.clone()
+    q_w = torch.randn(D, dtype=torch.bfloat16, device=DEV)
+    k_w = torch.randn(D, dtype=torch.bfloat16, device=DEV)
+    cos_t, sin_t = ref.rope_tables(512, D, 1e6)
+    cos_t, sin_t = cos_t.to(DEV), sin_t.to(DEV)
+    seq_ids = torch.zeros(T, dtype=torch.int32, device=DEV)
+    pos = torch.arange(40, 40 + T, dtype=torch.int32, device=DEV)
+    # unfused reference path
+    ops.qk_norm_rope(q2, k2, q_w, k_w, cos_t, sin_t, pos, Hq, Hk, D)
+    ops.write_kv(kcache2, vcache2, k2, v, bt, seq_ids, pos)
+    # fused
+    ops.qk_rope_write_kv(q, k, v, kcache, vcache, q_w, k_w, cos_t, sin_t,
+                         bt, seq_ids, pos)
+    assert torch.equal(q, q2)
+    assert torch.equal(vcache, vcache2)
+    assert bf16_close(kcache, kcache2, atol=1e-3)
