@@ -803,10 +803,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
 
     @app.get("/", response_class=HTMLResponse)
     async def index():
-        return ("<html><head><title>room_amd</title></head><body>"
-                "<h1>room_amd — MI355X-native agent swarm runtime</h1>"
-                "<p>API at /api/* (Bearer auth), WebSocket at /ws?token=...</p>"
-                "</body></html>")
+        from .dashboard import DASHBOARD_HTML
+        return DASHBOARD_HTML
 
     # ------------------------------------------------------------ websocket
 

@@ -1,0 +1,157 @@
+"""Built-in dashboard (reference ships a React SPA, src/ui/, served statically
+by the API server; that code is reference material and is not copied). This is
+a self-contained single-file dashboard over the same API + WS: rooms, status,
+workers, goals, decisions, activity, clerk chat, live event stream."""
+
+DASHBOARD_HTML = r"""<!doctype html>
+<html lang="en"><head><meta charset="utf-8">
+<title>room_amd</title>
+<style>
+ :root { color-scheme: dark; }
+ body { font-family: ui-monospace, Menlo, monospace; background:#0d1117;
+        color:#d0d7de; margin:0; }
+ header { padding:10px 16px; background:#161b22; display:flex; gap:16px;
+          align-items:center; border-bottom:1px solid #30363d; }
+ h1 { font-size:15px; margin:0; color:#7ee787; }
+ main { display:grid; grid-template-columns: 260px 1fr 330px; gap:10px;
+        padding:10px; height:calc(100vh - 58px); box-sizing:border-box; }
+ section { background:#161b22; border:1px solid #30363d; border-radius:6px;
+           padding:10px; overflow:auto; }
+ h2 { font-size:12px; text-transform:uppercase; color:#8b949e; margin:0 0 8px; }
+ .room { padding:6px 8px; border-radius:4px; cursor:pointer; }
+ .room:hover, .room.sel { background:#21262d; }
+ .tag { font-size:10px; padding:1px 6px; border-radius:8px; background:#21262d;
+        color:#7ee787; margin-left:6px; }
+ .tag.paused { color:#d29922; } .tag.stopped { color:#f85149; }
+ .ev { font-size:11px; padding:3px 0; border-bottom:1px dotted #21262d; }
+ .bar { background:#21262d; height:6px; border-radius:3px; margin-top:2px; }
+ .bar>div { background:#58a6ff; height:6px; border-radius:3px; }
+ button, input { background:#21262d; color:#d0d7de; border:1px solid #30363d;
+        border-radius:4px; padding:5px 8px; font:inherit; }
+ button:hover { background:#30363d; cursor:pointer; }
+ #chatlog { height:140px; overflow:auto; font-size:11px; }
+ table { width:100%; font-size:11px; border-collapse:collapse; }
+ td, th { text-align:left; padding:3px 4px; border-bottom:1px solid #21262d; }
+ .muted { color:#8b949e; }
+</style></head><body>
+<header>
+ <h1>room_amd</h1><span class="muted" id="status">connecting…</span>
+ <span style="flex:1"></span>
+ <input id="newroom" placeholder="new room name">
+ <button onclick="createRoom()">create room</button>
+</header>
+<main>
+ <section>
+  <h2>Rooms</h2><div id="rooms"></div>
+  <h2 style="margin-top:14px">Clerk</h2>
+  <div id="chatlog"></div>
+  <div style="display:flex;gap:4px;margin-top:6px">
+   <input id="chatin" style="flex:1" placeholder="ask the clerk…"
+          onkeydown="if(event.key==='Enter')clerkSend()">
+   <button onclick="clerkSend()">send</button>
+  </div>
+ </section>
+ <section>
+  <h2 id="roomtitle">Room</h2>
+  <div id="roomctl" style="margin-bottom:8px"></div>
+  <h2>Goals</h2><div id="goals" style="margin-bottom:10px"></div>
+  <h2>Workers</h2><table id="workers"></table>
+  <h2 style="margin-top:10px">Decisions</h2><table id="decisions"></table>
+ </section>
+ <section><h2>Live activity</h2><div id="events"></div></section>
+</main>
+<script>
+let token = localStorage.getItem('roomamd_token');
+let sel = null;
+const $$ = id => document.getElementById(id);
+async function api(path, opts) {
+  const r = await fetch('/api' + path, Object.assign({
+    headers: {'Authorization': 'Bearer ' + token,
+              'Content-Type': 'application/json'}}, opts));
+  if (r.status === 401) { await handshake(); return api(path, opts); }
+  return r.json();
+}
+async function handshake() {
+  const r = await fetch('/api/auth/handshake', {method: 'POST'});
+  token = (await r.json()).token;
+  localStorage.setItem('roomamd_token', token);
+}
+async function loadRooms() {
+  const rooms = await api('/rooms');
+  $$('rooms').innerHTML = rooms.map(r =>
+    `<div class="room ${sel===r.id?'sel':''}" onclick="select(${r.id})">` +
+    `${r.name}<span class="tag ${r.status}">${r.status}</span></div>`).join('');
+  if (sel === null && rooms.length) select(rooms[0].id);
+}
+async function select(id) {
+  sel = id; loadRooms();
+  const st = await api(`/rooms/${id}/status`);
+  $$('roomtitle').textContent = `${st.room.name} — ${st.room.goal || 'no goal'}`;
+  $$('roomctl').innerHTML =
+    `<button onclick="roomAct(${id},'start')">start</button> ` +
+    `<button onclick="roomAct(${id},'pause')">pause</button> ` +
+    `<button onclick="roomAct(${id},'restart')">restart</button>` +
+    ` <span class="muted">cycles: ${st.token_usage.cycles}</span>`;
+  $$('goals').innerHTML = st.goals.map(g =>
+    `<div class="ev">${g.description} <span class="muted">${g.status}</span>` +
+    `<div class="bar"><div style="width:${Math.round((g.progress||0)*100)}%"></div></div></div>`
+  ).join('') || '<span class="muted">none</span>';
+  $$('workers').innerHTML = '<tr><th>name</th><th>role</th><th>state</th></tr>' +
+    st.workers.map(w => `<tr><td>${w.name}</td><td>${w.role||''}</td>` +
+                        `<td>${w.agent_state}</td></tr>`).join('');
+  const decs = await api(`/rooms/${id}/decisions`);
+  $$('decisions').innerHTML = '<tr><th>proposal</th><th>status</th></tr>' +
+    decs.slice(0, 8).map(d => `<tr><td>${d.proposal}</td><td>${d.status}</td></tr>`).join('');
+}
+async function roomAct(id, act) { await api(`/rooms/${id}/${act}`, {method:'POST'}); select(id); }
+async function createRoom() {
+  const name = $$('newroom').value.trim();
+  if (!name) return;
+  await api('/rooms', {method:'POST', body: JSON.stringify({name})});
+  $$('newroom').value = ''; loadRooms();
+}
+async function clerkSend() {
+  const content = $$('chatin').value.trim();
+  if (!content) return;
+  $$('chatin').value = '';
+  chatLine('you', content);
+  const r = await api('/clerk/chat', {method:'POST',
+                                      body: JSON.stringify({content})});
+  chatLine('clerk', r.reply || '(no reply)');
+}
+function chatLine(who, text) {
+  $$('chatlog').innerHTML += `<div class="ev"><b>${who}:</b> ${text}</div>`;
+  $$('chatlog').scrollTop = 1e9;
+}
+function eventLine(e) {
+  $$('events').innerHTML =
+    `<div class="ev"><span class="muted">${e.channel}</span> ${e.type}` +
+    (e.data && e.data.content ? ' — ' + e.data.content : '') + '</div>' +
+    $$('events').innerHTML.slice(0, 40000);
+}
+async function connectWs() {
+  const ws = new WebSocket(
+    `${location.protocol==='https:'?'wss':'ws'}://${location.host}/ws?token=${token}`);
+  ws.onopen = () => {
+    $$('status').textContent = 'live';
+    ws.send(JSON.stringify({type:'subscribe', channel:'*'}));
+    ['rooms','runs','clerk'].forEach(c =>
+      ws.send(JSON.stringify({type:'subscribe', channel:c})));
+  };
+  ws.onmessage = m => {
+    const e = JSON.parse(m.data);
+    if (e.type === 'ping') return;
+    eventLine(e);
+    if (['room_created','cycle_finished'].includes(e.type)) loadRooms();
+  };
+  ws.onclose = () => { $$('status').textContent = 'reconnecting…';
+                       setTimeout(connectWs, 2000); };
+}
+(async () => {
+  if (!token) await handshake();
+  await loadRooms();
+  connectWs();
+  setInterval(() => { if (sel !== null) select(sel); }, 10000);
+})();
+</script></body></html>
+"""

@@ -253,3 +253,10 @@ def test_templates_and_prompt_sync_api(server, tmp_path, monkeypatch):
     imp = client.post(f"/api/rooms/{room['id']}/prompts/import", json={},
                       headers=h).json()
     assert len(imp) == 4
+
+
+def test_dashboard_served(server):
+    client, *_ = server
+    r = client.get("/")
+    assert r.status_code == 200
+    assert "room_amd" in r.text and "connectWs" in r.text
