@@ -74,9 +74,25 @@ def send_token(db: sqlite3.Connection, room_id: int, to_address: str, amount: st
     w = q.get_room_wallet(db, room_id)
     if w is None:
         raise ValueError(f"Room {room_id} has no wallet")
+    # build and sign the ERC-20 transfer (viem-equivalent: EIP-1559 type-2 tx,
+    # RFC6979 deterministic ECDSA); broadcast needs RPC, absent offline, so
+    # the raw tx travels in the pending row (nonce/fees are placeholders
+    # refreshed at broadcast time by a networked deployment)
+    from ..utils.crypto import erc20_transfer_calldata, sign_eip1559_tx
+    cfg = CHAIN_CONFIGS[chain]
+    token_addr = {"usdc": {"base": "0x833589fCD6eDb6E08f4c7C32D4f71b54bdA02913"},
+                  "usdt": {"base": "0xfde4C96c8593536E31F229EA8f37b2ADa2699bb2"}
+                  }.get(token, {}).get(chain, "0x" + "00" * 20)
+    units = int(round(value * 10 ** 6))  # USDC/USDT: 6 decimals
+    priv = bytes.fromhex(decrypt_private_key(w))
+    raw_tx = sign_eip1559_tx(priv, cfg["chainId"], nonce=0,
+                             max_priority_fee=10 ** 9, max_fee=10 ** 10,
+                             gas=100_000, to=token_addr, value=0,
+                             data=erc20_transfer_calldata(to_address, units))
     tx_id = q.log_wallet_tx(db, w["id"], "send", amount, counterparty=to_address,
-                            description=description, status="pending",
-                            category=token)
+                            description=f"{description or ''} raw:{raw_tx[:32]}…",
+                            status="pending", category=token)
     q.log_room_activity(db, room_id, "wallet",
                         f"Send {amount} {token.upper()} on {chain} → {to_address}")
-    return {"tx_id": tx_id, "status": "pending", "chain": chain, "token": token}
+    return {"tx_id": tx_id, "status": "pending", "chain": chain, "token": token,
+            "raw_tx": raw_tx}

@@ -323,3 +323,106 @@ def decrypt_gcm_hex(key: bytes, blob: str) -> str:
     pt = AESGCM(key).decrypt(bytes.fromhex(iv_hex), bytes.fromhex(ct_hex),
                              bytes.fromhex(tag_hex))
     return pt.decode()
+
+
+# ------------------------------------------------------------------ RLP
+
+def rlp_encode(item) -> bytes:
+    """Recursive-length-prefix encoding (Ethereum wire format)."""
+    if isinstance(item, int):
+        if item == 0:
+            item = b""
+        else:
+            item = item.to_bytes((item.bit_length() + 7) // 8, "big")
+    if isinstance(item, str):
+        item = bytes.fromhex(item[2:] if item.startswith("0x") else item)
+    if isinstance(item, (bytes, bytearray)):
+        if len(item) == 1 and item[0] < 0x80:
+            return bytes(item)
+        return _rlp_len(len(item), 0x80) + bytes(item)
+    if isinstance(item, (list, tuple)):
+        payload = b"".join(rlp_encode(x) for x in item)
+        return _rlp_len(len(payload), 0xC0) + payload
+    raise TypeError(f"cannot RLP-encode {type(item)}")
+
+
+def _rlp_len(n: int, offset: int) -> bytes:
+    if n < 56:
+        return bytes([offset + n])
+    blen = n.to_bytes((n.bit_length() + 7) // 8, "big")
+    return bytes([offset + 55 + len(blen)]) + blen
+
+
+# ---------------------------------------------------------- ECDSA (secp256k1)
+
+def _rfc6979_k(msg_hash: bytes, priv: bytes) -> int:
+    """Deterministic nonce per RFC 6979 (HMAC-SHA256)."""
+    V = b"\x01" * 32
+    K = b"\x00" * 32
+    K = hmac.new(K, V + b"\x00" + priv + msg_hash, hashlib.sha256).digest()
+    V = hmac.new(K, V, hashlib.sha256).digest()
+    K = hmac.new(K, V + b"\x01" + priv + msg_hash, hashlib.sha256).digest()
+    V = hmac.new(K, V, hashlib.sha256).digest()
+    while True:
+        V = hmac.new(K, V, hashlib.sha256).digest()
+        k = int.from_bytes(V, "big")
+        if 0 < k < _N:
+            return k
+        K = hmac.new(K, V + b"\x00", hashlib.sha256).digest()
+        V = hmac.new(K, V, hashlib.sha256).digest()
+
+
+def ecdsa_sign(msg_hash: bytes, priv: bytes) -> tuple[int, int, int]:
+    """Returns (r, s, y_parity) with low-s normalization (EIP-2)."""
+    z = int.from_bytes(msg_hash, "big")
+    d = int.from_bytes(priv, "big")
+    k = _rfc6979_k(msg_hash, priv)
+    R = _ec_mul(k)
+    r = R[0] % _N
+    s = (_inv(k, _N) * (z + r * d)) % _N
+    y_parity = R[1] & 1
+    if s > _N // 2:
+        s = _N - s
+        y_parity ^= 1
+    return r, s, y_parity
+
+
+def ecdsa_recover(msg_hash: bytes, r: int, s: int, y_parity: int) -> str:
+    """Recovers the signer's EVM address (used as a self-check in tests)."""
+    z = int.from_bytes(msg_hash, "big")
+    x = r
+    # y² = x³ + 7 mod p
+    y = pow((x * x * x + 7) % _P, (_P + 1) // 4, _P)
+    if y & 1 != y_parity:
+        y = _P - y
+    Rpt = (x, y)
+    r_inv = _inv(r, _N)
+    # Q = r⁻¹ (s·R − z·G)
+    sR = _ec_mul(s, Rpt)
+    zG = _ec_mul(z % _N)
+    neg_zG = (zG[0], _P - zG[1])
+    Q = _ec_mul(r_inv, _ec_add(sR, neg_zG))
+    raw = Q[0].to_bytes(32, "big") + Q[1].to_bytes(32, "big")
+    return to_checksum_address("0x" + keccak256(raw)[-20:].hex())
+
+
+# ------------------------------------------------------------ EIP-1559 tx
+
+def erc20_transfer_calldata(to_address: str, amount: int) -> bytes:
+    """transfer(address,uint256) calldata."""
+    selector = keccak256(b"transfer(address,uint256)")[:4]
+    addr = bytes.fromhex(to_address[2:].lower()).rjust(32, b"\x00")
+    return selector + addr + amount.to_bytes(32, "big")
+
+
+def sign_eip1559_tx(priv: bytes, chain_id: int, nonce: int,
+                    max_priority_fee: int, max_fee: int, gas: int,
+                    to: str, value: int, data: bytes = b"") -> str:
+    """Builds and signs a type-2 (EIP-1559) transaction; returns raw tx hex
+    ready for eth_sendRawTransaction."""
+    fields = [chain_id, nonce, max_priority_fee, max_fee, gas, to, value,
+              data, []]  # empty access list
+    unsigned = b"\x02" + rlp_encode(fields)
+    r, s, y = ecdsa_sign(keccak256(unsigned), priv)
+    signed = b"\x02" + rlp_encode(fields + [y, r, s])
+    return "0x" + signed.hex()

@@ -377,3 +377,34 @@ def test_session_survives_manager_restart(ldb):
     assert sess2["turn_count"] == 2
     # the second cycle continued the persisted history
     assert len(json.loads(sess2["messages_json"])) > n_msgs_1
+
+
+def test_ecdsa_sign_recover_roundtrip():
+    from room_amd.utils.crypto import (ecdsa_recover, ecdsa_sign,
+                                       generate_private_key, keccak256,
+                                       private_key_to_address)
+    priv = generate_private_key(b"test-seed")
+    addr = private_key_to_address(priv)
+    for msg in (b"hello", b"another message", b"\x00" * 32):
+        h = keccak256(msg)
+        r, s, y = ecdsa_sign(h, priv)
+        assert ecdsa_recover(h, r, s, y) == addr
+
+
+def test_rlp_known_vectors():
+    from room_amd.utils.crypto import rlp_encode
+    assert rlp_encode(b"dog") == b"\x83dog"
+    assert rlp_encode([b"cat", b"dog"]) == b"\xc8\x83cat\x83dog"
+    assert rlp_encode(b"") == b"\x80"
+    assert rlp_encode(0) == b"\x80"
+    assert rlp_encode(15) == b"\x0f"
+    assert rlp_encode(1024) == b"\x82\x04\x00"
+    assert rlp_encode([]) == b"\xc0"
+
+
+def test_send_token_produces_signed_raw_tx(db):
+    from room_amd.core.wallet import send_token
+    r = room.create_room(db, "signtx", worker_model="stub")
+    out = send_token(db, r["id"], "0x" + "ab" * 20, "12.5")
+    assert out["raw_tx"].startswith("0x02")
+    assert len(out["raw_tx"]) > 200
