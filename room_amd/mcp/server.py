@@ -440,6 +440,216 @@ class McpServer:
             with ldb as db:
                 return q.list_watches(db, room_id=args.get("room_id"))
 
+        # ---- additional surface (workers/goals/tasks/credentials/etc.)
+        @t("room_get_worker", "Read one worker.", {"p": {"worker_id": I}, "r": ["worker_id"]})
+        def _(args):
+            with ldb as db:
+                return q.get_worker(db, args["worker_id"]) or {"error": "not found"}
+
+        @t("room_update_worker", "Update worker fields (prompt/model/pacing).",
+           {"p": {"worker_id": I, "system_prompt": S, "model": S,
+                  "cycle_gap_ms": I, "max_turns": I}, "r": ["worker_id"]})
+        def _(args):
+            fields = {k: v for k, v in args.items() if k != "worker_id"}
+            with ldb as db:
+                return q.update_worker(db, args["worker_id"], **fields) or {}
+
+        @t("room_delete_worker", "Delete a worker.", {"p": {"worker_id": I}, "r": ["worker_id"]})
+        def _(args):
+            with ldb as db:
+                q.delete_worker(db, args["worker_id"])
+            return {"deleted": True}
+
+        @t("room_get_wip", "Read a worker's saved work-in-progress.",
+           {"p": {"worker_id": I}, "r": ["worker_id"]})
+        def _(args):
+            with ldb as db:
+                w = q.get_worker(db, args["worker_id"])
+            return {"wip": (w or {}).get("wip")}
+
+        @t("room_clear_wip", "Clear a worker's WIP.", {"p": {"worker_id": I}, "r": ["worker_id"]})
+        def _(args):
+            with ldb as db:
+                q.set_worker_wip(db, args["worker_id"], None)
+            return {"cleared": True}
+
+        @t("room_abandon_goal", "Abandon a goal with a reason.",
+           {"p": {"goal_id": I, "reason": S}, "r": ["goal_id"]})
+        def _(args):
+            with ldb as db:
+                g = goals_mod.abandon_goal(db, args["goal_id"], reason=args.get("reason"))
+            return {"goal_id": g["id"], "status": g["status"]}
+
+        @t("room_pause_task", "Pause a scheduled task.", {"p": {"task_id": I}, "r": ["task_id"]})
+        def _(args):
+            with ldb as db:
+                q.update_task(db, args["task_id"], status="paused")
+            return {"paused": True}
+
+        @t("room_resume_task", "Resume a paused task.", {"p": {"task_id": I}, "r": ["task_id"]})
+        def _(args):
+            with ldb as db:
+                q.update_task(db, args["task_id"], status="active")
+            return {"resumed": True}
+
+        @t("room_delete_task", "Delete a task.", {"p": {"task_id": I}, "r": ["task_id"]})
+        def _(args):
+            with ldb as db:
+                q.delete_task(db, args["task_id"])
+            return {"deleted": True}
+
+        @t("room_run_logs", "Console logs of a task run.",
+           {"p": {"run_id": I, "after_seq": I}, "r": ["run_id"]})
+        def _(args):
+            with ldb as db:
+                return q.get_console_logs(db, args["run_id"],
+                                          after_seq=args.get("after_seq", -1))
+
+        @t("room_set_credential", "Store an encrypted room credential.",
+           {"p": {"room_id": I, "name": S, "value": S, "type": S},
+            "r": ["room_id", "name", "value"]})
+        def _(args):
+            from ..core.secret_store import encrypt_secret
+            with ldb as db:
+                row = q.set_credential(db, args["room_id"], args["name"],
+                                       encrypt_secret(args["value"]),
+                                       cred_type=args.get("type", "other"))
+            return {"name": row["name"], "type": row["type"]}
+
+        @t("room_list_credentials", "List credential names (values hidden).",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return q.list_credentials(db, args["room_id"])
+
+        @t("room_delete_credential", "Delete a credential.",
+           {"p": {"room_id": I, "name": S}, "r": ["room_id", "name"]})
+        def _(args):
+            with ldb as db:
+                q.delete_credential(db, args["room_id"], args["name"])
+            return {"deleted": True}
+
+        @t("room_cycles", "Recent agent cycles of a room.",
+           {"p": {"room_id": I, "limit": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return q.list_room_cycles(db, args["room_id"],
+                                          limit=args.get("limit", 20))
+
+        @t("room_cycle_logs", "Streamed logs of one agent cycle.",
+           {"p": {"cycle_id": I, "after_seq": I}, "r": ["cycle_id"]})
+        def _(args):
+            with ldb as db:
+                return q.get_cycle_logs(db, args["cycle_id"],
+                                        after_seq=args.get("after_seq", -1))
+
+        @t("room_token_usage", "Token usage rollup for a room.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return q.get_room_token_usage(db, args["room_id"])
+
+        @t("room_list_templates", "Available room/worker templates.", {})
+        def _(args):
+            from ..core.templates import list_templates
+            return list_templates()
+
+        @t("room_create_from_template", "Instantiate a room template.",
+           {"p": {"template": S, "name": S, "worker_model": S},
+            "r": ["template", "name"]})
+        def _(args):
+            from ..core.templates import instantiate_room_template
+            with ldb as db:
+                r = instantiate_room_template(
+                    db, args["template"], args["name"],
+                    worker_model=args.get("worker_model", "qwen3-coder-30b"))
+            return {"room_id": r["id"]}
+
+        @t("room_export_prompts", "Export worker prompts as markdown files.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            from ..core.prompt_sync import export_worker_prompts
+            with ldb as db:
+                return {"files": export_worker_prompts(db, args["room_id"])}
+
+        @t("room_import_prompts", "Import worker prompts from markdown files.",
+           {"p": {"room_id": I, "force": B}, "r": ["room_id"]})
+        def _(args):
+            from ..core.prompt_sync import import_worker_prompts
+            with ldb as db:
+                return import_worker_prompts(db, args["room_id"],
+                                             force=args.get("force", False))
+
+        @t("room_public_feed", "Curated public activity feed.", {"p": {"limit": I}})
+        def _(args):
+            from ..core.public_feed import get_public_feed
+            with ldb as db:
+                return get_public_feed(db, limit=args.get("limit", 20))
+
+        @t("room_public_profile", "Public profile of a public room.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            from ..core.public_feed import get_public_room_profile
+            with ldb as db:
+                return get_public_room_profile(db, args["room_id"]) or                     {"error": "room not public"}
+
+        @t("room_notify_keeper", "Deliver a keeper notification (outbox/email).",
+           {"p": {"subject": S, "body": S, "room_id": I}, "r": ["subject", "body"]})
+        def _(args):
+            from ..core.notifications import notify_keeper
+            return notify_keeper(args["subject"], args["body"],
+                                 room_id=args.get("room_id"))
+
+        @t("room_read_outbox", "Read recent keeper notifications.", {"p": {"limit": I}})
+        def _(args):
+            from ..core.notifications import read_outbox
+            return read_outbox(limit=args.get("limit", 20))
+
+        @t("room_web_fetch", "Fetch a URL as readable text.",
+           {"p": {"url": S}, "r": ["url"]})
+        def _(args):
+            from ..core.web_tools import web_fetch
+            return web_fetch(args["url"])
+
+        @t("room_web_search", "Keyless web search.", {"p": {"query": S}, "r": ["query"]})
+        def _(args):
+            from ..core.web_tools import web_search
+            return web_search(args["query"])
+
+        @t("room_browser_action", "Persistent browser-session action.",
+           {"p": {"session_id": S, "action": S, "url": S},
+            "r": ["session_id", "action"]})
+        def _(args):
+            from ..core.web_tools import browser_action
+            return browser_action(args["session_id"], args["action"],
+                                  url=args.get("url"))
+
+        @t("room_clerk_chat", "Chat with the system-wide Clerk assistant.",
+           {"p": {"content": S}, "r": ["content"]})
+        def _(args):
+            from ..core.clerk import clerk_chat
+            return {"reply": clerk_chat(ldb, args["content"],
+                                        memory=self.memory)}
+
+        @t("room_delete_watch", "Remove a filesystem watch.",
+           {"p": {"watch_id": I}, "r": ["watch_id"]})
+        def _(args):
+            with ldb as db:
+                q.delete_watch(db, args["watch_id"])
+            return {"deleted": True}
+
+        @t("room_list_messages", "Inter-room messages of a room.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return q.list_room_messages(db, args["room_id"])
+
+        @t("room_self_mod_audit", "Self-modification audit log.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return q.list_self_mod_audit(db, args["room_id"])
+
     # ------------------------------------------------------------ JSON-RPC
 
     def handle(self, msg: dict) -> dict | None:
