@@ -405,9 +405,10 @@ class LocalEngine:
                                 or t in (tok.EOS, tok.IM_END)):
                             finished.append(r)
                 for r in finished:
-                    if r.session_key and r.session_key in self.sessions:
-                        self.sessions[r.session_key].tokens = (
-                            r.prompt_tokens + r.out_tokens[:-1])
+                    with self._lock:  # sessions map also mutated by release_session
+                        if r.session_key and r.session_key in self.sessions:
+                            self.sessions[r.session_key].tokens = (
+                                r.prompt_tokens + r.out_tokens[:-1])
                     self._active.remove(r)
                     r.done.set()
                 self.stats["decode_steps"] += steps
@@ -463,9 +464,10 @@ class LocalEngine:
         for r in finished:
             # the sampled token at r.pos is NOT yet in KV; it will be written
             # if the session continues (prompt extension re-runs it)
-            if r.session_key and r.session_key in self.sessions:
-                self.sessions[r.session_key].tokens = (
-                    r.prompt_tokens + r.out_tokens[:-1])
+            with self._lock:
+                if r.session_key and r.session_key in self.sessions:
+                    self.sessions[r.session_key].tokens = (
+                        r.prompt_tokens + r.out_tokens[:-1])
             self._active.remove(r)
             r.done.set()
 
