@@ -154,16 +154,15 @@ class Qwen3MoEModel:
                 ops.gemv(qkv, hbuf, layer.wqkv)
             else:
                 qkv = F.linear(hbuf, layer.wqkv)   # hipBLASLt GEMM
-            q = qkv[:, :qdim].reshape(T, cfg.num_q_heads, cfg.head_dim).contiguous()
-            k = qkv[:, qdim:qdim + kvdim].reshape(T, cfg.num_kv_heads,
-                                                  cfg.head_dim).contiguous()
-            v = qkv[:, qdim + kvdim:].reshape(T, cfg.num_kv_heads,
-                                              cfg.head_dim).contiguous()
-            ops.qk_rope_write_kv(q, k, v, kcaches[li], vcaches[li],
+            # q stays a strided view into qkv (the attention kernels take a
+            # row stride; .contiguous() copies profiled at 0.67 ms/step)
+            q = qkv[:, :qdim].view(T, cfg.num_q_heads, cfg.head_dim)
+            ops.qk_rope_write_kv(qkv, cfg.num_q_heads, kcaches[li], vcaches[li],
                                  layer.q_norm_w, layer.k_norm_w, self.cos_t,
                                  self.sin_t, block_table, seq_ids, q_pos,
                                  cfg.rms_eps)
-            attn = torch.empty_like(q)
+            attn = torch.empty(T, cfg.num_q_heads, cfg.head_dim,
+                               dtype=torch.bfloat16, device=dev)
             if decode:
                 ops.paged_attention_split(attn, q, kcaches[li], vcaches[li],
                                           block_table, seq_ids, q_pos, part,
@@ -186,8 +185,9 @@ class Qwen3MoEModel:
             else:
                 router_logits = F.linear(hbuf, layer.router_w).float()
             topk_ids, topk_w = ops.moe_router(router_logits, cfg.num_experts_per_tok)
-            moe_out_f32 = self._moe(hbuf, layer, topk_ids, topk_w)
-            moe_out = moe_out_f32.to(torch.bfloat16)
+            # f32 accumulator feeds fused_add_rmsnorm directly (templated
+            # input dtype — skips a cast kernel per layer)
+            moe_out = self._moe(hbuf, layer, topk_ids, topk_w)
 
         # final residual add + norm; skip lm_head entirely for logits-free
         # prefill chunks (an empty logits_rows means "no sampling this chunk")

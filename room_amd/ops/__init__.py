@@ -49,15 +49,17 @@ def qk_norm_rope(q: torch.Tensor, k: torch.Tensor, q_w: torch.Tensor,
                             n_qheads, n_kvheads, head_dim, eps)
 
 
-def qk_rope_write_kv(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+def qk_rope_write_kv(qkv: torch.Tensor, n_qheads: int,
                      kcache: torch.Tensor, vcache: torch.Tensor,
                      q_w: torch.Tensor, k_w: torch.Tensor,
                      cos_t: torch.Tensor, sin_t: torch.Tensor,
                      block_table: torch.Tensor, seq_ids: torch.Tensor,
                      positions: torch.Tensor, eps: float = 1e-6) -> None:
-    """Fused per-head QK RMSNorm + RoPE + paged KV scatter (one launch)."""
-    _require().qk_rope_write_kv(q, k, v, kcache, vcache, q_w, k_w, cos_t,
-                                sin_t, block_table, seq_ids, positions, eps)
+    """Fused per-head QK RMSNorm + RoPE + paged KV scatter, operating directly
+    on the packed [T, (Hq+2Hk)*D] projection output (no q/k/v copies)."""
+    _require().qk_rope_write_kv(qkv, kcache, vcache, q_w, k_w, cos_t,
+                                sin_t, block_table, seq_ids, positions,
+                                n_qheads, eps)
 
 
 def silu_mul(out: torch.Tensor, gateup: torch.Tensor) -> torch.Tensor:

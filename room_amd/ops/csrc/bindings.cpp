@@ -8,12 +8,11 @@ void qk_norm_rope(torch::Tensor q, torch::Tensor k, torch::Tensor q_w,
                   torch::Tensor k_w, torch::Tensor cos_t, torch::Tensor sin_t,
                   torch::Tensor positions, int64_t n_qheads, int64_t n_kvheads,
                   int64_t head_dim, double eps);
-void qk_rope_write_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                      torch::Tensor kcache, torch::Tensor vcache,
-                      torch::Tensor q_w, torch::Tensor k_w,
+void qk_rope_write_kv(torch::Tensor qkv, torch::Tensor kcache,
+                      torch::Tensor vcache, torch::Tensor q_w, torch::Tensor k_w,
                       torch::Tensor cos_t, torch::Tensor sin_t,
                       torch::Tensor block_table, torch::Tensor seq_ids,
-                      torch::Tensor positions, double eps);
+                      torch::Tensor positions, int64_t n_qheads, double eps);
 void silu_mul(torch::Tensor out, torch::Tensor gateup);
 void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                      torch::Tensor vcache, torch::Tensor block_table,

@@ -36,7 +36,8 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
                           const int* __restrict__ seq_ids,
                           const int* __restrict__ q_pos,
                           const int* __restrict__ tile_desc,  // [G, 2]: row0, n
-                          int n_kvheads, int max_blocks, float scale) {
+                          int n_kvheads, int max_blocks, int q_row_stride,
+                          float scale) {
   const int g = blockIdx.x;
   const int hk = blockIdx.y;
   const int row0 = tile_desc[g * 2 + 0];
@@ -65,7 +66,7 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
     bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
     if (tk < ntok)
       v = *reinterpret_cast<const bf16x8*>(
-          q + ((long)(row0 + tk) * n_qheads + hk * FP_QH + h) * FP_D + d8);
+          q + (long)(row0 + tk) * q_row_stride + (hk * FP_QH + h) * FP_D + d8);
     *reinterpret_cast<bf16x8*>(&q_s[h][tk][d8]) = v;
   }
   if (tid < FP_QTOK)
@@ -209,6 +210,7 @@ void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                    torch::Tensor tile_desc, double scale) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   TORCH_CHECK(q.size(-1) == FP_D);
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == FP_D);  // strided row-views OK
   const int n_kvheads = kcache.size(1);
   TORCH_CHECK(q.size(1) == n_kvheads * FP_QH);
   const int G = tile_desc.size(0);
@@ -221,6 +223,6 @@ void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                      (const short*)vcache.data_ptr(),
                      block_table.data_ptr<int>(), seq_ids.data_ptr<int>(),
                      q_pos.data_ptr<int>(), tile_desc.data_ptr<int>(),
-                     n_kvheads, max_blocks, (float)scale);
+                     n_kvheads, max_blocks, (int)q.stride(0), (float)scale);
   HIP_CHECK_KERNEL();
 }

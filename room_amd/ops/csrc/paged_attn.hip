@@ -33,7 +33,8 @@ void paged_attn_kernel(short* __restrict__ out,          // [T, Hq, D]
                        const int* __restrict__ block_table,  // [S, MB]
                        const int* __restrict__ seq_ids,      // [T]
                        const int* __restrict__ q_pos,        // [T] absolute pos
-                       int n_kvheads, int max_blocks, float scale) {
+                       int n_kvheads, int max_blocks, int q_row_stride,
+                       float scale) {
   const int t = blockIdx.x;       // query token
   const int hk = blockIdx.y;      // kv head
   const int tid = threadIdx.x;
@@ -51,7 +52,8 @@ void paged_attn_kernel(short* __restrict__ out,          // [T, Hq, D]
   // load q for the 8 heads of this group into LDS (fp32)
   for (int i = tid; i < QH_PER_KV * HEAD_DIM; i += blockDim.x) {
     int hh = i / HEAD_DIM, dd = i % HEAD_DIM;
-    q_s[hh][dd] = bf2f(q[((long)t * n_qheads + hk * QH_PER_KV + hh) * HEAD_DIM + dd]);
+    q_s[hh][dd] = bf2f(q[(long)t * q_row_stride
+                         + (hk * QH_PER_KV + hh) * HEAD_DIM + dd]);
   }
   __syncthreads();
 
@@ -172,7 +174,8 @@ void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                      (short*)out.data_ptr(), (const short*)q.data_ptr(),
                      (const short*)kcache.data_ptr(), (const short*)vcache.data_ptr(),
                      block_table.data_ptr<int>(), seq_ids.data_ptr<int>(),
-                     q_pos.data_ptr<int>(), n_kvheads, max_blocks, (float)scale);
+                     q_pos.data_ptr<int>(), n_kvheads, max_blocks,
+                     (int)q.stride(0), (float)scale);
   HIP_CHECK_KERNEL();
 }
 
@@ -209,7 +212,8 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
                              const int* __restrict__ block_table,
                              const int* __restrict__ seq_ids,
                              const int* __restrict__ q_pos,
-                             int n_kvheads, int max_blocks, float scale) {
+                             int n_kvheads, int max_blocks, int q_row_stride,
+                             float scale) {
   const int t = blockIdx.x;
   const int hk = blockIdx.y;
   const int split = blockIdx.z;
@@ -241,7 +245,8 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
   __shared__ float p_s[QH_PER_KV][CHUNK];
   for (int i = tid; i < QH_PER_KV * HEAD_DIM; i += blockDim.x) {
     int hh = i / HEAD_DIM, dd = i % HEAD_DIM;
-    q_s[hh][dd] = bf2f(q[((long)t * n_qheads + hk * QH_PER_KV + hh) * HEAD_DIM + dd]);
+    q_s[hh][dd] = bf2f(q[(long)t * q_row_stride
+                         + (hk * QH_PER_KV + hh) * HEAD_DIM + dd]);
   }
   __syncthreads();
 
@@ -338,6 +343,8 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
                            torch::Tensor block_table, torch::Tensor seq_ids,
                            torch::Tensor q_pos, torch::Tensor part,
                            torch::Tensor part_ml, double scale) {
+  // q may be a strided row-view into the packed qkv buffer
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
   const int T = q.size(0);
   const int n_kvheads = kcache.size(1);
   const int n_qheads = n_kvheads * QH_PER_KV;
@@ -350,7 +357,7 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
                      (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
                      (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
                      seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
-                     n_kvheads, max_blocks, (float)scale);
+                     n_kvheads, max_blocks, (int)q.stride(0), (float)scale);
   HIP_CHECK_KERNEL();
   dim3 g2(T, n_qheads);
   hipLaunchKernelGGL(paged_attn_merge_kernel, g2, dim3(HEAD_DIM), 0, s,

@@ -375,9 +375,12 @@ def test_qk_rope_write_kv_fused_matches_unfused():
     # unfused reference path
     ops.qk_norm_rope(q2, k2, q_w, k_w, cos_t, sin_t, pos, Hq, Hk, D)
     ops.write_kv(kcache2, vcache2, k2, v, bt, seq_ids, pos)
-    # fused
-    ops.qk_rope_write_kv(q, k, v, kcache, vcache, q_w, k_w, cos_t, sin_t,
+    # fused, on the packed qkv buffer
+    qkv = torch.cat([q.reshape(T, -1), k.reshape(T, -1), v.reshape(T, -1)],
+                    dim=1).contiguous()
+    ops.qk_rope_write_kv(qkv, Hq, kcache, vcache, q_w, k_w, cos_t, sin_t,
                          bt, seq_ids, pos)
-    assert torch.equal(q, q2)
+    q_out = qkv[:, :Hq * D].reshape(T, Hq, D)
+    assert torch.equal(q_out, q2)
     assert torch.equal(vcache, vcache2)
     assert bf16_close(kcache, kcache2, atol=1e-3)
