@@ -97,6 +97,82 @@ void v1(short* y, const short* x, const short* w, int B, int H, int N) {
   }
 }
 
+// v5: split-K — each output row is computed by 2 waves over half of H;
+// LDS combine. Doubles the wave count (ramp) at the cost of one barrier.
+__global__ __launch_bounds__(128)
+void v5(short* y, const short* x, const short* w, int B, int H, int N) {
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int n = blockIdx.x;
+  if (n >= N) return;
+  const int half = H / 2;
+  const short* wr = w + (long)n * H + wid * half;
+  const short* xb = x + wid * half;
+  float acc[8];
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) acc[b] = 0.f;
+  for (int base = lane * 8; base < half; base += 64 * 8) {
+    bf16x8 wv = *(const bf16x8*)(wr + base);
+    float wf[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) wf[j] = bf2f(wv[j]);
+    #pragma unroll
+    for (int b = 0; b < 8; ++b) if (b < B) {
+      bf16x8 xv = *(const bf16x8*)(xb + (long)b * H + base);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) acc[b] += wf[j] * bf2f(xv[j]);
+    }
+  }
+  __shared__ float partial[8];
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) if (b < B) {
+    float r = wsum(acc[b]);
+    if (lane == 0 && wid == 1) partial[b] = r;
+  }
+  __syncthreads();
+  if (wid == 0 && lane == 0) {
+    #pragma unroll
+    for (int b = 0; b < 8; ++b) if (b < B) {
+      float r = wsum(acc[b]);  // lane0 already has the wave sum from above
+    }
+  }
+  // recompute cleanly: lane 0 of wave 0 adds its wave total to partner total
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) if (b < B) {
+    float r = wsum(acc[b]);
+    if (wid == 0 && lane == 0)
+      y[(long)b * N + n] = f2bf(r + partial[b]);
+  }
+}
+
+// v6: nontemporal weight loads (stream past L2)
+__global__ __launch_bounds__(256)
+void v6(short* y, const short* x, const short* w, int B, int H, int N) {
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int n = blockIdx.x * 4 + wid;
+  if (n >= N) return;
+  float acc[8];
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) acc[b] = 0.f;
+  const short* wr = w + (long)n * H;
+  for (int base = lane * 8; base < H; base += 64 * 8) {
+    bf16x8 wv = __builtin_nontemporal_load((const bf16x8*)(wr + base));
+    float wf[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) wf[j] = bf2f(wv[j]);
+    #pragma unroll
+    for (int b = 0; b < 8; ++b) if (b < B) {
+      bf16x8 xv = *(const bf16x8*)(x + (long)b * H + base);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) acc[b] += wf[j] * bf2f(xv[j]);
+    }
+  }
+  #pragma unroll
+  for (int b = 0; b < 8; ++b) if (b < B) {
+    float r = wsum(acc[b]);
+    if (lane == 0) y[(long)b * N + n] = f2bf(r);
+  }
+}
+
 __global__ __launch_bounds__(256)
 void v3(short* y, const short* x, const short* w, int B, int H, int N) {
   // thread-per-output: serial 16B-chunk row read, no cross-lane reduce
@@ -132,8 +208,17 @@ void run(int variant, torch::Tensor y, torch::Tensor x, torch::Tensor w) {
   else if (variant == 2)
     hipLaunchKernelGGL((v0<8>), dim3((N + 7) / 8), dim3(512), 0, s,
       (short*)y.data_ptr(), (const short*)x.data_ptr(), (const short*)w.data_ptr(), B, H, N);
-  else
+  else if (variant == 3)
     hipLaunchKernelGGL(v3, dim3((N + 255) / 256), dim3(256), 0, s,
+      (short*)y.data_ptr(), (const short*)x.data_ptr(), (const short*)w.data_ptr(), B, H, N);
+  else if (variant == 4)  // 1-wave blocks (max block count)
+    hipLaunchKernelGGL((v0<1>), dim3(N), dim3(64), 0, s,
+      (short*)y.data_ptr(), (const short*)x.data_ptr(), (const short*)w.data_ptr(), B, H, N);
+  else if (variant == 5)
+    hipLaunchKernelGGL(v5, dim3(N), dim3(128), 0, s,
+      (short*)y.data_ptr(), (const short*)x.data_ptr(), (const short*)w.data_ptr(), B, H, N);
+  else
+    hipLaunchKernelGGL(v6, dim3((N + 3) / 4), dim3(256), 0, s,
       (short*)y.data_ptr(), (const short*)x.data_ptr(), (const short*)w.data_ptr(), B, H, N);
 }
 """
@@ -154,14 +239,14 @@ y = torch.empty(B, N, dtype=torch.bfloat16, device=DEV)
 
 # correctness vs torch
 ref = (x.float() @ ws[0].float().T).to(torch.bfloat16)
-for v in range(4):
+for v in range(7):
     mod.run(v, y, x, ws[0])
     torch.cuda.synchronize()
     assert torch.allclose(y.float(), ref.float(), atol=6e-2, rtol=6e-2), f"V{v}"
 print("all variants correct")
 
 ITER = 240
-for v in range(4):
+for v in range(7):
     for i in range(COPIES):
         mod.run(v, y, x, ws[i])
     torch.cuda.synchronize()
