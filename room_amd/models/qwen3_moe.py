@@ -138,6 +138,8 @@ class Qwen3MoEModel:
             obuf = torch.empty(T, cfg.hidden_size, dtype=torch.bfloat16, device=dev)
             router_logits = torch.empty(T, cfg.num_experts, dtype=torch.float32,
                                         device=dev)
+            x_alt = torch.empty_like(x)     # residual ping-pong for gemv_addnorm
+            empty_delta = torch.empty(0, dtype=torch.float32, device=dev)
             part = torch.empty(T, cfg.num_q_heads, 32, cfg.head_dim,
                                dtype=torch.float32, device=dev)
             part_ml = torch.empty(T, cfg.num_q_heads, 32, 2, dtype=torch.float32,
@@ -145,14 +147,23 @@ class Qwen3MoEModel:
 
         for li, layer in enumerate(self.layers):
             # --- attention block
-            if moe_out is None:
-                ops.rmsnorm(hbuf, x, layer.input_norm_w, cfg.rms_eps)
-            else:
-                ops.fused_add_rmsnorm(hbuf, x, moe_out, layer.input_norm_w,
-                                      cfg.rms_eps)
             if decode:
-                ops.gemv(qkv, hbuf, layer.wqkv)
+                # fused add(moe_prev)+RMSNorm+QKV-GEMV: one kernel replaces
+                # fused_add_rmsnorm + gemv (each removed kernel removes a
+                # launch+drain bubble from the 14-kernel decode chain)
+                if moe_out is None:
+                    ops.gemv_addnorm(qkv, x, empty_delta, x, layer.input_norm_w,
+                                     layer.wqkv, cfg.rms_eps)
+                else:
+                    ops.gemv_addnorm(qkv, x, moe_out, x_alt, layer.input_norm_w,
+                                     layer.wqkv, cfg.rms_eps)
+                    x, x_alt = x_alt, x
             else:
+                if moe_out is None:
+                    ops.rmsnorm(hbuf, x, layer.input_norm_w, cfg.rms_eps)
+                else:
+                    ops.fused_add_rmsnorm(hbuf, x, moe_out, layer.input_norm_w,
+                                          cfg.rms_eps)
                 qkv = F.linear(hbuf, layer.wqkv)   # hipBLASLt GEMM
             # q stays a strided view into qkv (the attention kernels take a
             # row stride; .contiguous() copies profiled at 0.67 ms/step)

@@ -125,6 +125,15 @@ def build_qtile_desc(segments: list, device) -> torch.Tensor:
     return torch.tensor(desc, dtype=torch.int32, device=device)
 
 
+def gemv_addnorm(y: torch.Tensor, x: torch.Tensor, delta: torch.Tensor,
+                 x_out: torch.Tensor, gamma: torch.Tensor, w: torch.Tensor,
+                 eps: float = 1e-6) -> torch.Tensor:
+    """y = rmsnorm(x + delta)·γ @ w^T; x_out = x + delta (pass delta with
+    numel 0 for a plain norm+gemv). Decode-path fusion (B ≤ 8)."""
+    _require().gemv_addnorm(y, x, delta, x_out, gamma, w, eps)
+    return y
+
+
 def moe_router(logits: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:
     T = logits.size(0)
     ids = torch.empty(T, k, dtype=torch.int32, device=logits.device)

@@ -54,6 +54,9 @@ void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                    torch::Tensor seq_ids, torch::Tensor q_pos,
                    torch::Tensor tile_desc, double scale);
 void gemv(torch::Tensor y, torch::Tensor x, torch::Tensor w);
+void gemv_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
+                  torch::Tensor x_out, torch::Tensor gamma, torch::Tensor w,
+                  double eps);
 void sample_tokens_v3(torch::Tensor out_tokens, torch::Tensor logits,
                       torch::Tensor seeds, int64_t top_k, double temperature,
                       double top_p);
@@ -87,6 +90,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "split-KV flash-decode paged attention");
   m.def("flash_prefill", &flash_prefill, "MFMA flash-attention prefill");
   m.def("gemv", &gemv, "dense skinny-batch GEMV (decode projections)");
+  m.def("gemv_addnorm", &gemv_addnorm,
+        "fused residual-add + RMSNorm + GEMV (decode)");
   m.def("sample_tokens_v3", &sample_tokens_v3, "register top-8 sampler");
   m.def("sample_scan_probe", &sample_scan_probe, "scan-cost probe");
   m.def("hist_append", &hist_append, "token-history append (multi-step decode)");

@@ -72,3 +72,137 @@ void gemv(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
   }
   HIP_CHECK_KERNEL();
 }
+
+// ------------------------------------------------ fused add+RMSNorm+GEMV
+// y[B,N] = rmsnorm(x + delta)·γ @ W^T, and (block 0 only) x_out = x + delta.
+// Folds the per-layer fused_add_rmsnorm kernel and the normed-activation
+// buffer round-trip into the projection itself: each wave already streams the
+// full activation row, so the row norm is one extra L1-hot pass (decode chains
+// ~14 dependent small kernels/layer; every removed kernel removes a
+// launch+fill/drain bubble — see profiles/PERF_NOTES.md).
+// Norm matches fused_add_rmsnorm semantics: computed over the bf16-rounded sum.
+template <bool F32OUT, bool DELTA_F32, bool HAS_DELTA>
+__global__ __launch_bounds__(256)
+void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
+                         const short* __restrict__ x,      // [B, H]
+                         const void* __restrict__ delta_,  // [B, H] or null
+                         short* __restrict__ x_out,        // [B, H]
+                         const short* __restrict__ gamma,  // [H]
+                         const short* __restrict__ w,      // [N, H]
+                         int B, int H, int N, float eps) {
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int n = blockIdx.x * 4 + wid;
+
+  // pass 1: row sums of squares (bf16-rounded, matching fused_add_rmsnorm)
+  float scale[GEMV_MAXB];
+  #pragma unroll
+  for (int b = 0; b < GEMV_MAXB; ++b) {
+    if (b < B) {
+      float ss = 0.f;
+      for (int base = lane * 8; base < H; base += WAVE * 8) {
+        bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float v = bf2f(xv[j]);
+          if (HAS_DELTA) {
+            if (DELTA_F32)
+              v += ((const float*)delta_)[(long)b * H + base + j];
+            else
+              v += bf2f(((const short*)delta_)[(long)b * H + base + j]);
+            v = bf2f(f2bf(v));
+          }
+          ss += v * v;
+        }
+      }
+      ss = wave_reduce_sum(ss);
+      scale[b] = rsqrtf(ss / H + eps);
+    }
+  }
+
+  // block 0 stores the updated residual (other blocks never read x_out here)
+  if (HAS_DELTA && blockIdx.x == 0) {
+    for (int i = threadIdx.x * 8; i < B * H; i += 256 * 8) {
+      bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i);
+      bf16x8 o;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = bf2f(xv[j]);
+        if (DELTA_F32) v += ((const float*)delta_)[i + j];
+        else v += bf2f(((const short*)delta_)[i + j]);
+        o[j] = f2bf(v);
+      }
+      *reinterpret_cast<bf16x8*>(x_out + i) = o;
+    }
+  }
+  if (n >= N) return;
+
+  // pass 2: the dot with inline normalization (x, delta, γ are L1-hot now)
+  float acc[GEMV_MAXB];
+  #pragma unroll
+  for (int b = 0; b < GEMV_MAXB; ++b) acc[b] = 0.f;
+  const short* wrow = w + (long)n * H;
+  for (int base = lane * 8; base < H; base += WAVE * 8) {
+    bf16x8 wv = *reinterpret_cast<const bf16x8*>(wrow + base);
+    bf16x8 gv = *reinterpret_cast<const bf16x8*>(gamma + base);
+    float wg[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) wg[j] = bf2f(wv[j]) * bf2f(gv[j]);
+    #pragma unroll
+    for (int b = 0; b < GEMV_MAXB; ++b) {
+      if (b < B) {
+        bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float v = bf2f(xv[j]);
+          if (HAS_DELTA) {
+            if (DELTA_F32)
+              v += ((const float*)delta_)[(long)b * H + base + j];
+            else
+              v += bf2f(((const short*)delta_)[(long)b * H + base + j]);
+            v = bf2f(f2bf(v));
+          }
+          acc[b] += wg[j] * v;
+        }
+      }
+    }
+  }
+  #pragma unroll
+  for (int b = 0; b < GEMV_MAXB; ++b) {
+    if (b < B) {
+      float r = wave_reduce_sum(acc[b]) * scale[b];
+      if (lane == 0) {
+        if (F32OUT) ((float*)y)[(long)b * N + n] = r;
+        else ((short*)y)[(long)b * N + n] = f2bf(r);
+      }
+    }
+  }
+}
+
+void gemv_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
+                  torch::Tensor x_out, torch::Tensor gamma, torch::Tensor w,
+                  double eps) {
+  const int B = x.size(0), H = x.size(1), N = w.size(0);
+  TORCH_CHECK(B <= GEMV_MAXB && H % (WAVE * 8) == 0);
+  TORCH_CHECK(x.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16);
+  const bool has_delta = delta.numel() > 0;
+  const bool delta_f32 = has_delta && delta.dtype() == torch::kFloat32;
+  const bool f32out = y.dtype() == torch::kFloat32;
+  dim3 grid((N + 3) / 4), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  const void* dptr = has_delta ? delta.data_ptr() : nullptr;
+  #define LAUNCH_AN(FO, DF, HD) hipLaunchKernelGGL( \
+      (gemv_addnorm_kernel<FO, DF, HD>), grid, block, 0, s, y.data_ptr(), \
+      (const short*)x.data_ptr(), dptr, (short*)x_out.data_ptr(), \
+      (const short*)gamma.data_ptr(), (const short*)w.data_ptr(), \
+      B, H, N, (float)eps)
+  if (!has_delta) {
+    if (f32out) LAUNCH_AN(true, false, false); else LAUNCH_AN(false, false, false);
+  } else if (delta_f32) {
+    if (f32out) LAUNCH_AN(true, true, true); else LAUNCH_AN(false, true, true);
+  } else {
+    if (f32out) LAUNCH_AN(true, false, true); else LAUNCH_AN(false, false, true);
+  }
+  #undef LAUNCH_AN
+  HIP_CHECK_KERNEL();
+}

@@ -384,3 +384,34 @@ def test_qk_rope_write_kv_fused_matches_unfused():
     assert torch.equal(q_out, q2)
     assert torch.equal(vcache, vcache2)
     assert bf16_close(kcache, kcache2, atol=1e-3)
+
+
+def test_gemv_addnorm_matches_composition():
+    torch.manual_seed(19)
+    B, H, N = 5, 2048, 1024
+    x = torch.randn(B, H, dtype=torch.bfloat16, device=DEV)
+    gamma = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(N, H, dtype=torch.bfloat16, device=DEV) * 0.05
+    for delta in (torch.randn(B, H, dtype=torch.float32, device=DEV) * 0.1,
+                  torch.randn(B, H, dtype=torch.bfloat16, device=DEV) * 0.1,
+                  torch.empty(0, dtype=torch.float32, device=DEV)):
+        x_in = x.clone()
+        x_out = torch.empty_like(x)
+        y = torch.empty(B, N, dtype=torch.bfloat16, device=DEV)
+        ops.gemv_addnorm(y, x_in, delta, x_out, gamma, w, 1e-6)
+        # reference composition: fused_add_rmsnorm (or rmsnorm) then gemv
+        if delta.numel():
+            res = x.clone()
+            hb = torch.empty_like(x)
+            ops.fused_add_rmsnorm(hb, res, delta.to(delta.dtype), gamma, 1e-6)
+            assert bf16_close(x_out, res)           # residual update matches
+        else:
+            hb = torch.empty_like(x)
+            ops.rmsnorm(hb, x, gamma, 1e-6)
+        y_ref = torch.empty(B, N, dtype=torch.bfloat16, device=DEV)
+        ops.gemv(y_ref, hb, w)
+        assert bf16_close(y, y_ref, atol=6e-2, rtol=6e-2)
+        # f32 output variant
+        yf = torch.empty(B, N, dtype=torch.float32, device=DEV)
+        ops.gemv_addnorm(yf, x.clone(), delta, torch.empty_like(x), gamma, w)
+        assert bf16_close(yf, y_ref, atol=6e-2, rtol=6e-2)
