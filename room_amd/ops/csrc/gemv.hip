@@ -94,13 +94,17 @@ void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
   const int lane = threadIdx.x & 63;
   const int n = blockIdx.x * 4 + wid;
 
-  // pass 1: row sums of squares (bf16-rounded, matching fused_add_rmsnorm)
-  float scale[GEMV_MAXB];
+  // pass 1 (block-cooperative): row sums of squares, bf16-rounded to match
+  // fused_add_rmsnorm. All 256 threads share the work (a per-wave pass was
+  // measured as a B=5 regression: 4× redundant VALU across the block).
+  __shared__ float ss_sh[GEMV_MAXB];
+  if (threadIdx.x < GEMV_MAXB) ss_sh[threadIdx.x] = 0.f;
+  __syncthreads();
   #pragma unroll
   for (int b = 0; b < GEMV_MAXB; ++b) {
     if (b < B) {
       float ss = 0.f;
-      for (int base = lane * 8; base < H; base += WAVE * 8) {
+      for (int base = threadIdx.x * 8; base < H; base += 256 * 8) {
         bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -116,9 +120,14 @@ void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
         }
       }
       ss = wave_reduce_sum(ss);
-      scale[b] = rsqrtf(ss / H + eps);
+      if (lane == 0) atomicAdd(&ss_sh[b], ss);
     }
   }
+  __syncthreads();
+  float scale[GEMV_MAXB];
+  #pragma unroll
+  for (int b = 0; b < GEMV_MAXB; ++b)
+    if (b < B) scale[b] = rsqrtf(ss_sh[b] / H + eps);
 
   // block 0 stores the updated residual (other blocks never read x_out here)
   if (HAS_DELTA && blockIdx.x == 0) {
