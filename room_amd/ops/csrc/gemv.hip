@@ -129,6 +129,34 @@ void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
   for (int b = 0; b < GEMV_MAXB; ++b)
     if (b < B) scale[b] = rsqrtf(ss_sh[b] / H + eps);
 
+  // pass 1.5 (block-cooperative): stage xn = (x+δ)·scale·γ into LDS once, so
+  // the dot loop is lean (w-load + LDS-load + FMA; no per-wave re-add/round)
+  extern __shared__ short xn_sh[];  // [B][H] bf16
+  #pragma unroll
+  for (int b = 0; b < GEMV_MAXB; ++b) {
+    if (b < B) {
+      for (int base = threadIdx.x * 8; base < H; base += 256 * 8) {
+        bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
+        bf16x8 gv = *reinterpret_cast<const bf16x8*>(gamma + base);
+        bf16x8 o;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float v = bf2f(xv[j]);
+          if (HAS_DELTA) {
+            if (DELTA_F32)
+              v += ((const float*)delta_)[(long)b * H + base + j];
+            else
+              v += bf2f(((const short*)delta_)[(long)b * H + base + j]);
+            v = bf2f(f2bf(v));
+          }
+          o[j] = f2bf(v * scale[b] * bf2f(gv[j]));
+        }
+        *reinterpret_cast<bf16x8*>(&xn_sh[b * H + base]) = o;
+      }
+    }
+  }
+  __syncthreads();
+
   // block 0 stores the updated residual (other blocks never read x_out here)
   if (HAS_DELTA && blockIdx.x == 0) {
     for (int i = threadIdx.x * 8; i < B * H; i += 256 * 8) {
@@ -146,40 +174,29 @@ void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
   }
   if (n >= N) return;
 
-  // pass 2: the dot with inline normalization (x, delta, γ are L1-hot now)
+  // pass 2: lean dot against the LDS-staged normed rows
   float acc[GEMV_MAXB];
   #pragma unroll
   for (int b = 0; b < GEMV_MAXB; ++b) acc[b] = 0.f;
   const short* wrow = w + (long)n * H;
   for (int base = lane * 8; base < H; base += WAVE * 8) {
     bf16x8 wv = *reinterpret_cast<const bf16x8*>(wrow + base);
-    bf16x8 gv = *reinterpret_cast<const bf16x8*>(gamma + base);
-    float wg[8];
+    float wf[8];
     #pragma unroll
-    for (int j = 0; j < 8; ++j) wg[j] = bf2f(wv[j]) * bf2f(gv[j]);
+    for (int j = 0; j < 8; ++j) wf[j] = bf2f(wv[j]);
     #pragma unroll
     for (int b = 0; b < GEMV_MAXB; ++b) {
       if (b < B) {
-        bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
+        bf16x8 xv = *reinterpret_cast<const bf16x8*>(&xn_sh[b * H + base]);
         #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float v = bf2f(xv[j]);
-          if (HAS_DELTA) {
-            if (DELTA_F32)
-              v += ((const float*)delta_)[(long)b * H + base + j];
-            else
-              v += bf2f(((const short*)delta_)[(long)b * H + base + j]);
-            v = bf2f(f2bf(v));
-          }
-          acc[b] += wg[j] * v;
-        }
+        for (int j = 0; j < 8; ++j) acc[b] += wf[j] * bf2f(xv[j]);
       }
     }
   }
   #pragma unroll
   for (int b = 0; b < GEMV_MAXB; ++b) {
     if (b < B) {
-      float r = wave_reduce_sum(acc[b]) * scale[b];
+      float r = wave_reduce_sum(acc[b]);
       if (lane == 0) {
         if (F32OUT) ((float*)y)[(long)b * N + n] = r;
         else ((short*)y)[(long)b * N + n] = f2bf(r);
@@ -198,10 +215,12 @@ void gemv_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
   const bool delta_f32 = has_delta && delta.dtype() == torch::kFloat32;
   const bool f32out = y.dtype() == torch::kFloat32;
   dim3 grid((N + 3) / 4), block(256);
+  const size_t lds = (size_t)B * H * sizeof(short);  // staged normed rows
+  TORCH_CHECK(lds <= 160 * 1024, "B*H too large for LDS staging");
   hipStream_t s = c10::hip::getCurrentHIPStream();
   const void* dptr = has_delta ? delta.data_ptr() : nullptr;
   #define LAUNCH_AN(FO, DF, HD) hipLaunchKernelGGL( \
-      (gemv_addnorm_kernel<FO, DF, HD>), grid, block, 0, s, y.data_ptr(), \
+      (gemv_addnorm_kernel<FO, DF, HD>), grid, block, lds, s, y.data_ptr(), \
       (const short*)x.data_ptr(), dptr, (short*)x_out.data_ptr(), \
       (const short*)gamma.data_ptr(), (const short*)w.data_ptr(), \
       B, H, N, (float)eps)
