@@ -147,10 +147,12 @@ class Qwen3MoEModel:
 
         for li, layer in enumerate(self.layers):
             # --- attention block
-            if decode:
+            if decode and T <= 2:
                 # fused add(moe_prev)+RMSNorm+QKV-GEMV: one kernel replaces
-                # fused_add_rmsnorm + gemv (each removed kernel removes a
-                # launch+drain bubble from the 14-kernel decode chain)
+                # fused_add_rmsnorm + gemv. Wins at B≤2 where the removed
+                # launch/drain bubble dominates (5.7 → 5.2 ms/step at B=1);
+                # at B≥3 the cooperative norm passes cost more than the bubble
+                # (measured 9.1 vs 8.8 at B=5) so the unfused path is used.
                 if moe_out is None:
                     ops.gemv_addnorm(qkv, x, empty_delta, x, layer.input_norm_w,
                                      layer.wqkv, cfg.rms_eps)
@@ -158,6 +160,13 @@ class Qwen3MoEModel:
                     ops.gemv_addnorm(qkv, x, moe_out, x_alt, layer.input_norm_w,
                                      layer.wqkv, cfg.rms_eps)
                     x, x_alt = x_alt, x
+            elif decode:
+                if moe_out is None:
+                    ops.rmsnorm(hbuf, x, layer.input_norm_w, cfg.rms_eps)
+                else:
+                    ops.fused_add_rmsnorm(hbuf, x, moe_out, layer.input_norm_w,
+                                          cfg.rms_eps)
+                ops.gemv(qkv, hbuf, layer.wqkv)
             else:
                 if moe_out is None:
                     ops.rmsnorm(hbuf, x, layer.input_norm_w, cfg.rms_eps)
