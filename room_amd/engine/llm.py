@@ -53,11 +53,12 @@ class GenRequest:
 
 
 class _Session:
-    __slots__ = ("slot", "tokens")
+    __slots__ = ("slot", "tokens", "last_used")
 
     def __init__(self, slot: int, tokens: list[int]):
         self.slot = slot
         self.tokens = tokens
+        self.last_used = time.time()
 
 
 class _DecodeGraph:
@@ -268,16 +269,33 @@ class LocalEngine:
                 self.sessions.pop(req.session_key, None)
                 sess = None
         if req.slot < 0:
+            if not cache.free_slots:
+                self._evict_lru_session()
             req.slot = cache.alloc_seq()
             req.pos = 0
             req.pending_prefill = list(req.prompt_tokens)
             if req.session_key:
                 self.sessions[req.session_key] = _Session(req.slot, [])
+        if req.session_key and req.session_key in self.sessions:
+            self.sessions[req.session_key].last_used = time.time()
         req.prefill_tokens_run = len(req.pending_prefill)
         if not req.pending_prefill:
             # prompt identical to cache (rare): re-run last token for logits
             req.pos = max(0, req.pos - 1)
             req.pending_prefill = [req.prompt_tokens[-1]]
+
+    def _evict_lru_session(self) -> None:
+        """Slot pressure: drop the least-recently-used idle session (its KV
+        blocks free; the durable agent_sessions row lets the next cycle
+        re-prefill). Sessions of currently-active requests are never evicted."""
+        active_slots = {r.slot for r in self._active}
+        candidates = [(s.last_used, key) for key, s in self.sessions.items()
+                      if s.slot not in active_slots]
+        if not candidates:
+            raise RuntimeError("KV cache: no evictable sessions")
+        _, key = min(candidates)
+        s = self.sessions.pop(key)
+        self.cache.free_seq(s.slot)
 
     def _scheduler_loop(self) -> None:
         idx = self.device.index if self.device.index is not None \

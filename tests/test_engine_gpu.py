@@ -93,3 +93,22 @@ def test_agent_cycle_on_gpu_engine(engine):
         from room_amd.db import queries as q
         cycles = q.list_room_cycles(db, r["id"])
         assert cycles[0]["status"] == "completed"
+
+
+def test_session_lru_eviction():
+    """More sessions than KV slots → LRU sessions evicted, not errors."""
+    from room_amd.engine.llm import LocalEngine
+    from room_amd.models.qwen3_moe import Qwen3MoEConfig
+    eng = LocalEngine(cfg=Qwen3MoEConfig.tiny(), kv_gb=1.0, max_seqs=4)
+    try:
+        # 3 usable slots (one is the graph pad slot); run 6 sessions
+        for i in range(6):
+            p = tok.encode(f"session {i} content " * 6)
+            r = eng.generate(p, max_new_tokens=2, session_key=f"lru-{i}")
+            assert len(r.out_tokens) == 2
+        assert len(eng.sessions) <= 3
+        # oldest sessions were evicted; the newest still has its slot
+        assert "lru-5" in eng.sessions
+        assert "lru-0" not in eng.sessions
+    finally:
+        eng.shutdown()
