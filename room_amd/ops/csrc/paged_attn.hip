@@ -241,14 +241,20 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
     return;
   }
 
+  // K/V chunks are staged through LDS cooperatively (fully-coalesced bulk
+  // HBM loads, one read per WG instead of one per q-head); the score and
+  // value phases then run out of LDS. Row pad of 8 bf16 keeps the per-lane
+  // row reads off a single bank (same layout as flash_prefill).
+  #define PA_PAD 8
   __shared__ float q_s[QH_PER_KV][HEAD_DIM];
   __shared__ float p_s[QH_PER_KV][CHUNK];
+  __shared__ short k_s[CHUNK][HEAD_DIM + PA_PAD];
+  __shared__ short v_s[CHUNK][HEAD_DIM + PA_PAD];
   for (int i = tid; i < QH_PER_KV * HEAD_DIM; i += blockDim.x) {
     int hh = i / HEAD_DIM, dd = i % HEAD_DIM;
     q_s[hh][dd] = bf2f(q[(long)t * q_row_stride
                          + (hk * QH_PER_KV + hh) * HEAD_DIM + dd]);
   }
-  __syncthreads();
 
   float acc[4] = {0.f, 0.f, 0.f, 0.f};
   float m_run = -INFINITY, l_run = 0.f;
@@ -256,17 +262,34 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
   const int* btab = block_table + (long)seq * max_blocks;
 
   for (int base = lo; base < hi; base += CHUNK) {
-    {
-      const int pos = base + sub;
-      float s = -INFINITY;
+    // ---- stage K and V [32][128] chunks: 512 vec8 each, 2 per thread
+    #pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = tid + it * 256;          // vec8 index 0..511
+      const int pos_l = (idx * 8) / HEAD_DIM;  // 0..31
+      const int d8 = (idx * 8) % HEAD_DIM;
+      const int pos = base + pos_l;
+      bf16x8 kv = {0, 0, 0, 0, 0, 0, 0, 0};
+      bf16x8 vv = {0, 0, 0, 0, 0, 0, 0, 0};
       if (pos < hi) {
         const int blk = btab[pos / BLOCK_SIZE];
-        const short* krow = kcache + (long)blk * kv_stride_block
-                            + ((long)hk * BLOCK_SIZE + (pos % BLOCK_SIZE)) * HEAD_DIM;
+        const long off = (long)blk * kv_stride_block
+                         + ((long)hk * BLOCK_SIZE + (pos % BLOCK_SIZE)) * HEAD_DIM + d8;
+        kv = *reinterpret_cast<const bf16x8*>(kcache + off);
+        vv = *reinterpret_cast<const bf16x8*>(vcache + off);
+      }
+      *reinterpret_cast<bf16x8*>(&k_s[pos_l][d8]) = kv;
+      *reinterpret_cast<bf16x8*>(&v_s[pos_l][d8]) = vv;
+    }
+    __syncthreads();
+
+    {   // score: thread (h, sub) dots q_s[h] with k_s[sub]
+      float s = -INFINITY;
+      if (base + sub < hi) {
         float dot = 0.f;
         #pragma unroll
         for (int v8 = 0; v8 < HEAD_DIM / 8; ++v8) {
-          bf16x8 kv = *reinterpret_cast<const bf16x8*>(krow + v8 * 8);
+          bf16x8 kv = *reinterpret_cast<const bf16x8*>(&k_s[sub][v8 * 8]);
           #pragma unroll
           for (int j = 0; j < 8; ++j) dot += q_s[h][v8 * 8 + j] * bf2f(kv[j]);
         }
@@ -288,12 +311,7 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
         for (int j = 0; j < lim; ++j) {
           const float w = __expf(p_s[h][j] - m_new);
           l_run += w;
-          const int pos = base + j;
-          const int blk = btab[pos / BLOCK_SIZE];
-          const short* vrow = vcache + (long)blk * kv_stride_block
-                              + ((long)hk * BLOCK_SIZE + (pos % BLOCK_SIZE)) * HEAD_DIM
-                              + sub * 4;
-          bf16x4 vv = *reinterpret_cast<const bf16x4*>(vrow);
+          bf16x4 vv = *reinterpret_cast<const bf16x4*>(&v_s[j][sub * 4]);
           acc[0] += w * bf2f(vv[0]);
           acc[1] += w * bf2f(vv[1]);
           acc[2] += w * bf2f(vv[2]);
@@ -302,7 +320,7 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
         m_run = m_new;
       }
     }
-    __syncthreads();
+    __syncthreads();  // before restaging K/V
   }
 
   if (sub == 0) { ml[0] = m_run; ml[1] = l_run; }
