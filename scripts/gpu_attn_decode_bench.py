@@ -5,11 +5,14 @@ KV lengths L (the bench's hot shape: B=5 agents, sessions ~4-6k tokens).
 Cold-cache rotation over independent KV pools so L2 doesn't flatter the loop.
 """
 import argparse
+import os
+import sys
 import time
 
 import torch
 
-from room_amd import ops
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from room_amd import ops  # noqa: E402
 
 HK, HQ, D, BS = 4, 32, 128, 16
 NSPLITS = 32
