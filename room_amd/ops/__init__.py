@@ -142,6 +142,16 @@ def moe_router(logits: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor
     return ids, w
 
 
+def router_topk(x: torch.Tensor, wr: torch.Tensor,
+                k: int) -> tuple[torch.Tensor, torch.Tensor]:
+    """Fused decode router: logits = x @ wr.T → softmax → top-k → renorm."""
+    T = x.size(0)
+    ids = torch.empty(T, k, dtype=torch.int32, device=x.device)
+    w = torch.empty(T, k, dtype=torch.float32, device=x.device)
+    _require().router_topk(ids, w, x, wr, k)
+    return ids, w
+
+
 def moe_gemv_h(h: torch.Tensor, x: torch.Tensor, w13: torch.Tensor,
                pair_token: torch.Tensor, pair_expert: torch.Tensor) -> torch.Tensor:
     _require().moe_gemv_h(h, x, w13, pair_token, pair_expert)

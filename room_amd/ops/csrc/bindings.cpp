@@ -22,6 +22,8 @@ void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
               torch::Tensor q_pos);
 void moe_router(torch::Tensor topk_ids, torch::Tensor topk_w, torch::Tensor logits,
                 int64_t K);
+void router_topk(torch::Tensor topk_ids, torch::Tensor topk_w, torch::Tensor x,
+                 torch::Tensor wr, int64_t K);
 void moe_gemv_h(torch::Tensor h, torch::Tensor x, torch::Tensor w13,
                 torch::Tensor pair_token, torch::Tensor pair_expert);
 void moe_gemv_down(torch::Tensor out, torch::Tensor h, torch::Tensor w2,
@@ -77,6 +79,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention", &paged_attention, "paged causal attention (GQA 8:1)");
   m.def("write_kv", &write_kv, "scatter K/V rows into paged cache");
   m.def("moe_router", &moe_router, "softmax top-k router");
+  m.def("router_topk", &router_topk, "fused router GEMV + softmax top-k (decode)");
   m.def("moe_gemv_h", &moe_gemv_h, "MoE gate/up GEMV + silu-mul (decode)");
   m.def("moe_gemv_down", &moe_gemv_down, "MoE down GEMV + weighted scatter-add");
   m.def("moe_grouped_gemm", &moe_grouped_gemm, "grouped MFMA GEMM (prefill)");

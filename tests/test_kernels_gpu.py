@@ -146,6 +146,25 @@ def test_moe_router():
             assert m[e] == pytest.approx(mr[e], abs=1e-4)
 
 
+def test_router_topk_fused():
+    """Fused router (GEMV + softmax + top-k) vs fp32 torch reference."""
+    torch.manual_seed(11)
+    for T, E, H, K in [(5, 128, 2048, 8), (1, 16, 512, 4)]:
+        x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV) * 0.5
+        wr = torch.randn(E, H, dtype=torch.bfloat16, device=DEV) * 0.05
+        ids, w = ops.router_topk(x, wr, K)
+        logits_ref = (x.float() @ wr.float().t())
+        ids_ref, w_ref = ref.moe_router_ref(logits_ref, K)
+        for t in range(T):
+            assert set(ids[t].tolist()) == set(ids_ref[t].tolist()), \
+                f"T={T} t={t}: {ids[t].tolist()} vs {ids_ref[t].tolist()}"
+            assert w[t].sum().item() == pytest.approx(1.0, abs=1e-3)
+            m = {int(i): float(v) for i, v in zip(ids[t], w[t])}
+            mr = {int(i): float(v) for i, v in zip(ids_ref[t], w_ref[t])}
+            for e in m:
+                assert m[e] == pytest.approx(mr[e], abs=5e-3)
+
+
 def _moe_setup(T=5, E=16, H=2048, I=768, K=8):
     x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV) * 0.5
     w13 = torch.randn(E, 2 * I, H, dtype=torch.bfloat16, device=DEV) * 0.02
