@@ -103,7 +103,7 @@ class _DecodeGraph:
                                    cache.block_table, cache.kcaches,
                                    cache.vcaches)
             toks = torch.empty(bucket, dtype=torch.int32, device=dev)
-            C.sample_tokens_v3(toks, logits, self.seeds, top_k, temperature,
+            C.sample_tokens_v4(toks, logits, self.seeds, top_k, temperature,
                                top_p)
             C.hist_append(self.hist, self.ctr, toks, self.KMAX)
             self.tok_in.copy_(toks)      # feed the next replay

@@ -233,7 +233,7 @@ def sample_tokens(logits: torch.Tensor, seeds: torch.Tensor, top_k: int = 40,
                   temperature: float = 0.7, top_p: float = 0.95) -> torch.Tensor:
     out = torch.empty(logits.size(0), dtype=torch.int32, device=logits.device)
     # v3: register-resident top-8 scan + tournament (4× the v2 LDS variant)
-    _require().sample_tokens_v3(out, logits, seeds, top_k, temperature, top_p)
+    _require().sample_tokens_v4(out, logits, seeds, top_k, temperature, top_p)
     return out
 
 

@@ -62,6 +62,9 @@ void gemv_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
 void sample_tokens_v3(torch::Tensor out_tokens, torch::Tensor logits,
                       torch::Tensor seeds, int64_t top_k, double temperature,
                       double top_p);
+void sample_tokens_v4(torch::Tensor out_tokens, torch::Tensor logits,
+                      torch::Tensor seeds, int64_t top_k, double temperature,
+                      double top_p);
 void sample_scan_probe(torch::Tensor out, torch::Tensor logits);
 void hist_append(torch::Tensor hist, torch::Tensor ctr, torch::Tensor toks,
                  int64_t kmax);
@@ -96,6 +99,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemv_addnorm", &gemv_addnorm,
         "fused residual-add + RMSNorm + GEMV (decode)");
   m.def("sample_tokens_v3", &sample_tokens_v3, "register top-8 sampler");
+  m.def("sample_tokens_v4", &sample_tokens_v4,
+        "two-stage parallel-scan sampler");
   m.def("sample_scan_probe", &sample_scan_probe, "scan-cost probe");
   m.def("hist_append", &hist_append, "token-history append (multi-step decode)");
   m.def("vs_topk", &vs_topk, "vector-store cosine top-k over bf16 matrix");

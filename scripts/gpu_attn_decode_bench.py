@@ -53,7 +53,7 @@ def run(B: int, L: int, iters: int = 50) -> float:
     us = (time.perf_counter() - t0) / iters * 1e6
     kv_mb = B * L * HK * D * 2 * 2 / 1e6
     print(f"B={B} L={L}: {us:7.1f} us/call  (KV {kv_mb:.0f} MB -> "
-          f"{kv_mb / us * 1e6 / 1e9:.2f} TB/s effective)")
+          f"{kv_mb / us:.2f} TB/s effective)")
     return us
 
 
