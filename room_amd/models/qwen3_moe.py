@@ -140,9 +140,10 @@ class Qwen3MoEModel:
                                         device=dev)
             x_alt = torch.empty_like(x)     # residual ping-pong for gemv_addnorm
             empty_delta = torch.empty(0, dtype=torch.float32, device=dev)
-            part = torch.empty(T, cfg.num_q_heads, 32, cfg.head_dim,
+            nsp = ops.attn_nsplits()
+            part = torch.empty(T, cfg.num_q_heads, nsp, cfg.head_dim,
                                dtype=torch.float32, device=dev)
-            part_ml = torch.empty(T, cfg.num_q_heads, 32, 2, dtype=torch.float32,
+            part_ml = torch.empty(T, cfg.num_q_heads, nsp, 2, dtype=torch.float32,
                                   device=dev)
 
         for li, layer in enumerate(self.layers):

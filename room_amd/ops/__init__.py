@@ -82,12 +82,19 @@ def write_kv(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
     _require().write_kv(kcache, vcache, k, v, block_table, seq_ids, q_pos)
 
 
+def attn_nsplits() -> int:
+    """Decode split-KV NSPLITS (compiled into the kernel)."""
+    if _C is not None and hasattr(_C, "attn_nsplits"):
+        return int(_C.attn_nsplits())
+    return 64
+
+
 def paged_attention_split(out: torch.Tensor, q: torch.Tensor, kcache: torch.Tensor,
                           vcache: torch.Tensor, block_table: torch.Tensor,
                           seq_ids: torch.Tensor, q_pos: torch.Tensor,
                           part: torch.Tensor, part_ml: torch.Tensor,
                           scale: float) -> torch.Tensor:
-    """Split-KV flash-decode: part [T, Hq, 32, 128] f32, part_ml [T, Hq, 32, 2]."""
+    """Split-KV flash-decode: part [T, Hq, NSPLITS, 128] f32, part_ml [.., 2]."""
     _require().paged_attention_split(out, q, kcache, vcache, block_table,
                                      seq_ids, q_pos, part, part_ml, scale)
     return out

@@ -51,6 +51,7 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
                            torch::Tensor block_table, torch::Tensor seq_ids,
                            torch::Tensor q_pos, torch::Tensor part,
                            torch::Tensor part_ml, double scale);
+int64_t attn_nsplits();
 void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                    torch::Tensor vcache, torch::Tensor block_table,
                    torch::Tensor seq_ids, torch::Tensor q_pos,
@@ -92,6 +93,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_combine", &moe_combine, "weighted scatter-add combine");
   m.def("moe_combine_gather", &moe_combine_gather, "atomics-free MoE combine");
   m.def("sample_tokens", &sample_tokens, "fused temperature/top-k/top-p sampling");
+  m.def("attn_nsplits", &attn_nsplits, "NSPLITS constant");
   m.def("paged_attention_split", &paged_attention_split,
         "split-KV flash-decode paged attention");
   m.def("flash_prefill", &flash_prefill, "MFMA flash-attention prefill");

@@ -201,7 +201,7 @@ void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
 // head); each writes an unnormalized partial (acc, m, l); a merge kernel
 // combines. NSPLITS is static so the decode step stays hipGraph-capturable.
 
-#define NSPLITS 32
+#define NSPLITS 64
 
 __global__ __launch_bounds__(256)
 void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
@@ -355,6 +355,8 @@ void paged_attn_merge_kernel(short* __restrict__ out,       // [T, Hq, D]
   out[((long)t * n_qheads + hq) * HEAD_DIM + d] =
       f2bf(l_tot > 0.f ? a / l_tot : 0.f);
 }
+
+int64_t attn_nsplits() { return NSPLITS; }
 
 void paged_attention_split(torch::Tensor out, torch::Tensor q,
                            torch::Tensor kcache, torch::Tensor vcache,

@@ -15,7 +15,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from room_amd import ops  # noqa: E402
 
 HK, HQ, D, BS = 4, 32, 128, 16
-NSPLITS = 32
+NSPLITS = ops.attn_nsplits()
 
 
 def run(B: int, L: int, iters: int = 50) -> float:

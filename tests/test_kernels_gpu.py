@@ -308,8 +308,9 @@ def test_paged_attention_split_matches_ref():
     seq_ids = torch.arange(B, dtype=torch.int32, device=DEV)
     q_pos = torch.tensor([l - 1 for l in lens], dtype=torch.int32, device=DEV)
     out = torch.empty_like(q)
-    part = torch.empty(B, Hq, 32, D, dtype=torch.float32, device=DEV)
-    part_ml = torch.empty(B, Hq, 32, 2, dtype=torch.float32, device=DEV)
+    nsp = ops.attn_nsplits()
+    part = torch.empty(B, Hq, nsp, D, dtype=torch.float32, device=DEV)
+    part_ml = torch.empty(B, Hq, nsp, 2, dtype=torch.float32, device=DEV)
     scale = D ** -0.5
     ops.paged_attention_split(out, q, kcache, vcache, bt, seq_ids, q_pos,
                               part, part_ml, scale)
