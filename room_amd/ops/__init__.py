@@ -86,7 +86,7 @@ def attn_nsplits() -> int:
     """Decode split-KV NSPLITS (compiled into the kernel)."""
     if _C is not None and hasattr(_C, "attn_nsplits"):
         return int(_C.attn_nsplits())
-    return 64
+    return 32
 
 
 def paged_attention_split(out: torch.Tensor, q: torch.Tensor, kcache: torch.Tensor,

@@ -201,7 +201,10 @@ void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
 // head); each writes an unnormalized partial (acc, m, l); a merge kernel
 // combines. NSPLITS is static so the decode step stays hipGraph-capturable.
 
-#define NSPLITS 64
+// NSPLITS=64 A/B: wins only at L>=16k (63.6->50.0 us B=1), loses at the
+// bench's 4-6k sessions (8.1->8.3 ms/step) and at B=1 (4.5->5.0): empty-split
+// dispatch + 2x merge traffic. Static 32 keeps the decode graph replayable.
+#define NSPLITS 32
 
 __global__ __launch_bounds__(256)
 void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
