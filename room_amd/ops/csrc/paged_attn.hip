@@ -207,10 +207,8 @@ void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
 #define NSPLITS 32
 
 __global__ __launch_bounds__(256)
-void paged_attn_split_kernel(short* __restrict__ out,     // [T, Hq, D] bf16
-                             float* __restrict__ part,   // [T, Hq, NSPLITS, D]
+void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
                              float* __restrict__ part_ml, // [T, Hq, NSPLITS, 2]
-                             int* __restrict__ ctr,       // [T, Hk] zeroed once
                              const short* __restrict__ q,
                              const short* __restrict__ kcache,
                              const short* __restrict__ vcache,
@@ -239,9 +237,13 @@ void paged_attn_split_kernel(short* __restrict__ out,     // [T, Hq, D] bf16
   float* acc_out = part + (((long)t * n_qheads + hq) * NSPLITS + split) * HEAD_DIM;
 
   if (lo >= hi) {  // empty split still writes a neutral partial
-    if (sub == 0) { ml[0] = -INFINITY; ml[1] = 0.f; }
-    *reinterpret_cast<f32x4*>(acc_out + sub * 4) = f32x4{0.f, 0.f, 0.f, 0.f};
-  } else {
+    if (h < QH_PER_KV) {
+      if (sub == 0) { ml[0] = -INFINITY; ml[1] = 0.f; }
+      *reinterpret_cast<f32x4*>(acc_out + sub * 4) = f32x4{0.f, 0.f, 0.f, 0.f};
+    }
+    return;
+  }
+
   // K/V chunks are staged through LDS cooperatively (fully-coalesced bulk
   // HBM loads, one read per WG instead of one per q-head); the score and
   // value phases then run out of LDS. Row pad of 8 bf16 keeps the per-lane
@@ -327,52 +329,34 @@ void paged_attn_split_kernel(short* __restrict__ out,     // [T, Hq, D] bf16
   if (sub == 0) { ml[0] = m_run; ml[1] = l_run; }
   *reinterpret_cast<f32x4*>(acc_out + sub * 4) =
       f32x4{acc[0], acc[1], acc[2], acc[3]};
-  }  // lo < hi
+}
 
-  // ---- last-WG-merges: the final split WG of (t, hk) combines all NSPLITS
-  // partials for its 8 q-heads in place of a separate merge kernel (removes
-  // one dependent launch + fill/drain bubble per layer). Counter self-resets
-  // so the buffer stays zeroed for the next (graph-replayed) launch.
-  __shared__ int amlast;
-  __threadfence();                 // partials visible device-wide
-  if (tid == 0) {
-    const int prev = atomicAdd(&ctr[(long)t * n_kvheads + hk], 1);
-    amlast = (prev == NSPLITS - 1);
-    if (amlast) ctr[(long)t * n_kvheads + hk] = 0;
-  }
-  __syncthreads();
-  if (!amlast) return;
-  __threadfence();                 // acquire other WGs' partials
+// merge: one wave per (t, hq); lane d-pairs combine the NSPLITS partials
+__global__ __launch_bounds__(128)
+void paged_attn_merge_kernel(short* __restrict__ out,       // [T, Hq, D]
+                             const float* __restrict__ part,
+                             const float* __restrict__ part_ml,
+                             int n_qheads) {
+  const int t = blockIdx.x;
+  const int hq = blockIdx.y;
+  const int d = threadIdx.x;  // 128 threads = one dim each
+  const float* ml = part_ml + (((long)t * n_qheads + hq) * NSPLITS) * 2;
+  const float* pacc = part + (((long)t * n_qheads + hq) * NSPLITS) * HEAD_DIM;
 
-  // 256 threads = 8 heads × 32 lanes × 4 dims
-  {
-    const int mh = tid >> 5;                 // head within the GQA group
-    const int d0 = (tid & 31) * 4;
-    const int mhq = hk * QH_PER_KV + mh;
-    const float* mml = part_ml + (((long)t * n_qheads + mhq) * NSPLITS) * 2;
-    const float* pacc = part + (((long)t * n_qheads + mhq) * NSPLITS) * HEAD_DIM;
-    float m_star = -INFINITY;
-    #pragma unroll
-    for (int s = 0; s < NSPLITS; ++s) m_star = fmaxf(m_star, mml[2 * s]);
-    float l_tot = 0.f;
-    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
-    #pragma unroll 4
-    for (int s = 0; s < NSPLITS; ++s) {
-      const float ms = mml[2 * s];
-      if (ms == -INFINITY) continue;
-      const float f = __expf(ms - m_star);
-      l_tot += mml[2 * s + 1] * f;
-      const f32x4 v = *reinterpret_cast<const f32x4*>(
-          pacc + (long)s * HEAD_DIM + d0);
-      a0 += v[0] * f; a1 += v[1] * f; a2 += v[2] * f; a3 += v[3] * f;
-    }
-    const float inv = (l_tot > 0.f) ? 1.0f / l_tot : 0.f;
-    bf16x4 o;
-    o[0] = f2bf(a0 * inv); o[1] = f2bf(a1 * inv);
-    o[2] = f2bf(a2 * inv); o[3] = f2bf(a3 * inv);
-    *reinterpret_cast<bf16x4*>(
-        out + ((long)t * n_qheads + mhq) * HEAD_DIM + d0) = o;
+  float m_star = -INFINITY;
+  #pragma unroll
+  for (int s = 0; s < NSPLITS; ++s) m_star = fmaxf(m_star, ml[2 * s]);
+  float l_tot = 0.f, a = 0.f;
+  #pragma unroll
+  for (int s = 0; s < NSPLITS; ++s) {
+    const float ms = ml[2 * s];
+    if (ms == -INFINITY) continue;
+    const float f = __expf(ms - m_star);
+    l_tot += ml[2 * s + 1] * f;
+    a += pacc[(long)s * HEAD_DIM + d] * f;
   }
+  out[((long)t * n_qheads + hq) * HEAD_DIM + d] =
+      f2bf(l_tot > 0.f ? a / l_tot : 0.f);
 }
 
 int64_t attn_nsplits() { return NSPLITS; }
@@ -389,25 +373,18 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
   const int n_qheads = n_kvheads * QH_PER_KV;
   const int max_blocks = block_table.size(1);
   TORCH_CHECK(part.size(0) >= T && part.size(2) == NSPLITS);
-  // persistent zeroed completion counters per (T, Hk): the kernel's last-WG
-  // protocol self-resets them, so they are allocated+zeroed exactly once —
-  // allocation-free on replay (hipGraph-capturable)
-  static std::unordered_map<long, torch::Tensor> ctrs;
-  const long key = (long)T * 1024 + n_kvheads;
-  auto it = ctrs.find(key);
-  if (it == ctrs.end())
-    it = ctrs.emplace(key, torch::zeros(
-        {(long)T * n_kvheads},
-        torch::TensorOptions().device(q.device()).dtype(torch::kInt32))).first;
   hipStream_t s = c10::hip::getCurrentHIPStream();
   dim3 g1(T, n_kvheads, NSPLITS);
   hipLaunchKernelGGL(paged_attn_split_kernel, g1, dim3(256), 0, s,
-                     (short*)out.data_ptr(),
                      part.data_ptr<float>(), part_ml.data_ptr<float>(),
-                     it->second.data_ptr<int>(),
                      (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
                      (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
                      seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
                      n_kvheads, max_blocks, (int)q.stride(0), (float)scale);
+  HIP_CHECK_KERNEL();
+  dim3 g2(T, n_qheads);
+  hipLaunchKernelGGL(paged_attn_merge_kernel, g2, dim3(HEAD_DIM), 0, s,
+                     (short*)out.data_ptr(), part.data_ptr<float>(),
+                     part_ml.data_ptr<float>(), n_qheads);
   HIP_CHECK_KERNEL();
 }
