@@ -64,3 +64,32 @@ def test_single_rank_fallbacks():
     assert ctx.broadcast_blob({"a": 1}) == {"a": 1}
     v, i = ctx.topk_merge(torch.tensor([0.1, 0.9]), torch.tensor([5, 6]), k=1)
     assert i.tolist() == [6]
+
+
+def test_bench_torchrun_world2_cpu():
+    """The driver's exact multi-GPU launch shape (torch.distributed.run,
+    nproc 2, 127.0.0.1) against bench.py on CPU/gloo + StubEngine: the rank-0
+    JSON contract line must come out well-formed with n_gpus=2."""
+    import json
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    port = _find_free_port()
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), str(repo / "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=240, cwd=repo)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    line = next(l for l in proc.stdout.splitlines()
+                if l.startswith("{") and '"metric"' in l)
+    out = json.loads(line)
+    assert out["metric"] == "agent-cycles/sec"
+    assert out["n_gpus"] == 2
+    assert out["steps"] == 2
+    assert out["value"] > 0
+    assert out["scaling"] == "weak"
+    assert out["config"]["agents_per_gpu"] == 5
