@@ -233,10 +233,11 @@ class Qwen3MoEModel:
         T = hbuf.size(0)
         K = cfg.num_experts_per_tok
         H, I = cfg.hidden_size, cfg.moe_intermediate_size
-        out = torch.zeros(T, H, dtype=torch.float32, device=hbuf.device)
 
         if T < cfg.moe_grouped_threshold:
+            out = torch.empty(T, H, dtype=torch.float32, device=hbuf.device)
             if self.moe_dedup:
+                out.zero_()
                 # expert-deduped GEMV (weight rows once per active expert);
                 # measured slower than the pair path at B<=8 on random routing
                 # -- kept behind ROOMAMD_MOE_DEDUP=1 pending a win
@@ -249,7 +250,8 @@ class Qwen3MoEModel:
             pair_w = topk_w.flatten().contiguous()
             P = pair_token.numel()
             h = torch.empty(P, I, dtype=torch.bfloat16, device=hbuf.device)
-            ops.moe_gemv_h(h, hbuf, layer.w13, pair_token, pair_expert)
+            ops.moe_gemv_h(h, hbuf, layer.w13, pair_token, pair_expert,
+                           out_zero=out)
             ops.moe_gemv_down(out, h, layer.w2, pair_w, pair_token, pair_expert)
             return out
 

@@ -160,8 +160,13 @@ def router_topk(x: torch.Tensor, wr: torch.Tensor,
 
 
 def moe_gemv_h(h: torch.Tensor, x: torch.Tensor, w13: torch.Tensor,
-               pair_token: torch.Tensor, pair_expert: torch.Tensor) -> torch.Tensor:
-    _require().moe_gemv_h(h, x, w13, pair_token, pair_expert)
+               pair_token: torch.Tensor, pair_expert: torch.Tensor,
+               out_zero: torch.Tensor | None = None) -> torch.Tensor:
+    """Gate/up GEMV + silu-mul per (token, expert) pair. out_zero: optional
+    f32 tensor zero-filled as a side job (the down kernel's accumulator)."""
+    if out_zero is None:
+        out_zero = torch.empty(0, dtype=torch.float32, device=h.device)
+    _require().moe_gemv_h(h, x, w13, pair_token, pair_expert, out_zero)
     return h
 
 

@@ -25,7 +25,8 @@ void moe_router(torch::Tensor topk_ids, torch::Tensor topk_w, torch::Tensor logi
 void router_topk(torch::Tensor topk_ids, torch::Tensor topk_w, torch::Tensor x,
                  torch::Tensor wr, int64_t K);
 void moe_gemv_h(torch::Tensor h, torch::Tensor x, torch::Tensor w13,
-                torch::Tensor pair_token, torch::Tensor pair_expert);
+                torch::Tensor pair_token, torch::Tensor pair_expert,
+                torch::Tensor out_zero);
 void moe_gemv_down(torch::Tensor out, torch::Tensor h, torch::Tensor w2,
                    torch::Tensor pair_w, torch::Tensor pair_token,
                    torch::Tensor pair_expert);

@@ -146,7 +146,17 @@ void moe_gemv_h_kernel(short* __restrict__ h,             // [P, I]
                        const short* __restrict__ w13,      // [E, 2I, H]
                        const int* __restrict__ pair_token,  // [P]
                        const int* __restrict__ pair_expert, // [P]
-                       int H, int I) {
+                       float* __restrict__ out_zero,        // [zero_n] or null
+                       long zero_n, int H, int I) {
+  // side job: zero the downstream accumulator (out of moe_gemv_down) here —
+  // the grid has ~2M threads vs ~10k floats to clear, and this kernel always
+  // runs right before the down kernel, so the separate fill launch it
+  // replaces (4.8 µs/layer, 2.8% of the decode step) is absorbed for free.
+  if (out_zero != nullptr) {
+    const long gid = ((long)blockIdx.y * gridDim.x + blockIdx.x) * blockDim.x
+                     + threadIdx.x;
+    if (gid < zero_n) out_zero[gid] = 0.f;
+  }
   const int p = blockIdx.x;
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -353,15 +363,25 @@ void router_topk(torch::Tensor topk_ids, torch::Tensor topk_w, torch::Tensor x,
 }
 
 void moe_gemv_h(torch::Tensor h, torch::Tensor x, torch::Tensor w13,
-                torch::Tensor pair_token, torch::Tensor pair_expert) {
+                torch::Tensor pair_token, torch::Tensor pair_expert,
+                torch::Tensor out_zero) {
   const int P = pair_token.size(0);
   const int H = x.size(-1), I = h.size(-1);
   dim3 grid(P, (I + 3) / 4), block(256);
+  float* zp = nullptr;
+  long zn = 0;
+  if (out_zero.defined() && out_zero.numel() > 0) {
+    TORCH_CHECK(out_zero.dtype() == torch::kFloat32);
+    zp = out_zero.data_ptr<float>();
+    zn = out_zero.numel();
+    TORCH_CHECK(zn <= (long)grid.x * grid.y * block.x,
+                "zero target larger than grid coverage");
+  }
   hipStream_t s = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(moe_gemv_h_kernel, grid, block, 0, s,
                      (short*)h.data_ptr(), (const short*)x.data_ptr(),
                      (const short*)w13.data_ptr(), pair_token.data_ptr<int>(),
-                     pair_expert.data_ptr<int>(), H, I);
+                     pair_expert.data_ptr<int>(), zp, zn, H, I);
   HIP_CHECK_KERNEL();
 }
 
