@@ -136,8 +136,6 @@ class Qwen3MoEModel:
         if decode:
             qkv = torch.empty(T, qdim + 2 * kvdim, dtype=torch.bfloat16, device=dev)
             obuf = torch.empty(T, cfg.hidden_size, dtype=torch.bfloat16, device=dev)
-            router_logits = torch.empty(T, cfg.num_experts, dtype=torch.float32,
-                                        device=dev)
             x_alt = torch.empty_like(x)     # residual ping-pong for gemv_addnorm
             empty_delta = torch.empty(0, dtype=torch.float32, device=dev)
             nsp = ops.attn_nsplits()
@@ -206,11 +204,13 @@ class Qwen3MoEModel:
             # per token serializes the 512 KB router-weight read that the
             # dense GEMV spreads over 128 waves. Two kernels win.
             if decode:
-                ops.gemv(router_logits, hbuf, layer.router_w)
+                # H-split partial GEMV (4x grid) + top-k summing the partials
+                topk_ids, topk_w = ops.router_gemv_topk(hbuf, layer.router_w,
+                                                        cfg.num_experts_per_tok)
             else:
                 router_logits = F.linear(hbuf, layer.router_w).float()
-            topk_ids, topk_w = ops.moe_router(router_logits,
-                                              cfg.num_experts_per_tok)
+                topk_ids, topk_w = ops.moe_router(router_logits,
+                                                  cfg.num_experts_per_tok)
             # f32 accumulator feeds fused_add_rmsnorm directly (templated
             # input dtype — skips a cast kernel per layer)
             moe_out = self._moe(hbuf, layer, topk_ids, topk_w)

@@ -149,6 +149,17 @@ def moe_router(logits: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor
     return ids, w
 
 
+def router_gemv_topk(x: torch.Tensor, wr: torch.Tensor,
+                     k: int) -> tuple[torch.Tensor, torch.Tensor]:
+    """Decode router: 4-way H-split partial GEMV (wide grid) + top-k kernel
+    that sums the partials on load. Replaces gemv + moe_router."""
+    T = x.size(0)
+    ids = torch.empty(T, k, dtype=torch.int32, device=x.device)
+    w = torch.empty(T, k, dtype=torch.float32, device=x.device)
+    _require().router_gemv_topk(ids, w, x, wr, k)
+    return ids, w
+
+
 def router_topk(x: torch.Tensor, wr: torch.Tensor,
                 k: int) -> tuple[torch.Tensor, torch.Tensor]:
     """Fused decode router: logits = x @ wr.T → softmax → top-k → renorm."""

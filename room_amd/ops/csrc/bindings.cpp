@@ -24,6 +24,8 @@ void moe_router(torch::Tensor topk_ids, torch::Tensor topk_w, torch::Tensor logi
                 int64_t K);
 void router_topk(torch::Tensor topk_ids, torch::Tensor topk_w, torch::Tensor x,
                  torch::Tensor wr, int64_t K);
+void router_gemv_topk(torch::Tensor topk_ids, torch::Tensor topk_w,
+                      torch::Tensor x, torch::Tensor wr, int64_t K);
 void moe_gemv_h(torch::Tensor h, torch::Tensor x, torch::Tensor w13,
                 torch::Tensor pair_token, torch::Tensor pair_expert,
                 torch::Tensor out_zero);
@@ -85,6 +87,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("write_kv", &write_kv, "scatter K/V rows into paged cache");
   m.def("moe_router", &moe_router, "softmax top-k router");
   m.def("router_topk", &router_topk, "fused router GEMV + softmax top-k (decode)");
+  m.def("router_gemv_topk", &router_gemv_topk,
+        "H-split router GEMV + top-k (two kernels, wide grid)");
   m.def("moe_gemv_h", &moe_gemv_h, "MoE gate/up GEMV + silu-mul (decode)");
   m.def("moe_gemv_down", &moe_gemv_down, "MoE down GEMV + weighted scatter-add");
   m.def("moe_grouped_gemm", &moe_grouped_gemm, "grouped MFMA GEMM (prefill)");

@@ -435,3 +435,19 @@ def test_gemv_addnorm_matches_composition():
         yf = torch.empty(B, N, dtype=torch.float32, device=DEV)
         ops.gemv_addnorm(yf, x.clone(), delta, torch.empty_like(x), gamma, w)
         assert bf16_close(yf, y_ref, atol=6e-2, rtol=6e-2)
+
+
+def test_router_gemv_topk_split():
+    """H-split router GEMV + top-k vs fp32 torch reference."""
+    torch.manual_seed(13)
+    for T, E, H, K in [(5, 128, 2048, 8), (1, 16, 512, 4), (8, 128, 2048, 8)]:
+        x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV) * 0.5
+        wr = torch.randn(E, H, dtype=torch.bfloat16, device=DEV) * 0.05
+        ids, w = ops.router_gemv_topk(x, wr, K)
+        ids_ref, w_ref = ref.moe_router_ref(x.float() @ wr.float().t(), K)
+        for t in range(T):
+            assert set(ids[t].tolist()) == set(ids_ref[t].tolist())
+            m = {int(i): float(v) for i, v in zip(ids[t], w[t])}
+            mr = {int(i): float(v) for i, v in zip(ids_ref[t], w_ref[t])}
+            for e in m:
+                assert m[e] == pytest.approx(mr[e], abs=5e-3)
