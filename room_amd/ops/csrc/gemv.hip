@@ -10,20 +10,23 @@
 
 #define GEMV_MAXB 8
 
-template <bool F32OUT>
+// BN = exact batch (compile-time): the generic MAXB=8 unroll carried 8 acc
+// registers + per-lane bounds branches at any B — register pressure blocked
+// software-pipelining of the 8 independent H-iterations.
+template <bool F32OUT, int BN>
 __global__ __launch_bounds__(256)
-void gemv_kernel(void* __restrict__ y,           // [B, N] bf16 or f32
-                 const short* __restrict__ x,     // [B, H]
+void gemv_kernel(void* __restrict__ y,           // [BN, N] bf16 or f32
+                 const short* __restrict__ x,     // [BN, H]
                  const short* __restrict__ w,     // [N, H]
-                 int B, int H, int N) {
+                 int H, int N) {
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int n = blockIdx.x * 4 + wid;
   if (n >= N) return;
 
-  float acc[GEMV_MAXB];
+  float acc[BN];
   #pragma unroll
-  for (int b = 0; b < GEMV_MAXB; ++b) acc[b] = 0.f;
+  for (int b = 0; b < BN; ++b) acc[b] = 0.f;
 
   const short* wrow = w + (long)n * H;
   for (int base = lane * 8; base < H; base += WAVE * 8) {
@@ -32,44 +35,52 @@ void gemv_kernel(void* __restrict__ y,           // [B, N] bf16 or f32
     #pragma unroll
     for (int j = 0; j < 8; ++j) wf[j] = bf2f(wv[j]);
     #pragma unroll
-    for (int b = 0; b < GEMV_MAXB; ++b) {
-      if (b < B) {
-        bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) acc[b] += wf[j] * bf2f(xv[j]);
-      }
+    for (int b = 0; b < BN; ++b) {
+      bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) acc[b] += wf[j] * bf2f(xv[j]);
     }
   }
   #pragma unroll
-  for (int b = 0; b < GEMV_MAXB; ++b) {
-    if (b < B) {
-      float r = wave_reduce_sum(acc[b]);
-      if (lane == 0) {
-        if (F32OUT)
-          ((float*)y)[(long)b * N + n] = r;
-        else
-          ((short*)y)[(long)b * N + n] = f2bf(r);
-      }
+  for (int b = 0; b < BN; ++b) {
+    float r = wave_reduce_sum(acc[b]);
+    if (lane == 0) {
+      if (F32OUT)
+        ((float*)y)[(long)b * N + n] = r;
+      else
+        ((short*)y)[(long)b * N + n] = f2bf(r);
     }
+  }
+}
+
+template <bool F32OUT>
+static void gemv_launch(void* y, const short* x, const short* w, int B, int H,
+                        int N, hipStream_t s) {
+  dim3 grid((N + 3) / 4), block(256);
+  switch (B) {
+#define GEMV_CASE(BN) \
+    case BN: hipLaunchKernelGGL((gemv_kernel<F32OUT, BN>), grid, block, 0, s, \
+                                y, x, w, H, N); break;
+    GEMV_CASE(1) GEMV_CASE(2) GEMV_CASE(3) GEMV_CASE(4)
+    GEMV_CASE(5) GEMV_CASE(6) GEMV_CASE(7) GEMV_CASE(8)
+#undef GEMV_CASE
+    default: TORCH_CHECK(false, "gemv handles B<=8");
   }
 }
 
 void gemv(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
   const int B = x.size(0), H = x.size(1), N = w.size(0);
-  TORCH_CHECK(B <= GEMV_MAXB, "gemv handles B<=8 (decode); use hipBLASLt above");
+  TORCH_CHECK(B >= 1 && B <= GEMV_MAXB,
+              "gemv handles B<=8 (decode); use hipBLASLt above");
   TORCH_CHECK(x.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16);
   TORCH_CHECK(H % (WAVE * 8) == 0, "H must be a multiple of 512");
-  dim3 grid((N + 3) / 4), block(256);
   hipStream_t s = c10::hip::getCurrentHIPStream();
-  if (y.dtype() == torch::kFloat32) {
-    hipLaunchKernelGGL(gemv_kernel<true>, grid, block, 0, s,
-                       y.data_ptr(), (const short*)x.data_ptr(),
-                       (const short*)w.data_ptr(), B, H, N);
-  } else {
-    hipLaunchKernelGGL(gemv_kernel<false>, grid, block, 0, s,
-                       y.data_ptr(), (const short*)x.data_ptr(),
-                       (const short*)w.data_ptr(), B, H, N);
-  }
+  if (y.dtype() == torch::kFloat32)
+    gemv_launch<true>(y.data_ptr(), (const short*)x.data_ptr(),
+                      (const short*)w.data_ptr(), B, H, N, s);
+  else
+    gemv_launch<false>(y.data_ptr(), (const short*)x.data_ptr(),
+                       (const short*)w.data_ptr(), B, H, N, s);
   HIP_CHECK_KERNEL();
 }
 
