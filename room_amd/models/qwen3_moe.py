@@ -222,9 +222,14 @@ class Qwen3MoEModel:
                                device=x.device)
         ops.fused_add_rmsnorm(hbuf, x, moe_out, self.final_norm_w, cfg.rms_eps)
         sel = hbuf if logits_rows is None else hbuf[logits_rows.long()]
-        # lm_head stays on hipBLASLt: at N=152k it sustains ~3 TB/s even for
-        # M≤8 (measured vs our gemv's 2.5 TB/s — scripts/gpu_op_microbench.py)
-        logits = F.linear(sel, self.lm_head).float()
+        if sel.size(0) <= 8:
+            # B-specialized GEMV (the earlier 2.5 TB/s vs hipBLASLt 3 TB/s
+            # comparison predates the exact-batch template)
+            logits = torch.empty(sel.size(0), cfg.vocab_size,
+                                 dtype=torch.float32, device=x.device)
+            ops.gemv(logits, sel.contiguous(), self.lm_head)
+        else:
+            logits = F.linear(sel, self.lm_head).float()
         return logits
 
     def _moe(self, hbuf: torch.Tensor, layer: Qwen3MoELayer,
