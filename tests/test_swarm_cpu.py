@@ -76,13 +76,17 @@ def test_bench_torchrun_world2_cpu():
     from pathlib import Path
 
     repo = Path(__file__).resolve().parent.parent
-    port = _find_free_port()
-    proc = subprocess.run(
-        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", str(port), str(repo / "bench.py"),
-         "--gpus", "2", "--steps", "2", "--warmup", "1"],
-        capture_output=True, text=True, timeout=240, cwd=repo)
+    proc = None
+    for attempt in range(3):  # master-port races with concurrent tests
+        port = _find_free_port()
+        proc = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", str(port), str(repo / "bench.py"),
+             "--gpus", "2", "--steps", "2", "--warmup", "1"],
+            capture_output=True, text=True, timeout=240, cwd=repo)
+        if proc.returncode == 0:
+            break
     assert proc.returncode == 0, proc.stderr[-2000:]
     line = next(l for l in proc.stdout.splitlines()
                 if l.startswith("{") and '"metric"' in l)
