@@ -92,15 +92,15 @@ void gemv(torch::Tensor y, torch::Tensor x, torch::Tensor w) {
 // ~14 dependent small kernels/layer; every removed kernel removes a
 // launch+fill/drain bubble — see profiles/PERF_NOTES.md).
 // Norm matches fused_add_rmsnorm semantics: computed over the bf16-rounded sum.
-template <bool F32OUT, bool DELTA_F32, bool HAS_DELTA>
+template <bool F32OUT, bool DELTA_F32, bool HAS_DELTA, int BN>
 __global__ __launch_bounds__(256)
-void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
-                         const short* __restrict__ x,      // [B, H]
-                         const void* __restrict__ delta_,  // [B, H] or null
-                         short* __restrict__ x_out,        // [B, H]
+void gemv_addnorm_kernel(void* __restrict__ y,            // [BN, N]
+                         const short* __restrict__ x,      // [BN, H]
+                         const void* __restrict__ delta_,  // [BN, H] or null
+                         short* __restrict__ x_out,        // [BN, H]
                          const short* __restrict__ gamma,  // [H]
                          const short* __restrict__ w,      // [N, H]
-                         int B, int H, int N, float eps) {
+                         int H, int N, float eps) {
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int n = blockIdx.x * 4 + wid;
@@ -108,12 +108,12 @@ void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
   // pass 1 (block-cooperative): row sums of squares, bf16-rounded to match
   // fused_add_rmsnorm. All 256 threads share the work (a per-wave pass was
   // measured as a B=5 regression: 4× redundant VALU across the block).
-  __shared__ float ss_sh[GEMV_MAXB];
-  if (threadIdx.x < GEMV_MAXB) ss_sh[threadIdx.x] = 0.f;
+  __shared__ float ss_sh[BN];
+  if (threadIdx.x < BN) ss_sh[threadIdx.x] = 0.f;
   __syncthreads();
   #pragma unroll
-  for (int b = 0; b < GEMV_MAXB; ++b) {
-    if (b < B) {
+  for (int b = 0; b < BN; ++b) {
+    {
       float ss = 0.f;
       for (int base = threadIdx.x * 8; base < H; base += 256 * 8) {
         bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
@@ -135,17 +135,17 @@ void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
     }
   }
   __syncthreads();
-  float scale[GEMV_MAXB];
+  float scale[BN];
   #pragma unroll
-  for (int b = 0; b < GEMV_MAXB; ++b)
-    if (b < B) scale[b] = rsqrtf(ss_sh[b] / H + eps);
+  for (int b = 0; b < BN; ++b)
+    scale[b] = rsqrtf(ss_sh[b] / H + eps);
 
   // pass 1.5 (block-cooperative): stage xn = (x+δ)·scale·γ into LDS once, so
   // the dot loop is lean (w-load + LDS-load + FMA; no per-wave re-add/round)
   extern __shared__ short xn_sh[];  // [B][H] bf16
   #pragma unroll
-  for (int b = 0; b < GEMV_MAXB; ++b) {
-    if (b < B) {
+  for (int b = 0; b < BN; ++b) {
+    {
       for (int base = threadIdx.x * 8; base < H; base += 256 * 8) {
         bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
         bf16x8 gv = *reinterpret_cast<const bf16x8*>(gamma + base);
@@ -170,7 +170,7 @@ void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
 
   // block 0 stores the updated residual (other blocks never read x_out here)
   if (HAS_DELTA && blockIdx.x == 0) {
-    for (int i = threadIdx.x * 8; i < B * H; i += 256 * 8) {
+    for (int i = threadIdx.x * 8; i < BN * H; i += 256 * 8) {
       bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i);
       bf16x8 o;
       #pragma unroll
@@ -186,9 +186,9 @@ void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
   if (n >= N) return;
 
   // pass 2: lean dot against the LDS-staged normed rows
-  float acc[GEMV_MAXB];
+  float acc[BN];
   #pragma unroll
-  for (int b = 0; b < GEMV_MAXB; ++b) acc[b] = 0.f;
+  for (int b = 0; b < BN; ++b) acc[b] = 0.f;
   const short* wrow = w + (long)n * H;
   for (int base = lane * 8; base < H; base += WAVE * 8) {
     bf16x8 wv = *reinterpret_cast<const bf16x8*>(wrow + base);
@@ -196,22 +196,18 @@ void gemv_addnorm_kernel(void* __restrict__ y,            // [B, N]
     #pragma unroll
     for (int j = 0; j < 8; ++j) wf[j] = bf2f(wv[j]);
     #pragma unroll
-    for (int b = 0; b < GEMV_MAXB; ++b) {
-      if (b < B) {
-        bf16x8 xv = *reinterpret_cast<const bf16x8*>(&xn_sh[b * H + base]);
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) acc[b] += wf[j] * bf2f(xv[j]);
-      }
+    for (int b = 0; b < BN; ++b) {
+      bf16x8 xv = *reinterpret_cast<const bf16x8*>(&xn_sh[b * H + base]);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) acc[b] += wf[j] * bf2f(xv[j]);
     }
   }
   #pragma unroll
-  for (int b = 0; b < GEMV_MAXB; ++b) {
-    if (b < B) {
-      float r = wave_reduce_sum(acc[b]);
-      if (lane == 0) {
-        if (F32OUT) ((float*)y)[(long)b * N + n] = r;
-        else ((short*)y)[(long)b * N + n] = f2bf(r);
-      }
+  for (int b = 0; b < BN; ++b) {
+    float r = wave_reduce_sum(acc[b]);
+    if (lane == 0) {
+      if (F32OUT) ((float*)y)[(long)b * N + n] = r;
+      else ((short*)y)[(long)b * N + n] = f2bf(r);
     }
   }
 }
@@ -230,11 +226,20 @@ void gemv_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
   TORCH_CHECK(lds <= 160 * 1024, "B*H too large for LDS staging");
   hipStream_t s = c10::hip::getCurrentHIPStream();
   const void* dptr = has_delta ? delta.data_ptr() : nullptr;
-  #define LAUNCH_AN(FO, DF, HD) hipLaunchKernelGGL( \
-      (gemv_addnorm_kernel<FO, DF, HD>), grid, block, lds, s, y.data_ptr(), \
+  #define LAUNCH_AN1(FO, DF, HD, BN) hipLaunchKernelGGL( \
+      (gemv_addnorm_kernel<FO, DF, HD, BN>), grid, block, lds, s, y.data_ptr(), \
       (const short*)x.data_ptr(), dptr, (short*)x_out.data_ptr(), \
       (const short*)gamma.data_ptr(), (const short*)w.data_ptr(), \
-      B, H, N, (float)eps)
+      H, N, (float)eps)
+  #define LAUNCH_AN(FO, DF, HD) do { switch (B) { \
+    case 1: LAUNCH_AN1(FO, DF, HD, 1); break; \
+    case 2: LAUNCH_AN1(FO, DF, HD, 2); break; \
+    case 3: LAUNCH_AN1(FO, DF, HD, 3); break; \
+    case 4: LAUNCH_AN1(FO, DF, HD, 4); break; \
+    case 5: LAUNCH_AN1(FO, DF, HD, 5); break; \
+    case 6: LAUNCH_AN1(FO, DF, HD, 6); break; \
+    case 7: LAUNCH_AN1(FO, DF, HD, 7); break; \
+    default: LAUNCH_AN1(FO, DF, HD, 8); break; } } while (0)
   if (!has_delta) {
     if (f32out) LAUNCH_AN(true, false, false); else LAUNCH_AN(false, false, false);
   } else if (delta_f32) {
@@ -243,5 +248,6 @@ void gemv_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
     if (f32out) LAUNCH_AN(true, false, true); else LAUNCH_AN(false, false, true);
   }
   #undef LAUNCH_AN
+  #undef LAUNCH_AN1
   HIP_CHECK_KERNEL();
 }
