@@ -291,10 +291,16 @@ void moe_gemv_down_kernel(float* __restrict__ out,         // [T, H] f32
                           const int* __restrict__ pair_token,
                           const int* __restrict__ pair_expert,
                           int H, int I) {
+  // 16 lanes per output, 4 outputs per wave: the I=768 row is 1.5 full-wave
+  // bursts (the second one 50% idle) and a 6-shuffle reduction per 1.5 KB in
+  // the wave-per-output shape; 16-lane groups keep every lane busy (6 bursts
+  // of 128 elems) and cut the reduction to 4 shuffles per output.
   const int p = blockIdx.x;
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int o = blockIdx.y * 4 + wid;
+  const int grp = lane >> 4;       // output within the wave
+  const int sub = lane & 15;       // 16 lanes per output
+  const int o = (blockIdx.y * 4 + wid) * 4 + grp;
   if (o >= H) return;
   const int t = pair_token[p];
   const int e = pair_expert[p];
@@ -302,14 +308,15 @@ void moe_gemv_down_kernel(float* __restrict__ out,         // [T, H] f32
   const short* wrow = w2 + ((long)e * H + o) * I;
 
   float d = 0.f;
-  for (int base = lane * 8; base < I; base += WAVE * 8) {
+  for (int base = sub * 8; base < I; base += 16 * 8) {
     bf16x8 hv = *reinterpret_cast<const bf16x8*>(hrow + base);
     bf16x8 wv = *reinterpret_cast<const bf16x8*>(wrow + base);
     #pragma unroll
     for (int q_ = 0; q_ < 8; ++q_) d += bf2f(hv[q_]) * bf2f(wv[q_]);
   }
-  d = wave_reduce_sum(d);
-  if (lane == 0)
+  #pragma unroll
+  for (int off = 8; off > 0; off >>= 1) d += __shfl_xor(d, off, WAVE);
+  if (sub == 0)
     atomicAdd(out + (long)t * H + o, d * pair_w[p]);
 }
 
@@ -513,7 +520,8 @@ void moe_gemv_down(torch::Tensor out, torch::Tensor h, torch::Tensor w2,
   const int P = pair_token.size(0);
   const int H = out.size(-1), I = h.size(-1);
   TORCH_CHECK(out.dtype() == torch::kFloat32);
-  dim3 grid(P, (H + 3) / 4), block(256);
+  TORCH_CHECK(I % (16 * 8) == 0, "I must be a multiple of 128");
+  dim3 grid(P, (H + 15) / 16), block(256);   // 16 outputs per WG
   hipStream_t s = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(moe_gemv_down_kernel, grid, block, 0, s,
                      out.data_ptr<float>(), (const short*)h.data_ptr(),
