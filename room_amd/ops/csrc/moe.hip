@@ -251,14 +251,10 @@ void moe_gemv_h_kernel(short* __restrict__ h,             // [P, I]
                      + threadIdx.x;
     if (gid < zero_n) out_zero[gid] = 0.f;
   }
-  // 16-lane output groups (see moe_gemv_down): 4 outputs per wave, 4-shuffle
-  // reductions, 256 B-per-row parallel streaming.
   const int p = blockIdx.x;
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int grp = lane >> 4;
-  const int sub = lane & 15;
-  const int j = (blockIdx.y * 4 + wid) * 4 + grp;
+  const int j = blockIdx.y * 4 + wid;
   if (j >= I) return;
   const int t = pair_token[p];
   const int e = pair_expert[p];
@@ -267,7 +263,7 @@ void moe_gemv_h_kernel(short* __restrict__ h,             // [P, I]
   const short* urow = w13 + ((long)e * 2 * I + I + j) * H;
 
   float dg = 0.f, du = 0.f;
-  for (int base = sub * 8; base < H; base += 16 * 8) {
+  for (int base = lane * 8; base < H; base += WAVE * 8) {
     bf16x8 xv = *reinterpret_cast<const bf16x8*>(xrow + base);
     bf16x8 gv = *reinterpret_cast<const bf16x8*>(grow + base);
     bf16x8 uv = *reinterpret_cast<const bf16x8*>(urow + base);
@@ -278,12 +274,9 @@ void moe_gemv_h_kernel(short* __restrict__ h,             // [P, I]
       du += xf * bf2f(uv[q_]);
     }
   }
-  #pragma unroll
-  for (int off = 8; off > 0; off >>= 1) {
-    dg += __shfl_xor(dg, off, WAVE);
-    du += __shfl_xor(du, off, WAVE);
-  }
-  if (sub == 0) {
+  dg = wave_reduce_sum(dg);
+  du = wave_reduce_sum(du);
+  if (lane == 0) {
     float s = dg / (1.0f + __expf(-dg));
     h[(long)p * I + j] = f2bf(s * du);
   }
@@ -503,8 +496,7 @@ void moe_gemv_h(torch::Tensor h, torch::Tensor x, torch::Tensor w13,
                 torch::Tensor out_zero) {
   const int P = pair_token.size(0);
   const int H = x.size(-1), I = h.size(-1);
-  TORCH_CHECK(H % (16 * 8) == 0, "H must be a multiple of 128");
-  dim3 grid(P, (I + 15) / 16), block(256);   // 16 outputs per WG
+  dim3 grid(P, (I + 3) / 4), block(256);
   float* zp = nullptr;
   long zn = 0;
   if (out_zero.defined() && out_zero.numel() > 0) {
