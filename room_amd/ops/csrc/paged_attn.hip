@@ -264,32 +264,29 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
   const long kv_stride_block = (long)n_kvheads * BLOCK_SIZE * HEAD_DIM;
   const int* btab = block_table + (long)seq * max_blocks;
 
-  // software pipeline: V(i) is staged DURING score(i) (which only needs K),
-  // K(i+1) is staged DURING value(i) (K free once scores are out) — the
-  // global→LDS latency hides behind compute instead of serializing the loop.
-  // Same LDS footprint and the same two barriers per chunk.
-  #define PA_STAGE(dst, src, chunk_base)                                     \
-    _Pragma("unroll")                                                        \
-    for (int it = 0; it < 2; ++it) {                                         \
-      const int idx = tid + it * 256;                                        \
-      const int pos_l = (idx * 8) / HEAD_DIM;                                \
-      const int d8 = (idx * 8) % HEAD_DIM;                                   \
-      const int pos = (chunk_base) + pos_l;                                  \
-      bf16x8 vv = {0, 0, 0, 0, 0, 0, 0, 0};                                  \
-      if (pos < hi) {                                                        \
-        const int blk = btab[pos / BLOCK_SIZE];                              \
-        const long off = (long)blk * kv_stride_block                         \
-            + ((long)hk * BLOCK_SIZE + (pos % BLOCK_SIZE)) * HEAD_DIM + d8;  \
-        vv = *reinterpret_cast<const bf16x8*>((src) + off);                  \
-      }                                                                      \
-      *reinterpret_cast<bf16x8*>(&dst[pos_l][d8]) = vv;                      \
-    }
-
-  PA_STAGE(k_s, kcache, lo)
-  __syncthreads();
-
   for (int base = lo; base < hi; base += CHUNK) {
-    {   // score: thread (h, sub) dots q_s[h] with k_s[sub]; stage V(i)
+    // ---- stage K and V [32][128] chunks: 512 vec8 each, 2 per thread
+    #pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = tid + it * 256;          // vec8 index 0..511
+      const int pos_l = (idx * 8) / HEAD_DIM;  // 0..31
+      const int d8 = (idx * 8) % HEAD_DIM;
+      const int pos = base + pos_l;
+      bf16x8 kv = {0, 0, 0, 0, 0, 0, 0, 0};
+      bf16x8 vv = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (pos < hi) {
+        const int blk = btab[pos / BLOCK_SIZE];
+        const long off = (long)blk * kv_stride_block
+                         + ((long)hk * BLOCK_SIZE + (pos % BLOCK_SIZE)) * HEAD_DIM + d8;
+        kv = *reinterpret_cast<const bf16x8*>(kcache + off);
+        vv = *reinterpret_cast<const bf16x8*>(vcache + off);
+      }
+      *reinterpret_cast<bf16x8*>(&k_s[pos_l][d8]) = kv;
+      *reinterpret_cast<bf16x8*>(&v_s[pos_l][d8]) = vv;
+    }
+    __syncthreads();
+
+    {   // score: thread (h, sub) dots q_s[h] with k_s[sub]
       float s = -INFINITY;
       if (base + sub < hi) {
         float dot = 0.f;
@@ -302,11 +299,9 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
         s = dot * scale;
       }
       p_s[h][sub] = s;
-      PA_STAGE(v_s, vcache, base)
     }
     __syncthreads();
-    {   // softmax + value from LDS; stage K(i+1) (k_s free after scores)
-      PA_STAGE(k_s, kcache, base + CHUNK)
+    {
       float chunk_max = -INFINITY;
       #pragma unroll
       for (int j = 0; j < CHUNK; ++j) chunk_max = fmaxf(chunk_max, p_s[h][j]);
@@ -328,9 +323,8 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
         m_run = m_new;
       }
     }
-    __syncthreads();
+    __syncthreads();  // before restaging K/V
   }
-  #undef PA_STAGE
 
   if (sub == 0) { ml[0] = m_run; ml[1] = l_run; }
   *reinterpret_cast<f32x4*>(acc_out + sub * 4) =
