@@ -2,10 +2,12 @@
 
 The reference pinned `qwen3-coder:30b` served by an external Ollama sidecar
 (src/shared/local-model.ts:3-5); here the model runs in-process: weights are
-bf16 tensors resident in HBM3E, the hot ops are the CDNA4 HIP kernels in
-room_amd/ops (RMSNorm, fused QK-norm+RoPE, paged attention, MoE router/GEMV/
-grouped-MFMA-GEMM, fused sampling), and the plain projections (QKV/O/router/
-lm_head) go through hipBLASLt via torch.nn.functional.linear.
+bf16 tensors resident in HBM3E, and the hot ops are the CDNA4 HIP kernels in
+room_amd/ops (RMSNorm, fused QK-norm+RoPE+KV-scatter, split-KV paged
+attention, MFMA flash prefill, H-split router + top-k, MoE pair GEMV /
+grouped MFMA GEMM, BN-specialized dense GEMV, two-stage sampling). At decode
+every projection (QKV/O/router/lm_head) runs on the hand-written GEMVs;
+prefill's large GEMMs go through hipBLASLt via torch.nn.functional.linear.
 
 Inference-only (no autograd); a training path is out of scope for the
 reference's semantics (it never trains).
