@@ -41,6 +41,31 @@ def nudge_worker(worker_id: int) -> bool:
         return False  # fail-silent like the reference nudge
 
 
+def _server_port() -> int | None:
+    from ..server.auth import data_dir
+    try:
+        return int((data_dir() / "api.port").read_text().strip())
+    except Exception:
+        return None
+
+
+def _server_post(path: str) -> bool:
+    """Authenticated POST to the local HTTP server; False if unreachable."""
+    import urllib.request
+
+    from ..server.auth import data_dir
+    try:
+        port = _server_port()
+        token = (data_dir() / "api.token").read_text().strip()
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{port}{path}", method="POST",
+            headers={"Authorization": f"Bearer {token}"}, data=b"{}")
+        urllib.request.urlopen(req, timeout=3)
+        return True
+    except Exception:
+        return False
+
+
 class McpServer:
     def __init__(self, ldb: LockedDb, memory=None, nudge=nudge_worker):
         self.ldb = ldb
@@ -649,6 +674,299 @@ class McpServer:
         def _(args):
             with ldb as db:
                 return q.list_self_mod_audit(db, args["room_id"])
+
+        # ---- parity surface (reference mcp/tools: quorum/skills/room/memory/
+        # credentials/identity/invite/scheduler/wallet/resources details)
+
+        @t("room_decision_detail", "A decision with its votes.",
+           {"p": {"decision_id": I}, "r": ["decision_id"]})
+        def _(args):
+            with ldb as db:
+                d = q.get_decision(db, args["decision_id"])
+                if d is None:
+                    return {"error": f"decision {args['decision_id']} not found"}
+                votes = q.get_votes(db, args["decision_id"])
+            return {**d, "votes": [{"worker_id": v["worker_id"], "vote": v["vote"],
+                                    "reasoning": v["reasoning"]} for v in votes]}
+
+        @t("room_activate_skill", "Enable auto-activation for a skill.",
+           {"p": {"skill_id": I}, "r": ["skill_id"]})
+        def _(args):
+            with ldb as db:
+                s = q.update_skill(db, args["skill_id"], auto_activate=True)
+            return {"activated": s is not None}
+
+        @t("room_deactivate_skill", "Disable auto-activation for a skill.",
+           {"p": {"skill_id": I}, "r": ["skill_id"]})
+        def _(args):
+            with ldb as db:
+                s = q.update_skill(db, args["skill_id"], auto_activate=False)
+            return {"deactivated": s is not None}
+
+        @t("room_delete_skill", "Delete a skill.",
+           {"p": {"skill_id": I}, "r": ["skill_id"]})
+        def _(args):
+            with ldb as db:
+                q.delete_skill(db, args["skill_id"])
+            return {"deleted": True}
+
+        @t("room_configure_room", "Update a room's quorum/pacing config "
+           "(threshold, timeoutMinutes, tieBreaker, autoApprove, minCycleGapMs...).",
+           {"p": {"room_id": I, "config": {"type": "object"}},
+            "r": ["room_id", "config"]})
+        def _(args):
+            with ldb as db:
+                room = q.get_room(db, args["room_id"])
+                if room is None:
+                    return {"error": "room not found"}
+                raw = room.get("config") or {}
+                cfg = json.loads(raw) if isinstance(raw, str) else dict(raw)
+                cfg.update(args["config"])
+                q.update_room(db, args["room_id"], config=json.dumps(cfg))
+            return {"config": cfg}
+
+        @t("room_delete_room", "Permanently delete a room and its data.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                room_mod.delete_room(db, args["room_id"])
+            return {"deleted": True}
+
+        @t("room_get_credential", "Retrieve a credential's decrypted value "
+           "(room_list_credentials shows masked names only).",
+           {"p": {"room_id": I, "name": S}, "r": ["room_id", "name"]})
+        def _(args):
+            from ..core.secret_store import decrypt_secret
+            with ldb as db:
+                row = q.get_credential(db, args["room_id"], args["name"])
+            if row is None:
+                return {"error": f"credential '{args['name']}' not found"}
+            return {"name": row["name"], "type": row["type"],
+                    "value": decrypt_secret(row["value_encrypted"])}
+
+        @t("room_forget", "Delete a memory entity (and its observations/"
+           "relations/embedding).", {"p": {"entity_id": I}, "r": ["entity_id"]})
+        def _(args):
+            with ldb as db:
+                ent = q.get_entity(db, args["entity_id"])
+                if ent is None:
+                    return {"error": f"memory {args['entity_id']} not found"}
+                q.delete_entity(db, args["entity_id"])
+            if self.memory is not None:
+                try:
+                    self.memory.store.remove(args["entity_id"])
+                except Exception:
+                    pass  # GPU index rebuilds from SQLite
+            return {"forgot": ent["name"]}
+
+        @t("room_memory_list", "List a room's memory entities (newest first).",
+           {"p": {"room_id": I, "limit": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                rows = db.execute(
+                    "SELECT id, name, type, category, created_at FROM entities"
+                    " WHERE room_id = ? ORDER BY id DESC LIMIT ?",
+                    (args["room_id"], args.get("limit", 50))).fetchall()
+            return rows
+
+        @t("room_identity_get", "The room's on-chain identity (ERC-8004).",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                w = q.get_room_wallet(db, args["room_id"])
+            if w is None:
+                return {"error": "room has no wallet"}
+            return {"address": w["address"],
+                    "agent_id": w.get("identity_agent_id"),
+                    "registered": bool(w.get("identity_agent_id"))}
+
+        @t("room_identity_update", "Rebuild and re-register the identity "
+           "metadata URI from current room state.",
+           {"p": {"room_id": I, "chain": S}, "r": ["room_id"]})
+        def _(args):
+            from ..core import identity as identity_mod
+            with ldb as db:
+                return identity_mod.register_identity(
+                    db, args["room_id"], chain=args.get("chain", "base"))
+
+        @t("room_invite_create", "Create a cloud invite link for this room's "
+           "network.", {"p": {"room_id": I, "max_uses": I, "expires_in_days": I},
+                        "r": ["room_id"]})
+        def _(args):
+            from ..core import cloud_sync
+            if cloud_sync.cloud_api() is None:
+                return {"error": "Cloud is not configured (ROOMAMD_CLOUD_API); "
+                                 "invites need the cloud relay."}
+            token = cloud_sync.register_with_cloud(ldb, args["room_id"])
+            out = cloud_sync._post(
+                f"/rooms/{args['room_id']}/invites",
+                {"maxUses": args.get("max_uses"),
+                 "expiresInDays": args.get("expires_in_days")}, token)
+            return out or {"error": "Failed to create invite (cloud unavailable)."}
+
+        @t("room_invite_list", "List this room's invite links.",
+           {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            from ..core import cloud_sync
+            if cloud_sync.cloud_api() is None:
+                return {"error": "Cloud is not configured (ROOMAMD_CLOUD_API)."}
+            token = cloud_sync.load_room_tokens().get(str(args["room_id"]))
+            out = cloud_sync._post(f"/rooms/{args['room_id']}/invites/list",
+                                   {}, token)
+            return out or {"error": "Cloud unavailable."}
+
+        @t("room_invite_network", "Rooms that joined through this room's "
+           "invites.", {"p": {"room_id": I}, "r": ["room_id"]})
+        def _(args):
+            from ..core import cloud_sync
+            if cloud_sync.cloud_api() is None:
+                return {"error": "Cloud is not configured (ROOMAMD_CLOUD_API)."}
+            token = cloud_sync.load_room_tokens().get(str(args["room_id"]))
+            out = cloud_sync._post(f"/rooms/{args['room_id']}/network", {}, token)
+            return out or {"error": "Cloud unavailable."}
+
+        @t("room_pause_watch", "Pause a filesystem watch.",
+           {"p": {"watch_id": I}, "r": ["watch_id"]})
+        def _(args):
+            with ldb as db:
+                db.execute("UPDATE watches SET status = 'paused' WHERE id = ?",
+                           (args["watch_id"],))
+            return {"paused": True}
+
+        @t("room_resume_watch", "Resume a paused filesystem watch.",
+           {"p": {"watch_id": I}, "r": ["watch_id"]})
+        def _(args):
+            with ldb as db:
+                db.execute("UPDATE watches SET status = 'active' WHERE id = ?",
+                           (args["watch_id"],))
+            return {"resumed": True}
+
+        @t("room_reset_session", "Clear a task's session so the next run "
+           "starts a fresh conversation.", {"p": {"task_id": I}, "r": ["task_id"]})
+        def _(args):
+            with ldb as db:
+                task = q.get_task(db, args["task_id"])
+                if task is None:
+                    return {"error": f"task {args['task_id']} not found"}
+                q.update_task(db, args["task_id"], session_id=None)
+            return {"reset": task["name"]}
+
+        @t("room_run_task", "Execute a task immediately (returns right away; "
+           "use room_task_runs for status).", {"p": {"task_id": I}, "r": ["task_id"]})
+        def _(args):
+            with ldb as db:
+                task = q.get_task(db, args["task_id"])
+                if task is None:
+                    return {"error": f"task {args['task_id']} not found"}
+            # cross-process: ask the HTTP server to run it now; fall back to
+            # marking it due so the 15 s runtime loop picks it up
+            if _server_post(f"/api/tasks/{args['task_id']}/run"):
+                return {"started": task["name"]}
+            with ldb as db:
+                q.update_task(db, args["task_id"], status="active",
+                              scheduled_at=q.now_iso())
+            return {"queued": task["name"],
+                    "note": "server offline; task due on next scheduler pass"}
+
+        @t("room_wallet_balance", "On-chain token balance of the room wallet "
+           "(USDC/USDT across Base, Ethereum, Arbitrum, Optimism, Polygon).",
+           {"p": {"room_id": I, "network": S, "token": S}, "r": ["room_id"]})
+        def _(args):
+            with ldb as db:
+                return wallet_mod.get_on_chain_balance(
+                    db, args["room_id"], chain=args.get("network", "base"),
+                    token=args.get("token", "usdc"))
+
+        @t("room_wallet_topup", "A top-up URL the keeper can use to fund the "
+           "room wallet by card (USDC on Base).",
+           {"p": {"room_id": I, "amount": N}, "r": ["room_id"]})
+        def _(args):
+            from ..core import cloud_sync
+            with ldb as db:
+                w = q.get_room_wallet(db, args["room_id"])
+            if w is None:
+                return {"error": "room has no wallet"}
+            if cloud_sync.cloud_api() is not None:
+                token = cloud_sync.load_room_tokens().get(str(args["room_id"]))
+                out = cloud_sync._post(f"/rooms/{args['room_id']}/onramp",
+                                       {"address": w["address"],
+                                        "amount": args.get("amount")}, token)
+                if out and out.get("onrampUrl"):
+                    return {"onramp_url": out["onrampUrl"]}
+            return {"address": w["address"],
+                    "note": "Cloud on-ramp unavailable; send USDC on Base "
+                            "directly to this address."}
+
+        @t("room_webhook_url", "Webhook URL for a task (external trigger) or "
+           "room (message + queen wake).",
+           {"p": {"task_id": I, "room_id": I, "generate_if_missing": B}})
+        def _(args):
+            import secrets as _secrets
+            port = _server_port()
+            base = f"http://127.0.0.1:{port}" if port else "http://<server>"
+            with ldb as db:
+                if args.get("task_id"):
+                    task = q.get_task(db, args["task_id"])
+                    if task is None:
+                        return {"error": f"task {args['task_id']} not found"}
+                    tok = task.get("webhook_token")
+                    if not tok and args.get("generate_if_missing"):
+                        tok = _secrets.token_hex(16)
+                        q.update_task(db, args["task_id"], webhook_token=tok)
+                    if not tok:
+                        return {"error": "task has no webhook token "
+                                         "(pass generate_if_missing)"}
+                    return {"url": f"{base}/api/hooks/task/{tok}"}
+                if args.get("room_id"):
+                    room = q.get_room(db, args["room_id"])
+                    if room is None:
+                        return {"error": "room not found"}
+                    tok = room.get("webhook_token")
+                    if not tok and args.get("generate_if_missing"):
+                        tok = _secrets.token_hex(16)
+                        q.update_room(db, args["room_id"], webhook_token=tok)
+                    if not tok:
+                        return {"error": "room has no webhook token "
+                                         "(pass generate_if_missing)"}
+                    return {"url": f"{base}/api/hooks/queen/{tok}"}
+            return {"error": "pass task_id or room_id"}
+
+        @t("room_resources", "Local machine + GPU resource usage (decide "
+           "whether the swarm needs more capacity).", {})
+        def _(args):
+            import os as _os
+            load1, load5, _ = _os.getloadavg()
+            cpus = _os.cpu_count() or 1
+            mem = {}
+            try:
+                with open("/proc/meminfo") as f:
+                    for line in f:
+                        k, v = line.split(":", 1)
+                        if k in ("MemTotal", "MemAvailable"):
+                            mem[k] = int(v.strip().split()[0]) * 1024
+            except OSError:
+                pass
+            used_pct = (round((1 - mem["MemAvailable"] / mem["MemTotal"]) * 100)
+                        if mem else None)
+            out = {"cpu_load_1m": load1, "cpu_load_5m": load5, "cpus": cpus,
+                   "cpu_pct_of_capacity": round(load1 / cpus * 100),
+                   "ram_used_pct": used_pct}
+            try:
+                import torch
+                if torch.cuda.is_available():
+                    free_b, total_b = torch.cuda.mem_get_info()
+                    out["gpu_hbm_used_pct"] = round((1 - free_b / total_b) * 100)
+                    out["gpu_count"] = torch.cuda.device_count()
+            except Exception:
+                pass
+            with ldb as db:
+                tasks = q.list_tasks(db)
+                out["active_tasks"] = sum(1 for tk in tasks
+                                          if tk["status"] == "active")
+            high = (load1 / cpus > 0.8) or (used_pct or 0) > 85
+            out["summary"] = ("HIGH LOAD — consider scaling swarm capacity"
+                              if high else "Normal load")
+            return out
 
     # ------------------------------------------------------------ JSON-RPC
 

@@ -110,3 +110,92 @@ def test_unknown_tool_and_method(mcp):
     assert resp["error"]["code"] == -32602
     resp = mcp.handle({"jsonrpc": "2.0", "id": 6, "method": "bogus"})
     assert resp["error"]["code"] == -32601
+
+
+def test_parity_surface_tools(mcp):
+    """The reference's full 76-tool surface: detail/config/forget/session/
+    webhook/resources tools round-trip (mcp/tools/* parity)."""
+    out = call(mcp, "room_create_room", {"name": "parity", "goal": "g",
+                                         "worker_model": "stub"})
+    rid = out["room_id"]
+
+    # decision detail with votes
+    ann = call(mcp, "room_announce", {"room_id": rid, "proposal": "d1",
+                                      "decision_type": "custom"})
+    did = ann["decision_id"]
+    w0 = call(mcp, "room_create_worker", {"room_id": rid, "name": "voter",
+                                          "role": "executor"})
+    call(mcp, "room_object", {"decision_id": did,
+                              "worker_id": w0["worker_id"], "reason": "nope"})
+    detail = call(mcp, "room_decision_detail", {"decision_id": did})
+    assert detail["id"] == did and detail["status"] == "objected"
+    assert "votes" in detail
+
+    # skills: activate / deactivate / delete
+    sk = call(mcp, "room_create_skill", {"room_id": rid, "name": "s1",
+                                         "content": "recipe"})
+    sid = sk["skill_id"]
+    assert call(mcp, "room_activate_skill", {"skill_id": sid})["activated"]
+    assert call(mcp, "room_deactivate_skill", {"skill_id": sid})["deactivated"]
+    assert call(mcp, "room_delete_skill", {"skill_id": sid})["deleted"]
+
+    # configure room merges config
+    cfg = call(mcp, "room_configure_room",
+               {"room_id": rid, "config": {"threshold": "unanimous"}})
+    assert cfg["config"]["threshold"] == "unanimous"
+
+    # credentials get returns the decrypted value
+    call(mcp, "room_set_credential", {"room_id": rid, "name": "api_key",
+                                      "value": "sk-123"})
+    got = call(mcp, "room_get_credential", {"room_id": rid, "name": "api_key"})
+    assert got["value"] == "sk-123"
+
+    # memory list + forget
+    m = call(mcp, "room_remember", {"room_id": rid, "name": "fact",
+                                    "content": "water is wet"})
+    lst = call(mcp, "room_memory_list", {"room_id": rid})
+    assert any(e["name"] == "fact" for e in lst)
+    fg = call(mcp, "room_forget", {"entity_id": m["entity_id"]})
+    assert fg["forgot"] == "fact"
+    assert not any(e["name"] == "fact"
+                   for e in call(mcp, "room_memory_list", {"room_id": rid}))
+
+    # identity get (wallet auto-created with the room)
+    ident = call(mcp, "room_identity_get", {"room_id": rid})
+    assert ident["address"].startswith("0x")
+
+    # invites fail gracefully offline
+    inv = call(mcp, "room_invite_create", {"room_id": rid})
+    assert "error" in inv
+
+    # watch pause/resume
+    w = call(mcp, "room_watch_path", {"room_id": rid, "path": "/tmp/x",
+                                      "action_prompt": "check"})
+    call(mcp, "room_pause_watch", {"watch_id": w["watch_id"]})
+    call(mcp, "room_resume_watch", {"watch_id": w["watch_id"]})
+
+    # task session reset + webhook url + run-now fallback
+    task = call(mcp, "room_create_task",
+                {"name": "t1", "prompt": "p", "room_id": rid,
+                 "trigger_type": "manual"})
+    tid = task["task_id"]
+    assert "reset" in call(mcp, "room_reset_session", {"task_id": tid})
+    url = call(mcp, "room_webhook_url", {"task_id": tid,
+                                         "generate_if_missing": True})
+    assert "/api/hooks/task/" in url["url"]
+    run = call(mcp, "room_run_task", {"task_id": tid})
+    assert "started" in run or "queued" in run
+
+    # wallet balance (offline → structured error/zero) + topup fallback
+    bal = call(mcp, "room_wallet_balance", {"room_id": rid})
+    assert isinstance(bal, dict)
+    top = call(mcp, "room_wallet_topup", {"room_id": rid, "amount": 25})
+    assert "onramp_url" in top or top["address"].startswith("0x")
+
+    # resources
+    res = call(mcp, "room_resources", {})
+    assert res["cpus"] >= 1 and "summary" in res
+
+    # the full surface is now >= the reference's 76 tools
+    lst = mcp.handle({"jsonrpc": "2.0", "id": 9, "method": "tools/list"})
+    assert len(lst["result"]["tools"]) >= 90
