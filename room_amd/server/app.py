@@ -799,6 +799,375 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
         with ldb as db:
             return get_public_feed(db, limit=limit)
 
+    # ------------------------------------------- route-shape parity surface
+    # The reference exposes several flat/detail routes alongside the nested
+    # ones (routes/*.ts); these aliases + detail endpoints keep
+    # a reference API client working unchanged.
+
+    @app.get("/api/decisions/{decision_id}")
+    async def decision_detail(decision_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            d = q.get_decision(db, decision_id)
+            if d is None:
+                raise HTTPException(404, "decision not found")
+            return {**d, "votes": q.get_votes(db, decision_id)}
+
+    @app.post("/api/decisions/{decision_id}/resolve")
+    async def decision_resolve(decision_id: int, body: dict = Body(default={}),
+                               role: str = Depends(rw)):
+        with ldb as db:
+            q.resolve_decision(db, decision_id, body.get("result", "approved"),
+                               body.get("reason"))
+            return q.get_decision(db, decision_id)
+
+    @app.get("/api/goals/{goal_id}")
+    async def goal_detail(goal_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            g = q.get_goal(db, goal_id)
+            if g is None:
+                raise HTTPException(404, "goal not found")
+            return g
+
+    @app.get("/api/goals/{goal_id}/subgoals")
+    async def goal_subgoals(goal_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            return db.execute("SELECT * FROM goals WHERE parent_goal_id = ?"
+                              " ORDER BY id", (goal_id,)).fetchall()
+
+    @app.get("/api/goals/{goal_id}/updates")
+    async def goal_updates(goal_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            return db.execute("SELECT * FROM goal_updates WHERE goal_id = ?"
+                              " ORDER BY id DESC", (goal_id,)).fetchall()
+
+    @app.post("/api/goals/{goal_id}/updates")
+    async def goal_add_update(goal_id: int, body: dict = Body(...),
+                              role: str = Depends(rw)):
+        with ldb as db:
+            q.add_goal_update(db, goal_id, body.get("observation", ""),
+                              metric_value=body.get("progress"))
+            return q.get_goal(db, goal_id)
+
+    @app.delete("/api/goals/{goal_id}")
+    async def goal_delete(goal_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            db.execute("DELETE FROM goals WHERE id = ?", (goal_id,))
+        return {"deleted": True}
+
+    @app.get("/api/memory/entities")
+    async def memory_entities(room_id: int | None = None, limit: int = 100,
+                              role: str = Depends(get_role)):
+        with ldb as db:
+            if room_id is not None:
+                return db.execute(
+                    "SELECT * FROM entities WHERE room_id = ?"
+                    " ORDER BY id DESC LIMIT ?", (room_id, limit)).fetchall()
+            return db.execute("SELECT * FROM entities ORDER BY id DESC"
+                              " LIMIT ?", (limit,)).fetchall()
+
+    @app.patch("/api/memory/entities/{entity_id}")
+    async def memory_update(entity_id: int, body: dict = Body(...),
+                            role: str = Depends(rw)):
+        with ldb as db:
+            if "name" in body:
+                db.execute("UPDATE entities SET name = ? WHERE id = ?",
+                           (body["name"], entity_id))
+            if "content" in body:
+                q.add_observation(db, entity_id, body["content"])
+            ent = q.get_entity(db, entity_id)
+            if ent is None:
+                raise HTTPException(404, "entity not found")
+            return ent
+
+    @app.post("/api/memory/relations")
+    async def memory_relate(body: dict = Body(...),
+                            role: str = Depends(rw)):
+        with ldb as db:
+            return q.create_relation(db, body["from_entity"],
+                                     body["to_entity"],
+                                     body.get("relation_type", "related_to"))
+
+    @app.delete("/api/memory/relations/{relation_id}")
+    async def memory_unrelate(relation_id: int,
+                              role: str = Depends(rw)):
+        with ldb as db:
+            db.execute("DELETE FROM relations WHERE id = ?", (relation_id,))
+        return {"deleted": True}
+
+    @app.delete("/api/memory/observations/{obs_id}")
+    async def memory_del_obs(obs_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            db.execute("DELETE FROM observations WHERE id = ?", (obs_id,))
+        return {"deleted": True}
+
+    @app.get("/api/memory/stats")
+    async def memory_stats(role: str = Depends(get_role)):
+        with ldb as db:
+            ents = db.execute("SELECT COUNT(*) AS n FROM entities").fetchone()["n"]
+            obs = db.execute("SELECT COUNT(*) AS n FROM observations").fetchone()["n"]
+            rels = db.execute("SELECT COUNT(*) AS n FROM relations").fetchone()["n"]
+            emb = db.execute("SELECT COUNT(*) AS n FROM embeddings").fetchone()["n"]
+        gpu_rows = (memory.store.size if memory is not None
+                    and hasattr(memory, "store") else None)
+        return {"entities": ents, "observations": obs, "relations": rels,
+                "embeddings": emb, "gpu_index_rows": gpu_rows}
+
+    @app.get("/api/messages/{message_id}")
+    async def message_detail(message_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            m = db.execute("SELECT * FROM room_messages WHERE id = ?",
+                           (message_id,)).fetchone()
+            if m is None:
+                raise HTTPException(404, "message not found")
+            db.execute("UPDATE room_messages SET status = 'read' WHERE id = ?",
+                       (message_id,))
+            return m
+
+    @app.post("/api/messages/{message_id}/reply")
+    async def message_reply(message_id: int, body: dict = Body(...),
+                            role: str = Depends(rw)):
+        with ldb as db:
+            m = db.execute("SELECT * FROM room_messages WHERE id = ?",
+                           (message_id,)).fetchone()
+            if m is None:
+                raise HTTPException(404, "message not found")
+            out = q.create_room_message(
+                db, m["room_id"], "outbound",
+                to_room_id=m["from_room_id"],
+                subject=f"Re: {m['subject']}", body=body.get("body", ""))
+            return out
+
+    @app.delete("/api/messages/{message_id}")
+    async def message_delete(message_id: int,
+                             role: str = Depends(rw)):
+        with ldb as db:
+            db.execute("DELETE FROM room_messages WHERE id = ?", (message_id,))
+        return {"deleted": True}
+
+    @app.get("/api/rooms/queen-states")
+    async def queen_states(role: str = Depends(get_role)):
+        with ldb as db:
+            rows = db.execute(
+                "SELECT r.id AS room_id, r.name, w.id AS queen_id,"
+                " w.agent_state FROM rooms r JOIN workers w"
+                " ON w.id = r.queen_worker_id").fetchall()
+        return rows
+
+    @app.get("/api/rooms/{room_id}/queen")
+    async def room_queen(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            room = q.get_room(db, room_id)
+            if room is None:
+                raise HTTPException(404, "room not found")
+            return q.get_worker(db, room["queen_worker_id"])
+
+    @app.post("/api/rooms/{room_id}/queen/start")
+    async def queen_start(room_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            room = q.get_room(db, room_id)
+            if room is None:
+                raise HTTPException(404, "room not found")
+            qid = room["queen_worker_id"]
+        if loop_mgr is not None:
+            loop_mgr.resume_agent(qid)
+        return {"started": qid}
+
+    @app.post("/api/rooms/{room_id}/queen/stop")
+    async def queen_stop(room_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            room = q.get_room(db, room_id)
+            if room is None:
+                raise HTTPException(404, "room not found")
+            qid = room["queen_worker_id"]
+        if loop_mgr is not None:
+            loop_mgr.pause_agent(qid)
+        return {"stopped": qid}
+
+    @app.get("/api/rooms/{room_id}/usage")
+    async def room_usage(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            row = db.execute(
+                "SELECT COUNT(*) AS cycles,"
+                " COALESCE(SUM(input_tokens),0) AS input_tokens,"
+                " COALESCE(SUM(output_tokens),0) AS output_tokens"
+                " FROM worker_cycles WHERE room_id = ?", (room_id,)).fetchone()
+        return row
+
+    @app.get("/api/runs")
+    async def runs_all(limit: int = 50, role: str = Depends(get_role)):
+        with ldb as db:
+            return db.execute("SELECT * FROM task_runs ORDER BY id DESC"
+                              " LIMIT ?", (limit,)).fetchall()
+
+    # flat aliases for nested collections (reference routes/workers.ts,
+    # skills.ts, self-mod.ts, prompt-sync, tasks.ts shapes)
+    @app.get("/api/workers")
+    async def workers_all(room_id: int | None = None,
+                          role: str = Depends(get_role)):
+        with ldb as db:
+            if room_id is not None:
+                return q.list_room_workers(db, room_id)
+            return db.execute("SELECT * FROM workers ORDER BY id").fetchall()
+
+    @app.post("/api/workers")
+    async def workers_create(body: dict = Body(...),
+                             role: str = Depends(rw)):
+        return await create_worker(body["room_id"], body, role)  # type: ignore
+
+    @app.get("/api/skills")
+    async def skills_all(room_id: int | None = None,
+                         role: str = Depends(get_role)):
+        with ldb as db:
+            if room_id is not None:
+                return q.list_room_skills(db, room_id)
+            return db.execute("SELECT * FROM skills ORDER BY id").fetchall()
+
+    @app.get("/api/skills/{skill_id}")
+    async def skill_detail(skill_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            s = q.get_skill(db, skill_id)
+            if s is None:
+                raise HTTPException(404, "skill not found")
+            return s
+
+    @app.post("/api/skills")
+    async def skills_create(body: dict = Body(...),
+                            role: str = Depends(rw)):
+        from ..core import skills as skills_mod
+        with ldb as db:
+            return skills_mod.create_agent_skill(
+                db, body.get("room_id"), body["name"], body["content"],
+                activation_context=body.get("activation_context"),
+                auto_activate=body.get("auto_activate", False))
+
+    @app.get("/api/self-mod/audit")
+    async def self_mod_audit_all(room_id: int | None = None,
+                                 role: str = Depends(get_role)):
+        with ldb as db:
+            if room_id is not None:
+                return q.list_self_mod_audit(db, room_id)
+            return db.execute("SELECT * FROM self_mod_audit ORDER BY id DESC"
+                              " LIMIT 200").fetchall()
+
+    @app.post("/api/self-mod/audit/{audit_id}/revert")
+    async def self_mod_revert_alias(audit_id: int,
+                                    role: str = Depends(rw)):
+        from ..core import self_mod as self_mod_mod
+        with ldb as db:
+            return self_mod_mod.revert_modification(db, audit_id)
+
+    @app.post("/api/tasks/{task_id}/pause")
+    async def task_pause(task_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            t_ = q.update_task(db, task_id, status="paused")
+            if t_ is None:
+                raise HTTPException(404, "task not found")
+            return t_
+
+    @app.post("/api/tasks/{task_id}/resume")
+    async def task_resume(task_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            t_ = q.update_task(db, task_id, status="active")
+            if t_ is None:
+                raise HTTPException(404, "task not found")
+            return t_
+
+    @app.post("/api/tasks/{task_id}/reset-session")
+    async def task_reset_session(task_id: int,
+                                 role: str = Depends(rw)):
+        with ldb as db:
+            t_ = q.update_task(db, task_id, session_id=None)
+            if t_ is None:
+                raise HTTPException(404, "task not found")
+            return {"reset": t_["name"]}
+
+    @app.post("/api/workers/prompts/export")
+    async def prompts_export_all(role: str = Depends(rw)):
+        from ..core.prompt_sync import export_worker_prompts
+        with ldb as db:
+            rooms = q.list_rooms(db)
+            return {"exported": sum(
+                len(export_worker_prompts(db, r["id"])) for r in rooms)}
+
+    @app.post("/api/workers/prompts/import")
+    async def prompts_import_all(body: dict = Body(default={}),
+                                 role: str = Depends(rw)):
+        from ..core.prompt_sync import import_worker_prompts
+        with ldb as db:
+            rooms = q.list_rooms(db)
+            return {"imported": sum(
+                len(import_worker_prompts(db, r["id"],
+                                          force=body.get("force", False)))
+                for r in rooms)}
+
+    # clerk status/usage + contact/provider/local-model status (the reference
+    # manages external providers and an Ollama sidecar here; the engine is
+    # in-process so these report the native equivalent)
+    @app.get("/api/clerk/status")
+    async def clerk_status(role: str = Depends(get_role)):
+        with ldb as db:
+            n = db.execute("SELECT COUNT(*) AS n FROM chat_messages"
+                           " WHERE room_id IS NULL").fetchone()["n"]
+        return {"available": True, "model": "qwen3-coder-30b (in-process)",
+                "messages": n}
+
+    @app.get("/api/clerk/usage")
+    async def clerk_usage(role: str = Depends(get_role)):
+        with ldb as db:
+            return db.execute(
+                "SELECT source, model, SUM(input_tokens) AS input_tokens,"
+                " SUM(output_tokens) AS output_tokens, COUNT(*) AS calls"
+                " FROM clerk_usage GROUP BY source, model").fetchall()
+
+    @app.post("/api/clerk/reset")
+    async def clerk_reset(role: str = Depends(rw)):
+        with ldb as db:
+            db.execute("DELETE FROM chat_messages WHERE room_id IS NULL")
+        return {"reset": True}
+
+    @app.post("/api/clerk/typing")
+    async def clerk_typing(role: str = Depends(get_role)):
+        bus.emit("clerk", "typing", {})
+        return {"ok": True}
+
+    @app.post("/api/clerk/presence")
+    async def clerk_presence(role: str = Depends(get_role)):
+        bus.emit("clerk", "presence", {})
+        return {"ok": True}
+
+    @app.get("/api/contacts/status")
+    async def contacts_status(role: str = Depends(get_role)):
+        import os as _os
+        with ldb as db:
+            email = q.get_setting(db, "keeper_email")
+            tg = q.get_setting(db, "keeper_telegram_chat_id")
+        return {"email": {"configured": bool(email or
+                                             _os.environ.get("ROOMAMD_KEEPER_EMAIL")),
+                          "address": email},
+                "telegram": {"configured": bool(tg)}}
+
+    @app.get("/api/providers/status")
+    async def providers_status(role: str = Depends(get_role)):
+        import torch as _torch
+        return {"providers": [{
+            "id": "local", "name": "room_amd in-process engine",
+            "model": "qwen3-coder-30b", "ready": True,
+            "gpu": _torch.cuda.is_available()}]}
+
+    @app.get("/api/local-model/status")
+    async def local_model_status(role: str = Depends(get_role)):
+        import torch as _torch
+        ok = _torch.cuda.is_available()
+        return {"installed": True, "running": ok,
+                "model": "qwen3-coder-30b",
+                "backend": "in-process CDNA4 engine (no sidecar)",
+                "detail": None if ok else "no GPU visible in this process"}
+
+    @app.get("/api/settings/referral")
+    async def settings_referral(role: str = Depends(get_role)):
+        with ldb as db:
+            return {"code": q.get_setting(db, "keeper_referral_code")}
+
     # -------------------------------------------------------------- UI stub
 
     @app.get("/", response_class=HTMLResponse)
@@ -854,5 +1223,14 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
         finally:
             send_task.cancel()
             unsub()
+
+    # Starlette matches routes in registration order; float the late-added
+    # literal paths above their parameterized siblings (/api/rooms/{room_id}
+    # would otherwise shadow /api/rooms/queen-states with a 422).
+    literals = [r for r in app.router.routes
+                if getattr(r, "path", "") in ("/api/rooms/queen-states",)]
+    for r in literals:
+        app.router.routes.remove(r)
+        app.router.routes.insert(0, r)
 
     return app

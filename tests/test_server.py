@@ -260,3 +260,112 @@ def test_dashboard_served(server):
     r = client.get("/")
     assert r.status_code == 200
     assert "room_amd" in r.text and "connectWs" in r.text
+
+
+def test_route_shape_parity_surface(server):
+    """The reference's flat/detail route shapes round-trip."""
+    client, h, *_ = server
+    room = client.post("/api/rooms", json={"name": "parity", "goal": "g",
+                                           "worker_model": "stub"},
+                       headers=h).json()
+    rid = room["id"]
+
+    # queen-states literal route must not be shadowed by /api/rooms/{id}
+    qs = client.get("/api/rooms/queen-states", headers=h)
+    assert qs.status_code == 200 and qs.json()[0]["room_id"] == rid
+    queen = client.get(f"/api/rooms/{rid}/queen", headers=h).json()
+    assert queen["id"] == room["queen_worker_id"]
+    assert client.post(f"/api/rooms/{rid}/queen/stop", headers=h).status_code == 200
+    assert client.post(f"/api/rooms/{rid}/queen/start", headers=h).status_code == 200
+
+    # decision detail + resolve
+    d = client.post(f"/api/rooms/{rid}/decisions",
+                    json={"proposal": "x", "decision_type": "custom"},
+                    headers=h).json()
+    det = client.get(f"/api/decisions/{d['id']}", headers=h).json()
+    assert det["proposal"] == "x" and "votes" in det
+    res = client.post(f"/api/decisions/{d['id']}/resolve",
+                      json={"result": "approved"}, headers=h).json()
+    assert res["status"] == "approved"
+
+    # goal detail/subgoals/updates/delete
+    g = client.post(f"/api/rooms/{rid}/goals", json={"description": "root"},
+                    headers=h).json()
+    sub = client.post(f"/api/rooms/{rid}/goals",
+                      json={"description": "child",
+                            "parent_goal_id": g["id"]}, headers=h).json()
+    assert client.get(f"/api/goals/{g['id']}", headers=h).json()["id"] == g["id"]
+    subs = client.get(f"/api/goals/{g['id']}/subgoals", headers=h).json()
+    assert [s["id"] for s in subs] == [sub["id"]]
+    client.post(f"/api/goals/{g['id']}/updates",
+                json={"observation": "progress note"}, headers=h)
+    ups = client.get(f"/api/goals/{g['id']}/updates", headers=h).json()
+    assert ups and ups[0]["observation"] == "progress note"
+    assert client.delete(f"/api/goals/{sub['id']}",
+                         headers=h).json()["deleted"]
+
+    # memory list / patch / relations / stats
+    e1 = client.post("/api/memory/entities",
+                     json={"room_id": rid, "name": "a", "content": "A"},
+                     headers=h).json()
+    e2 = client.post("/api/memory/entities",
+                     json={"room_id": rid, "name": "b", "content": "B"},
+                     headers=h).json()
+    lst = client.get("/api/memory/entities", params={"room_id": rid},
+                     headers=h).json()
+    assert {x["name"] for x in lst} >= {"a", "b"}
+    client.patch(f"/api/memory/entities/{e1['entity_id']}",
+                 json={"content": "more"}, headers=h)
+    rel = client.post("/api/memory/relations",
+                      json={"from_entity": e1["entity_id"],
+                            "to_entity": e2["entity_id"],
+                            "relation_type": "supports"}, headers=h).json()
+    stats = client.get("/api/memory/stats", headers=h).json()
+    assert stats["entities"] >= 2 and stats["relations"] >= 1
+    client.delete(f"/api/memory/relations/{rel if isinstance(rel, int) else rel['id']}", headers=h)
+
+    # flat collections + aliases
+    assert any(w["id"] == queen["id"]
+               for w in client.get("/api/workers", headers=h).json())
+    sk = client.post("/api/skills", json={"room_id": rid, "name": "s",
+                                          "content": "c"}, headers=h).json()
+    assert client.get(f"/api/skills/{sk['id']}", headers=h).json()["name"] == "s"
+    assert client.get("/api/skills", params={"room_id": rid},
+                      headers=h).status_code == 200
+    assert client.get("/api/self-mod/audit", headers=h).status_code == 200
+    assert client.get("/api/runs", headers=h).status_code == 200
+
+    # task pause/resume/reset-session aliases
+    t = client.post("/api/tasks", json={"name": "t", "prompt": "p",
+                                        "trigger_type": "manual",
+                                        "room_id": rid}, headers=h).json()
+    assert client.post(f"/api/tasks/{t['id']}/pause",
+                       headers=h).json()["status"] == "paused"
+    assert client.post(f"/api/tasks/{t['id']}/resume",
+                       headers=h).json()["status"] == "active"
+    assert client.post(f"/api/tasks/{t['id']}/reset-session",
+                       headers=h).json()["reset"] == "t"
+
+    # messages detail/reply/delete
+    m = client.post(f"/api/rooms/{rid}/messages",
+                    json={"direction": "inbound", "from_room_id": "ext-1",
+                          "subject": "hi", "body": "hello"}, headers=h).json()
+    det = client.get(f"/api/messages/{m['id']}", headers=h).json()
+    assert det["subject"] == "hi"
+    rep = client.post(f"/api/messages/{m['id']}/reply",
+                      json={"body": "yo"}, headers=h).json()
+    assert rep["subject"] == "Re: hi"
+    assert client.delete(f"/api/messages/{m['id']}",
+                         headers=h).json()["deleted"]
+
+    # status/usage/contact/provider/local-model/clerk surfaces
+    assert "cycles" in client.get(f"/api/rooms/{rid}/usage", headers=h).json()
+    assert client.get("/api/contacts/status", headers=h).json()["email"] is not None
+    prov = client.get("/api/providers/status", headers=h).json()["providers"][0]
+    assert prov["model"] == "qwen3-coder-30b"
+    lm = client.get("/api/local-model/status", headers=h).json()
+    assert lm["installed"] and "in-process" in lm["backend"]
+    assert client.get("/api/clerk/status", headers=h).json()["available"]
+    assert client.get("/api/clerk/usage", headers=h).status_code == 200
+    assert client.post("/api/clerk/typing", headers=h).json()["ok"]
+    assert client.get("/api/settings/referral", headers=h).status_code == 200
