@@ -55,6 +55,31 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     app = FastAPI(title="room_amd", version="0.1.0")
     app.state.ctx = state
 
+    # opt-in HTTP profiling (reference: QUOROOM_PROFILE_HTTP=1,
+    # server/index.ts:289-320 — per-request hrtime, normalized endpoint
+    # buckets, slow-request threshold log)
+    import os as _os
+    if _os.environ.get("ROOMAMD_PROFILE_HTTP") == "1":
+        import logging
+        _plog = logging.getLogger("room_amd.http_profile")
+        state.http_profile = defaultdict(lambda: {"count": 0, "total_ms": 0.0,
+                                                  "max_ms": 0.0})
+
+        @app.middleware("http")
+        async def _profile_http(request: Request, call_next):
+            t0 = time.perf_counter()
+            response = await call_next(request)
+            ms = (time.perf_counter() - t0) * 1000
+            route = request.scope.get("route")
+            bucket = f"{request.method} {route.path if route else request.url.path}"
+            b = state.http_profile[bucket]
+            b["count"] += 1
+            b["total_ms"] += ms
+            b["max_ms"] = max(b["max_ms"], ms)
+            if ms > 500:
+                _plog.warning("slow request: %s took %.0f ms", bucket, ms)
+            return response
+
     # ---------------------------------------------------------------- auth
 
     def get_role(request: Request) -> str:
@@ -727,6 +752,16 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
         return {"key": key, "value": payload.get("value")}
 
     # --------------------------------------------------------------- status
+
+    @app.get("/api/status/http-profile")
+    async def http_profile(role: str = Depends(get_role)):
+        prof = getattr(state, "http_profile", None)
+        if prof is None:
+            return {"enabled": False,
+                    "hint": "set ROOMAMD_PROFILE_HTTP=1 before boot"}
+        return {"enabled": True, "endpoints": {
+            k: {**v, "avg_ms": round(v["total_ms"] / max(v["count"], 1), 2)}
+            for k, v in sorted(prof.items())}}
 
     @app.get("/api/status")
     async def status(role: str = Depends(get_role)):

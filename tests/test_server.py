@@ -369,3 +369,25 @@ def test_route_shape_parity_surface(server):
     assert client.get("/api/clerk/usage", headers=h).status_code == 200
     assert client.post("/api/clerk/typing", headers=h).json()["ok"]
     assert client.get("/api/settings/referral", headers=h).status_code == 200
+
+
+def test_http_profiling_optin(monkeypatch):
+    """ROOMAMD_PROFILE_HTTP=1 → per-endpoint timing buckets (reference
+    QUOROOM_PROFILE_HTTP, index.ts:289-320)."""
+    monkeypatch.setenv("ROOMAMD_PROFILE_HTTP", "1")
+    from fastapi.testclient import TestClient
+
+    from room_amd.db import LockedDb, init_test_db
+    from room_amd.server.app import create_app
+    from room_amd.server.auth import AuthManager
+
+    auth = AuthManager(skip_token_file=True)
+    app = create_app(LockedDb(init_test_db()), auth=auth)
+    client = TestClient(app)
+    h = {"Authorization": f"Bearer {auth.agent_token}"}
+    client.get("/api/rooms", headers=h)
+    client.get("/api/rooms", headers=h)
+    prof = client.get("/api/status/http-profile", headers=h).json()
+    assert prof["enabled"]
+    assert prof["endpoints"]["GET /api/rooms"]["count"] == 2
+    assert prof["endpoints"]["GET /api/rooms"]["avg_ms"] >= 0
