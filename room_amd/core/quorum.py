@@ -98,11 +98,20 @@ def tally(db: sqlite3.Connection, decision_id: int) -> dict:
 
 
 def resolve_voting_decision(db: sqlite3.Connection, decision_id: int) -> dict:
-    """Majority resolution with queen tie-breaker semantics."""
+    """Majority resolution with queen tie-breaker semantics. min_voters on
+    the decision blocks resolution until enough votes are in; non-voters
+    are charged a missed vote for voter-health accounting."""
     decision = q.get_decision(db, decision_id)
     if decision is None:
         raise ValueError(f"Decision {decision_id} not found")
     t = tally(db, decision_id)
+    if t["total"] < (decision.get("min_voters") or 0):
+        return decision  # quorum not met yet — stays open
+    voted = {v["worker_id"] for v in q.get_votes(db, decision_id)}
+    for w in q.list_room_workers(db, decision["room_id"]):
+        if w["id"] not in voted and w["id"] != decision.get("proposer_id"):
+            db.execute("UPDATE workers SET votes_missed = votes_missed + 1"
+                       " WHERE id = ?", (w["id"],))
     if t["yes"] > t["no"]:
         q.resolve_decision(db, decision_id, "approved", f"{t['yes']}-{t['no']}")
     elif t["no"] > t["yes"]:

@@ -216,12 +216,16 @@ def get_room_activity(db: sqlite3.Connection, room_id: int, limit: int = 50) -> 
 
 def create_decision(db: sqlite3.Connection, room_id: int, proposer_id: int | None,
                     proposal: str, decision_type: str, threshold: str = "majority",
-                    timeout_minutes: int = 60) -> dict:
+                    timeout_minutes: int = 60, min_voters: int = 0,
+                    sealed: bool = False) -> dict:
+    # reference db-queries.ts:1266-1274 (min_voters / sealed on the row)
     timeout_at = _iso_in_ms(timeout_minutes * 60_000)
     cur = db.execute(
         "INSERT INTO quorum_decisions (room_id, proposer_id, proposal, decision_type,"
-        " status, threshold, timeout_at) VALUES (?,?,?,?,'voting',?,?)",
-        (room_id, proposer_id, proposal, decision_type, threshold, timeout_at),
+        " status, threshold, timeout_at, min_voters, sealed)"
+        " VALUES (?,?,?,?,'voting',?,?,?,?)",
+        (room_id, proposer_id, proposal, decision_type, threshold, timeout_at,
+         min_voters, int(sealed)),
     )
     return get_decision(db, cur.lastrowid)
 
@@ -285,6 +289,24 @@ def cast_vote(db: sqlite3.Connection, decision_id: int, worker_id: int, vote: st
     return db.execute(
         "SELECT * FROM quorum_votes WHERE decision_id = ? AND worker_id = ?",
         (decision_id, worker_id)).fetchone()
+
+
+def get_voter_health(db: sqlite3.Connection, room_id: int,
+                     threshold: float = 0.5) -> list[dict]:
+    """Per-worker quorum participation (reference db-queries.ts:1368-1383;
+    unlike the reference, votes_cast/votes_missed are actually incremented
+    on vote cast / decision resolution here)."""
+    out = []
+    for w in list_room_workers(db, room_id):
+        total = (w["votes_cast"] or 0) + (w["votes_missed"] or 0)
+        rate = 1.0 if total == 0 else w["votes_cast"] / total
+        out.append({"worker_id": w["id"], "worker_name": w["name"],
+                    "votes_cast": w["votes_cast"],
+                    "votes_missed": w["votes_missed"],
+                    "total_decisions": total,
+                    "participation_rate": rate,
+                    "is_healthy": rate >= threshold})
+    return out
 
 
 def get_votes(db: sqlite3.Connection, decision_id: int) -> list[dict]:

@@ -422,9 +422,13 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.get("/api/decisions/{decision_id}/votes")
     async def decision_votes(decision_id: int, role: str = Depends(get_role)):
         with ldb as db:
-            return q.get_votes(db, decision_id)
-
-    # ---------------------------------------------------------------- tasks
+            d = q.get_decision(db, decision_id)
+            votes = q.get_votes(db, decision_id)
+        # sealed ballot: redact values while the vote is open
+        # (reference routes/decisions.ts:103-107)
+        if d and d.get("sealed") and d["status"] == "voting":
+            return [{**v, "vote": "sealed", "reasoning": None} for v in votes]
+        return votes
 
     @app.get("/api/tasks")
     async def list_tasks(room_id: Optional[int] = None,
@@ -1017,6 +1021,15 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
         if loop_mgr is not None:
             loop_mgr.pause_agent(qid)
         return {"stopped": qid}
+
+    @app.get("/api/rooms/{room_id}/voter-health")
+    async def voter_health(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            room = q.get_room(db, room_id)
+            if room is None:
+                raise HTTPException(404, "room not found")
+            thr = (room.get("config") or {}).get("voterHealthThreshold", 0.5)
+            return q.get_voter_health(db, room_id, threshold=thr)
 
     @app.get("/api/rooms/{room_id}/usage")
     async def room_usage(room_id: int, role: str = Depends(get_role)):

@@ -408,3 +408,30 @@ def test_send_token_produces_signed_raw_tx(db):
     out = send_token(db, r["id"], "0x" + "ab" * 20, "12.5")
     assert out["raw_tx"].startswith("0x02")
     assert len(out["raw_tx"]) > 200
+
+
+def test_sealed_ballot_min_voters_voter_health(db):
+    """Sealed-ballot redaction, min_voters quorum floor, and voter-health
+    accounting (reference decisions.ts:97-117, db-queries.ts:1368-1383)."""
+    from room_amd.core import quorum, room as room_mod
+    from room_amd.db import queries as q
+
+    r = room_mod.create_room(db, "vh", goal="g", worker_model="stub")
+    w1 = q.create_worker(db, "a", "p", room_id=r["id"])
+    w2 = q.create_worker(db, "b", "p", room_id=r["id"])
+    d = q.create_decision(db, r["id"], r["queen_worker_id"], "ship it",
+                          "custom", min_voters=2, sealed=True)
+
+    quorum.vote(db, d["id"], w1["id"], "yes")
+    # min_voters=2: one vote must NOT resolve
+    out = quorum.resolve_voting_decision(db, d["id"])
+    assert out["status"] == "voting"
+
+    quorum.vote(db, d["id"], w2["id"], "yes")
+    out = quorum.resolve_voting_decision(db, d["id"])
+    assert out["status"] == "approved"
+
+    health = q.get_voter_health(db, r["id"])
+    by_id = {h["worker_id"]: h for h in health}
+    assert by_id[w1["id"]]["votes_cast"] == 1
+    assert by_id[w1["id"]]["is_healthy"]
