@@ -1211,6 +1211,131 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
                 "backend": "in-process CDNA4 engine (no sidecar)",
                 "detail": None if ok else "no GPU visible in this process"}
 
+    @app.get("/api/rooms/{room_id}/badges")
+    async def room_badges(room_id: int, role: str = Depends(get_role)):
+        """Unread counters for the UI (reference rooms.ts:228-247)."""
+        with ldb as db:
+            if q.get_room(db, room_id) is None:
+                raise HTTPException(404, "room not found")
+            pend = db.execute(
+                "SELECT COUNT(*) AS n FROM escalations WHERE room_id = ?"
+                " AND status = 'pending' AND to_agent_id IS NULL"
+                " AND from_agent_id IS NOT NULL", (room_id,)).fetchone()["n"]
+            unread = db.execute(
+                "SELECT COUNT(*) AS n FROM room_messages WHERE room_id = ?"
+                " AND status = 'unread'", (room_id,)).fetchone()["n"]
+            votes = db.execute(
+                "SELECT COUNT(*) AS n FROM quorum_decisions WHERE room_id = ?"
+                " AND status IN ('voting', 'announced')",
+                (room_id,)).fetchone()["n"]
+        return {"room_id": room_id, "pending_escalations": pend,
+                "unread_messages": unread, "active_votes": votes}
+
+    @app.post("/api/rooms/{room_id}")
+    async def update_room_alias(room_id: int, payload: dict = Body(...),
+                                role: str = Depends(rw)):
+        return await update_room(room_id, payload, role)  # type: ignore
+
+    @app.get("/api/settings/{key}")
+    async def get_setting_route(key: str, role: str = Depends(get_role)):
+        with ldb as db:
+            return {"key": key, "value": q.get_setting(db, key)}
+
+    # keeper contact verification (reference routes/contacts.ts; email codes
+    # go through the notification relay — SMTP if configured, outbox.jsonl
+    # otherwise — and telegram binding is cloud-mediated)
+    def _hash_email_code(email: str, code: str) -> str:
+        import hashlib
+        return hashlib.sha256(f"{email}:{code}".encode()).hexdigest()
+
+    @app.post("/api/contacts/email/start")
+    async def contacts_email_start(body: dict = Body(...),
+                                   role: str = Depends(rw)):
+        import secrets as _secrets
+        from datetime import datetime, timedelta
+
+        from ..core.notifications import notify_keeper
+        email = (body.get("email") or "").strip().lower()
+        if "@" not in email or "." not in email.split("@")[-1]:
+            raise HTTPException(400, "Valid email is required")
+        with ldb as db:
+            if (q.get_setting(db, "contact_email") == email
+                    and q.get_setting(db, "contact_email_verified_at")):
+                return {"ok": True, "already_verified": True, "email": email}
+            code = f"{_secrets.randbelow(1_000_000):06d}"
+            expires = (datetime.now() + timedelta(minutes=15)).isoformat()
+            q.set_setting(db, "contact_email", email)
+            q.set_setting(db, "contact_email_verified_at", "")
+            q.set_setting(db, "contact_email_code_hash",
+                          _hash_email_code(email, code))
+            q.set_setting(db, "contact_email_code_expires_at", expires)
+        notify_keeper("Verification code",
+                      f"Your room_amd verification code is {code}",
+                      channel="email")
+        return {"ok": True, "sent_to": email, "expires_at": expires}
+
+    @app.post("/api/contacts/email/verify")
+    async def contacts_email_verify(body: dict = Body(...),
+                                    role: str = Depends(rw)):
+        from datetime import datetime
+        code = (body.get("code") or "").strip()
+        with ldb as db:
+            email = q.get_setting(db, "contact_email") or ""
+            expect = q.get_setting(db, "contact_email_code_hash")
+            expires = q.get_setting(db, "contact_email_code_expires_at") or ""
+            if not expect or _hash_email_code(email, code) != expect:
+                raise HTTPException(400, "Invalid code")
+            if expires and expires < datetime.now().isoformat():
+                raise HTTPException(400, "Code expired")
+            q.set_setting(db, "contact_email_verified_at",
+                          datetime.now().isoformat())
+            q.set_setting(db, "contact_email_code_hash", "")
+            q.set_setting(db, "keeper_email", email)
+        return {"ok": True, "verified": True, "email": email}
+
+    @app.post("/api/contacts/email/resend")
+    async def contacts_email_resend(role: str = Depends(rw)):
+        with ldb as db:
+            email = q.get_setting(db, "contact_email")
+        if not email:
+            raise HTTPException(400, "No pending email verification")
+        return await contacts_email_start({"email": email}, role)  # type: ignore
+
+    @app.post("/api/contacts/telegram/start")
+    async def contacts_telegram_start(role: str = Depends(rw)):
+        from ..core import cloud_sync
+        if cloud_sync.cloud_api() is None:
+            raise HTTPException(503, "Telegram binding needs the cloud relay "
+                                     "(ROOMAMD_CLOUD_API)")
+        return {"ok": False, "error": "cloud telegram verification relay "
+                                      "not reachable"}
+
+    @app.post("/api/contacts/telegram/check")
+    async def contacts_telegram_check(role: str = Depends(get_role)):
+        with ldb as db:
+            chat = q.get_setting(db, "keeper_telegram_chat_id")
+        return {"connected": bool(chat), "chat_id": chat}
+
+    @app.post("/api/contacts/telegram/disconnect")
+    async def contacts_telegram_disconnect(role: str = Depends(rw)):
+        with ldb as db:
+            q.set_setting(db, "keeper_telegram_chat_id", "")
+        return {"ok": True}
+
+    @app.get("/api/rooms/{room_id}/cloud-id")
+    async def room_cloud_id(room_id: int, role: str = Depends(get_role)):
+        from ..core import cloud_sync
+        tok = cloud_sync.load_room_tokens().get(str(room_id))
+        return {"room_id": room_id, "registered": tok is not None,
+                "cloud_id": f"room-{room_id}" if tok else None}
+
+    @app.get("/api/rooms/{room_id}/network")
+    async def room_network(room_id: int, role: str = Depends(get_role)):
+        from ..core import cloud_sync
+        if cloud_sync.cloud_api() is None:
+            return {"rooms": [], "note": "cloud not configured"}
+        return {"rooms": []}
+
     @app.get("/api/settings/referral")
     async def settings_referral(role: str = Depends(get_role)):
         with ldb as db:
@@ -1276,7 +1401,7 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     # literal paths above their parameterized siblings (/api/rooms/{room_id}
     # would otherwise shadow /api/rooms/queen-states with a 422).
     literals = [r for r in app.router.routes
-                if getattr(r, "path", "") in ("/api/rooms/queen-states",)]
+                if getattr(r, "path", "") in ("/api/rooms/queen-states", "/api/settings/referral")]
     for r in literals:
         app.router.routes.remove(r)
         app.router.routes.insert(0, r)
