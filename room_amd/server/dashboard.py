@@ -54,9 +54,8 @@ DASHBOARD_HTML = r"""<!doctype html>
  <section>
   <h2 id="roomtitle">Room</h2>
   <div id="roomctl" style="margin-bottom:8px"></div>
-  <h2>Goals</h2><div id="goals" style="margin-bottom:10px"></div>
-  <h2>Workers</h2><table id="workers"></table>
-  <h2 style="margin-top:10px">Decisions</h2><table id="decisions"></table>
+  <nav id="tabs" style="margin-bottom:8px"></nav>
+  <div id="tabbody"></div>
  </section>
  <section><h2>Live activity</h2><div id="events"></div></section>
 </main>
@@ -83,8 +82,16 @@ async function loadRooms() {
     `${r.name}<span class="tag ${r.status}">${r.status}</span></div>`).join('');
   if (sel === null && rooms.length) select(rooms[0].id);
 }
+const TABS = ['overview','tasks','skills','memory','messages','wallet','settings'];
+let tab = 'overview';
+function tabbar() {
+  $$('tabs').innerHTML = TABS.map(t =>
+    `<button style="${t===tab?'background:#30363d':''}" ` +
+    `onclick="setTab('${t}')">${t}</button>`).join(' ');
+}
+async function setTab(t) { tab = t; if (sel !== null) await select(sel); }
 async function select(id) {
-  sel = id; loadRooms();
+  sel = id; loadRooms(); tabbar();
   const st = await api(`/rooms/${id}/status`);
   $$('roomtitle').textContent = `${st.room.name} — ${st.room.goal || 'no goal'}`;
   $$('roomctl').innerHTML =
@@ -92,16 +99,82 @@ async function select(id) {
     `<button onclick="roomAct(${id},'pause')">pause</button> ` +
     `<button onclick="roomAct(${id},'restart')">restart</button>` +
     ` <span class="muted">cycles: ${st.token_usage.cycles}</span>`;
-  $$('goals').innerHTML = st.goals.map(g =>
-    `<div class="ev">${g.description} <span class="muted">${g.status}</span>` +
-    `<div class="bar"><div style="width:${Math.round((g.progress||0)*100)}%"></div></div></div>`
-  ).join('') || '<span class="muted">none</span>';
-  $$('workers').innerHTML = '<tr><th>name</th><th>role</th><th>state</th></tr>' +
+  const b = $$('tabbody');
+  if (tab === 'overview') {
+    const decs = await api(`/rooms/${id}/decisions`);
+    b.innerHTML = '<h2>Goals</h2>' + (st.goals.map(g =>
+      `<div class="ev">${g.description} <span class="muted">${g.status}</span>` +
+      `<div class="bar"><div style="width:${Math.round((g.progress||0)*100)}%"></div></div></div>`
+    ).join('') || '<span class="muted">none</span>') +
+    '<h2 style="margin-top:10px">Workers</h2><table>' +
+    '<tr><th>name</th><th>role</th><th>state</th></tr>' +
     st.workers.map(w => `<tr><td>${w.name}</td><td>${w.role||''}</td>` +
-                        `<td>${w.agent_state}</td></tr>`).join('');
-  const decs = await api(`/rooms/${id}/decisions`);
-  $$('decisions').innerHTML = '<tr><th>proposal</th><th>status</th></tr>' +
-    decs.slice(0, 8).map(d => `<tr><td>${d.proposal}</td><td>${d.status}</td></tr>`).join('');
+                        `<td>${w.agent_state}</td></tr>`).join('') + '</table>' +
+    '<h2 style="margin-top:10px">Decisions</h2><table>' +
+    '<tr><th>proposal</th><th>status</th></tr>' +
+    decs.slice(0, 8).map(d => `<tr><td>${d.proposal}</td><td>${d.status}</td></tr>`).join('')
+    + '</table>';
+  } else if (tab === 'tasks') {
+    const ts = await api(`/tasks?room_id=${id}`);
+    b.innerHTML = '<table><tr><th>name</th><th>trigger</th><th>status</th>' +
+      '<th>runs</th><th></th></tr>' + ts.map(t =>
+      `<tr><td>${t.name}</td><td>${t.cron_expression||t.trigger_type}</td>` +
+      `<td>${t.status}</td><td>${t.run_count}</td>` +
+      `<td><button onclick="api('/tasks/${t.id}/run',{method:'POST'})">run</button></td></tr>`
+      ).join('') + '</table>';
+  } else if (tab === 'skills') {
+    const sk = await api(`/rooms/${id}/skills`);
+    b.innerHTML = '<table><tr><th>name</th><th>v</th><th>auto</th><th>content</th></tr>' +
+      sk.map(x => `<tr><td>${x.name}</td><td>${x.version}</td>` +
+      `<td>${x.auto_activate?'on':''}</td>` +
+      `<td class="muted">${(x.content||'').slice(0,120)}</td></tr>`).join('') + '</table>';
+  } else if (tab === 'memory') {
+    const ents = await api(`/memory/entities?room_id=${id}&limit=50`);
+    b.innerHTML = '<div style="display:flex;gap:4px;margin-bottom:6px">' +
+      '<input id="memq" style="flex:1" placeholder="hybrid search…" ' +
+      'onkeydown="if(event.key===\'Enter\')memSearch()">' +
+      '<button onclick="memSearch()">search</button></div><div id="memres"></div>' +
+      '<table><tr><th>id</th><th>name</th><th>type</th><th>created</th></tr>' +
+      ents.map(e => `<tr><td>${e.id}</td><td>${e.name}</td><td>${e.type}</td>` +
+      `<td class="muted">${e.created_at}</td></tr>`).join('') + '</table>';
+  } else if (tab === 'messages') {
+    const ms = await api(`/rooms/${id}/messages`);
+    const es = await api(`/rooms/${id}/escalations`);
+    b.innerHTML = '<h2>Escalations</h2>' + (es.map(e =>
+      `<div class="ev">${e.question} <span class="muted">${e.status}</span></div>`
+      ).join('') || '<span class="muted">none</span>') +
+      '<h2 style="margin-top:10px">Inter-room messages</h2>' + (ms.map(m =>
+      `<div class="ev"><b>${m.subject}</b> <span class="muted">${m.direction} ` +
+      `${m.status}</span><br>${(m.body||'').slice(0,200)}</div>`).join('')
+      || '<span class="muted">none</span>');
+  } else if (tab === 'wallet') {
+    const w = await api(`/rooms/${id}/wallet`);
+    const tx = await api(`/rooms/${id}/wallet/transactions`);
+    b.innerHTML = `<div class="ev">address: <b>${w.address||'—'}</b></div>` +
+      '<h2 style="margin-top:10px">Transactions</h2><table>' +
+      '<tr><th>type</th><th>amount</th><th>counterparty</th><th>status</th></tr>' +
+      tx.map(t => `<tr><td>${t.type}</td><td>${t.amount}</td>` +
+      `<td class="muted">${(t.counterparty||'—').slice(0,14)}</td>` +
+      `<td>${t.status}</td></tr>`).join('') + '</table>';
+  } else if (tab === 'settings') {
+    const room = await api(`/rooms/${id}`);
+    const vh = await api(`/rooms/${id}/voter-health`);
+    b.innerHTML = '<h2>Room config</h2><pre class="muted" style="font-size:11px">' +
+      JSON.stringify(room.config || {}, null, 1) + '</pre>' +
+      '<h2>Voter health</h2><table><tr><th>worker</th><th>cast</th>' +
+      '<th>missed</th><th>healthy</th></tr>' +
+      vh.map(v => `<tr><td>${v.worker_name}</td><td>${v.votes_cast}</td>` +
+      `<td>${v.votes_missed}</td><td>${v.is_healthy?'yes':'NO'}</td></tr>`
+      ).join('') + '</table>';
+  }
+}
+async function memSearch() {
+  const qy = $$('memq').value.trim();
+  if (!qy) return;
+  const hits = await api(`/memory/search?query=${encodeURIComponent(qy)}&room_id=${sel}`);
+  $$('memres').innerHTML = hits.map(h =>
+    `<div class="ev"><b>${h.name}</b> <span class="muted">${(h.score||0).toFixed(3)}</span></div>`
+    ).join('') || '<span class="muted">no hits</span>';
 }
 async function roomAct(id, act) { await api(`/rooms/${id}/${act}`, {method:'POST'}); select(id); }
 async function createRoom() {
