@@ -9,6 +9,10 @@ import os
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
 from setuptools import setup
+# On ROCm builds torch.utils.cpp_extension exposes the HIP toolchain through
+# the CUDAExtension class name (its "nvcc" args feed hipcc); the sources are
+# plain HIP written for gfx950 — the pipeline's mechanical hipify pass is an
+# identity transform over them (its *_hip.hip copies are gitignored).
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
 SOURCES = [
