@@ -97,3 +97,35 @@ def test_bench_torchrun_world2_cpu():
     assert out["value"] > 0
     assert out["scaling"] == "weak"
     assert out["config"]["agents_per_gpu"] == 5
+
+
+def _worker4(rank: int, world: int, port: int, fail_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world)
+        os.environ["LOCAL_RANK"] = str(rank)
+        ctx = SwarmContext.from_env(device=torch.device("cpu"))
+        votes = torch.tensor([1 if rank % 2 == 0 else -1] * 5,
+                             dtype=torch.int8)
+        tally = ctx.quorum_tally(votes)
+        assert tally == {"yes": 10, "no": 10, "abstain": 0, "total": 20}, tally
+        got = ctx.broadcast_blob({"g": rank} if rank == 0 else None, src=0)
+        assert got == {"g": 0}
+        v, i = ctx.topk_merge(torch.tensor([float(rank), float(rank) - 10.0]),
+                              torch.tensor([rank * 2, rank * 2 + 1]), k=4)
+        assert i.tolist() == [6, 4, 2, 0], i.tolist()  # ranks desc
+        ctx.barrier()
+        torch.distributed.destroy_process_group()
+    except Exception as e:
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}")
+        raise
+
+
+def test_swarm_collectives_world4():
+    """Driver-shape world=4: quorum tally, broadcast, top-k merge."""
+    port = _find_free_port()
+    fail_q = mp.get_context("spawn").SimpleQueue()
+    mp.spawn(_worker4, args=(4, port, fail_q), nprocs=4, join=True)
+    assert fail_q.empty()
