@@ -276,11 +276,17 @@ class AgentLoopManager:
             if result.success and result.text and not (w and w.get("wip")) \
                     and result.tool_calls_executed == 0:
                 q.set_worker_wip(db, worker_id, f"[auto] {result.text[:500]}")
-            # persist session
+            # persist session; trim at MESSAGE granularity so the stored
+            # blob is always valid JSON (a char-level cut would make the next
+            # cycle's json.loads fail and silently reset the session)
+            msgs = result.messages or []
+            blob = json.dumps(msgs)
+            while len(blob) > 200_000 and len(msgs) > 2:
+                msgs = msgs[2:]            # drop the oldest exchange
+                blob = json.dumps(msgs)
             q.save_agent_session(
                 db, worker_id, session_id=result.session_id,
-                messages_json=json.dumps(result.messages or [])[:200_000],
-                model=model, turn_count=turn_count + 1)
+                messages_json=blob, model=model, turn_count=turn_count + 1)
             q.prune_old_cycles(db, room_id)
 
         # stuck detector: same summary 3 cycles in a row (:605-617)
