@@ -272,6 +272,15 @@ class LocalEngine:
             if not cache.free_slots:
                 self._evict_lru_session()
             req.slot = cache.alloc_seq()
+        # block-pressure relief: if the pool cannot hold this request's
+        # prompt + generation budget, evict idle LRU sessions until it can
+        # (their durable agent_sessions rows re-prefill on the next cycle)
+        need = (len(req.prompt_tokens) + req.max_new_tokens) // 16 + 2
+        while cache.blocks_free() < need:
+            try:
+                self._evict_lru_session(exclude_slot=req.slot)
+            except RuntimeError:
+                break  # nothing evictable — ensure_capacity will raise
             req.pos = 0
             req.pending_prefill = list(req.prompt_tokens)
             if req.session_key:
@@ -284,11 +293,12 @@ class LocalEngine:
             req.pos = max(0, req.pos - 1)
             req.pending_prefill = [req.prompt_tokens[-1]]
 
-    def _evict_lru_session(self) -> None:
-        """Slot pressure: drop the least-recently-used idle session (its KV
-        blocks free; the durable agent_sessions row lets the next cycle
-        re-prefill). Sessions of currently-active requests are never evicted."""
-        active_slots = {r.slot for r in self._active}
+    def _evict_lru_session(self, exclude_slot: int = -1) -> None:
+        """Slot/block pressure: drop the least-recently-used idle session
+        (its KV blocks free; the durable agent_sessions row lets the next
+        cycle re-prefill). Sessions of currently-active requests — and the
+        request being admitted (exclude_slot) — are never evicted."""
+        active_slots = {r.slot for r in self._active} | {exclude_slot}
         candidates = [(s.last_used, key) for key, s in self.sessions.items()
                       if s.slot not in active_slots]
         if not candidates:
