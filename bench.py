@@ -180,7 +180,9 @@ def main() -> None:
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": "qwen3-coder-30b" if args.model_config == "30b" else "tiny",
+                "model": (model_tag if not use_gpu
+                          else ("qwen3-coder-30b" if args.model_config == "30b"
+                                else "tiny")),
                 "agents_per_gpu": len(agent_ids),
                 "decode_tokens_per_cycle": args.decode_tokens,
                 "global_batch": n_gpus * len(agent_ids),
