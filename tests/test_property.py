@@ -53,3 +53,40 @@ def test_chat_encoding_monotone(roles_contents):
     assert all(0 <= i < tok.VOCAB_SIZE for i in ids)
     longer = tok.encode_chat(msgs + [{"role": "user", "content": "more text"}])
     assert len(longer) > len(ids)
+
+
+@settings(max_examples=150, deadline=None)
+@given(st.integers(min_value=0, max_value=59),
+       st.integers(min_value=0, max_value=23),
+       st.integers(min_value=1, max_value=28),
+       st.integers(min_value=1, max_value=12))
+def test_cron_next_after_always_matches(minute, hour, dom, month):
+    """next_after of a fully-pinned cron lands exactly on a matching time,
+    strictly in the future."""
+    from datetime import datetime
+
+    from room_amd.core.cron import Cron
+    c = Cron(f"{minute} {hour} {dom} {month} *")
+    base = datetime(2026, 1, 1, 0, 0)
+    nxt = c.next_after(base)
+    assert nxt is not None and nxt > base
+    assert (nxt.minute, nxt.hour, nxt.day, nxt.month) == (minute, hour, dom,
+                                                          month)
+    assert c.matches(nxt)
+
+
+@settings(max_examples=150, deadline=None)
+@given(st.integers(min_value=2, max_value=30))
+def test_cron_step_interval_spacing(step):
+    """*/N minute crons fire at multiples of N, and consecutive fires are
+    exactly the step (or the hour rollover remainder) apart."""
+    from datetime import datetime
+
+    from room_amd.core.cron import Cron
+    c = Cron(f"*/{step} * * * *")
+    t = datetime(2026, 3, 5, 10, 1)
+    a = c.next_after(t)
+    b = c.next_after(a)
+    assert a.minute % step == 0 and b.minute % step == 0
+    assert c.matches(a) and c.matches(b)
+    assert (b - a).total_seconds() <= step * 60 + 3600  # step or rollover
