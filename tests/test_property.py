@@ -65,7 +65,7 @@ def test_cron_next_after_always_matches(minute, hour, dom, month):
     strictly in the future."""
     from datetime import datetime
 
-    from room_amd.core.cron import Cron
+    from room_amd.core.cron import CronExpression as Cron
     c = Cron(f"{minute} {hour} {dom} {month} *")
     base = datetime(2026, 1, 1, 0, 0)
     nxt = c.next_after(base)
@@ -82,7 +82,7 @@ def test_cron_step_interval_spacing(step):
     exactly the step (or the hour rollover remainder) apart."""
     from datetime import datetime
 
-    from room_amd.core.cron import Cron
+    from room_amd.core.cron import CronExpression as Cron
     c = Cron(f"*/{step} * * * *")
     t = datetime(2026, 3, 5, 10, 1)
     a = c.next_after(t)
