@@ -82,7 +82,7 @@ async function loadRooms() {
     `${r.name}<span class="tag ${r.status}">${r.status}</span></div>`).join('');
   if (sel === null && rooms.length) select(rooms[0].id);
 }
-const TABS = ['overview','tasks','skills','memory','messages','wallet','settings'];
+const TABS = ['overview','tasks','skills','memory','messages','wallet','credentials','settings','status','help'];
 let tab = 'overview';
 function tabbar() {
   $$('tabs').innerHTML = TABS.map(t =>
@@ -156,6 +156,45 @@ async function select(id) {
       tx.map(t => `<tr><td>${t.type}</td><td>${t.amount}</td>` +
       `<td class="muted">${(t.counterparty||'—').slice(0,14)}</td>` +
       `<td>${t.status}</td></tr>`).join('') + '</table>';
+  } else if (tab === 'credentials') {
+    const cs = await api(`/rooms/${id}/credentials`);
+    b.innerHTML = '<div style="display:flex;gap:4px;margin-bottom:6px">' +
+      '<input id="credname" placeholder="name">' +
+      '<input id="credval" placeholder="value" type="password" style="flex:1">' +
+      '<button onclick="credAdd()">add</button></div>' +
+      '<table><tr><th>name</th><th>type</th><th>by</th><th>created</th></tr>' +
+      cs.map(x => `<tr><td>${x.name}</td><td>${x.type}</td>` +
+      `<td>${x.provided_by}</td><td class="muted">${x.created_at}</td></tr>`
+      ).join('') + '</table>' +
+      '<div class="muted" style="margin-top:6px">values are AES-256-GCM ' +
+      'encrypted at rest; agents read them with room_get_credential</div>';
+  } else if (tab === 'status') {
+    const st2 = await api('/status');
+    const prof = await api('/status/http-profile');
+    const lm = await api('/local-model/status');
+    b.innerHTML = '<h2>Server</h2><pre class="muted" style="font-size:11px">' +
+      JSON.stringify(st2, null, 1) + '</pre>' +
+      '<h2>Engine</h2><pre class="muted" style="font-size:11px">' +
+      JSON.stringify(lm, null, 1) + '</pre>' +
+      '<h2>HTTP profile</h2><pre class="muted" style="font-size:11px">' +
+      JSON.stringify(prof, null, 1).slice(0, 3000) + '</pre>';
+  } else if (tab === 'help') {
+    b.innerHTML = `
+      <h2>room_amd — MI355X-native agent swarm</h2>
+      <div class="ev">Each room runs a <b>queen</b> (control plane: goals,
+      delegation, quorum announcements) and <b>workers</b> (executors).
+      Agents cycle observe → prompt → decode on the in-process CDNA4
+      engine → tools → persist. Decisions use announce-and-object with
+      keeper override.</div>
+      <div class="ev"><b>Start here:</b> create a room, give it a goal,
+      press start. Watch live activity on the right; answer escalations in
+      the messages tab; fund the wallet from the wallet tab.</div>
+      <div class="ev"><b>Interfaces:</b> this dashboard, the REST API
+      (docs/API.md), the MCP stdio server (<code>room-amd mcp</code>, 95
+      tools) and webhooks (tasks + queen wake).</div>
+      <div class="ev"><b>Keeper contact:</b> verify an email under
+      contacts so escalations reach you; or read the outbox at
+      ~/.roomamd/outbox.jsonl.</div>`;
   } else if (tab === 'settings') {
     const room = await api(`/rooms/${id}`);
     const vh = await api(`/rooms/${id}/voter-health`);
@@ -167,6 +206,13 @@ async function select(id) {
       `<td>${v.votes_missed}</td><td>${v.is_healthy?'yes':'NO'}</td></tr>`
       ).join('') + '</table>';
   }
+}
+async function credAdd() {
+  const name = $$('credname').value.trim(), value = $$('credval').value;
+  if (!name || !value) return;
+  await api(`/rooms/${sel}/credentials`, {method:'POST',
+            body: JSON.stringify({name, value})});
+  setTab('credentials');
 }
 async function memSearch() {
   const qy = $$('memq').value.trim();
