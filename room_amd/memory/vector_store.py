@@ -54,6 +54,8 @@ class GpuVectorStore:
             for i, eid in enumerate(entity_ids):
                 row = self._id_to_row.get(eid)
                 if row is None:
+                    if self.size >= self.capacity:
+                        raise RuntimeError("vector store full")
                     row = self.size
                     self.size += 1
                     self._id_to_row[eid] = row
@@ -162,6 +164,12 @@ class MemoryService:
                 vec = self.embedder.embed(text)
                 q.upsert_embedding(db, ent["id"], vec,
                                    self.embedder.text_hash(text))
-                self.store.upsert(ent["id"], vec)
+                try:
+                    self.store.upsert(ent["id"], vec)
+                except RuntimeError:
+                    # GPU index at capacity: SQLite row is still embedded
+                    # (durable + FTS-searchable); hot-index misses degrade
+                    # recall to keyword-only for the overflow, not an error
+                    pass
                 done += 1
         return done
