@@ -43,6 +43,14 @@ class ServerState:
         self.runtime = runtime
         self.started_at = time.time()
         self.webhook_hits: dict[str, deque] = defaultdict(deque)
+        # strong refs for fire-and-forget task executions (asyncio keeps only
+        # weak refs; an untracked create_task can be GC'd mid-run)
+        self.bg_tasks: set = set()
+
+    def spawn(self, coro) -> None:
+        t = asyncio.create_task(coro)
+        self.bg_tasks.add(t)
+        t.add_done_callback(self.bg_tasks.discard)
 
 
 def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
@@ -139,7 +147,7 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
         if task is None:
             raise HTTPException(404, "unknown webhook token")
         if runner is not None:
-            asyncio.create_task(runner.execute_task(task["id"]))
+            state.spawn(runner.execute_task(task["id"]))
         return {"queued": True, "task_id": task["id"]}
 
     @app.post("/api/hooks/queen/{token}")
@@ -492,7 +500,7 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     async def run_task(task_id: int, role: str = Depends(rw)):
         if runner is None:
             raise HTTPException(503, "task runner not available")
-        asyncio.create_task(runner.execute_task(task_id))
+        state.spawn(runner.execute_task(task_id))
         return {"queued": True}
 
     @app.get("/api/tasks/{task_id}/runs")

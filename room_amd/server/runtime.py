@@ -36,6 +36,14 @@ class ServerRuntime:
         self._tasks: list[asyncio.Task] = []
         self._stop = asyncio.Event()
         self._last_cron_minute: dict[int, str] = {}
+        # fire-and-forget executions need a strong ref: asyncio holds only
+        # weak refs to tasks, so an untracked create_task can be GC'd mid-run
+        self._bg: set[asyncio.Task] = set()
+
+    def _spawn(self, coro) -> None:
+        t = asyncio.create_task(coro)
+        self._bg.add(t)
+        t.add_done_callback(self._bg.discard)
 
     async def start(self) -> None:
         with self.ldb as db:
@@ -86,12 +94,12 @@ class ServerRuntime:
                     try:
                         if CronExpression(t["cron_expression"]).matches(now):
                             self._last_cron_minute[t["id"]] = minute_key
-                            asyncio.create_task(self.runner.execute_task(t["id"]))
+                            self._spawn(self.runner.execute_task(t["id"]))
                     except ValueError:
                         pass
                 for t in due_once:
                     if not self.runner.is_task_running(t["id"]):
-                        asyncio.create_task(self.runner.execute_task(t["id"]))
+                        self._spawn(self.runner.execute_task(t["id"]))
             except Exception:
                 pass
             if await self._sleep(CRON_REFRESH_S):
