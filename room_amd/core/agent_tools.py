@@ -99,6 +99,13 @@ _QUEEN_ONLY_TOOLS = [
     ToolDef("room_send_token", "Send tokens from the room wallet.",
             _obj({"to_address": _S, "amount": _S, "chain": _S, "token": _S},
                  ["to_address", "amount"])),
+    ToolDef("room_update_worker", "Update a worker's prompt, role or pacing.",
+            _obj({"worker_id": _I, "system_prompt": _S, "role": _S,
+                  "cycle_gap_ms": _I, "max_turns": _I}, ["worker_id"])),
+    ToolDef("room_web_search", "Keyless web search for research.",
+            _obj({"query": _S}, ["query"])),
+    ToolDef("room_web_fetch", "Fetch a URL as readable text.",
+            _obj({"url": _S}, ["url"])),
 ]
 
 QUEEN_TOOLS: list[ToolDef] = _QUEEN_ONLY_TOOLS + _COMMON_TOOLS
@@ -116,6 +123,23 @@ def execute_agent_tool(db: sqlite3.Connection, room_id: int, worker_id: int,
     """In-process tool dispatch. Returns a string result fed back to the model."""
     name, args = call.name, call.arguments
     try:
+        if name == "room_update_worker":
+            fields = {k: v for k, v in args.items()
+                      if k in ("system_prompt", "role", "cycle_gap_ms",
+                               "max_turns") and v is not None}
+            w = q.update_worker(db, args["worker_id"], **fields)
+            if w is None:
+                return json.dumps({"error": "worker not found"})
+            return json.dumps({"updated": w["id"], "fields": list(fields)})
+
+        if name == "room_web_search":
+            from .web_tools import web_search
+            return json.dumps(web_search(args["query"]))[:4000]
+
+        if name == "room_web_fetch":
+            from .web_tools import web_fetch
+            return json.dumps(web_fetch(args["url"]))[:6000]
+
         if name == "room_set_goal":
             g = q.create_goal(db, room_id, args["description"],
                               parent_goal_id=args.get("parent_goal_id"),

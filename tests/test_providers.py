@@ -108,3 +108,30 @@ def test_engine_error_propagates():
     res = execute_agent(AgentExecutionOptions(prompt="x", model="boom"))
     assert not res.success
     assert "engine exploded" in res.error
+
+
+def test_queen_update_worker_and_web_tools(db):
+    """Queen-only tools: room_update_worker applies fields; web tools degrade
+    to structured errors offline (reference queen-tools.ts web/update set)."""
+    import json
+
+    from room_amd.core import room as room_mod
+    from room_amd.core.agent_tools import execute_agent_tool
+    from room_amd.db import queries as q
+    from room_amd.engine.types import ToolCall
+
+    r = room_mod.create_room(db, "qt", goal="g", worker_model="stub")
+    w = q.create_worker(db, "e", "old prompt", room_id=r["id"])
+    out = json.loads(execute_agent_tool(
+        db, r["id"], r["queen_worker_id"],
+        ToolCall("room_update_worker",
+                 {"worker_id": w["id"], "system_prompt": "new prompt",
+                  "max_turns": 5})))
+    assert out["updated"] == w["id"]
+    w2 = q.get_worker(db, w["id"])
+    assert w2["system_prompt"] == "new prompt" and w2["max_turns"] == 5
+
+    res = json.loads(execute_agent_tool(
+        db, r["id"], r["queen_worker_id"],
+        ToolCall("room_web_search", {"query": "anything"})))
+    assert isinstance(res, dict)  # offline → {"error": ...} or results
