@@ -106,6 +106,10 @@ _QUEEN_ONLY_TOOLS = [
             _obj({"query": _S}, ["query"])),
     ToolDef("room_web_fetch", "Fetch a URL as readable text.",
             _obj({"url": _S}, ["url"])),
+    ToolDef("room_browser", "Persistent browser session action "
+            "(navigate/snapshot/click/type).",
+            _obj({"session_id": _S, "action": _S, "url": _S, "selector": _S,
+                  "text": _S}, ["session_id", "action"])),
 ]
 
 QUEEN_TOOLS: list[ToolDef] = _QUEEN_ONLY_TOOLS + _COMMON_TOOLS
@@ -139,6 +143,12 @@ def execute_agent_tool(db: sqlite3.Connection, room_id: int, worker_id: int,
         if name == "room_web_fetch":
             from .web_tools import web_fetch
             return json.dumps(web_fetch(args["url"]))[:6000]
+
+        if name == "room_browser":
+            from .web_tools import browser_action
+            return json.dumps(browser_action(
+                args["session_id"], args["action"], url=args.get("url"),
+                selector=args.get("selector"), text=args.get("text")))[:6000]
 
         if name == "room_set_goal":
             g = q.create_goal(db, room_id, args["description"],
