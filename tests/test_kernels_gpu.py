@@ -287,7 +287,10 @@ def test_vector_store_topk():
 
 def test_gemv_vs_matmul():
     torch.manual_seed(13)
-    for B, H, N in [(1, 2048, 5120), (5, 4096, 2048), (8, 2048, 128)]:
+    # cover every BN template specialization (1..8)
+    for B, H, N in [(1, 2048, 5120), (2, 2048, 512), (3, 512, 1024),
+                    (4, 4096, 512), (5, 4096, 2048), (6, 2048, 512),
+                    (7, 512, 512), (8, 2048, 128)]:
         x = torch.randn(B, H, dtype=torch.bfloat16, device=DEV)
         w = torch.randn(N, H, dtype=torch.bfloat16, device=DEV) * 0.05
         y = torch.empty(B, N, dtype=torch.bfloat16, device=DEV)
