@@ -38,6 +38,18 @@ CLERK_TOOLS = [
             _obj({"room_id": _I, "body": _S}, ["room_id", "body"])),
     ToolDef("clerk_room_status", "Get a room's status summary.",
             _obj({"room_id": _I}, ["room_id"])),
+    ToolDef("clerk_restart_room", "Restart a room (clears goals/decisions).",
+            _obj({"room_id": _I}, ["room_id"])),
+    ToolDef("clerk_delete_room", "Permanently delete a room.",
+            _obj({"room_id": _I}, ["room_id"])),
+    ToolDef("clerk_list_tasks", "List scheduled tasks (optionally per room).",
+            _obj({"room_id": _I})),
+    ToolDef("clerk_keeper_vote", "Cast the keeper's override vote on a "
+            "decision.", _obj({"decision_id": _I, "vote": _S},
+                              ["decision_id", "vote"])),
+    ToolDef("clerk_answer_escalation", "Answer a pending escalation on the "
+            "keeper's behalf.", _obj({"escalation_id": _I, "answer": _S},
+                                     ["escalation_id", "answer"])),
 ]
 
 
@@ -66,6 +78,27 @@ def execute_clerk_tool(ldb: LockedDb, call: ToolCall) -> str:
             if name == "clerk_send_message":
                 e = q.create_escalation(db, args["room_id"], args["body"])
                 return json.dumps({"escalation_id": e["id"]})
+            if name == "clerk_restart_room":
+                room_mod.restart_room(db, args["room_id"])
+                return json.dumps({"restarted": True})
+            if name == "clerk_delete_room":
+                room_mod.delete_room(db, args["room_id"])
+                return json.dumps({"deleted": True})
+            if name == "clerk_list_tasks":
+                ts = q.list_tasks(db, room_id=args.get("room_id"))
+                return json.dumps([{"id": t["id"], "name": t["name"],
+                                    "status": t["status"],
+                                    "trigger": t["trigger_type"]}
+                                   for t in ts])
+            if name == "clerk_keeper_vote":
+                from . import quorum as quorum_mod
+                d = quorum_mod.keeper_vote(db, args["decision_id"],
+                                           args["vote"])
+                return json.dumps({"decision_id": d["id"],
+                                   "status": d["status"]})
+            if name == "clerk_answer_escalation":
+                q.answer_escalation(db, args["escalation_id"], args["answer"])
+                return json.dumps({"answered": True})
             if name == "clerk_room_status":
                 s = room_mod.get_room_status(db, args["room_id"])
                 return json.dumps({"room": s["room"]["name"],
