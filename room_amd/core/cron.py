@@ -40,6 +40,11 @@ class CronExpression:
             raise ValueError(f"cron needs 5 fields, got {len(fields)}: {expr!r}")
         self.minute, self.hour, self.dom, self.month, self.dow = (
             _parse_field(f, lo, hi) for f, (lo, hi) in zip(fields, FIELD_RANGES))
+        # standard cron (and node-cron): when BOTH day-of-month and
+        # day-of-week are restricted, a time matches if EITHER matches
+        # ('0 0 13 * 5' = every Friday AND every 13th, not Friday-the-13th)
+        self._dom_restricted = fields[2].strip() != "*"
+        self._dow_restricted = fields[4].strip() != "*"
 
     def matches(self, dt: datetime) -> bool:
         return self._full_match(dt)
@@ -48,10 +53,14 @@ class CronExpression:
         cron_dow = (dt.weekday() + 1) % 7  # python Mon=0 → cron Sun=0
         return cron_dow in self.dow
 
+    def _day_match(self, dt: datetime) -> bool:
+        if self._dom_restricted and self._dow_restricted:
+            return dt.day in self.dom or self._dow_match(dt)
+        return dt.day in self.dom and self._dow_match(dt)
+
     def _full_match(self, dt: datetime) -> bool:
         return (dt.minute in self.minute and dt.hour in self.hour
-                and dt.day in self.dom and dt.month in self.month
-                and self._dow_match(dt))
+                and dt.month in self.month and self._day_match(dt))
 
     def next_after(self, dt: datetime, limit_days: int = 366) -> datetime | None:
         t = dt.replace(second=0, microsecond=0) + timedelta(minutes=1)

@@ -24,6 +24,7 @@ import torch
 from .. import ops
 from ..models.qwen3_moe import Qwen3MoEConfig, Qwen3MoEModel
 from . import tokenizer as tok
+from .admission import SessionAdmitter
 from .kv_cache import BLOCK_SIZE, PagedKVCache
 from .types import AgentExecutionOptions, ToolDef
 
@@ -45,20 +46,12 @@ class GenRequest:
     out_tokens: list[int] = field(default_factory=list)
     error: Optional[str] = None
     prefill_tokens_run: int = 0   # actually-prefilled (after cache reuse)
+    cancelled: bool = False       # set by the caller on timeout; scheduler drops
     # scheduler state
     slot: int = -1
     pos: int = 0                  # next position to write
     pending_prefill: list[int] = field(default_factory=list)
     last_token: int = -1
-
-
-class _Session:
-    __slots__ = ("slot", "tokens", "last_used")
-
-    def __init__(self, slot: int, tokens: list[int]):
-        self.slot = slot
-        self.tokens = tokens
-        self.last_used = time.time()
 
 
 class _DecodeGraph:
@@ -181,7 +174,7 @@ class LocalEngine:
         self.graphs_enabled = (os.environ.get("ROOMAMD_NO_GRAPHS") != "1")
         self._graphs: dict[tuple, _DecodeGraph] = {}
         self._graphs_broken = False
-        self.sessions: dict[str, _Session] = {}
+        self.admitter = SessionAdmitter(self.cache, cfg.max_position)
         self._queue: "queue.Queue[GenRequest]" = queue.Queue()
         self._active: list[GenRequest] = []
         self._lock = threading.Lock()
@@ -192,6 +185,10 @@ class LocalEngine:
         self._thread = threading.Thread(target=self._scheduler_loop, daemon=True,
                                         name="room-amd-engine")
         self._thread.start()
+
+    @property
+    def sessions(self):
+        return self.admitter.sessions
 
     # ------------------------------------------------------------ public API
 
@@ -206,7 +203,10 @@ class LocalEngine:
                          top_p=top_p, top_k=top_k, session_key=session_key)
         self._queue.put(req)
         if not req.done.wait(timeout):
-            req.error = "generation timeout"
+            # mark cancelled: the scheduler drops it (admit or next step
+            # boundary) and frees its slot — it must NOT keep decoding to
+            # max_new_tokens holding KV blocks and GPU time
+            req.cancelled = True
             raise TimeoutError("generation timed out")
         if req.error:
             raise RuntimeError(req.error)
@@ -225,9 +225,7 @@ class LocalEngine:
 
     def release_session(self, session_key: str) -> None:
         with self._lock:
-            s = self.sessions.pop(session_key, None)
-            if s is not None:
-                self.cache.free_seq(s.slot)
+            self.admitter.release(session_key)
 
     def shutdown(self) -> None:
         self._stop = True
@@ -236,76 +234,23 @@ class LocalEngine:
     # ------------------------------------------------------------ scheduler
 
     def _admit(self, req: GenRequest) -> None:
-        cache = self.cache
         # map token ids into this model's vocab (the synthetic tokenizer emits
         # Qwen-range ids; reduced test configs have a smaller embedding table —
         # out-of-range ids would be an out-of-bounds gather on the GPU)
         V = self.cfg.vocab_size
         req.prompt_tokens = [t % V for t in req.prompt_tokens]
-        # context-window guard: middle-truncate prompts that cannot fit (keep
-        # the head [system prompt] and the recent tail), mirroring the
-        # reference's context-overflow recovery (agent-loop.ts:773-782)
-        budget = self.cfg.max_position - req.max_new_tokens - 8
-        if len(req.prompt_tokens) > budget:
-            head = budget // 4
-            tail = budget - head
-            req.prompt_tokens = (req.prompt_tokens[:head]
-                                 + req.prompt_tokens[-tail:])
-        sess = self.sessions.get(req.session_key) if req.session_key else None
-        if sess is not None:
-            # reuse the longest common token prefix: the session keeps its slot
-            # and blocks; position rolls back to the divergence point.
-            cached = sess.tokens
-            common = 0
-            limit = min(len(cached), len(req.prompt_tokens) - 1)
-            while common < limit and cached[common] == req.prompt_tokens[common]:
-                common += 1
-            if common > 0:
-                req.slot = sess.slot
-                req.pos = common
-                req.pending_prefill = req.prompt_tokens[common:]
-            else:
-                cache.free_seq(sess.slot)
-                self.sessions.pop(req.session_key, None)
-                sess = None
-        if req.slot < 0:
-            if not cache.free_slots:
-                self._evict_lru_session()
-            req.slot = cache.alloc_seq()
-        # block-pressure relief: if the pool cannot hold this request's
-        # prompt + generation budget, evict idle LRU sessions until it can
-        # (their durable agent_sessions rows re-prefill on the next cycle)
-        need = (len(req.prompt_tokens) + req.max_new_tokens) // 16 + 2
-        while cache.blocks_free() < need:
-            try:
-                self._evict_lru_session(exclude_slot=req.slot)
-            except RuntimeError:
-                break  # nothing evictable — ensure_capacity will raise
-            req.pos = 0
-            req.pending_prefill = list(req.prompt_tokens)
-            if req.session_key:
-                self.sessions[req.session_key] = _Session(req.slot, [])
-        if req.session_key and req.session_key in self.sessions:
-            self.sessions[req.session_key].last_used = time.time()
-        req.prefill_tokens_run = len(req.pending_prefill)
-        if not req.pending_prefill:
-            # prompt identical to cache (rare): re-run last token for logits
-            req.pos = max(0, req.pos - 1)
-            req.pending_prefill = [req.prompt_tokens[-1]]
+        # slot/session/block bookkeeping lives in the device-free admitter
+        # (admission.py) so it has CPU unit-test coverage
+        self.admitter.admit(req)
 
-    def _evict_lru_session(self, exclude_slot: int = -1) -> None:
-        """Slot/block pressure: drop the least-recently-used idle session
-        (its KV blocks free; the durable agent_sessions row lets the next
-        cycle re-prefill). Sessions of currently-active requests — and the
-        request being admitted (exclude_slot) — are never evicted."""
-        active_slots = {r.slot for r in self._active} | {exclude_slot}
-        candidates = [(s.last_used, key) for key, s in self.sessions.items()
-                      if s.slot not in active_slots]
-        if not candidates:
-            raise RuntimeError("KV cache: no evictable sessions")
-        _, key = min(candidates)
-        s = self.sessions.pop(key)
-        self.cache.free_seq(s.slot)
+    def _complete(self, req: GenRequest, ok: bool) -> None:
+        """Take `req` out of the active set and resolve its future. ok=False
+        (cancelled/error) drops the session so the slot/blocks free."""
+        kv_tokens = (req.prompt_tokens + req.out_tokens[:-1]) if ok else None
+        with self._lock:
+            self.admitter.finish(req, kv_tokens)
+        self._active.remove(req)
+        req.done.set()
 
     def _scheduler_loop(self) -> None:
         idx = self.device.index if self.device.index is not None \
@@ -318,10 +263,9 @@ class LocalEngine:
                 import sys
                 import traceback
                 traceback.print_exc(file=sys.stderr)
-                for r in self._active:
+                for r in list(self._active):
                     r.error = f"engine error: {e}"
-                    r.done.set()
-                self._active.clear()
+                    self._complete(r, ok=False)
 
     def _scheduler_iteration(self) -> None:
         # admit new requests; an admit failure must resolve that future, never
@@ -331,6 +275,9 @@ class LocalEngine:
             while True:
                 req = self._queue.get(timeout=0.05 if block else 0)
                 block = False
+                if req.cancelled:  # caller gave up before admission
+                    req.done.set()
+                    continue
                 try:
                     with self._lock:
                         self._admit(req)
@@ -340,6 +287,11 @@ class LocalEngine:
                     req.done.set()
         except queue.Empty:
             pass
+        # drop requests whose caller timed out (a timed-out
+        # request must not keep decoding and holding a KV slot)
+        for r in [r for r in self._active if r.cancelled]:
+            r.error = "cancelled"
+            self._complete(r, ok=False)
         if not self._active:
             return
 
@@ -433,12 +385,7 @@ class LocalEngine:
                                 or t in (tok.EOS, tok.IM_END)):
                             finished.append(r)
                 for r in finished:
-                    with self._lock:  # sessions map also mutated by release_session
-                        if r.session_key and r.session_key in self.sessions:
-                            self.sessions[r.session_key].tokens = (
-                                r.prompt_tokens + r.out_tokens[:-1])
-                    self._active.remove(r)
-                    r.done.set()
+                    self._complete(r, ok=True)
                 self.stats["decode_steps"] += steps
                 self.stats["decode_tokens"] += steps * len(reqs)
                 self.stats["decode_time"] += time.time() - t0
@@ -460,28 +407,32 @@ class LocalEngine:
         logits = self.model.forward(tokens_t, seq_t, pos_t,
                                     self.cache.block_table,
                                     self.cache.kcaches, self.cache.vcaches)
-        seeds = torch.randint(1, 2**62, (logits.size(0),), dtype=torch.int64,
-                              device=dev)
-        toks_t = ops.sample_tokens(logits, seeds, top_k=r0.top_k,
-                                   temperature=r0.temperature, top_p=r0.top_p)
-        self._finish_tokens(reqs, toks_t)
+        self._sample_and_append(reqs, logits)
         self.stats["decode_steps"] += 1
         self.stats["decode_tokens"] += len(reqs)
         self.stats["decode_time"] += time.time() - t0
 
     def _sample_and_append(self, reqs: list[GenRequest], logits: torch.Tensor) -> None:
+        # sample per sampling-param group (concurrent requests
+        # with different params must not inherit the first request's settings)
         dev = self.device
-        temps = {(r.temperature, r.top_p, r.top_k) for r in reqs}
-        # all agents share sampling params in practice; take the first
-        temperature, top_p, top_k = next(iter(temps))
-        seeds = torch.randint(1, 2**62, (logits.size(0),), dtype=torch.int64,
-                              device=dev)
-        toks = ops.sample_tokens(logits, seeds, top_k=top_k,
-                                 temperature=temperature, top_p=top_p)
-        self._finish_tokens(reqs, toks)
+        groups: dict[tuple, list[int]] = {}
+        for i, r in enumerate(reqs):
+            groups.setdefault((r.temperature, r.top_p, r.top_k), []).append(i)
+        toks_host = [0] * len(reqs)
+        for (temperature, top_p, top_k), idxs in groups.items():
+            rows = (logits if len(idxs) == len(reqs)
+                    else logits[torch.tensor(idxs, device=dev)])
+            seeds = torch.randint(1, 2**62, (rows.size(0),), dtype=torch.int64,
+                                  device=dev)
+            toks = ops.sample_tokens(rows, seeds, top_k=top_k,
+                                     temperature=temperature, top_p=top_p)
+            for i, t in zip(idxs, toks.tolist()):
+                toks_host[i] = int(t)
+        self._finish_host_tokens(reqs, toks_host)
 
-    def _finish_tokens(self, reqs: list[GenRequest], toks: torch.Tensor) -> None:
-        toks_host = toks.tolist()  # one small sync per step
+    def _finish_host_tokens(self, reqs: list[GenRequest],
+                            toks_host: list[int]) -> None:
         finished = []
         for r, t in zip(reqs, toks_host):
             r.last_token = int(t)
@@ -492,12 +443,7 @@ class LocalEngine:
         for r in finished:
             # the sampled token at r.pos is NOT yet in KV; it will be written
             # if the session continues (prompt extension re-runs it)
-            with self._lock:
-                if r.session_key and r.session_key in self.sessions:
-                    self.sessions[r.session_key].tokens = (
-                        r.prompt_tokens + r.out_tokens[:-1])
-            self._active.remove(r)
-            r.done.set()
+            self._complete(r, ok=True)
 
 
 # ------------------------------------------------------------ singleton

@@ -32,6 +32,22 @@ def test_cron_next_after():
     assert nxt == datetime(2026, 9, 12, 6, 30)
 
 
+def test_cron_dom_dow_or_semantics():
+    """Standard cron: when both day-of-month and day-of-week are restricted,
+    either may match ('0 0 13 * 5' fires every Friday AND every 13th)."""
+    c = CronExpression("0 0 13 * 5")
+    assert c.matches(datetime(2026, 9, 13, 0, 0))   # the 13th (a Sunday)
+    assert c.matches(datetime(2026, 9, 18, 0, 0))   # a Friday (the 18th)
+    assert not c.matches(datetime(2026, 9, 14, 0, 0))  # Monday the 14th
+    # only one restricted → AND as usual
+    c2 = CronExpression("0 0 13 * *")
+    assert c2.matches(datetime(2026, 9, 13, 0, 0))
+    assert not c2.matches(datetime(2026, 9, 18, 0, 0))
+    c3 = CronExpression("0 0 * * 5")
+    assert c3.matches(datetime(2026, 9, 18, 0, 0))
+    assert not c3.matches(datetime(2026, 9, 13, 0, 0))
+
+
 def test_cron_validate():
     assert validate_cron("0 9 * * *")
     assert not validate_cron("not a cron")
