@@ -90,11 +90,24 @@ def vote(db: sqlite3.Connection, decision_id: int, worker_id: int, vote_value: s
 
 
 def tally(db: sqlite3.Connection, decision_id: int) -> dict:
+    """Vote tally (quorum.ts:100-110 semantics). When a SwarmContext is
+    installed (multi-GPU swarm, one room shard per rank), per-shard counts
+    are summed with an RCCL all-reduce over xGMI so the tally covers votes
+    cast on every GPU's shard — the collective is part of the quorum system,
+    not a bench bolt-on. Local SQLite rows remain the durable record."""
     votes = q.get_votes(db, decision_id)
     yes = sum(1 for v in votes if v["vote"] == "yes")
     no = sum(1 for v in votes if v["vote"] == "no")
     abstain = sum(1 for v in votes if v["vote"] == "abstain")
-    return {"yes": yes, "no": no, "abstain": abstain, "total": len(votes)}
+    local = {"yes": yes, "no": no, "abstain": abstain, "total": len(votes)}
+    from ..parallel.swarm import get_swarm_context
+    ctx = get_swarm_context()
+    if ctx is not None and ctx.is_distributed:
+        agg = ctx.tally_allreduce(local)
+        agg["source"] = f"rccl-allreduce world={ctx.world_size}"
+        return agg
+    local["source"] = "local"
+    return local
 
 
 def resolve_voting_decision(db: sqlite3.Connection, decision_id: int) -> dict:
@@ -112,10 +125,13 @@ def resolve_voting_decision(db: sqlite3.Connection, decision_id: int) -> dict:
         if w["id"] not in voted and w["id"] != decision.get("proposer_id"):
             db.execute("UPDATE workers SET votes_missed = votes_missed + 1"
                        " WHERE id = ?", (w["id"],))
+    src = t.get("source", "local")
     if t["yes"] > t["no"]:
-        q.resolve_decision(db, decision_id, "approved", f"{t['yes']}-{t['no']}")
+        q.resolve_decision(db, decision_id, "approved",
+                           f"{t['yes']}-{t['no']} ({src})")
     elif t["no"] > t["yes"]:
-        q.resolve_decision(db, decision_id, "rejected", f"{t['yes']}-{t['no']}")
+        q.resolve_decision(db, decision_id, "rejected",
+                           f"{t['yes']}-{t['no']} ({src})")
     else:
         room = q.get_room(db, decision["room_id"])
         tie_breaker = room["config"].get("tieBreaker", "queen") if room else "queen"

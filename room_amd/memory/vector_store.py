@@ -143,12 +143,37 @@ class MemoryService:
         return ent["id"]
 
     def recall(self, room_id: int | None, query: str, limit: int = 5) -> list[dict]:
+        """Hybrid recall. In the multi-GPU swarm each rank owns a memory
+        shard: every shard fuses its local FTS+cosine hits, then the fused
+        top-k rides one RCCL all-gather and the global list is re-ranked by
+        score — the topk-merge semantics of SURVEY §2c, integrated here so
+        every agent recall is swarm-wide."""
         from ..db import queries as q
         qvec = self.embedder.embed(query)
         semantic = self.store.search(qvec, k=20)
         with self.ldb as db:
-            return q.hybrid_search(db, query, qvec, limit=limit, room_id=room_id,
+            hits = q.hybrid_search(db, query, qvec, limit=limit, room_id=room_id,
                                    semantic_hits=semantic)
+        from ..parallel.swarm import get_swarm_context
+        ctx = get_swarm_context()
+        if ctx is not None and ctx.is_distributed:
+            payload = [{"name": h.get("name"), "category": h.get("category"),
+                        "score": h.get("score"),
+                        "observations": h.get("observations", [])[:5],
+                        "shard": ctx.rank} for h in hits]
+            merged = [h for shard in ctx.allgather_obj(payload) for h in shard]
+            merged.sort(key=lambda h: -(h.get("score") or 0.0))
+            # dedupe by name (same entity remembered on several shards)
+            seen, out = set(), []
+            for h in merged:
+                if h["name"] in seen:
+                    continue
+                seen.add(h["name"])
+                out.append(h)
+                if len(out) >= limit:
+                    break
+            return out
+        return hits
 
     def index_pending(self, batch: int = 64) -> int:
         """Background indexing of unembedded entities (reference:
