@@ -135,17 +135,18 @@ def main() -> None:
         # shard, agents vote as SQLite rows, and resolve_voting_decision's
         # tally rides an RCCL all-reduce across every GPU shard — the
         # system's own vote path, not a bench bolt-on
-        with ldb as db:
-            d = q.create_decision(db, room_id, agent_ids[0],
-                                  f"step {step_no[0]} plan", "low_impact")
-            for wid in agent_ids[1:]:
-                quorum.vote(db, d["id"], wid, "yes")
-            resolved = quorum.resolve_voting_decision(db, d["id"])
-            assert resolved["status"] in ("approved", "rejected"), resolved
-        # control-plane refresh: goal/skill/WIP broadcast from the queen rank
-        sync.step(room_id)
-        # swarm-wide memory recall: per-shard hybrid top-k + all-gather merge
-        memsvc.recall(room_id, "benchmark synthetic observation", limit=3)
+        with ctx.step_scope():
+            with ldb as db:
+                d = q.create_decision(db, room_id, agent_ids[0],
+                                      f"step {step_no[0]} plan", "low_impact")
+                for wid in agent_ids[1:]:
+                    quorum.vote(db, d["id"], wid, "yes")
+                resolved = quorum.resolve_voting_decision(db, d["id"])
+                assert resolved["status"] in ("approved", "rejected"), resolved
+            # control-plane refresh: goal/skill/WIP broadcast from queen rank
+            sync.step(room_id)
+            # swarm-wide memory recall: per-shard top-k + all-gather merge
+            memsvc.recall(room_id, "benchmark synthetic observation", limit=3)
         # launch the next vote-vector all-gather WITHOUT fencing compute; the
         # next step's decode kernels overlap with it
         votes = torch.ones(len(agent_ids), dtype=torch.int8,

@@ -102,7 +102,7 @@ def tally(db: sqlite3.Connection, decision_id: int) -> dict:
     local = {"yes": yes, "no": no, "abstain": abstain, "total": len(votes)}
     from ..parallel.swarm import get_swarm_context
     ctx = get_swarm_context()
-    if ctx is not None and ctx.is_distributed:
+    if ctx is not None and ctx.collective_safe:
         agg = ctx.tally_allreduce(local)
         agg["source"] = f"rccl-allreduce world={ctx.world_size}"
         return agg

@@ -156,7 +156,7 @@ class MemoryService:
                                    semantic_hits=semantic)
         from ..parallel.swarm import get_swarm_context
         ctx = get_swarm_context()
-        if ctx is not None and ctx.is_distributed:
+        if ctx is not None and ctx.collective_safe:
             payload = [{"name": h.get("name"), "category": h.get("category"),
                         "score": h.get("score"),
                         "observations": h.get("observations", [])[:5],

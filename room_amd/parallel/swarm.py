@@ -53,6 +53,30 @@ class SwarmContext:
         self.device = device
         self.group = group
         self.is_distributed = world_size > 1 and dist.is_initialized()
+        # collective calls must be symmetric across ranks; in_step marks the
+        # swarm-step regions where every rank runs the same control flow.
+        # Outside them (e.g. an HTTP route handler that fires on one rank
+        # only) core modules fall back to shard-local behavior instead of
+        # blocking the whole swarm on a lone all-reduce.
+        self.in_step = False
+
+    def step_scope(self):
+        """`with ctx.step_scope():` — marks a symmetric swarm-step region."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def _scope():
+            prev = self.in_step
+            self.in_step = True
+            try:
+                yield self
+            finally:
+                self.in_step = prev
+        return _scope()
+
+    @property
+    def collective_safe(self) -> bool:
+        return self.is_distributed and self.in_step
 
     # ------------------------------------------------------------ init
 

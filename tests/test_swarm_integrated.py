@@ -27,6 +27,7 @@ def _integrated_worker(rank: int, world: int, port: int, fail_q):
         os.environ["WORLD_SIZE"] = str(world)
         os.environ["LOCAL_RANK"] = str(rank)
         ctx = SwarmContext.from_env(device=torch.device("cpu"))
+        ctx.in_step = True  # whole worker body is rank-symmetric
 
         from room_amd.core import quorum, room as room_mod
         from room_amd.db import LockedDb, init_test_db
