@@ -96,6 +96,9 @@ class Qwen3MoEModel:
         import os as _os
         self._moe_scratch: dict = {}   # dedup-GEMV persistent buffers
         self.moe_dedup = _os.environ.get('ROOMAMD_MOE_DEDUP') == '1'
+        # fused decode kernels (attn_merge_o / router_addnorm); env kill
+        # switch kept for A/B and fallback
+        self.fused_decode = _os.environ.get('ROOMAMD_NO_FUSED_DECODE') != '1'
         cos_t, sin_t = rope_tables(cfg.max_position, cfg.head_dim, cfg.rope_theta)
         self.cos_t = cos_t.to(self.device)
         self.sin_t = sin_t.to(self.device)
@@ -137,14 +140,24 @@ class Qwen3MoEModel:
         self._arangeP = torch.arange(T * K, device=dev, dtype=torch.int32)
         if decode:
             qkv = torch.empty(T, qdim + 2 * kvdim, dtype=torch.bfloat16, device=dev)
-            obuf = torch.empty(T, cfg.hidden_size, dtype=torch.bfloat16, device=dev)
-            x_alt = torch.empty_like(x)     # residual ping-pong for gemv_addnorm
+            x_alt = torch.empty_like(x)     # residual ping-pong for fused norms
             empty_delta = torch.empty(0, dtype=torch.float32, device=dev)
             nsp = ops.attn_nsplits()
             part = torch.empty(T, cfg.num_q_heads, nsp, cfg.head_dim,
                                dtype=torch.float32, device=dev)
             part_ml = torch.empty(T, cfg.num_q_heads, nsp, 2, dtype=torch.float32,
                                   device=dev)
+            # fused decode path (attn_merge_o + router_addnorm): collapses
+            # merge+O-GEMV and add-norm+router-partial into single kernels
+            fused = self.fused_decode
+            if fused:
+                o_accum = torch.empty(T, cfg.hidden_size, dtype=torch.float32,
+                                      device=dev)
+                rlogits = torch.empty(T, cfg.num_experts, dtype=torch.float32,
+                                      device=dev)
+            else:
+                obuf = torch.empty(T, cfg.hidden_size, dtype=torch.bfloat16,
+                                   device=dev)
 
         for li, layer in enumerate(self.layers):
             # --- attention block
@@ -182,6 +195,23 @@ class Qwen3MoEModel:
                                  layer.q_norm_w, layer.k_norm_w, self.cos_t,
                                  self.sin_t, block_table, seq_ids, q_pos,
                                  cfg.rms_eps)
+            if decode and fused:
+                # split partials (+zero the O accumulator as a side job), then
+                # one kernel merges per-head and projects through Wo with f32
+                # atomics — no bf16 attn round-trip, two launches fewer
+                ops.paged_attention_splitk(part, part_ml, q, kcaches[li],
+                                           vcaches[li], block_table, seq_ids,
+                                           q_pos, self.scale, o_accum)
+                ops.attn_merge_o(o_accum, part, part_ml, layer.wo)
+                # --- MoE block (fused): add+norm+router logits in one kernel
+                ops.router_addnorm(rlogits, x, o_accum, x_alt, hbuf,
+                                   layer.post_attn_norm_w, layer.router_w,
+                                   cfg.rms_eps)
+                x, x_alt = x_alt, x
+                topk_ids, topk_w = ops.moe_router(rlogits,
+                                                  cfg.num_experts_per_tok)
+                moe_out = self._moe(hbuf, layer, topk_ids, topk_w)
+                continue
             attn = torch.empty(T, cfg.num_q_heads, cfg.head_dim,
                                dtype=torch.bfloat16, device=dev)
             if decode:

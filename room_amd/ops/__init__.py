@@ -106,6 +106,38 @@ def gemv(y: torch.Tensor, x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return y
 
 
+def paged_attention_splitk(part: torch.Tensor, part_ml: torch.Tensor,
+                           q: torch.Tensor, kcache: torch.Tensor,
+                           vcache: torch.Tensor, block_table: torch.Tensor,
+                           seq_ids: torch.Tensor, q_pos: torch.Tensor,
+                           scale: float, o_zero: torch.Tensor) -> None:
+    """Split-KV decode attention, partials only (no bf16 merge output); zeros
+    `o_zero` (the attn_merge_o f32 accumulator) as a side job."""
+    _require().paged_attention_splitk(part, part_ml, q, kcache, vcache,
+                                      block_table, seq_ids, q_pos, scale,
+                                      o_zero)
+
+
+def attn_merge_o(o_accum: torch.Tensor, part: torch.Tensor,
+                 part_ml: torch.Tensor, wo: torch.Tensor) -> torch.Tensor:
+    """Fused split-partial merge + O-projection: o_accum[B, H] (f32,
+    pre-zeroed) += Wo @ merged-attention. Replaces the standalone merge
+    kernel + O GEMV on the decode path."""
+    _require().attn_merge_o(o_accum, part, part_ml, wo)
+    return o_accum
+
+
+def router_addnorm(y: torch.Tensor, x: torch.Tensor, delta: torch.Tensor,
+                   x_out: torch.Tensor, xn_out: torch.Tensor,
+                   gamma: torch.Tensor, w: torch.Tensor,
+                   eps: float = 1e-6) -> torch.Tensor:
+    """x_out = x + delta (f32); xn_out = rmsnorm(x_out)·γ; y = xn @ w^T
+    (f32 router logits). One kernel replaces fused_add_rmsnorm + the H-split
+    router partial GEMV on the decode path. x_out must not alias x."""
+    _require().router_addnorm(y, x, delta, x_out, xn_out, gamma, w, eps)
+    return y
+
+
 def flash_prefill(out: torch.Tensor, q: torch.Tensor, kcache: torch.Tensor,
                   vcache: torch.Tensor, block_table: torch.Tensor,
                   seq_ids: torch.Tensor, q_pos: torch.Tensor,

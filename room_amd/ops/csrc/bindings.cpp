@@ -55,6 +55,16 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
                            torch::Tensor q_pos, torch::Tensor part,
                            torch::Tensor part_ml, double scale);
 int64_t attn_nsplits();
+void paged_attention_splitk(torch::Tensor part, torch::Tensor part_ml,
+                            torch::Tensor q, torch::Tensor kcache,
+                            torch::Tensor vcache, torch::Tensor block_table,
+                            torch::Tensor seq_ids, torch::Tensor q_pos,
+                            double scale, torch::Tensor o_zero);
+void attn_merge_o(torch::Tensor o_accum, torch::Tensor part,
+                  torch::Tensor part_ml, torch::Tensor wo);
+void router_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
+                    torch::Tensor x_out, torch::Tensor xn_out,
+                    torch::Tensor gamma, torch::Tensor w, double eps);
 void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                    torch::Tensor vcache, torch::Tensor block_table,
                    torch::Tensor seq_ids, torch::Tensor q_pos,
@@ -99,6 +109,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_combine_gather", &moe_combine_gather, "atomics-free MoE combine");
   m.def("sample_tokens", &sample_tokens, "fused temperature/top-k/top-p sampling");
   m.def("attn_nsplits", &attn_nsplits, "NSPLITS constant");
+  m.def("paged_attention_splitk", &paged_attention_splitk,
+        "split-KV decode attention, partials only (+O-accum zero side-job)");
+  m.def("attn_merge_o", &attn_merge_o,
+        "fused split-partial merge + O-projection (f32 atomic accum)");
+  m.def("router_addnorm", &router_addnorm,
+        "fused residual-add + RMSNorm + router logits (+residual/xn out)");
   m.def("paged_attention_split", &paged_attention_split,
         "split-KV flash-decode paged attention");
   m.def("flash_prefill", &flash_prefill, "MFMA flash-attention prefill");

@@ -1,0 +1,262 @@
+// Decode-path fusion kernels (round-2: collapse the ~11 dependent launches
+// per layer; each removed kernel removes a launch + grid fill/drain bubble —
+// profiles/PERF_NOTES.md puts the per-layer bubble cost at 30-40 µs).
+//
+// 1) attn_merge_o: split-KV partial merge + O-projection in ONE kernel.
+//    The standalone merge (10 µs) writes a bf16 [T,Hq,D] buffer that the
+//    O-GEMV (9.4 µs) immediately re-reads. Here each workgroup owns one
+//    (kv-merge of one q-head) × (one 128-wide output tile of Wo): it merges
+//    the head's NSPLITS partials into LDS (f32, no bf16 hop), stages its Wo
+//    tile through LDS (each Wo byte is read exactly once across the grid),
+//    and atomically accumulates the per-head contribution into the f32
+//    output accumulator — the same f32-delta the fused add-norm consumes.
+//
+// 2) router_addnorm: residual-add + RMSNorm + router logits in ONE kernel,
+//    also emitting the updated residual and the normed activations (the MoE
+//    GEMV input). Replaces fused_add_rmsnorm + H-split router partial GEMV.
+//    N=128 outputs → 32 workgroups; each does the (cheap, L2-hot) norm pass
+//    into LDS and four 2048-wide dots per wave.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include "common.h"
+
+#define HEAD_DIM 128
+#define NSPLITS 32          // must match paged_attn.hip
+#define AMO_NT 128          // output-tile width of attn_merge_o
+
+// ------------------------------------------------------------ attn_merge_o
+// grid: (N / AMO_NT, Hq); block 256.
+// o_accum[b, n] += sum_d Wo[n, hq*128+d] * merged[b, hq, d]
+template <int BN>
+__global__ __launch_bounds__(256)
+void attn_merge_o_kernel(float* __restrict__ o_accum,      // [BN, N] (pre-zeroed)
+                         const float* __restrict__ part,    // [BN, Hq, NS, D]
+                         const float* __restrict__ part_ml, // [BN, Hq, NS, 2]
+                         const short* __restrict__ wo,      // [N, Hq*D]
+                         int n_qheads, int N) {
+  const int n0 = blockIdx.x * AMO_NT;
+  const int hq = blockIdx.y;
+  const int tid = threadIdx.x;
+
+  __shared__ float merged[BN][HEAD_DIM + 4];
+  // +2 bf16 row pad → 260 B row stride = 65 banks: consecutive rows land on
+  // consecutive banks, so the 32 lanes reading distinct rows never conflict
+  __shared__ short w_s[AMO_NT][HEAD_DIM + 2];
+
+  // ---- phase A: merge this head's NSPLITS partials for every batch row.
+  // thread (bslot = tid>>5, d4 = (tid&31)*4) handles a f32x4 of dims;
+  // 8 bslots cover BN<=8.
+  {
+    const int bslot = tid >> 5;
+    const int d4 = (tid & 31) * 4;
+    if (bslot < BN) {
+      const int b = bslot;
+      const float* ml = part_ml + (((long)b * n_qheads + hq) * NSPLITS) * 2;
+      const float* pacc = part + (((long)b * n_qheads + hq) * NSPLITS) * HEAD_DIM;
+      float m_star = -INFINITY;
+      #pragma unroll
+      for (int s = 0; s < NSPLITS; ++s) m_star = fmaxf(m_star, ml[2 * s]);
+      float l_tot = 0.f;
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      #pragma unroll 4
+      for (int s = 0; s < NSPLITS; ++s) {
+        const float ms = ml[2 * s];
+        if (ms == -INFINITY) continue;
+        const float f = __expf(ms - m_star);
+        l_tot += ml[2 * s + 1] * f;
+        const f32x4 v = *reinterpret_cast<const f32x4*>(
+            pacc + (long)s * HEAD_DIM + d4);
+        acc[0] += v[0] * f; acc[1] += v[1] * f;
+        acc[2] += v[2] * f; acc[3] += v[3] * f;
+      }
+      const float inv = (l_tot > 0.f) ? 1.0f / l_tot : 0.f;
+      merged[b][d4 + 0] = acc[0] * inv;
+      merged[b][d4 + 1] = acc[1] * inv;
+      merged[b][d4 + 2] = acc[2] * inv;
+      merged[b][d4 + 3] = acc[3] * inv;
+    }
+  }
+
+  // ---- stage the Wo tile: rows n0..n0+NT, k-slice [hq*128, hq*128+128).
+  // thread pair (r = tid>>1, half = tid&1) loads 128 B of row r.
+  {
+    const int r = tid >> 1;
+    const int half = (tid & 1) * 64;              // 64 bf16 = 128 B
+    const short* wrow = wo + ((long)(n0 + r) * n_qheads + hq) * HEAD_DIM + half;
+    #pragma unroll
+    for (int v8 = 0; v8 < 8; ++v8)
+      *reinterpret_cast<bf16x8*>(&w_s[r][half + v8 * 8]) =
+          *reinterpret_cast<const bf16x8*>(wrow + v8 * 8);
+  }
+  __syncthreads();
+
+  // ---- phase B: per-head O contribution. wave w covers n-local
+  // [w*32, w*32+32); lane l: n_local = w*32 + (l&31), b parity = l>>5.
+  const int wv = tid >> 6;
+  const int lane = tid & 63;
+  const int nl = wv * 32 + (lane & 31);
+  const int bpar = lane >> 5;                     // 0: even b, 1: odd b
+  float acc[(BN + 1) / 2];
+  #pragma unroll
+  for (int i = 0; i < (BN + 1) / 2; ++i) acc[i] = 0.f;
+  #pragma unroll 4
+  for (int d = 0; d < HEAD_DIM; ++d) {
+    const float wval = bf2f(w_s[nl][d]);
+    #pragma unroll
+    for (int i = 0; i < (BN + 1) / 2; ++i) {
+      const int b = 2 * i + bpar;
+      if (b < BN) acc[i] += wval * merged[b][d];
+    }
+  }
+  #pragma unroll
+  for (int i = 0; i < (BN + 1) / 2; ++i) {
+    const int b = 2 * i + bpar;
+    if (b < BN) atomicAdd(&o_accum[(long)b * N + n0 + nl], acc[i]);
+  }
+}
+
+void attn_merge_o(torch::Tensor o_accum, torch::Tensor part,
+                  torch::Tensor part_ml, torch::Tensor wo) {
+  const int B = o_accum.size(0), N = o_accum.size(1);
+  const int n_qheads = part.size(1);
+  TORCH_CHECK(B >= 1 && B <= 8, "attn_merge_o handles B<=8 (decode)");
+  TORCH_CHECK(o_accum.dtype() == torch::kFloat32 && part.dtype() == torch::kFloat32);
+  TORCH_CHECK(wo.dtype() == torch::kBFloat16);
+  TORCH_CHECK(part.size(2) == NSPLITS && part.size(3) == HEAD_DIM);
+  TORCH_CHECK(N % AMO_NT == 0);
+  TORCH_CHECK(wo.size(0) == N && wo.size(1) == n_qheads * HEAD_DIM);
+  dim3 grid(N / AMO_NT, n_qheads), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  switch (B) {
+#define AMO_CASE(BN) \
+    case BN: hipLaunchKernelGGL((attn_merge_o_kernel<BN>), grid, block, 0, s, \
+        o_accum.data_ptr<float>(), part.data_ptr<float>(), \
+        part_ml.data_ptr<float>(), (const short*)wo.data_ptr(), \
+        n_qheads, N); break;
+    AMO_CASE(1) AMO_CASE(2) AMO_CASE(3) AMO_CASE(4)
+    AMO_CASE(5) AMO_CASE(6) AMO_CASE(7) AMO_CASE(8)
+#undef AMO_CASE
+  }
+  HIP_CHECK_KERNEL();
+}
+
+// ------------------------------------------------------------ router_addnorm
+// x_out = x + delta (f32 attn accumulator, bf16-rounded like
+// fused_add_rmsnorm); xn = rmsnorm(x_out)·γ; y = xn @ Wr^T (f32 logits).
+// Writes x_out and xn once (block 0); every block stages xn in LDS for its
+// four router dots. grid: N/4 (=32 for E=128); block 256.
+template <int BN>
+__global__ __launch_bounds__(256)
+void router_addnorm_kernel(float* __restrict__ y,          // [BN, N]
+                           const short* __restrict__ x,     // [BN, H]
+                           const float* __restrict__ delta, // [BN, H]
+                           short* __restrict__ x_out,       // [BN, H]
+                           short* __restrict__ xn_out,      // [BN, H]
+                           const short* __restrict__ gamma, // [H]
+                           const short* __restrict__ w,     // [N, H]
+                           int H, int N, float eps) {
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int n = blockIdx.x * 4 + wid;
+
+  extern __shared__ short xn_sh[];                // [BN][H] bf16
+  __shared__ float ss_sh[BN];
+  if (tid < BN) ss_sh[tid] = 0.f;
+  __syncthreads();
+
+  // pass A: xsum = bf16(x + delta) staged to LDS; accumulate sum-of-squares
+  #pragma unroll
+  for (int b = 0; b < BN; ++b) {
+    float ss = 0.f;
+    for (int base = tid * 8; base < H; base += 256 * 8) {
+      bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
+      bf16x8 o;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = bf2f(xv[j]) + delta[(long)b * H + base + j];
+        o[j] = f2bf(v);
+        v = bf2f(o[j]);                            // norm over the rounded sum
+        ss += v * v;
+      }
+      *reinterpret_cast<bf16x8*>(&xn_sh[b * H + base]) = o;
+    }
+    ss = wave_reduce_sum(ss);
+    if (lane == 0) atomicAdd(&ss_sh[b], ss);
+  }
+  __syncthreads();
+  float scale[BN];
+  #pragma unroll
+  for (int b = 0; b < BN; ++b) scale[b] = rsqrtf(ss_sh[b] / H + eps);
+
+  // pass B: block 0 persists the residual; all blocks scale LDS in place
+  // (x_out write must read xsum before the in-place overwrite → same loop)
+  #pragma unroll
+  for (int b = 0; b < BN; ++b) {
+    for (int base = tid * 8; base < H; base += 256 * 8) {
+      bf16x8 sv = *reinterpret_cast<const bf16x8*>(&xn_sh[b * H + base]);
+      if (blockIdx.x == 0)
+        *reinterpret_cast<bf16x8*>(x_out + (long)b * H + base) = sv;
+      bf16x8 gv = *reinterpret_cast<const bf16x8*>(gamma + base);
+      bf16x8 o;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = f2bf(bf2f(sv[j]) * scale[b] * bf2f(gv[j]));
+      *reinterpret_cast<bf16x8*>(&xn_sh[b * H + base]) = o;
+      if (blockIdx.x == 0)
+        *reinterpret_cast<bf16x8*>(xn_out + (long)b * H + base) = o;
+    }
+  }
+  __syncthreads();
+
+  // pass C: four 2048-wide dots per block out of LDS
+  if (n >= N) return;
+  float acc[BN];
+  #pragma unroll
+  for (int b = 0; b < BN; ++b) acc[b] = 0.f;
+  const short* wrow = w + (long)n * H;
+  for (int base = lane * 8; base < H; base += WAVE * 8) {
+    bf16x8 wv = *reinterpret_cast<const bf16x8*>(wrow + base);
+    float wf[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) wf[j] = bf2f(wv[j]);
+    #pragma unroll
+    for (int b = 0; b < BN; ++b) {
+      bf16x8 xv = *reinterpret_cast<const bf16x8*>(&xn_sh[b * H + base]);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) acc[b] += wf[j] * bf2f(xv[j]);
+    }
+  }
+  #pragma unroll
+  for (int b = 0; b < BN; ++b) {
+    const float r = wave_reduce_sum(acc[b]);
+    if (lane == 0) y[(long)b * N + n] = r;
+  }
+}
+
+void router_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
+                    torch::Tensor x_out, torch::Tensor xn_out,
+                    torch::Tensor gamma, torch::Tensor w, double eps) {
+  const int B = x.size(0), H = x.size(1), N = w.size(0);
+  TORCH_CHECK(B >= 1 && B <= 8, "router_addnorm handles B<=8 (decode)");
+  TORCH_CHECK(x.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16);
+  TORCH_CHECK(delta.dtype() == torch::kFloat32 && delta.is_contiguous());
+  TORCH_CHECK(y.dtype() == torch::kFloat32);
+  TORCH_CHECK(H % (WAVE * 8) == 0);
+  const size_t lds = (size_t)B * H * sizeof(short);
+  TORCH_CHECK(lds <= 160 * 1024, "B*H too large for LDS staging");
+  dim3 grid((N + 3) / 4), block(256);
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  switch (B) {
+#define RAN_CASE(BN) \
+    case BN: hipLaunchKernelGGL((router_addnorm_kernel<BN>), grid, block, lds, s, \
+        y.data_ptr<float>(), (const short*)x.data_ptr(), \
+        delta.data_ptr<float>(), (short*)x_out.data_ptr(), \
+        (short*)xn_out.data_ptr(), (const short*)gamma.data_ptr(), \
+        (const short*)w.data_ptr(), H, N, (float)eps); break;
+    RAN_CASE(1) RAN_CASE(2) RAN_CASE(3) RAN_CASE(4)
+    RAN_CASE(5) RAN_CASE(6) RAN_CASE(7) RAN_CASE(8)
+#undef RAN_CASE
+  }
+  HIP_CHECK_KERNEL();
+}

@@ -216,11 +216,21 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
                              const int* __restrict__ seq_ids,
                              const int* __restrict__ q_pos,
                              int n_kvheads, int max_blocks, int q_row_stride,
-                             float scale) {
+                             float scale,
+                             float* __restrict__ o_zero, long zero_n) {
   const int t = blockIdx.x;
   const int hk = blockIdx.y;
   const int split = blockIdx.z;
   const int tid = threadIdx.x;
+  // side job: zero the fused attn_merge_o accumulator (runs right before it
+  // in stream order; the grid has ~160k threads vs ~10k floats to clear, so
+  // the separate fill launch is absorbed for free — same pattern as
+  // moe_gemv_h's out_zero)
+  if (o_zero != nullptr) {
+    const long gid = (((long)blockIdx.z * gridDim.y + blockIdx.y) * gridDim.x
+                      + blockIdx.x) * blockDim.x + tid;
+    if (gid < zero_n) o_zero[gid] = 0.f;
+  }
   const int h = tid >> 5;
   const int sub = tid & 31;
   const int seq = seq_ids[t];
@@ -380,11 +390,37 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
                      (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
                      (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
                      seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
-                     n_kvheads, max_blocks, (int)q.stride(0), (float)scale);
+                     n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
+                     (float*)nullptr, 0L);
   HIP_CHECK_KERNEL();
   dim3 g2(T, n_qheads);
   hipLaunchKernelGGL(paged_attn_merge_kernel, g2, dim3(HEAD_DIM), 0, s,
                      (short*)out.data_ptr(), part.data_ptr<float>(),
                      part_ml.data_ptr<float>(), n_qheads);
+  HIP_CHECK_KERNEL();
+}
+
+// split-only variant for the fused attn_merge_o path: no bf16 merge output;
+// zeroes the downstream f32 O accumulator as a side job.
+void paged_attention_splitk(torch::Tensor part, torch::Tensor part_ml,
+                            torch::Tensor q, torch::Tensor kcache,
+                            torch::Tensor vcache, torch::Tensor block_table,
+                            torch::Tensor seq_ids, torch::Tensor q_pos,
+                            double scale, torch::Tensor o_zero) {
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
+  const int T = q.size(0);
+  const int n_kvheads = kcache.size(1);
+  const int max_blocks = block_table.size(1);
+  TORCH_CHECK(part.size(0) >= T && part.size(2) == NSPLITS);
+  TORCH_CHECK(o_zero.dtype() == torch::kFloat32 && o_zero.is_contiguous());
+  hipStream_t s = c10::hip::getCurrentHIPStream();
+  dim3 g1(T, n_kvheads, NSPLITS);
+  hipLaunchKernelGGL(paged_attn_split_kernel, g1, dim3(256), 0, s,
+                     part.data_ptr<float>(), part_ml.data_ptr<float>(),
+                     (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
+                     (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
+                     seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
+                     n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
+                     o_zero.data_ptr<float>(), (long)o_zero.numel());
   HIP_CHECK_KERNEL();
 }

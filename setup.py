@@ -21,6 +21,7 @@ SOURCES = [
     "room_amd/ops/csrc/paged_attn.hip",
     "room_amd/ops/csrc/moe.hip",
     "room_amd/ops/csrc/gemv.hip",
+    "room_amd/ops/csrc/fused_decode.hip",
     "room_amd/ops/csrc/flash_prefill.hip",
     "room_amd/ops/csrc/sampling.hip",
     "room_amd/ops/csrc/vector_store.hip",

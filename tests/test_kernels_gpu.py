@@ -455,3 +455,67 @@ def test_router_gemv_topk_split():
             mr = {int(i): float(v) for i, v in zip(ids_ref[t], w_ref[t])}
             for e in m:
                 assert m[e] == pytest.approx(mr[e], abs=5e-3)
+
+
+@pytest.mark.parametrize("B", [1, 5, 8])
+def test_attn_merge_o_matches_composition(B):
+    """Fused split-merge + O-projection vs the two-kernel path
+    (paged_attention_split's merge + gemv), which is itself ref-validated."""
+    torch.manual_seed(23)
+    Hq, Hk, D = 32, 4, 128
+    H = 2048
+    lens = [1 + 97 * i for i in range(B)]
+    kcache, vcache, bt = _setup_cache(B, 1024)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+    wo = torch.randn(H, Hq * D, dtype=torch.bfloat16, device=DEV) * 0.05
+    seq_ids = torch.arange(B, dtype=torch.int32, device=DEV)
+    q_pos = torch.tensor([l - 1 for l in lens], dtype=torch.int32, device=DEV)
+    nsp = ops.attn_nsplits()
+    part = torch.empty(B, Hq, nsp, D, dtype=torch.float32, device=DEV)
+    part_ml = torch.empty(B, Hq, nsp, 2, dtype=torch.float32, device=DEV)
+    scale = D ** -0.5
+    # fused path
+    o_accum = torch.full((B, H), 777.0, dtype=torch.float32, device=DEV)
+    ops.paged_attention_splitk(part, part_ml, q, kcache, vcache, bt,
+                               seq_ids, q_pos, scale, o_accum)
+    ops.attn_merge_o(o_accum, part, part_ml, wo)
+    # two-kernel reference path
+    attn = torch.empty_like(q)
+    ops.paged_attention_split(attn, q, kcache, vcache, bt, seq_ids, q_pos,
+                              part, part_ml, scale)
+    o_ref = torch.empty(B, H, dtype=torch.float32, device=DEV)
+    ops.gemv(o_ref, attn.reshape(B, Hq * D), wo)
+    assert bf16_close(o_accum, o_ref, atol=6e-2, rtol=6e-2)
+    # fp32 torch reference end-to-end
+    merged = ref.paged_attention_ref(q, kcache, vcache, bt, seq_ids, q_pos,
+                                     scale)
+    o_t = merged.reshape(B, Hq * D).float() @ wo.float().T
+    assert bf16_close(o_accum, o_t, atol=0.25, rtol=6e-2)
+
+
+@pytest.mark.parametrize("B,E", [(1, 128), (5, 128), (8, 16)])
+def test_router_addnorm_matches_composition(B, E):
+    torch.manual_seed(24)
+    H = 2048 if E == 128 else 512
+    x = torch.randn(B, H, dtype=torch.bfloat16, device=DEV)
+    delta = torch.randn(B, H, dtype=torch.float32, device=DEV) * 0.1
+    gamma = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(E, H, dtype=torch.bfloat16, device=DEV) * 0.05
+    x_out = torch.empty_like(x)
+    xn_out = torch.empty_like(x)
+    y = torch.empty(B, E, dtype=torch.float32, device=DEV)
+    ops.router_addnorm(y, x, delta, x_out, xn_out, gamma, w, 1e-6)
+    # composition reference: fused_add_rmsnorm then f32 gemv
+    res = x.clone()
+    hb = torch.empty_like(x)
+    ops.fused_add_rmsnorm(hb, res, delta, gamma, 1e-6)
+    assert bf16_close(x_out, res)
+    assert bf16_close(xn_out, hb)
+    y_ref = torch.empty(B, E, dtype=torch.float32, device=DEV)
+    ops.gemv(y_ref, hb, w)
+    assert bf16_close(y, y_ref, atol=6e-2, rtol=6e-2)
+    # fp32 torch reference
+    s = (x.float() + delta).to(torch.bfloat16).float()
+    xn_t = s * torch.rsqrt(s.pow(2).mean(-1, keepdim=True) + 1e-6) * gamma.float()
+    y_t = xn_t @ w.float().T
+    assert bf16_close(y, y_t, atol=0.3, rtol=6e-2)
