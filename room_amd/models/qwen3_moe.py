@@ -96,9 +96,12 @@ class Qwen3MoEModel:
         import os as _os
         self._moe_scratch: dict = {}   # dedup-GEMV persistent buffers
         self.moe_dedup = _os.environ.get('ROOMAMD_MOE_DEDUP') == '1'
-        # fused decode kernels (attn_merge_o / router_addnorm); env kill
-        # switch kept for A/B and fallback
-        self.fused_decode = _os.environ.get('ROOMAMD_NO_FUSED_DECODE') != '1'
+        # fused decode kernels (attn_merge_o / router_addnorm); separate env
+        # kill switches for A/B and fallback
+        nofuse = _os.environ.get('ROOMAMD_NO_FUSED_DECODE') == '1'
+        self.fuse_o = not (nofuse or _os.environ.get('ROOMAMD_NO_FUSE_O') == '1')
+        self.fuse_router = not (nofuse
+                                or _os.environ.get('ROOMAMD_NO_FUSE_ROUTER') == '1')
         cos_t, sin_t = rope_tables(cfg.max_position, cfg.head_dim, cfg.rope_theta)
         self.cos_t = cos_t.to(self.device)
         self.sin_t = sin_t.to(self.device)
@@ -147,17 +150,18 @@ class Qwen3MoEModel:
                                dtype=torch.float32, device=dev)
             part_ml = torch.empty(T, cfg.num_q_heads, nsp, 2, dtype=torch.float32,
                                   device=dev)
-            # fused decode path (attn_merge_o + router_addnorm): collapses
+            # fused decode path (attn_merge_o / router_addnorm): collapses
             # merge+O-GEMV and add-norm+router-partial into single kernels
-            fused = self.fused_decode
-            if fused:
+            fuse_o, fuse_router = self.fuse_o, self.fuse_router
+            if fuse_o:
                 o_accum = torch.empty(T, cfg.hidden_size, dtype=torch.float32,
-                                      device=dev)
-                rlogits = torch.empty(T, cfg.num_experts, dtype=torch.float32,
                                       device=dev)
             else:
                 obuf = torch.empty(T, cfg.hidden_size, dtype=torch.bfloat16,
                                    device=dev)
+            if fuse_router:
+                rlogits = torch.empty(T, cfg.num_experts, dtype=torch.float32,
+                                      device=dev)
 
         for li, layer in enumerate(self.layers):
             # --- attention block
@@ -195,32 +199,45 @@ class Qwen3MoEModel:
                                  layer.q_norm_w, layer.k_norm_w, self.cos_t,
                                  self.sin_t, block_table, seq_ids, q_pos,
                                  cfg.rms_eps)
-            if decode and fused:
-                # split partials (+zero the O accumulator as a side job), then
-                # one kernel merges per-head and projects through Wo with f32
-                # atomics — no bf16 attn round-trip, two launches fewer
-                ops.paged_attention_splitk(part, part_ml, q, kcaches[li],
-                                           vcaches[li], block_table, seq_ids,
-                                           q_pos, self.scale, o_accum)
-                ops.attn_merge_o(o_accum, part, part_ml, layer.wo)
-                # --- MoE block (fused): add+norm+router logits in one kernel
-                ops.router_addnorm(rlogits, x, o_accum, x_alt, hbuf,
-                                   layer.post_attn_norm_w, layer.router_w,
-                                   cfg.rms_eps)
-                x, x_alt = x_alt, x
-                topk_ids, topk_w = ops.moe_router(rlogits,
-                                                  cfg.num_experts_per_tok)
+            if decode:
+                if fuse_o:
+                    # split partials (+zero the O accumulator as a side job),
+                    # then one kernel merges per-head and projects through Wo
+                    # with f32 atomics — no bf16 attn round-trip
+                    ops.paged_attention_splitk(part, part_ml, q, kcaches[li],
+                                               vcaches[li], block_table,
+                                               seq_ids, q_pos, self.scale,
+                                               o_accum)
+                    ops.attn_merge_o(o_accum, part, part_ml, layer.wo)
+                    o = o_accum                   # f32 delta
+                else:
+                    attn = torch.empty(T, cfg.num_q_heads, cfg.head_dim,
+                                       dtype=torch.bfloat16, device=dev)
+                    ops.paged_attention_split(attn, q, kcaches[li],
+                                              vcaches[li], block_table,
+                                              seq_ids, q_pos, part, part_ml,
+                                              self.scale)
+                    ops.gemv(obuf, attn.reshape(T, qdim), layer.wo)
+                    o = obuf
+                # --- MoE block
+                if fuse_router:
+                    # add+norm+router logits in one kernel
+                    ops.router_addnorm(rlogits, x, o, x_alt, hbuf,
+                                       layer.post_attn_norm_w, layer.router_w,
+                                       cfg.rms_eps)
+                    x, x_alt = x_alt, x
+                    topk_ids, topk_w = ops.moe_router(rlogits,
+                                                      cfg.num_experts_per_tok)
+                else:
+                    ops.fused_add_rmsnorm(hbuf, x, o, layer.post_attn_norm_w,
+                                          cfg.rms_eps)
+                    topk_ids, topk_w = ops.router_gemv_topk(
+                        hbuf, layer.router_w, cfg.num_experts_per_tok)
                 moe_out = self._moe(hbuf, layer, topk_ids, topk_w)
                 continue
             attn = torch.empty(T, cfg.num_q_heads, cfg.head_dim,
                                dtype=torch.bfloat16, device=dev)
-            if decode:
-                ops.paged_attention_split(attn, q, kcaches[li], vcaches[li],
-                                          block_table, seq_ids, q_pos, part,
-                                          part_ml, self.scale)
-                ops.gemv(obuf, attn.reshape(T, qdim), layer.wo)
-                o = obuf
-            elif qtile_desc is not None:
+            if qtile_desc is not None:
                 ops.flash_prefill(attn, q, kcaches[li], vcaches[li], block_table,
                                   seq_ids, q_pos, qtile_desc, self.scale)
                 o = F.linear(attn.reshape(T, qdim), layer.wo)
@@ -229,20 +246,11 @@ class Qwen3MoEModel:
                                     seq_ids, q_pos, self.scale)
                 o = F.linear(attn.reshape(T, qdim), layer.wo)
 
-            # --- MoE block
+            # --- MoE block (prefill; decode handled above)
             ops.fused_add_rmsnorm(hbuf, x, o, layer.post_attn_norm_w, cfg.rms_eps)
-            # NOTE: a single-kernel fused router (ops.router_topk) was measured
-            # SLOWER here (decode 8.5 -> 11.0 ms/step at B=5): one workgroup
-            # per token serializes the 512 KB router-weight read that the
-            # dense GEMV spreads over 128 waves. Two kernels win.
-            if decode:
-                # H-split partial GEMV (4x grid) + top-k summing the partials
-                topk_ids, topk_w = ops.router_gemv_topk(hbuf, layer.router_w,
-                                                        cfg.num_experts_per_tok)
-            else:
-                router_logits = F.linear(hbuf, layer.router_w).float()
-                topk_ids, topk_w = ops.moe_router(router_logits,
-                                                  cfg.num_experts_per_tok)
+            router_logits = F.linear(hbuf, layer.router_w).float()
+            topk_ids, topk_w = ops.moe_router(router_logits,
+                                              cfg.num_experts_per_tok)
             # f32 accumulator feeds fused_add_rmsnorm directly (templated
             # input dtype — skips a cast kernel per layer)
             moe_out = self._moe(hbuf, layer, topk_ids, topk_w)

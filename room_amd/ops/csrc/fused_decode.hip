@@ -145,11 +145,11 @@ void attn_merge_o(torch::Tensor o_accum, torch::Tensor part,
 // fused_add_rmsnorm); xn = rmsnorm(x_out)·γ; y = xn @ Wr^T (f32 logits).
 // Writes x_out and xn once (block 0); every block stages xn in LDS for its
 // four router dots. grid: N/4 (=32 for E=128); block 256.
-template <int BN>
+template <int BN, bool DF32>
 __global__ __launch_bounds__(256)
 void router_addnorm_kernel(float* __restrict__ y,          // [BN, N]
                            const short* __restrict__ x,     // [BN, H]
-                           const float* __restrict__ delta, // [BN, H]
+                           const void* __restrict__ delta_, // [BN, H] f32|bf16
                            short* __restrict__ x_out,       // [BN, H]
                            short* __restrict__ xn_out,      // [BN, H]
                            const short* __restrict__ gamma, // [H]
@@ -174,7 +174,9 @@ void router_addnorm_kernel(float* __restrict__ y,          // [BN, N]
       bf16x8 o;
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float v = bf2f(xv[j]) + delta[(long)b * H + base + j];
+        float d = DF32 ? ((const float*)delta_)[(long)b * H + base + j]
+                       : bf2f(((const short*)delta_)[(long)b * H + base + j]);
+        float v = bf2f(xv[j]) + d;
         o[j] = f2bf(v);
         v = bf2f(o[j]);                            // norm over the rounded sum
         ss += v * v;
@@ -240,23 +242,28 @@ void router_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
   const int B = x.size(0), H = x.size(1), N = w.size(0);
   TORCH_CHECK(B >= 1 && B <= 8, "router_addnorm handles B<=8 (decode)");
   TORCH_CHECK(x.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16);
-  TORCH_CHECK(delta.dtype() == torch::kFloat32 && delta.is_contiguous());
+  const bool df32 = delta.dtype() == torch::kFloat32;
+  TORCH_CHECK(df32 || delta.dtype() == torch::kBFloat16);
+  TORCH_CHECK(delta.is_contiguous());
   TORCH_CHECK(y.dtype() == torch::kFloat32);
   TORCH_CHECK(H % (WAVE * 8) == 0);
   const size_t lds = (size_t)B * H * sizeof(short);
   TORCH_CHECK(lds <= 160 * 1024, "B*H too large for LDS staging");
   dim3 grid((N + 3) / 4), block(256);
   hipStream_t s = c10::hip::getCurrentHIPStream();
-  switch (B) {
-#define RAN_CASE(BN) \
-    case BN: hipLaunchKernelGGL((router_addnorm_kernel<BN>), grid, block, lds, s, \
+  switch (B * 2 + (df32 ? 1 : 0)) {
+#define RAN_CASE1(BN, DF) \
+    case BN * 2 + (DF ? 1 : 0): \
+      hipLaunchKernelGGL((router_addnorm_kernel<BN, DF>), grid, block, lds, s, \
         y.data_ptr<float>(), (const short*)x.data_ptr(), \
-        delta.data_ptr<float>(), (short*)x_out.data_ptr(), \
+        delta.data_ptr(), (short*)x_out.data_ptr(), \
         (short*)xn_out.data_ptr(), (const short*)gamma.data_ptr(), \
         (const short*)w.data_ptr(), H, N, (float)eps); break;
+#define RAN_CASE(BN) RAN_CASE1(BN, true) RAN_CASE1(BN, false)
     RAN_CASE(1) RAN_CASE(2) RAN_CASE(3) RAN_CASE(4)
     RAN_CASE(5) RAN_CASE(6) RAN_CASE(7) RAN_CASE(8)
 #undef RAN_CASE
+#undef RAN_CASE1
   }
   HIP_CHECK_KERNEL();
 }
