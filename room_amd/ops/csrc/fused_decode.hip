@@ -44,49 +44,83 @@ void attn_merge_o_kernel(float* __restrict__ o_accum,      // [BN, N] (pre-zeroe
   __shared__ short w_s[AMO_NT][HEAD_DIM + 2];
 
   // ---- phase A: merge this head's NSPLITS partials for every batch row.
-  // thread (bslot = tid>>5, d4 = (tid&31)*4) handles a f32x4 of dims;
-  // 8 bslots cover BN<=8.
-  {
-    const int bslot = tid >> 5;
-    const int d4 = (tid & 31) * 4;
-    if (bslot < BN) {
-      const int b = bslot;
-      const float* ml = part_ml + (((long)b * n_qheads + hq) * NSPLITS) * 2;
-      const float* pacc = part + (((long)b * n_qheads + hq) * NSPLITS) * HEAD_DIM;
-      float m_star = -INFINITY;
-      #pragma unroll
-      for (int s = 0; s < NSPLITS; ++s) m_star = fmaxf(m_star, ml[2 * s]);
-      float l_tot = 0.f;
-      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-      #pragma unroll 4
-      for (int s = 0; s < NSPLITS; ++s) {
-        const float ms = ml[2 * s];
-        if (ms == -INFINITY) continue;
-        const float f = __expf(ms - m_star);
-        l_tot += ml[2 * s + 1] * f;
-        const f32x4 v = *reinterpret_cast<const f32x4*>(
-            pacc + (long)s * HEAD_DIM + d4);
-        acc[0] += v[0] * f; acc[1] += v[1] * f;
-        acc[2] += v[2] * f; acc[3] += v[3] * f;
-      }
-      const float inv = (l_tot > 0.f) ? 1.0f / l_tot : 0.f;
-      merged[b][d4 + 0] = acc[0] * inv;
-      merged[b][d4 + 1] = acc[1] * inv;
-      merged[b][d4 + 2] = acc[2] * inv;
-      merged[b][d4 + 3] = acc[3] * inv;
-    }
-  }
+  // All 256 threads cooperate per row: thread (sg = tid>>5, d4 = (tid&31)*4)
+  // covers split group sg (NSPLITS/8 = 4 splits) × one f32x4 of dims — the
+  // 4 loads per thread are independent (one latency window, vs a 32-long
+  // strided chain in the first version); the 8 split-group partials reduce
+  // through LDS.
+  __shared__ float red[8][32][4];
 
-  // ---- stage the Wo tile: rows n0..n0+NT, k-slice [hq*128, hq*128+128).
-  // thread pair (r = tid>>1, half = tid&1) loads 128 B of row r.
+  // issue the Wo tile loads FIRST (registers), so the HBM latency overlaps
+  // phase A's merge work; the LDS store + barrier happen after phase A
+  bf16x8 wreg[8];
   {
     const int r = tid >> 1;
     const int half = (tid & 1) * 64;              // 64 bf16 = 128 B
     const short* wrow = wo + ((long)(n0 + r) * n_qheads + hq) * HEAD_DIM + half;
     #pragma unroll
     for (int v8 = 0; v8 < 8; ++v8)
-      *reinterpret_cast<bf16x8*>(&w_s[r][half + v8 * 8]) =
-          *reinterpret_cast<const bf16x8*>(wrow + v8 * 8);
+      wreg[v8] = *reinterpret_cast<const bf16x8*>(wrow + v8 * 8);
+  }
+
+  {
+    const int sg = tid >> 5;                       // split group 0..7
+    const int d32 = tid & 31;
+    const int d4 = d32 * 4;
+    constexpr int SPG = NSPLITS / 8;               // splits per group
+    #pragma unroll
+    for (int b = 0; b < BN; ++b) {
+      const float* ml = part_ml + (((long)b * n_qheads + hq) * NSPLITS) * 2;
+      const float* pacc = part + (((long)b * n_qheads + hq) * NSPLITS) * HEAD_DIM;
+      float m_star = -INFINITY;
+      #pragma unroll
+      for (int s = 0; s < NSPLITS; ++s) m_star = fmaxf(m_star, ml[2 * s]);
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      #pragma unroll
+      for (int i = 0; i < SPG; ++i) {
+        const int s = sg * SPG + i;
+        const float ms = ml[2 * s];
+        if (ms == -INFINITY) continue;
+        const float f = __expf(ms - m_star);
+        const f32x4 v = *reinterpret_cast<const f32x4*>(
+            pacc + (long)s * HEAD_DIM + d4);
+        acc[0] += v[0] * f; acc[1] += v[1] * f;
+        acc[2] += v[2] * f; acc[3] += v[3] * f;
+      }
+      red[sg][d32][0] = acc[0]; red[sg][d32][1] = acc[1];
+      red[sg][d32][2] = acc[2]; red[sg][d32][3] = acc[3];
+      __syncthreads();
+      // l_tot: every thread recomputes the full sum from ml (L1-hot, cheap)
+      if (sg == 0) {
+        float lt = 0.f;
+        #pragma unroll
+        for (int s = 0; s < NSPLITS; ++s) {
+          const float ms = ml[2 * s];
+          if (ms != -INFINITY) lt += ml[2 * s + 1] * __expf(ms - m_star);
+        }
+        const float inv = (lt > 0.f) ? 1.0f / lt : 0.f;
+        f32x4 tot = {0.f, 0.f, 0.f, 0.f};
+        #pragma unroll
+        for (int g = 0; g < 8; ++g) {
+          tot[0] += red[g][d32][0]; tot[1] += red[g][d32][1];
+          tot[2] += red[g][d32][2]; tot[3] += red[g][d32][3];
+        }
+        merged[b][d4 + 0] = tot[0] * inv;
+        merged[b][d4 + 1] = tot[1] * inv;
+        merged[b][d4 + 2] = tot[2] * inv;
+        merged[b][d4 + 3] = tot[3] * inv;
+      }
+      __syncthreads();
+    }
+  }
+
+  // ---- stage the Wo tile from registers (loads issued before phase A)
+  {
+    const int r = tid >> 1;
+    const int half = (tid & 1) * 64;
+    #pragma unroll
+    for (int v8 = 0; v8 < 8; ++v8)
+      *reinterpret_cast<bf16x8*>(&w_s[r][half + v8 * 8]) = wreg[v8];
   }
   __syncthreads();
 
@@ -171,12 +205,24 @@ void router_addnorm_kernel(float* __restrict__ y,          // [BN, N]
     float ss = 0.f;
     for (int base = tid * 8; base < H; base += 256 * 8) {
       bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * H + base);
+      float dv[8];
+      if (DF32) {  // two f32x4 vector loads, not 8 scalar loads
+        const f32x4 d0 = *reinterpret_cast<const f32x4*>(
+            (const float*)delta_ + (long)b * H + base);
+        const f32x4 d1 = *reinterpret_cast<const f32x4*>(
+            (const float*)delta_ + (long)b * H + base + 4);
+        dv[0] = d0[0]; dv[1] = d0[1]; dv[2] = d0[2]; dv[3] = d0[3];
+        dv[4] = d1[0]; dv[5] = d1[1]; dv[6] = d1[2]; dv[7] = d1[3];
+      } else {
+        const bf16x8 db = *reinterpret_cast<const bf16x8*>(
+            (const short*)delta_ + (long)b * H + base);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) dv[j] = bf2f(db[j]);
+      }
       bf16x8 o;
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float d = DF32 ? ((const float*)delta_)[(long)b * H + base + j]
-                       : bf2f(((const short*)delta_)[(long)b * H + base + j]);
-        float v = bf2f(xv[j]) + d;
+        float v = bf2f(xv[j]) + dv[j];
         o[j] = f2bf(v);
         v = bf2f(o[j]);                            // norm over the rounded sum
         ss += v * v;
