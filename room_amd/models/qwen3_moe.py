@@ -96,12 +96,14 @@ class Qwen3MoEModel:
         import os as _os
         self._moe_scratch: dict = {}   # dedup-GEMV persistent buffers
         self.moe_dedup = _os.environ.get('ROOMAMD_MOE_DEDUP') == '1'
-        # fused decode kernels (attn_merge_o / router_addnorm); separate env
-        # kill switches for A/B and fallback
-        nofuse = _os.environ.get('ROOMAMD_NO_FUSED_DECODE') == '1'
-        self.fuse_o = not (nofuse or _os.environ.get('ROOMAMD_NO_FUSE_O') == '1')
-        self.fuse_router = not (nofuse
-                                or _os.environ.get('ROOMAMD_NO_FUSE_ROUTER') == '1')
+        # fused decode kernels (attn_merge_o / router_addnorm): measured
+        # SLOWER than the two-kernel paths at B=5 (profiles/PERF_NOTES.md
+        # round-2 negative results — attn_merge_o re-reads the split partials
+        # once per output tile, 16× redundant HBM traffic, 31.8 µs vs the
+        # 19.4 µs merge+GEMV pair; router_addnorm 10.9 vs 9.75 on a
+        # latency-starved 32-WG grid). Kept behind env for the record.
+        self.fuse_o = _os.environ.get('ROOMAMD_FUSE_O') == '1'
+        self.fuse_router = _os.environ.get('ROOMAMD_FUSE_ROUTER') == '1'
         cos_t, sin_t = rope_tables(cfg.max_position, cfg.head_dim, cfg.rope_theta)
         self.cos_t = cos_t.to(self.device)
         self.sin_t = sin_t.to(self.device)

@@ -156,6 +156,17 @@ __global__ void write_kv_kernel(short* __restrict__ kcache,
   }
 }
 
+// ROOMAMD_ATTN_CHUNK=64 selects the larger-chunk split kernel (fewer
+// barriers per span, more LDS); read once, process-stable so hipGraph
+// replay stays consistent.
+static int attn_chunk() {
+  static int c = [] {
+    const char* e = getenv("ROOMAMD_ATTN_CHUNK");
+    return (e && atoi(e) == 64) ? 64 : 32;
+  }();
+  return c;
+}
+
 // ---------------------------------------------------------------- wrappers
 
 void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
@@ -206,6 +217,9 @@ void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
 // dispatch + 2x merge traffic. Static 32 keeps the decode graph replayable.
 #define NSPLITS 32
 
+// CH: kv positions staged/scored per iteration. 64 halves the barrier count
+// per span (3 barriers per chunk) at the cost of LDS (41 KB vs 24 KB).
+template <int CH>
 __global__ __launch_bounds__(256)
 void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
                              float* __restrict__ part_ml, // [T, Hq, NSPLITS, 2]
@@ -238,8 +252,8 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
   const int n_qheads = n_kvheads * QH_PER_KV;
   const int hq = hk * QH_PER_KV + h;
 
-  // split range, CHUNK-aligned
-  const int span = ((bound + NSPLITS - 1) / NSPLITS + CHUNK - 1) & ~(CHUNK - 1);
+  // split range, CH-aligned
+  const int span = ((bound + NSPLITS - 1) / NSPLITS + CH - 1) & ~(CH - 1);
   const int lo = split * span;
   const int hi = min(bound, lo + span);
 
@@ -260,9 +274,9 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
   // row reads off a single bank (same layout as flash_prefill).
   #define PA_PAD 8
   __shared__ float q_s[QH_PER_KV][HEAD_DIM];
-  __shared__ float p_s[QH_PER_KV][CHUNK];
-  __shared__ short k_s[CHUNK][HEAD_DIM + PA_PAD];
-  __shared__ short v_s[CHUNK][HEAD_DIM + PA_PAD];
+  __shared__ float p_s[QH_PER_KV][CH];
+  __shared__ short k_s[CH][HEAD_DIM + PA_PAD];
+  __shared__ short v_s[CH][HEAD_DIM + PA_PAD];
   for (int i = tid; i < QH_PER_KV * HEAD_DIM; i += blockDim.x) {
     int hh = i / HEAD_DIM, dd = i % HEAD_DIM;
     q_s[hh][dd] = bf2f(q[(long)t * q_row_stride
@@ -274,12 +288,12 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
   const long kv_stride_block = (long)n_kvheads * BLOCK_SIZE * HEAD_DIM;
   const int* btab = block_table + (long)seq * max_blocks;
 
-  for (int base = lo; base < hi; base += CHUNK) {
-    // ---- stage K and V [32][128] chunks: 512 vec8 each, 2 per thread
+  for (int base = lo; base < hi; base += CH) {
+    // ---- stage K and V [CH][128] chunks: CH*16 vec8 each
     #pragma unroll
-    for (int it = 0; it < 2; ++it) {
-      const int idx = tid + it * 256;          // vec8 index 0..511
-      const int pos_l = (idx * 8) / HEAD_DIM;  // 0..31
+    for (int it = 0; it < CH / 16; ++it) {
+      const int idx = tid + it * 256;          // vec8 index 0..CH*16-1
+      const int pos_l = (idx * 8) / HEAD_DIM;  // 0..CH-1
       const int d8 = (idx * 8) % HEAD_DIM;
       const int pos = base + pos_l;
       bf16x8 kv = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -296,31 +310,35 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
     }
     __syncthreads();
 
-    {   // score: thread (h, sub) dots q_s[h] with k_s[sub]
-      float s = -INFINITY;
-      if (base + sub < hi) {
-        float dot = 0.f;
-        #pragma unroll
-        for (int v8 = 0; v8 < HEAD_DIM / 8; ++v8) {
-          bf16x8 kv = *reinterpret_cast<const bf16x8*>(&k_s[sub][v8 * 8]);
+    {   // score: thread (h, sub) dots q_s[h] with k_s[sub + r*32]
+      #pragma unroll
+      for (int r = 0; r < CH / 32; ++r) {
+        const int p = sub + r * 32;
+        float s = -INFINITY;
+        if (base + p < hi) {
+          float dot = 0.f;
           #pragma unroll
-          for (int j = 0; j < 8; ++j) dot += q_s[h][v8 * 8 + j] * bf2f(kv[j]);
+          for (int v8 = 0; v8 < HEAD_DIM / 8; ++v8) {
+            bf16x8 kv = *reinterpret_cast<const bf16x8*>(&k_s[p][v8 * 8]);
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) dot += q_s[h][v8 * 8 + j] * bf2f(kv[j]);
+          }
+          s = dot * scale;
         }
-        s = dot * scale;
+        p_s[h][p] = s;
       }
-      p_s[h][sub] = s;
     }
     __syncthreads();
     {
       float chunk_max = -INFINITY;
       #pragma unroll
-      for (int j = 0; j < CHUNK; ++j) chunk_max = fmaxf(chunk_max, p_s[h][j]);
+      for (int j = 0; j < CH; ++j) chunk_max = fmaxf(chunk_max, p_s[h][j]);
       const float m_new = fmaxf(m_run, chunk_max);
       if (m_new != -INFINITY) {
         const float rescale = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
         acc[0] *= rescale; acc[1] *= rescale; acc[2] *= rescale; acc[3] *= rescale;
         l_run *= rescale;
-        const int lim = min(CHUNK, hi - base);
+        const int lim = min(CH, hi - base);
         for (int j = 0; j < lim; ++j) {
           const float w = __expf(p_s[h][j] - m_new);
           l_run += w;
@@ -385,13 +403,22 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(part.size(0) >= T && part.size(2) == NSPLITS);
   hipStream_t s = c10::hip::getCurrentHIPStream();
   dim3 g1(T, n_kvheads, NSPLITS);
-  hipLaunchKernelGGL(paged_attn_split_kernel, g1, dim3(256), 0, s,
-                     part.data_ptr<float>(), part_ml.data_ptr<float>(),
-                     (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
-                     (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
-                     seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
-                     n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
-                     (float*)nullptr, 0L);
+  if (attn_chunk() == 64)
+    hipLaunchKernelGGL(paged_attn_split_kernel<64>, g1, dim3(256), 0, s,
+                       part.data_ptr<float>(), part_ml.data_ptr<float>(),
+                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
+                       (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
+                       seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
+                       n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
+                       (float*)nullptr, 0L);
+  else
+    hipLaunchKernelGGL(paged_attn_split_kernel<32>, g1, dim3(256), 0, s,
+                       part.data_ptr<float>(), part_ml.data_ptr<float>(),
+                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
+                       (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
+                       seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
+                       n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
+                       (float*)nullptr, 0L);
   HIP_CHECK_KERNEL();
   dim3 g2(T, n_qheads);
   hipLaunchKernelGGL(paged_attn_merge_kernel, g2, dim3(HEAD_DIM), 0, s,
@@ -415,12 +442,21 @@ void paged_attention_splitk(torch::Tensor part, torch::Tensor part_ml,
   TORCH_CHECK(o_zero.dtype() == torch::kFloat32 && o_zero.is_contiguous());
   hipStream_t s = c10::hip::getCurrentHIPStream();
   dim3 g1(T, n_kvheads, NSPLITS);
-  hipLaunchKernelGGL(paged_attn_split_kernel, g1, dim3(256), 0, s,
-                     part.data_ptr<float>(), part_ml.data_ptr<float>(),
-                     (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
-                     (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
-                     seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
-                     n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
-                     o_zero.data_ptr<float>(), (long)o_zero.numel());
+  if (attn_chunk() == 64)
+    hipLaunchKernelGGL(paged_attn_split_kernel<64>, g1, dim3(256), 0, s,
+                       part.data_ptr<float>(), part_ml.data_ptr<float>(),
+                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
+                       (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
+                       seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
+                       n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
+                       o_zero.data_ptr<float>(), (long)o_zero.numel());
+  else
+    hipLaunchKernelGGL(paged_attn_split_kernel<32>, g1, dim3(256), 0, s,
+                       part.data_ptr<float>(), part_ml.data_ptr<float>(),
+                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
+                       (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
+                       seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
+                       n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
+                       o_zero.data_ptr<float>(), (long)o_zero.numel());
   HIP_CHECK_KERNEL();
 }
