@@ -151,12 +151,15 @@ def flash_prefill(out: torch.Tensor, q: torch.Tensor, kcache: torch.Tensor,
 
 def build_qtile_desc(segments: list, device) -> torch.Tensor:
     """Q-tile descriptors from host-known prefill segments [(row0, count), ...]
-    (one sequence per segment) — no device sync."""
+    (one sequence per segment) — no device sync. Tile rows match the
+    flash_prefill kernel's FP_QTOK."""
+    qt = int(_C.flash_prefill_qtile()) if (
+        _C is not None and hasattr(_C, "flash_prefill_qtile")) else 32
     desc = []
     for row0, count in segments:
         r = row0
         while r < row0 + count:
-            n = min(16, row0 + count - r)
+            n = min(qt, row0 + count - r)
             desc.append((r, n))
             r += n
     if not desc:

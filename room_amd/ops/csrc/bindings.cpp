@@ -65,6 +65,7 @@ void attn_merge_o(torch::Tensor o_accum, torch::Tensor part,
 void router_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
                     torch::Tensor x_out, torch::Tensor xn_out,
                     torch::Tensor gamma, torch::Tensor w, double eps);
+int64_t flash_prefill_qtile();
 void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                    torch::Tensor vcache, torch::Tensor block_table,
                    torch::Tensor seq_ids, torch::Tensor q_pos,
@@ -118,6 +119,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_split", &paged_attention_split,
         "split-KV flash-decode paged attention");
   m.def("flash_prefill", &flash_prefill, "MFMA flash-attention prefill");
+  m.def("flash_prefill_qtile", &flash_prefill_qtile, "q-tile rows constant");
   m.def("gemv", &gemv, "dense skinny-batch GEMV (decode projections)");
   m.def("gemv_addnorm", &gemv_addnorm,
         "fused residual-add + RMSNorm + GEMV (decode)");

@@ -1,17 +1,19 @@
 // MFMA flash-attention prefill over the paged KV cache.
 //
 // Replaces the vector-ALU prefill path (paged_attn_kernel) which profiled at
-// 62% of prefill time (3.06 ms/layer @ T=2048 — scripts/gpu_prefill_attrib.py):
-// per-thread 128-dim dots re-read K 8× per head group and the value phase
-// serialized on L1 latency. This kernel is flash-style: Q-tile × KV-tile with
-// online softmax, QK^T and PV on mfma_f32_16x16x32_bf16, K/V staged through
-// LDS once per chunk.
+// 62% of prefill time (3.06 ms/layer @ T=2048 — scripts/gpu_prefill_attrib.py).
+// Flash-style: Q-tile × KV-chunk with online softmax, QK^T and PV on
+// mfma_f32_16x16x32_bf16, K/V staged through LDS once per chunk.
 //
-// Geometry: one workgroup = (q-tile of ≤16 tokens of ONE sequence) × (kv head).
-//   512 threads = 8 waves; wave w = q-head w of the GQA group.
-//   M = 16 tokens (wave-local), N = 32 kv positions/chunk, D = 128.
-//   QK^T: C[16,32] = Q[16,128]·K^T — 2 n-frags × 4 k-steps = 8 MFMA/chunk.
-//   PV:   O[16,128] += P[16,32]·V — 8 n-frags × 1 k-step = 8 MFMA/chunk.
+// Round-2 geometry (v1 was M=16/N=32 and measured MFMA:VALU 1:16 — the
+// softmax bookkeeping drowned the matrix pipe): one workgroup =
+// (q-tile of ≤32 tokens of ONE sequence) × (kv head); 512 threads = 8 waves;
+// wave w = q-head w of the GQA group.
+//   M = 32 tokens (2 row-fragments), N = 64 kv positions/chunk (4 col-frags),
+//   D = 128. Per chunk per wave: QK^T 2m×4n×4k = 32 MFMA, PV 2m×8n×2k = 32.
+//   4× the MFMA work per staged chunk of v1, while the cross-lane max/sum
+//   reductions stay one pair per row per chunk (amortized over 4 col-frags)
+//   and the 3 block barriers cover 2048 scores instead of 512.
 // C-fragment mapping (verified): row=(lane>>4)*4+r, col=lane&15.
 // LDS rows padded so b128 reads across rows are ≤2-way bank conflicts (free).
 #include <torch/extension.h>
@@ -19,8 +21,8 @@
 #include "common.h"
 
 #define FP_BS 16          // paged KV block size
-#define FP_QTOK 16        // q tokens per tile
-#define FP_CHUNK 32       // kv positions per chunk
+#define FP_QTOK 32        // q tokens per tile (2 MFMA row-blocks)
+#define FP_CHUNK 64       // kv positions per chunk (4 MFMA col-blocks)
 #define FP_D 128
 #define FP_QH 8           // GQA group
 #define FP_PADK 8         // bf16 pad for K/V/P LDS rows
@@ -48,16 +50,16 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
   const int n_qheads = n_kvheads * FP_QH;
   const int hq = hk * FP_QH + wid;
 
-  __shared__ short q_s[FP_QH][FP_QTOK][FP_D + FP_PADK];
-  __shared__ short k_s[FP_CHUNK][FP_D + FP_PADK];
-  __shared__ short vt_s[FP_D][FP_CHUNK + FP_PADK];
-  __shared__ short p_s[FP_QH][FP_QTOK][FP_CHUNK + FP_PADK];
+  __shared__ short q_s[FP_QH][FP_QTOK][FP_D + FP_PADK];       // 69.6 KB
+  __shared__ short k_s[FP_CHUNK][FP_D + FP_PADK];             // 17.4 KB
+  __shared__ short vt_s[FP_D][FP_CHUNK + FP_PADK];            // 18.4 KB
+  __shared__ short p_s[FP_QH][FP_QTOK][FP_CHUNK + FP_PADK];   // 36.9 KB
   __shared__ int qp_s[FP_QTOK];
   __shared__ int seq_s;
 
-  // ---- load Q tile: 8h × 16tok × 128d = 2048 vec8 → 4 per thread
+  // ---- load Q tile: 8h × 32tok × 128d = 4096 vec8 → 8 per thread
   #pragma unroll
-  for (int it = 0; it < 4; ++it) {
+  for (int it = 0; it < 8; ++it) {
     const int idx = tid + it * 512;           // vec8 index
     const int h = idx / (FP_QTOK * FP_D / 8);
     const int rem = idx % (FP_QTOK * FP_D / 8);
@@ -80,19 +82,24 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
   const int* btab = block_table + (long)seq * max_blocks;
   const long kv_stride_block = (long)n_kvheads * FP_BS * FP_D;
 
-  // per-lane row stats: rows (lane>>4)*4 + r
-  float m_run[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
-  float l_run[4] = {0.f, 0.f, 0.f, 0.f};
-  fpfrag_t oacc[8];
+  // per-lane row stats: m-tile mt rows mt*16 + (lane>>4)*4 + r
+  float m_run[2][4], l_run[2][4];
+  fpfrag_t oacc[2][8];
   #pragma unroll
-  for (int nf = 0; nf < 8; ++nf) oacc[nf] = fpfrag_t{0.f, 0.f, 0.f, 0.f};
+  for (int mt = 0; mt < 2; ++mt) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) { m_run[mt][r] = -INFINITY; l_run[mt][r] = 0.f; }
+    #pragma unroll
+    for (int nf = 0; nf < 8; ++nf) oacc[mt][nf] = fpfrag_t{0.f, 0.f, 0.f, 0.f};
+  }
 
   for (int base = 0; base < bound_max; base += FP_CHUNK) {
-    // ---- stage K chunk [32][128] and V^T [128][32] cooperatively
-    // 32×128/8 = 512 vec8: one per thread
-    {
-      const int pos_l = (tid * 8) / FP_D;      // 0..31
-      const int d8 = (tid * 8) % FP_D;
+    // ---- stage K chunk [64][128] and V^T [128][64]: 1024 vec8 → 2/thread
+    #pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = tid + it * 512;
+      const int pos_l = (idx * 8) / FP_D;      // 0..63
+      const int d8 = (idx * 8) % FP_D;
       const int pos = base + pos_l;
       bf16x8 kv = {0, 0, 0, 0, 0, 0, 0, 0};
       bf16x8 vv = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -109,82 +116,88 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
     }
     __syncthreads();
 
-    // ---- QK^T: C[16,32] per wave
-    fpfrag_t sfrag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+    // ---- QK^T: C[32,64] per wave = 2 m-tiles × 4 n-tiles × 4 k-steps
+    fpfrag_t sfrag[2][4];
+    #pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+      #pragma unroll
+      for (int nt = 0; nt < 4; ++nt)
+        sfrag[mt][nt] = fpfrag_t{0.f, 0.f, 0.f, 0.f};
     #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
       const int akoff = kk * 32 + (lane >> 4) * 8;
-      bf16x8 a = *reinterpret_cast<const bf16x8*>(&q_s[wid][lane & 15][akoff]);
+      bf16x8 a0 = *reinterpret_cast<const bf16x8*>(&q_s[wid][lane & 15][akoff]);
+      bf16x8 a1 = *reinterpret_cast<const bf16x8*>(
+          &q_s[wid][16 + (lane & 15)][akoff]);
       #pragma unroll
-      for (int nf = 0; nf < 2; ++nf) {
+      for (int nt = 0; nt < 4; ++nt) {
         bf16x8 b = *reinterpret_cast<const bf16x8*>(
-            &k_s[nf * 16 + (lane & 15)][akoff]);
-        sfrag[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, sfrag[nf], 0, 0, 0);
+            &k_s[nt * 16 + (lane & 15)][akoff]);
+        sfrag[0][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, sfrag[0][nt], 0, 0, 0);
+        sfrag[1][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, sfrag[1][nt], 0, 0, 0);
       }
     }
 
-    // ---- causal mask + online softmax (per row r: token (lane>>4)*4+r)
-    float p_vals[2][4];
-    float rescale[4];
+    // ---- causal mask + online softmax; one max/sum lane-reduce pair per
+    // row per chunk, amortized over the 4 col-frags
+    const int colL = lane & 15;
     #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int tk = (lane >> 4) * 4 + r;
-      const int bound = qp_s[tk] + 1;  // -1+1=0 for pad rows → all masked
-      float s0 = sfrag[0][r] * scale;
-      float s1 = sfrag[1][r] * scale;
-      const int pos0 = base + (lane & 15);
-      const int pos1 = pos0 + 16;
-      if (pos0 >= bound) s0 = -INFINITY;
-      if (pos1 >= bound) s1 = -INFINITY;
-      // row max across the 16-lane group (cols)
-      float mx = fmaxf(s0, s1);
-      #pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
-      const float m_new = fmaxf(m_run[r], mx);
-      float p0 = 0.f, p1 = 0.f, rs = 1.f;
-      if (m_new != -INFINITY) {
-        rs = (m_run[r] == -INFINITY) ? 1.f : __expf(m_run[r] - m_new);
-        p0 = (s0 == -INFINITY) ? 0.f : __expf(s0 - m_new);
-        p1 = (s1 == -INFINITY) ? 0.f : __expf(s1 - m_new);
-        float lsum = p0 + p1;
-        #pragma unroll
-        for (int off = 8; off > 0; off >>= 1)
-          lsum += __shfl_xor(lsum, off, WAVE);
-        l_run[r] = l_run[r] * rs + lsum;
-        m_run[r] = m_new;
-      }
-      rescale[r] = rs;
-      p_vals[0][r] = p0;
-      p_vals[1][r] = p1;
-    }
-
-    // ---- write P to LDS (transpose C-layout → A-layout)
-    #pragma unroll
-    for (int nf = 0; nf < 2; ++nf) {
+    for (int mt = 0; mt < 2; ++mt) {
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int tk = (lane >> 4) * 4 + r;
-        p_s[wid][tk][nf * 16 + (lane & 15)] = f2bf(p_vals[nf][r]);
+        const int tk = mt * 16 + (lane >> 4) * 4 + r;
+        const int bound = qp_s[tk] + 1;  // -1+1=0 for pad rows → all masked
+        float sv[4];
+        float mx = -INFINITY;
+        #pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          float s = sfrag[mt][nt][r] * scale;
+          if (base + nt * 16 + colL >= bound) s = -INFINITY;
+          sv[nt] = s;
+          mx = fmaxf(mx, s);
+        }
+        #pragma unroll
+        for (int off = 8; off > 0; off >>= 1)
+          mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+        const float m_new = fmaxf(m_run[mt][r], mx);
+        float rs = 1.f, lsum = 0.f;
+        float pv[4] = {0.f, 0.f, 0.f, 0.f};
+        if (m_new != -INFINITY) {
+          rs = (m_run[mt][r] == -INFINITY) ? 1.f : __expf(m_run[mt][r] - m_new);
+          #pragma unroll
+          for (int nt = 0; nt < 4; ++nt) {
+            pv[nt] = (sv[nt] == -INFINITY) ? 0.f : __expf(sv[nt] - m_new);
+            lsum += pv[nt];
+          }
+          #pragma unroll
+          for (int off = 8; off > 0; off >>= 1)
+            lsum += __shfl_xor(lsum, off, WAVE);
+          l_run[mt][r] = l_run[mt][r] * rs + lsum;
+          m_run[mt][r] = m_new;
+        }
+        // write P row to LDS (C-layout → A-layout) + rescale O row
+        #pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+          p_s[wid][tk][nt * 16 + colL] = f2bf(pv[nt]);
+        #pragma unroll
+        for (int nf = 0; nf < 8; ++nf) oacc[mt][nf][r] *= rs;
       }
-    }
-    // rescale O accumulators (row r factor applies to oacc[*][r])
-    #pragma unroll
-    for (int nf = 0; nf < 8; ++nf) {
-      #pragma unroll
-      for (int r = 0; r < 4; ++r) oacc[nf][r] *= rescale[r];
     }
     __syncthreads();  // P + Vt visible
 
-    // ---- PV: O[16,128] += P[16,32] · V[32,128]
-    {
-      bf16x8 a = *reinterpret_cast<const bf16x8*>(
-          &p_s[wid][lane & 15][(lane >> 4) * 8]);
+    // ---- PV: O[32,128] += P[32,64] · V[64,128]
+    #pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int akoff = kk * 32 + (lane >> 4) * 8;
+      bf16x8 a0 = *reinterpret_cast<const bf16x8*>(&p_s[wid][lane & 15][akoff]);
+      bf16x8 a1 = *reinterpret_cast<const bf16x8*>(
+          &p_s[wid][16 + (lane & 15)][akoff]);
       #pragma unroll
       for (int nf = 0; nf < 8; ++nf) {
         bf16x8 b = *reinterpret_cast<const bf16x8*>(
-            &vt_s[nf * 16 + (lane & 15)][(lane >> 4) * 8]);
-        oacc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, oacc[nf], 0, 0, 0);
+            &vt_s[nf * 16 + (lane & 15)][akoff]);
+        oacc[0][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, oacc[0][nf], 0, 0, 0);
+        oacc[1][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, oacc[1][nf], 0, 0, 0);
       }
     }
     __syncthreads();  // before restaging K/V
@@ -192,14 +205,17 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
 
   // ---- epilogue: normalize and store
   #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int tk = (lane >> 4) * 4 + r;
-    if (tk >= ntok) continue;
-    const float inv_l = (l_run[r] > 0.f) ? 1.0f / l_run[r] : 0.f;
+  for (int mt = 0; mt < 2; ++mt) {
     #pragma unroll
-    for (int nf = 0; nf < 8; ++nf) {
-      out[((long)(row0 + tk) * n_qheads + hq) * FP_D + nf * 16 + (lane & 15)] =
-          f2bf(oacc[nf][r] * inv_l);
+    for (int r = 0; r < 4; ++r) {
+      const int tk = mt * 16 + (lane >> 4) * 4 + r;
+      if (tk >= ntok) continue;
+      const float inv_l = (l_run[mt][r] > 0.f) ? 1.0f / l_run[mt][r] : 0.f;
+      #pragma unroll
+      for (int nf = 0; nf < 8; ++nf) {
+        out[((long)(row0 + tk) * n_qheads + hq) * FP_D + nf * 16 + (lane & 15)] =
+            f2bf(oacc[mt][nf][r] * inv_l);
+      }
     }
   }
 }
@@ -226,3 +242,5 @@ void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                      n_kvheads, max_blocks, (int)q.stride(0), (float)scale);
   HIP_CHECK_KERNEL();
 }
+
+int64_t flash_prefill_qtile() { return FP_QTOK; }
