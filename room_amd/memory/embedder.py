@@ -27,8 +27,33 @@ def _features(text: str) -> list[str]:
     return feats
 
 
+def _mode() -> str:
+    """Embedder selection: 'encoder' (MiniLM-shaped transformer, encoder.py)
+    or 'hash' (n-gram projection). Default: encoder on GPU hosts (the GPU
+    runs the forward; SURVEY §2b calls for a GPU encoder), hash on CPU-only
+    hosts (the 6-layer fp32 forward is too slow per-call on one core).
+    Override with ROOMAMD_EMBEDDER=encoder|hash."""
+    import os
+    m = os.environ.get("ROOMAMD_EMBEDDER")
+    if m in ("encoder", "hash"):
+        return m
+    try:
+        import torch
+        return "encoder" if torch.cuda.is_available() else "hash"
+    except Exception:
+        return "hash"
+
+
 def embed(text: str) -> list[float]:
-    """Deterministic 384-dim L2-normalized embedding."""
+    """384-dim L2-normalized embedding (deterministic per host mode)."""
+    if _mode() == "encoder":
+        from . import encoder
+        return encoder.encode_texts([text])[0]
+    return embed_hash(text)
+
+
+def embed_hash(text: str) -> list[float]:
+    """Deterministic hashed n-gram projection (CPU fallback / baseline)."""
     vec = [0.0] * EMBEDDING_DIM
     for f in _features(text):
         h = hashlib.blake2s(f.encode(), digest_size=8).digest()
@@ -43,7 +68,10 @@ def embed(text: str) -> list[float]:
 
 
 def embed_batch(texts: list[str]) -> list[list[float]]:
-    return [embed(t) for t in texts]
+    if _mode() == "encoder":
+        from . import encoder
+        return encoder.encode_texts(texts)
+    return [embed_hash(t) for t in texts]
 
 
 def cosine_similarity(a: list[float], b: list[float]) -> float:
