@@ -526,26 +526,47 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     # --------------------------------------------------------------- memory
 
     @app.get("/api/memory/search")
-    async def memory_search(query: str, room_id: Optional[int] = None,
+    async def memory_search(query: Optional[str] = None,
+                            q_param: Optional[str] = Query(None, alias="q"),
+                            room_id: Optional[int] = None,
+                            roomId: Optional[int] = None,
                             limit: int = 5, role: str = Depends(get_role)):
+        # the SPA client sends ?q= (client.ts:319); agent tools send ?query=
+        text = query or q_param
+        if not text:
+            raise HTTPException(400, "q is required")
+        rid = room_id if room_id is not None else roomId
         if memory is not None:
-            return memory.recall(room_id, query, limit=limit)
+            return memory.recall(rid, text, limit=limit)
         with ldb as db:
-            return q.hybrid_search(db, query, None, limit=limit, room_id=room_id)
+            return q.hybrid_search(db, text, None, limit=limit, room_id=rid)
 
     @app.post("/api/memory/entities")
     async def remember(payload: dict = Body(...), role: str = Depends(rw)):
-        if memory is not None:
-            eid = memory.remember(payload.get("room_id"), payload["name"],
-                                  payload["content"],
+        """Accepts both the SPA client shape ({name, type, category, roomId},
+        client.ts:313 — content optional) and the agent-tool shape
+        ({name, content, room_id}). Returns the full entity row plus the
+        entity_id alias older callers read."""
+        room_id = payload.get("room_id", payload.get("roomId"))
+        name = payload.get("name")
+        if not name:
+            raise HTTPException(400, "name is required")
+        content = payload.get("content")
+        obs = payload.get("observations") or ([content] if content else [])
+        if memory is not None and content:
+            eid = memory.remember(room_id, name, content,
                                   category=payload.get("category"))
-            return {"entity_id": eid}
+        else:
+            with ldb as db:
+                ent = q.create_entity(db, name,
+                                      entity_type=payload.get("type", "fact"),
+                                      category=payload.get("category"),
+                                      room_id=room_id, observations=obs)
+                eid = ent["id"]
         with ldb as db:
-            ent = q.create_entity(db, payload["name"],
-                                  category=payload.get("category"),
-                                  room_id=payload.get("room_id"),
-                                  observations=[payload["content"]])
-        return {"entity_id": ent["id"]}
+            row = dict(q.get_entity(db, eid))
+        row["entity_id"] = eid
+        return row
 
     @app.get("/api/memory/entities/{entity_id}")
     async def get_entity(entity_id: int, role: str = Depends(get_role)):
@@ -706,11 +727,18 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/rooms/{room_id}/messages")
     async def send_message(room_id: int, payload: dict = Body(...),
                            role: str = Depends(rw)):
+        """SPA client shape {toRoomId, body, subject} (client.ts:770-771 →
+        outbound, room-messages.ts validation) or agent shape
+        {direction, body, subject, from_room_id}."""
+        body_text = str(payload.get("body") or payload.get("content") or "").strip()
+        if not body_text:
+            raise HTTPException(400, "body is required")
+        direction = payload.get("direction",
+                                "outbound" if payload.get("toRoomId") else "inbound")
         with ldb as db:
-            m = q.create_room_message(db, room_id,
-                                      payload.get("direction", "inbound"),
-                                      payload.get("subject", ""),
-                                      payload["body"],
+            m = q.create_room_message(db, room_id, direction,
+                                      payload.get("subject") or "(no subject)",
+                                      body_text,
                                       from_room_id=payload.get("from_room_id"))
         bus.emit(f"room:{room_id}", "message", {"id": m["id"]})
         return m
@@ -795,6 +823,12 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
                               "hbm_total_gb": round(total_b / 2**30, 1)}
         except Exception:
             pass
+        # update availability (updateChecker.ts surfaces this in /api/status)
+        uc = getattr(state, "update_checker", None)
+        if uc is None:
+            from ..core.update_checker import UpdateChecker
+            uc = state.update_checker = UpdateChecker(out["version"])
+        out["update"] = uc.status()
         return out
 
     @app.get("/api/templates")
@@ -1350,6 +1384,335 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
             return {"code": q.get_setting(db, "keeper_referral_code")}
 
     # -------------------------------------------------------------- UI stub
+
+    # -------------------------------------------- UI client-contract routes
+    # Completes the endpoint set the reference SPA's typed client calls
+    # (docs/ui_client_contract.json, extracted from src/ui/lib/client.ts).
+
+    @app.get("/api/credentials/{cred_id}")
+    async def get_credential_by_id(cred_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            row = db.execute(
+                "SELECT id, room_id, name, type, provided_by, created_at"
+                " FROM credentials WHERE id = ?", (cred_id,)).fetchone()
+        if row is None:
+            raise HTTPException(404, "credential not found")
+        return row
+
+    @app.delete("/api/credentials/{cred_id}")
+    async def delete_credential_by_id(cred_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            db.execute("DELETE FROM credentials WHERE id = ?", (cred_id,))
+        return {"ok": True}
+
+    @app.post("/api/rooms/{room_id}/credentials/validate")
+    async def validate_credential(room_id: int, payload: dict = Body(...),
+                                  role: str = Depends(rw)):
+        """Shape check only — there is no egress to hit provider APIs
+        (reference validates against the provider; we validate format)."""
+        name = str(payload.get("name", ""))
+        value = str(payload.get("value", ""))
+        ok = bool(name) and len(value) >= 8 and not value.isspace()
+        return {"ok": ok,
+                "error": None if ok else "value too short to be a credential"}
+
+    @app.get("/api/memory/entities/{entity_id}/observations")
+    async def entity_observations(entity_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            if q.get_entity(db, entity_id) is None:
+                raise HTTPException(404, "entity not found")
+            return q.get_observations(db, entity_id)
+
+    @app.post("/api/memory/entities/{entity_id}/observations")
+    async def add_entity_observation(entity_id: int, payload: dict = Body(...),
+                                     role: str = Depends(rw)):
+        with ldb as db:
+            if q.get_entity(db, entity_id) is None:
+                raise HTTPException(404, "entity not found")
+            oid = q.add_observation(db, entity_id, payload["content"],
+                                    source=payload.get("source", "keeper"))
+            return {"id": oid, "entity_id": entity_id,
+                    "content": payload["content"]}
+
+    @app.get("/api/memory/entities/{entity_id}/relations")
+    async def entity_relations(entity_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            return db.execute(
+                "SELECT * FROM relations WHERE from_entity = ? OR to_entity = ?"
+                " ORDER BY id", (entity_id, entity_id)).fetchall()
+
+    @app.post("/api/rooms/{room_id}/messages/{message_id}/read")
+    async def mark_message_read(room_id: int, message_id: int,
+                                role: str = Depends(rw)):
+        with ldb as db:
+            q.mark_room_message_read(db, message_id)
+        return {"ok": True}
+
+    @app.post("/api/rooms/{room_id}/messages/read-all")
+    async def mark_all_messages_read(room_id: int, role: str = Depends(rw)):
+        with ldb as db:
+            cur = db.execute(
+                "UPDATE room_messages SET status = 'read' WHERE room_id = ?"
+                " AND status = 'unread'", (room_id,))
+        return {"ok": True, "marked": cur.rowcount}
+
+    @app.post("/api/escalations/{escalation_id}/resolve")
+    async def resolve_escalation(escalation_id: int, payload: dict = Body(...),
+                                 role: str = Depends(rw)):
+        """Reference routes/escalations.ts:40-63: answer + wake sender and
+        queen. Same semantics as our /answer alias."""
+        answer = payload.get("answer")
+        if not answer or not isinstance(answer, str):
+            raise HTTPException(400, "answer is required")
+        with ldb as db:
+            row = db.execute("SELECT * FROM escalations WHERE id = ?",
+                             (escalation_id,)).fetchone()
+            if row is None:
+                raise HTTPException(404, "Escalation not found")
+            q.answer_escalation(db, escalation_id, answer)
+            updated = db.execute("SELECT * FROM escalations WHERE id = ?",
+                                 (escalation_id,)).fetchone()
+            room = q.get_room(db, row["room_id"])
+        bus.emit(f"room:{row['room_id']}", "escalation:resolved", dict(updated))
+        if loop_mgr is not None:
+            if row["from_agent_id"]:
+                loop_mgr.trigger_agent(row["from_agent_id"])
+            if room and room.get("queen_worker_id") and \
+                    room["queen_worker_id"] != row["from_agent_id"]:
+                loop_mgr.trigger_agent(room["queen_worker_id"])
+        return updated
+
+    # ------------------------------------------------- wallet (UI contract)
+
+    @app.get("/api/rooms/{room_id}/wallet/summary")
+    async def wallet_summary(room_id: int, role: str = Depends(get_role)):
+        """Revenue summary (db-queries.ts:2210-2231 shape)."""
+        with ldb as db:
+            w = q.get_room_wallet(db, room_id)
+            if w is None:
+                return {"totalIncome": 0, "totalExpenses": 0, "netProfit": 0,
+                        "transactionCount": 0}
+            inc = db.execute(
+                "SELECT COALESCE(SUM(CAST(amount AS REAL)), 0) AS t FROM"
+                " wallet_transactions WHERE wallet_id = ? AND type IN"
+                " ('receive', 'fund')", (w["id"],)).fetchone()["t"]
+            exp = db.execute(
+                "SELECT COALESCE(SUM(CAST(amount AS REAL)), 0) AS t FROM"
+                " wallet_transactions WHERE wallet_id = ? AND type IN"
+                " ('send', 'purchase')", (w["id"],)).fetchone()["t"]
+            cnt = db.execute(
+                "SELECT COUNT(*) AS c FROM wallet_transactions WHERE"
+                " wallet_id = ?", (w["id"],)).fetchone()["c"]
+        return {"totalIncome": inc, "totalExpenses": exp,
+                "netProfit": inc - exp, "transactionCount": cnt}
+
+    @app.get("/api/rooms/{room_id}/wallet/balance")
+    async def wallet_balance(room_id: int, role: str = Depends(get_role)):
+        """On-chain balance. No egress here → the zero-balance shape the
+        reference returns when every chain RPC fails (wallet.ts:103-148)."""
+        import datetime as _dt
+        with ldb as db:
+            w = q.get_room_wallet(db, room_id)
+        if w is None:
+            return None
+        try:
+            with ldb as db:
+                bal = wallet_mod.get_on_chain_balance(db, room_id)
+        except Exception:
+            bal = None  # RPC unreachable (offline) → zero-balance shape
+        if bal and "totalBalance" in bal:
+            return bal
+        by_chain = ({bal["chain"]: bal["balance"]}
+                    if bal and bal.get("balance") is not None else {})
+        return {"totalBalance": sum(v for v in by_chain.values() if v) or 0,
+                "byChain": by_chain, "address": w["address"],
+                "fetchedAt": _dt.datetime.utcnow().isoformat() + "Z"}
+
+    @app.get("/api/rooms/{room_id}/wallet/onramp-url")
+    async def wallet_onramp_url(room_id: int, role: str = Depends(get_role)):
+        with ldb as db:
+            w = q.get_room_wallet(db, room_id)
+        if w is None:
+            raise HTTPException(400, "Room has no wallet")
+        raise HTTPException(503, "On-ramp unavailable")  # cloud-mediated; offline
+
+    @app.post("/api/rooms/{room_id}/wallet/withdraw")
+    async def wallet_withdraw(room_id: int, payload: dict = Body(...),
+                              role: str = Depends(rw)):
+        """Reference wallet.ts:162-230 validation, then the send path."""
+        import re as _re
+        to = str(payload.get("to", "")).strip()
+        amount = str(payload.get("amount", "")).strip()
+        if not to or not amount:
+            raise HTTPException(400, "Missing required fields: to, amount")
+        if not _re.fullmatch(r"0x[0-9a-fA-F]{40}", to):
+            raise HTTPException(400, "Invalid address")
+        try:
+            parsed = float(amount)
+        except ValueError:
+            raise HTTPException(400, "Invalid amount")
+        if not (parsed > 0):
+            raise HTTPException(400, "Invalid amount")
+        with ldb as db:
+            if q.get_room(db, room_id) is None:
+                raise HTTPException(404, "Room not found")
+            if q.get_room_wallet(db, room_id) is None:
+                raise HTTPException(400, "Room has no wallet")
+            try:
+                return wallet_mod.send_token(
+                    db, room_id, to, amount,
+                    chain=payload.get("chain", "base"),
+                    token=payload.get("token", "usdc"),
+                    description="keeper withdrawal")
+            except ValueError as e:
+                raise HTTPException(422, str(e))
+
+    # --------------------------------------------------- clerk (UI contract)
+
+    @app.post("/api/clerk/api-key")
+    async def clerk_api_key(payload: dict = Body(...), role: str = Depends(rw)):
+        provider = str(payload.get("provider", "")).strip()
+        key = str(payload.get("key", "")).strip()
+        if provider not in ("openai_api", "anthropic_api", "gemini_api"):
+            raise HTTPException(
+                400, "provider must be openai_api, anthropic_api, or gemini_api")
+        if not key:
+            raise HTTPException(400, "key is required")
+        with ldb as db:
+            q.set_setting(db, f"clerk_api_key_{provider}", encrypt_secret(key))
+        # no egress: stored without provider-side validation
+        return {"ok": True, "validated": False,
+                "note": "stored; provider validation requires network"}
+
+    @app.put("/api/clerk/settings")
+    async def clerk_settings(payload: dict = Body(...), role: str = Depends(rw)):
+        """clerk.ts:561-596: clerk model doubles as default queen model."""
+        with ldb as db:
+            if "model" in payload:
+                model = str(payload["model"])
+                q.set_setting(db, "clerk_model", model)
+                q.set_setting(db, "queen_model", model)
+            if "commentary" in payload:
+                q.set_setting(db, "clerk_commentary",
+                              "on" if payload["commentary"] else "off")
+            if "pace" in payload:
+                q.set_setting(db, "clerk_commentary_pace", str(payload["pace"]))
+        return {"ok": True}
+
+    # ------------------------------------------------------- update checker
+
+    @app.post("/api/status/check-update")
+    async def check_update(role: str = Depends(rw)):
+        uc = getattr(state, "update_checker", None)
+        if uc is None:
+            from ..core.update_checker import UpdateChecker
+            uc = state.update_checker = UpdateChecker("0.1.0")
+        return uc.check()
+
+    # ------------------------------------- provider / installer (UI contract)
+    # The reference manages external CLI/Ollama installs with streamed
+    # sessions (provider-install.ts, local-model.ts). This build's engine is
+    # in-process — these endpoints preserve the UI flow with sessions that
+    # complete immediately and a status that reports the built-in engine.
+
+    def _session_registry() -> dict:
+        reg = getattr(state, "provider_sessions", None)
+        if reg is None:
+            reg = state.provider_sessions = {"next_id": 1, "sessions": {}}
+        return reg
+
+    def _mk_session(kind: str, provider: str) -> dict:
+        reg = _session_registry()
+        sid = reg["next_id"]
+        reg["next_id"] += 1
+        sess = {"id": sid, "kind": kind, "provider": provider,
+                "status": "completed",
+                "lines": [{"id": 1, "stream": "system",
+                           "text": "in-process engine: nothing to install",
+                           "timestamp": time.strftime("%Y-%m-%dT%H:%M:%SZ")}],
+                "exitCode": 0}
+        reg["sessions"][sid] = sess
+        return sess
+
+    @app.post("/api/providers/{provider}/connect")
+    async def provider_connect(provider: str, role: str = Depends(rw)):
+        return {"session": _mk_session("connect", provider)}
+
+    @app.post("/api/providers/{provider}/disconnect")
+    async def provider_disconnect(provider: str, role: str = Depends(rw)):
+        return {"ok": True}
+
+    @app.post("/api/providers/{provider}/install")
+    async def provider_install(provider: str, role: str = Depends(rw)):
+        return {"session": _mk_session("install", provider)}
+
+    @app.get("/api/providers/{provider}/session")
+    async def provider_session(provider: str, role: str = Depends(get_role)):
+        reg = _session_registry()
+        for s in reversed(list(reg["sessions"].values())):
+            if s["provider"] == provider and s["kind"] == "connect":
+                return {"session": s}
+        return {"session": None}
+
+    @app.get("/api/providers/{provider}/install-session")
+    async def provider_install_session(provider: str,
+                                       role: str = Depends(get_role)):
+        reg = _session_registry()
+        for s in reversed(list(reg["sessions"].values())):
+            if s["provider"] == provider and s["kind"] == "install":
+                return {"session": s}
+        return {"session": None}
+
+    @app.get("/api/providers/sessions/{sid}")
+    async def provider_session_by_id(sid: int, role: str = Depends(get_role)):
+        s = _session_registry()["sessions"].get(sid)
+        if s is None:
+            raise HTTPException(404, "session not found")
+        return {"session": s}
+
+    @app.get("/api/providers/install-sessions/{sid}")
+    async def provider_install_session_by_id(sid: int,
+                                             role: str = Depends(get_role)):
+        return await provider_session_by_id(sid, role)  # type: ignore
+
+    @app.post("/api/providers/sessions/{sid}/cancel")
+    async def provider_session_cancel(sid: int, role: str = Depends(rw)):
+        s = _session_registry()["sessions"].get(sid)
+        if s is None:
+            raise HTTPException(404, "session not found")
+        if s["status"] in ("starting", "running"):
+            s["status"] = "canceled"
+        return {"session": s}
+
+    @app.post("/api/providers/install-sessions/{sid}/cancel")
+    async def provider_install_cancel(sid: int, role: str = Depends(rw)):
+        return await provider_session_cancel(sid, role)  # type: ignore
+
+    @app.post("/api/local-model/install")
+    async def local_model_install(role: str = Depends(rw)):
+        return {"started": False, "alreadyInstalled": True,
+                "session": _mk_session("install", "local")}
+
+    @app.get("/api/local-model/install-session")
+    async def local_model_install_session(role: str = Depends(get_role)):
+        reg = _session_registry()
+        for s in reversed(list(reg["sessions"].values())):
+            if s["provider"] == "local":
+                return {"session": s}
+        return {"session": None}
+
+    @app.post("/api/local-model/install-sessions/{sid}/cancel")
+    async def local_model_install_cancel(sid: int, role: str = Depends(rw)):
+        return await provider_session_cancel(sid, role)  # type: ignore
+
+    @app.post("/api/local-model/apply-all")
+    async def local_model_apply_all(role: str = Depends(rw)):
+        """Set every room's worker_model to the in-process engine default."""
+        with ldb as db:
+            rooms = q.list_rooms(db)
+            for r in rooms:
+                q.update_room(db, r["id"], worker_model="qwen3-coder-30b")
+        return {"applied": len(rooms), "model": "qwen3-coder-30b"}
 
     @app.get("/", response_class=HTMLResponse)
     async def index():

@@ -107,10 +107,16 @@ class ServerRuntime:
 
     # --- maintenance
     async def _maintenance_loop(self) -> None:
+        from ..core.update_checker import UpdateChecker
+        self.update_checker = UpdateChecker("0.1.0")
+        self.update_checker.boot_health_check()
         while not self._stop.is_set():
             try:
                 with self.ldb as db:
                     q.cleanup_stale_runs(db)
+                # release poll on the updateChecker.ts 4h cadence (offline →
+                # state recorded, never raises)
+                await asyncio.to_thread(self.update_checker.maybe_check)
             except Exception:
                 pass
             if await self._sleep(MAINTENANCE_S):
