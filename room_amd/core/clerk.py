@@ -134,35 +134,87 @@ def clerk_chat(ldb: LockedDb, content: str, memory=None,
 
 
 class CommentaryEngine:
-    """Narrates swarm activity (reference: clerk-commentary.ts — subscribes to
-    cycle events on the bus, buffers up to 200 entries, paced generation)."""
+    """Narrates swarm activity (reference: clerk-commentary.ts).
+
+    Full pacing semantics (clerk-commentary.ts:21-33):
+    - active pace: a random 8–30 s interval between lines while the keeper
+      is present (presence heartbeat within the last 90 s);
+    - light pace: 2–3 h intervals when the keeper is away or the
+      `clerk_commentary_mode` setting is 'light';
+    - keeper-pause: any keeper chat message silences commentary until 60 s
+      of keeper silence have passed;
+    - `clerk_commentary` setting 'off' disables generation entirely;
+    - buffer capped at 200 entries.
+    """
+
+    ACTIVE_MIN_S, ACTIVE_MAX_S = 8.0, 30.0
+    LIGHT_MIN_S, LIGHT_MAX_S = 2 * 3600.0, 3 * 3600.0
+    PRESENCE_TIMEOUT_S = 90.0
+    SILENCE_THRESHOLD_S = 60.0
 
     def __init__(self, ldb: LockedDb, bus, model: str = "stub",
-                 pace_s: float = 8.0, buffer_cap: int = 200):
+                 buffer_cap: int = 200, time_source=time.time):
+        import random
         self.ldb = ldb
         self.bus = bus
         self.model = model
-        self.pace_s = pace_s
         self.buffer: list[str] = []
         self.buffer_cap = buffer_cap
-        self._last_emit = 0.0
+        self.time = time_source
+        self._rng = random.Random(0xC0FFEE)
+        self._last_presence = 0.0
+        self._last_keeper_msg = 0.0
+        self._next_due = self.time() + self.ACTIVE_MIN_S
         self._unsub = bus.on("*", self._on_event)
+        # back-compat: the runtime loop polls on this cadence
+        self.pace_s = self.ACTIVE_MIN_S
 
     def _on_event(self, channel: str, event: dict) -> None:
-        if event.get("type") in ("cycle_finished", "decision", "escalation",
-                                 "run_finished"):
-            self.buffer.append(f"{channel}: {event['type']}")
+        etype = event.get("type")
+        if channel == "clerk":
+            if etype == "presence":
+                self._last_presence = self.time()
+            elif etype in ("keeper_message", "typing"):
+                self._last_keeper_msg = self.time()
+            return
+        if etype in ("cycle_finished", "decision", "escalation",
+                     "run_finished"):
+            self.buffer.append(f"{channel}: {etype}")
             if len(self.buffer) > self.buffer_cap:
                 self.buffer = self.buffer[-self.buffer_cap:]
 
+    # ------------------------------------------------------------- pacing
+
+    def current_pace(self) -> str:
+        """'active' | 'light' (mode setting + presence window)."""
+        with self.ldb as db:
+            mode = q.get_setting(db, "clerk_commentary_mode") or "auto"
+        if mode == "light":
+            return "light"
+        present = self.time() - self._last_presence < self.PRESENCE_TIMEOUT_S
+        return "active" if present else "light"
+
+    def _schedule_next(self) -> None:
+        if self.current_pace() == "active":
+            lo, hi = self.ACTIVE_MIN_S, self.ACTIVE_MAX_S
+        else:
+            lo, hi = self.LIGHT_MIN_S, self.LIGHT_MAX_S
+        self._next_due = self.time() + self._rng.uniform(lo, hi)
+
     def tick(self) -> Optional[str]:
         """Generate one commentary line if due (call from a runtime loop)."""
-        now = time.time()
-        if not self.buffer or now - self._last_emit < self.pace_s:
+        now = self.time()
+        with self.ldb as db:
+            if (q.get_setting(db, "clerk_commentary") or "on") == "off":
+                return None
+        # keeper-pause: stay silent until 60 s after the last keeper message
+        if now - self._last_keeper_msg < self.SILENCE_THRESHOLD_S:
+            return None
+        if not self.buffer or now < self._next_due:
             return None
         events = self.buffer[-20:]
         self.buffer = []
-        self._last_emit = now
+        self._schedule_next()
         result = execute_agent(AgentExecutionOptions(
             prompt="Recent swarm events:\n" + "\n".join(events)
                    + "\nGive one short, lively commentary line.",

@@ -30,6 +30,21 @@ def notify_keeper(subject: str, body: str, room_id: int | None = None,
         pass
     if bus is not None:
         bus.emit("clerk", "keeper_notification", entry)
+    tg_token = os.environ.get("ROOMAMD_TELEGRAM_BOT_TOKEN")
+    tg_chat = os.environ.get("ROOMAMD_TELEGRAM_CHAT_ID")
+    if tg_token and tg_chat:
+        try:  # telegram relay (clerk-notifications.ts); fail-silent offline
+            import urllib.parse
+            import urllib.request
+            data = urllib.parse.urlencode(
+                {"chat_id": tg_chat,
+                 "text": f"{subject}\n{body[:1000]}"}).encode()
+            urllib.request.urlopen(
+                f"https://api.telegram.org/bot{tg_token}/sendMessage",
+                data=data, timeout=5)
+            entry["telegram_delivered"] = True
+        except Exception:
+            entry["telegram_delivered"] = False
     smtp = os.environ.get("ROOMAMD_SMTP_URL")
     if smtp:
         try:  # fire-and-forget relay; never raises

@@ -772,6 +772,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/clerk/chat")
     async def clerk_chat(payload: dict = Body(...), role: str = Depends(rw)):
         from ..core.clerk import clerk_chat as do_chat
+        # keeper message → commentary engine pauses for its silence window
+        bus.emit("clerk", "keeper_message", {})
         reply = await asyncio.to_thread(do_chat, ldb, payload["content"],
                                         memory)
         bus.emit("clerk", "message", {"content": reply})
@@ -1353,7 +1355,7 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
                                       "not reachable"}
 
     @app.post("/api/contacts/telegram/check")
-    async def contacts_telegram_check(role: str = Depends(get_role)):
+    async def contacts_telegram_check(role: str = Depends(rw)):
         with ldb as db:
             chat = q.get_setting(db, "keeper_telegram_chat_id")
         return {"connected": bool(chat), "chat_id": chat}

@@ -24,13 +24,16 @@ ROLE_AGENT = "agent"
 ROLE_USER = "user"
 ROLE_MEMBER = "member"
 
-# member whitelist: read-only plus these write endpoints (access.ts:13-24)
+# member whitelist: read-only plus these write endpoints (access.ts:13-24),
+# plus the harmless presence/typing heartbeats the SPA posts for any viewer
 MEMBER_WRITE_WHITELIST = {
     ("POST", "/api/rooms/{room_id}/messages"),
     ("POST", "/api/rooms/{room_id}/escalations"),
     ("POST", "/api/decisions/{decision_id}/vote"),
     ("POST", "/api/decisions/{decision_id}/object"),
     ("POST", "/api/rooms/{room_id}/chat"),
+    ("POST", "/api/clerk/presence"),
+    ("POST", "/api/clerk/typing"),
 }
 
 

@@ -57,6 +57,7 @@ class ServerRuntime:
         ]
         if self.commentary is not None:
             self._tasks.append(asyncio.create_task(self._commentary_loop()))
+        self._tasks.append(asyncio.create_task(self._alert_relay_loop()))
         self._tasks.append(asyncio.create_task(self._watcher_loop()))
         self._watch_mtimes: dict[int, float] = {}
 
@@ -139,6 +140,34 @@ class ServerRuntime:
             except Exception:
                 pass
             if await self._sleep(INBOX_POLL_S):
+                return
+
+    # --- clerk alert relay: pending keeper-facing escalations ride the
+    # notification outbox (email/telegram when configured) on a 15 s poll
+    # (reference runtime.ts clerk alert relay + clerk-notifications.ts)
+    async def _alert_relay_loop(self) -> None:
+        from ..core.notifications import notify_keeper
+        self._alerted: set[int] = getattr(self, "_alerted", set())
+        while not self._stop.is_set():
+            try:
+                with self.ldb as db:
+                    rooms = q.list_rooms(db)
+                    pend = []
+                    for room in rooms:
+                        for e in q.list_escalations(db, room["id"],
+                                                    status="pending"):
+                            if e["to_agent_id"] is None and \
+                                    e["id"] not in self._alerted:
+                                pend.append((room, e))
+                for room, e in pend:
+                    self._alerted.add(e["id"])
+                    await asyncio.to_thread(
+                        notify_keeper,
+                        f"[{room['name']}] escalation #{e['id']}",
+                        e["question"], room_id=room["id"], bus=self.bus)
+            except Exception:
+                pass
+            if await self._sleep(15.0):
                 return
 
     # --- clerk commentary narration
