@@ -169,6 +169,11 @@ class CommentaryEngine:
         # back-compat: the runtime loop polls on this cadence
         self.pace_s = self.ACTIVE_MIN_S
 
+    def stop(self) -> None:
+        if self._unsub:
+            self._unsub()
+            self._unsub = None
+
     def _on_event(self, channel: str, event: dict) -> None:
         etype = event.get("type")
         if channel == "clerk":

@@ -140,8 +140,12 @@ def test_commentary_engine(db):
     from room_amd.core.events import EventBus
     ldb = LockedDb(db)
     bus = EventBus()
-    eng = CommentaryEngine(ldb, bus, model="stub", pace_s=0.0)
-    bus.emit("room:1", "cycle_finished", {"cycle_id": 1})
+    clock = [1000.0]
+    eng = CommentaryEngine(ldb, bus, model="stub",
+                           time_source=lambda: clock[0])
+    bus.emit("room:1", "cycle_finished",
+             {"type": "cycle_finished", "cycle_id": 1})
+    clock[0] += 31  # past the max active interval
     line = eng.tick()
     assert line is not None
     with ldb as conn:
