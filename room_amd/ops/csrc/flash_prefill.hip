@@ -66,9 +66,14 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
     const int tk = rem / (FP_D / 8);
     const int d8 = (rem % (FP_D / 8)) * 8;
     bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-    if (tk < ntok)
+    if (tk < ntok) {
       v = *reinterpret_cast<const bf16x8*>(
           q + (long)(row0 + tk) * q_row_stride + (hk * FP_QH + h) * FP_D + d8);
+      // fold the softmax scale into Q here: saves one VALU mul per score
+      // in the hot per-chunk loop (8 per lane per chunk)
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = f2bf(bf2f(v[j]) * scale);
+    }
     *reinterpret_cast<bf16x8*>(&q_s[h][tk][d8]) = v;
   }
   if (tid < FP_QTOK)
@@ -77,8 +82,11 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
   __syncthreads();
 
   const int seq = seq_s;
-  int bound_max = 0;
-  for (int i = 0; i < ntok; ++i) bound_max = max(bound_max, qp_s[i] + 1);
+  int bound_max = 0, bound_min = INT_MAX;
+  for (int i = 0; i < ntok; ++i) {
+    bound_max = max(bound_max, qp_s[i] + 1);
+    bound_min = min(bound_min, qp_s[i] + 1);
+  }
   const int* btab = block_table + (long)seq * max_blocks;
   const long kv_stride_block = (long)n_kvheads * FP_BS * FP_D;
 
@@ -139,8 +147,11 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
     }
 
     // ---- causal mask + online softmax; one max/sum lane-reduce pair per
-    // row per chunk, amortized over the 4 col-frags
+    // row per chunk, amortized over the 4 col-frags. Chunks entirely below
+    // every row's causal bound (the common case for full tiles deep in the
+    // context) skip the per-score mask compares.
     const int colL = lane & 15;
+    const bool interior = (ntok == FP_QTOK) && (base + FP_CHUNK <= bound_min);
     #pragma unroll
     for (int mt = 0; mt < 2; ++mt) {
       #pragma unroll
@@ -149,12 +160,20 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
         const int bound = qp_s[tk] + 1;  // -1+1=0 for pad rows → all masked
         float sv[4];
         float mx = -INFINITY;
-        #pragma unroll
-        for (int nt = 0; nt < 4; ++nt) {
-          float s = sfrag[mt][nt][r] * scale;
-          if (base + nt * 16 + colL >= bound) s = -INFINITY;
-          sv[nt] = s;
-          mx = fmaxf(mx, s);
+        if (interior) {
+          #pragma unroll
+          for (int nt = 0; nt < 4; ++nt) {
+            sv[nt] = sfrag[mt][nt][r];   // scale folded into Q at staging
+            mx = fmaxf(mx, sv[nt]);
+          }
+        } else {
+          #pragma unroll
+          for (int nt = 0; nt < 4; ++nt) {
+            float s = sfrag[mt][nt][r];
+            if (base + nt * 16 + colL >= bound) s = -INFINITY;
+            sv[nt] = s;
+            mx = fmaxf(mx, s);
+          }
         }
         #pragma unroll
         for (int off = 8; off > 0; off >>= 1)
@@ -175,12 +194,17 @@ void flash_prefill_kernel(short* __restrict__ out,         // [T, Hq, D]
           l_run[mt][r] = l_run[mt][r] * rs + lsum;
           m_run[mt][r] = m_new;
         }
-        // write P row to LDS (C-layout → A-layout) + rescale O row
+        // write P row to LDS (C-layout → A-layout) + rescale O row.
+        // rs==1 is the common case once the running max stabilizes; the rs
+        // check is row-uniform across the 16 col lanes, so divergence cost
+        // stays within the wave's row groups.
         #pragma unroll
         for (int nt = 0; nt < 4; ++nt)
           p_s[wid][tk][nt * 16 + colL] = f2bf(pv[nt]);
-        #pragma unroll
-        for (int nf = 0; nf < 8; ++nf) oacc[mt][nf][r] *= rs;
+        if (rs != 1.f) {
+          #pragma unroll
+          for (int nf = 0; nf < 8; ++nf) oacc[mt][nf][r] *= rs;
+        }
       }
     }
     __syncthreads();  // P + Vt visible
