@@ -1,4 +1,4 @@
-"""CLI entry: room-amd mcp | serve [port] | status | bench | uninstall
+"""CLI entry: room-amd mcp | serve [port] | status | update | uninstall
 (reference: src/cli/index.ts — quoroom mcp|serve|update|uninstall)."""
 from __future__ import annotations
 
@@ -22,6 +22,8 @@ def main(argv: list[str] | None = None) -> int:
 
     st = sub.add_parser("status", help="query the running server's status")
     st.add_argument("--port", type=int, default=None)
+
+    sub.add_parser("update", help="check for (and stage) a newer release")
 
     sub.add_parser("uninstall", help="remove ~/.roomamd data (asks first)")
 
@@ -54,6 +56,26 @@ def main(argv: list[str] | None = None) -> int:
         except Exception as e:
             print(f"server not reachable: {e}", file=sys.stderr)
             return 1
+
+    if args.cmd == "update":
+        # reference: quoroom update (cli/update.ts) — poll the release feed,
+        # report, and stage a user-space update when one is available
+        from ..core.update_checker import UpdateChecker
+        from ..server.auth import data_dir
+        uc = UpdateChecker("0.1.0", data_dir=data_dir())
+        uc.boot_health_check()
+        st = uc.check()
+        if st["state"] in ("offline", "error"):
+            print(f"update check failed ({st['state']}): {st['error']}",
+                  file=sys.stderr)
+            return 1
+        if not st["updateAvailable"]:
+            print(f"up to date (v{st['currentVersion']})")
+            return 0
+        print(f"update available: v{st['latestVersion']} ({st['releaseUrl']})")
+        out = uc.stage_update()
+        print(f"staged: {out}")
+        return 0
 
     if args.cmd == "uninstall":
         from ..server.auth import data_dir

@@ -215,7 +215,8 @@ class TaskRunner:
             q.update_task(db, task_id, **updates)
 
             # learned-context distillation every 3 runs (learned-context.ts:20-31)
-            if result.success and should_distill(new_count):
+            if result.success and should_distill(
+                    new_count, has_context=bool(task.get("learned_context"))):
                 memo = distill_learned_context(db, task_id, model=model)
                 if memo:
                     q.update_task(db, task_id, learned_context=memo[:1500])
