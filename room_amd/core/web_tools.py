@@ -50,23 +50,32 @@ def web_search(query: str, limit: int = 5) -> dict:
 
 
 def browser_action(session_id: str, action: str, **kwargs) -> dict:
-    """Persistent browser-session contract (reference keeps Chromium sessions
-    with 30-min idle GC). No browser runtime is present in this image; the
-    session bookkeeping is preserved so callers get stable semantics."""
-    import time
-    sess = _BROWSER_SESSIONS.setdefault(
-        session_id, {"created": time.time(), "history": []})
-    sess["last_used"] = time.time()
-    sess["history"].append(action)
-    # idle GC
-    for sid in list(_BROWSER_SESSIONS):
-        if time.time() - _BROWSER_SESSIONS[sid]["last_used"] > 1800:
-            del _BROWSER_SESSIONS[sid]
-    if action == "goto" and kwargs.get("url"):
-        return web_fetch(kwargs["url"])
-    return {"ok": False, "error": "no browser runtime in this environment",
+    """Persistent browser-session action (reference browserActionPersistent,
+    web-tools.ts:456-599). Drives headless Chromium over CDP when a binary
+    exists (core/browser.py: persistent user-data-dir sessions, 30-min idle
+    GC, accessibility-style snapshots); otherwise degrades to an urllib
+    fetch for goto/navigate and an explicit error for interactions — the
+    reference's exact 'Chromium not installed' fallback semantics."""
+    from . import browser as br
+    url = kwargs.get("url", "")
+    if br.find_chromium() is not None:
+        acts = [] if action in ("goto", "navigate", "snapshot") else [
+            {"type": action, **kwargs}]
+        out = br.browser_action(url, acts, session_id=session_id or None)
+        ok = not out["snapshot"].startswith(("browser error",
+                                             "Chromium not installed"))
+        return {"ok": ok, "session": out["sessionId"] or session_id,
+                "action": action, "url": out["url"],
+                "snapshot": out["snapshot"]}
+    if action in ("goto", "navigate") and url:
+        return web_fetch(url)
+    return {"ok": False, "error": "no browser runtime in this environment "
+                                  "(install chromium or set ROOMAMD_CHROMIUM)",
             "session": session_id, "action": action}
 
 
 def close_browser(session_id: str) -> bool:
-    return _BROWSER_SESSIONS.pop(session_id, None) is not None
+    from . import browser as br
+    had = session_id in br._sessions
+    br.close_session(session_id)
+    return had or _BROWSER_SESSIONS.pop(session_id, None) is not None
