@@ -30,10 +30,16 @@ def _tokens_path() -> Path:
 
 
 def load_room_tokens() -> dict:
+    """Token store file format preserved from the reference:
+    {"rooms": {"<roomId>": "<token>"}} (cloud-sync.ts:20-52; the on-disk
+    contract in SURVEY §2d). A legacy flat dict is still read."""
     p = _tokens_path()
     if p.exists():
         try:
-            return json.loads(p.read_text())
+            data = json.loads(p.read_text())
+            if isinstance(data, dict) and isinstance(data.get("rooms"), dict):
+                return data["rooms"]
+            return data if isinstance(data, dict) else {}
         except (ValueError, OSError):
             return {}
     return {}
@@ -45,7 +51,7 @@ def save_room_token(room_id: int, token: str) -> None:
     p = _tokens_path()
     try:
         p.parent.mkdir(parents=True, exist_ok=True)
-        p.write_text(json.dumps(tokens))
+        p.write_text(json.dumps({"rooms": tokens}))
         p.chmod(0o600)
     except OSError:
         pass
@@ -73,7 +79,9 @@ def register_with_cloud(ldb: LockedDb, room_id: int) -> str | None:
         room = q.get_room(db, room_id)
     if room is None:
         return None
-    out = _post("/api/runtime/register",
+    # endpoint paths mirror cloud-sync.ts (/rooms/register,
+    # /rooms/<id>/heartbeat, /rooms/message)
+    out = _post("/rooms/register",
                 {"machine_id": get_machine_id(), "room_id": room_id,
                  "name": room["name"]})
     if out and out.get("token"):
@@ -89,7 +97,7 @@ def send_heartbeat(ldb: LockedDb, room_id: int) -> bool:
     with ldb as db:
         usage = q.get_room_token_usage(db, room_id)
         workers = q.list_room_workers(db, room_id)
-    out = _post("/api/runtime/heartbeat",
+    out = _post(f"/rooms/{room_id}/heartbeat",
                 {"room_id": room_id, "ts": int(time.time()),
                  "cycles": usage["cycles"], "workers": len(workers)},
                 token=token)
@@ -99,7 +107,7 @@ def send_heartbeat(ldb: LockedDb, room_id: int) -> bool:
 def send_cloud_room_message(ldb: LockedDb, room_id: int, to_room: str,
                             subject: str, body: str) -> bool:
     token = load_room_tokens().get(str(room_id))
-    out = _post("/api/runtime/messages",
+    out = _post("/rooms/message",
                 {"to": to_room, "subject": subject, "body": body}, token=token)
     with ldb as db:  # outbound row recorded regardless (durable intent)
         q.create_room_message(db, room_id, "outbound", subject, body,
@@ -115,7 +123,7 @@ def fetch_cloud_room_messages(ldb: LockedDb, room_id: int) -> int:
         return 0
     try:
         req = urllib.request.Request(
-            f"{api}/api/runtime/messages?room_id={room_id}",
+            f"{api}/rooms/{room_id}/messages",
             headers={"Authorization": f"Bearer {token}"})
         with urllib.request.urlopen(req, timeout=5) as resp:
             msgs = json.load(resp)
