@@ -167,6 +167,14 @@ static int attn_chunk() {
   return c;
 }
 
+static bool attn_pipe() {
+  static bool p = [] {
+    const char* e = getenv("ROOMAMD_NO_ATTN_PIPE");
+    return !(e && e[0] == '1');
+  }();
+  return p;
+}
+
 // ---------------------------------------------------------------- wrappers
 
 void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
@@ -219,7 +227,13 @@ void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
 
 // CH: kv positions staged/scored per iteration. 64 halves the barrier count
 // per span (3 barriers per chunk) at the cost of LDS (41 KB vs 24 KB).
-template <int CH>
+// PIPE: double-buffered K/V staging — the next chunk's global loads are
+// issued right after the current chunk's LDS stores become visible, so the
+// HBM latency overlaps the score+value phases instead of stalling at the
+// top of each iteration (the round-1 in-phase interleave attempt failed
+// because it stretched the phases' critical paths; this version keeps the
+// phases untouched and only moves the load-issue point).
+template <int CH, bool PIPE>
 __global__ __launch_bounds__(256)
 void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
                              float* __restrict__ part_ml, // [T, Hq, NSPLITS, 2]
@@ -273,10 +287,11 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
   // value phases then run out of LDS. Row pad of 8 bf16 keeps the per-lane
   // row reads off a single bank (same layout as flash_prefill).
   #define PA_PAD 8
+  constexpr int NBUF = PIPE ? 2 : 1;
   __shared__ float q_s[QH_PER_KV][HEAD_DIM];
   __shared__ float p_s[QH_PER_KV][CH];
-  __shared__ short k_s[CH][HEAD_DIM + PA_PAD];
-  __shared__ short v_s[CH][HEAD_DIM + PA_PAD];
+  __shared__ short k_s[NBUF][CH][HEAD_DIM + PA_PAD];
+  __shared__ short v_s[NBUF][CH][HEAD_DIM + PA_PAD];
   for (int i = tid; i < QH_PER_KV * HEAD_DIM; i += blockDim.x) {
     int hh = i / HEAD_DIM, dd = i % HEAD_DIM;
     q_s[hh][dd] = bf2f(q[(long)t * q_row_stride
@@ -288,29 +303,47 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
   const long kv_stride_block = (long)n_kvheads * BLOCK_SIZE * HEAD_DIM;
   const int* btab = block_table + (long)seq * max_blocks;
 
-  for (int base = lo; base < hi; base += CH) {
-    // ---- stage K and V [CH][128] chunks: CH*16 vec8 each
+  // per-thread staging registers (CH/16 vec8 pairs)
+  bf16x8 kreg[CH / 16], vreg[CH / 16];
+  auto issue_loads = [&](int base) {
     #pragma unroll
     for (int it = 0; it < CH / 16; ++it) {
-      const int idx = tid + it * 256;          // vec8 index 0..CH*16-1
-      const int pos_l = (idx * 8) / HEAD_DIM;  // 0..CH-1
+      const int idx = tid + it * 256;
+      const int pos_l = (idx * 8) / HEAD_DIM;
       const int d8 = (idx * 8) % HEAD_DIM;
       const int pos = base + pos_l;
-      bf16x8 kv = {0, 0, 0, 0, 0, 0, 0, 0};
-      bf16x8 vv = {0, 0, 0, 0, 0, 0, 0, 0};
+      kreg[it] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      vreg[it] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       if (pos < hi) {
         const int blk = btab[pos / BLOCK_SIZE];
         const long off = (long)blk * kv_stride_block
                          + ((long)hk * BLOCK_SIZE + (pos % BLOCK_SIZE)) * HEAD_DIM + d8;
-        kv = *reinterpret_cast<const bf16x8*>(kcache + off);
-        vv = *reinterpret_cast<const bf16x8*>(vcache + off);
+        kreg[it] = *reinterpret_cast<const bf16x8*>(kcache + off);
+        vreg[it] = *reinterpret_cast<const bf16x8*>(vcache + off);
       }
-      *reinterpret_cast<bf16x8*>(&k_s[pos_l][d8]) = kv;
-      *reinterpret_cast<bf16x8*>(&v_s[pos_l][d8]) = vv;
     }
-    __syncthreads();
+  };
+  auto store_regs = [&](int buf) {
+    #pragma unroll
+    for (int it = 0; it < CH / 16; ++it) {
+      const int idx = tid + it * 256;
+      const int pos_l = (idx * 8) / HEAD_DIM;
+      const int d8 = (idx * 8) % HEAD_DIM;
+      *reinterpret_cast<bf16x8*>(&k_s[buf][pos_l][d8]) = kreg[it];
+      *reinterpret_cast<bf16x8*>(&v_s[buf][pos_l][d8]) = vreg[it];
+    }
+  };
 
-    {   // score: thread (h, sub) dots q_s[h] with k_s[sub + r*32]
+  issue_loads(lo);
+  int iter = 0;
+  for (int base = lo; base < hi; base += CH, ++iter) {
+    const int cur = PIPE ? (iter & 1) : 0;
+    store_regs(cur);
+    __syncthreads();                 // staged chunk visible to all waves
+    if (PIPE && base + CH < hi)
+      issue_loads(base + CH);        // HBM latency overlaps score+value
+
+    {   // score: thread (h, sub) dots q_s[h] with k_s[cur][sub + r*32]
       #pragma unroll
       for (int r = 0; r < CH / 32; ++r) {
         const int p = sub + r * 32;
@@ -319,7 +352,7 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
           float dot = 0.f;
           #pragma unroll
           for (int v8 = 0; v8 < HEAD_DIM / 8; ++v8) {
-            bf16x8 kv = *reinterpret_cast<const bf16x8*>(&k_s[p][v8 * 8]);
+            bf16x8 kv = *reinterpret_cast<const bf16x8*>(&k_s[cur][p][v8 * 8]);
             #pragma unroll
             for (int j = 0; j < 8; ++j) dot += q_s[h][v8 * 8 + j] * bf2f(kv[j]);
           }
@@ -342,7 +375,7 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
         for (int j = 0; j < lim; ++j) {
           const float w = __expf(p_s[h][j] - m_new);
           l_run += w;
-          bf16x4 vv = *reinterpret_cast<const bf16x4*>(&v_s[j][sub * 4]);
+          bf16x4 vv = *reinterpret_cast<const bf16x4*>(&v_s[cur][j][sub * 4]);
           acc[0] += w * bf2f(vv[0]);
           acc[1] += w * bf2f(vv[1]);
           acc[2] += w * bf2f(vv[2]);
@@ -351,7 +384,7 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
         m_run = m_new;
       }
     }
-    __syncthreads();  // before restaging K/V
+    __syncthreads();  // value reads done before the next store_regs
   }
 
   if (sub == 0) { ml[0] = m_run; ml[1] = l_run; }
@@ -403,22 +436,24 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(part.size(0) >= T && part.size(2) == NSPLITS);
   hipStream_t s = c10::hip::getCurrentHIPStream();
   dim3 g1(T, n_kvheads, NSPLITS);
-  if (attn_chunk() == 64)
-    hipLaunchKernelGGL(paged_attn_split_kernel<64>, g1, dim3(256), 0, s,
-                       part.data_ptr<float>(), part_ml.data_ptr<float>(),
-                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
-                       (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
-                       seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
-                       n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
-                       (float*)nullptr, 0L);
-  else
-    hipLaunchKernelGGL(paged_attn_split_kernel<32>, g1, dim3(256), 0, s,
-                       part.data_ptr<float>(), part_ml.data_ptr<float>(),
-                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
-                       (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
-                       seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
-                       n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
-                       (float*)nullptr, 0L);
+  #define PA_SPLIT_ARGS part.data_ptr<float>(), part_ml.data_ptr<float>(), \
+      (const short*)q.data_ptr(), (const short*)kcache.data_ptr(), \
+      (const short*)vcache.data_ptr(), block_table.data_ptr<int>(), \
+      seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(), \
+      n_kvheads, max_blocks, (int)q.stride(0), (float)scale, \
+      (float*)nullptr, 0L
+  if (attn_chunk() == 64) {
+    if (attn_pipe())
+      hipLaunchKernelGGL((paged_attn_split_kernel<64, true>), g1, dim3(256), 0, s, PA_SPLIT_ARGS);
+    else
+      hipLaunchKernelGGL((paged_attn_split_kernel<64, false>), g1, dim3(256), 0, s, PA_SPLIT_ARGS);
+  } else {
+    if (attn_pipe())
+      hipLaunchKernelGGL((paged_attn_split_kernel<32, true>), g1, dim3(256), 0, s, PA_SPLIT_ARGS);
+    else
+      hipLaunchKernelGGL((paged_attn_split_kernel<32, false>), g1, dim3(256), 0, s, PA_SPLIT_ARGS);
+  }
+  #undef PA_SPLIT_ARGS
   HIP_CHECK_KERNEL();
   dim3 g2(T, n_qheads);
   hipLaunchKernelGGL(paged_attn_merge_kernel, g2, dim3(HEAD_DIM), 0, s,
@@ -442,21 +477,23 @@ void paged_attention_splitk(torch::Tensor part, torch::Tensor part_ml,
   TORCH_CHECK(o_zero.dtype() == torch::kFloat32 && o_zero.is_contiguous());
   hipStream_t s = c10::hip::getCurrentHIPStream();
   dim3 g1(T, n_kvheads, NSPLITS);
-  if (attn_chunk() == 64)
-    hipLaunchKernelGGL(paged_attn_split_kernel<64>, g1, dim3(256), 0, s,
-                       part.data_ptr<float>(), part_ml.data_ptr<float>(),
-                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
-                       (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
-                       seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
-                       n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
-                       o_zero.data_ptr<float>(), (long)o_zero.numel());
-  else
-    hipLaunchKernelGGL(paged_attn_split_kernel<32>, g1, dim3(256), 0, s,
-                       part.data_ptr<float>(), part_ml.data_ptr<float>(),
-                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
-                       (const short*)vcache.data_ptr(), block_table.data_ptr<int>(),
-                       seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(),
-                       n_kvheads, max_blocks, (int)q.stride(0), (float)scale,
-                       o_zero.data_ptr<float>(), (long)o_zero.numel());
+  #define PA_SPLITK_ARGS part.data_ptr<float>(), part_ml.data_ptr<float>(), \
+      (const short*)q.data_ptr(), (const short*)kcache.data_ptr(), \
+      (const short*)vcache.data_ptr(), block_table.data_ptr<int>(), \
+      seq_ids.data_ptr<int>(), q_pos.data_ptr<int>(), \
+      n_kvheads, max_blocks, (int)q.stride(0), (float)scale, \
+      o_zero.data_ptr<float>(), (long)o_zero.numel()
+  if (attn_chunk() == 64) {
+    if (attn_pipe())
+      hipLaunchKernelGGL((paged_attn_split_kernel<64, true>), g1, dim3(256), 0, s, PA_SPLITK_ARGS);
+    else
+      hipLaunchKernelGGL((paged_attn_split_kernel<64, false>), g1, dim3(256), 0, s, PA_SPLITK_ARGS);
+  } else {
+    if (attn_pipe())
+      hipLaunchKernelGGL((paged_attn_split_kernel<32, true>), g1, dim3(256), 0, s, PA_SPLITK_ARGS);
+    else
+      hipLaunchKernelGGL((paged_attn_split_kernel<32, false>), g1, dim3(256), 0, s, PA_SPLITK_ARGS);
+  }
+  #undef PA_SPLITK_ARGS
   HIP_CHECK_KERNEL();
 }
