@@ -168,9 +168,12 @@ static int attn_chunk() {
 }
 
 static bool attn_pipe() {
+  // measured SLOWER in-bench (decode 10.70 -> 10.95 s over 6 steps): the
+  // staging registers + duplicated LDS buffers cost more occupancy than the
+  // hidden HBM latency buys. Off by default; ROOMAMD_ATTN_PIPE=1 re-enables.
   static bool p = [] {
-    const char* e = getenv("ROOMAMD_NO_ATTN_PIPE");
-    return !(e && e[0] == '1');
+    const char* e = getenv("ROOMAMD_ATTN_PIPE");
+    return e && e[0] == '1';
   }();
   return p;
 }
