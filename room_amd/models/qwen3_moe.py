@@ -104,6 +104,9 @@ class Qwen3MoEModel:
         # latency-starved 32-WG grid). Kept behind env for the record.
         self.fuse_o = _os.environ.get('ROOMAMD_FUSE_O') == '1'
         self.fuse_router = _os.environ.get('ROOMAMD_FUSE_ROUTER') == '1'
+        # grouped-GEMM m-tile: 256 halves expert-panel re-reads when experts
+        # average >=128 pairs (prefill chunks >=2k tokens)
+        self.moe_bm = 256 if _os.environ.get('ROOMAMD_MOE_BM') == '256' else 128
         cos_t, sin_t = rope_tables(cfg.max_position, cfg.head_dim, cfg.rope_theta)
         self.cos_t = cos_t.to(self.device)
         self.sin_t = sin_t.to(self.device)
@@ -311,13 +314,14 @@ class Qwen3MoEModel:
         inv_order = torch.empty_like(order)
         inv_order[order] = torch.arange(P, device=hbuf.device)
         inv_order = inv_order.int().contiguous()
-        desc = ops.moe_build_desc_device(pair_expert, cfg.num_experts, bm=128)
+        bm = self.moe_bm
+        desc = ops.moe_build_desc_device(pair_expert, cfg.num_experts, bm=bm)
         gateup = torch.empty(P, 2 * I, dtype=torch.bfloat16, device=hbuf.device)
-        ops.moe_grouped_gemm128(gateup, hbuf, layer.w13, pair_token, desc)
+        ops.moe_grouped_gemm128(gateup, hbuf, layer.w13, pair_token, desc, bm=bm)
         h = torch.empty(P, I, dtype=torch.bfloat16, device=hbuf.device)
         ops.silu_mul(h, gateup)
         z = torch.empty(P, H, dtype=torch.bfloat16, device=hbuf.device)
-        ops.moe_grouped_gemm128(z, h, layer.w2, self._arangeP, desc)
+        ops.moe_grouped_gemm128(z, h, layer.w2, self._arangeP, desc, bm=bm)
         out_bf = torch.empty(T, H, dtype=torch.bfloat16, device=hbuf.device)
         ops.moe_combine_gather(out_bf, z, topk_w.contiguous(), inv_order)
         return out_bf

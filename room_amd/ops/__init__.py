@@ -230,10 +230,12 @@ def moe_grouped_gemm(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
 
 
 def moe_grouped_gemm128(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
-                        pair_token: torch.Tensor,
-                        tile_desc: torch.Tensor) -> torch.Tensor:
-    """BM=128 variant: expert weight panels read once per 128-row m-tile."""
-    _require().moe_grouped_gemm128(out, x, w, pair_token, tile_desc)
+                        pair_token: torch.Tensor, tile_desc: torch.Tensor,
+                        bm: int = 128) -> torch.Tensor:
+    """Grouped MFMA GEMM: expert weight panels read once per BM-row m-tile
+    (bm must match the tile_desc's build bm; 256 halves W traffic when
+    experts average ≥128 pairs)."""
+    _require().moe_grouped_gemm128(out, x, w, pair_token, tile_desc, bm)
     return out
 
 

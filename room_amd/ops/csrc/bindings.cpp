@@ -35,7 +35,8 @@ void moe_gemv_down(torch::Tensor out, torch::Tensor h, torch::Tensor w2,
 void moe_grouped_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                       torch::Tensor pair_token, torch::Tensor tile_desc);
 void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
-                         torch::Tensor pair_token, torch::Tensor tile_desc);
+                         torch::Tensor pair_token, torch::Tensor tile_desc,
+                         int64_t bm);
 void moe_gemv_dedup(torch::Tensor out, torch::Tensor x, torch::Tensor w13,
                     torch::Tensor w2, torch::Tensor topk_ids,
                     torch::Tensor topk_w, torch::Tensor counts,

@@ -569,7 +569,7 @@ void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
 
 // BN templated: wider N-tiles cut the X-tile re-read factor (X is re-read
 // once per N-tile; gateup N=1536 at BN=64 → 24×, at BN=128 → 12×).
-template <int BN>
+template <int BM, int BN>
 __global__ __launch_bounds__(512)
 void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
                                 const short* __restrict__ x,   // [T, H]
@@ -585,23 +585,26 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
   const int n0 = blockIdx.y * BN;         // n-tile from the grid
 
   const int tid = threadIdx.x;
-  const int wid = tid >> 6;       // 0..7: wave's 16-row m-slice
+  const int wid = tid >> 6;       // wave index: MF 16-row m-slices each
   const int lane = tid & 63;
 
-  __shared__ short xs[G2_BM][G2_BK + G2_PAD];
+  __shared__ short xs[BM][G2_BK + G2_PAD];
   __shared__ short ws[BN][G2_BK + G2_PAD];
 
   const short* wbase = w + (long)e * N * H;
   constexpr int NF = BN / 16;             // n-fragments per wave
+  constexpr int MF = BM / 128;            // m-fragments per wave (1 or 2)
 
-  cfrag_t acc[NF];
+  cfrag_t acc[MF][NF];
   #pragma unroll
-  for (int nf = 0; nf < NF; ++nf) acc[nf] = cfrag_t{0.f, 0.f, 0.f, 0.f};
+  for (int mf = 0; mf < MF; ++mf)
+    #pragma unroll
+    for (int nf = 0; nf < NF; ++nf) acc[mf][nf] = cfrag_t{0.f, 0.f, 0.f, 0.f};
 
   for (int k0 = 0; k0 < H; k0 += G2_BK) {
-    // stage X tile: 128 rows × 64 k = 1024 vec8 → 2 per thread
+    // stage X tile: BM rows × 64 k = BM*8 vec8 → BM/64 per thread
     #pragma unroll
-    for (int it = 0; it < 2; ++it) {
+    for (int it = 0; it < BM / 64; ++it) {
       const int idx = tid + it * 512;
       const int r = (idx * 8) / G2_BK;
       const int c = (idx * 8) % G2_BK;
@@ -626,34 +629,41 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
 
     #pragma unroll
     for (int kk = 0; kk < G2_BK / 32; ++kk) {
-      const int arow = wid * 16 + (lane & 15);
       const int akoff = kk * 32 + (lane >> 4) * 8;
-      bf16x8 a = *reinterpret_cast<const bf16x8*>(&xs[arow][akoff]);
       #pragma unroll
-      for (int nf = 0; nf < NF; ++nf) {
-        const int bcol = nf * 16 + (lane & 15);
-        bf16x8 b = *reinterpret_cast<const bf16x8*>(&ws[bcol][akoff]);
-        acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nf], 0, 0, 0);
+      for (int mf = 0; mf < MF; ++mf) {
+        const int arow = mf * 128 + wid * 16 + (lane & 15);
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(&xs[arow][akoff]);
+        #pragma unroll
+        for (int nf = 0; nf < NF; ++nf) {
+          const int bcol = nf * 16 + (lane & 15);
+          bf16x8 b = *reinterpret_cast<const bf16x8*>(&ws[bcol][akoff]);
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mf][nf], 0, 0, 0);
+        }
       }
     }
     __syncthreads();
   }
 
-  const int crow = wid * 16 + (lane >> 4) * 4;
   const int ccol_base = lane & 15;
   #pragma unroll
-  for (int nf = 0; nf < NF; ++nf) {
+  for (int mf = 0; mf < MF; ++mf) {
+    const int crow = mf * 128 + wid * 16 + (lane >> 4) * 4;
     #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int m = crow + r;
-      if (m < msize)
-        out[(long)(row0 + m) * N + n0 + nf * 16 + ccol_base] = f2bf(acc[nf][r]);
+    for (int nf = 0; nf < NF; ++nf) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = crow + r;
+        if (m < msize)
+          out[(long)(row0 + m) * N + n0 + nf * 16 + ccol_base] = f2bf(acc[mf][nf][r]);
+      }
     }
   }
 }
 
 void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
-                         torch::Tensor pair_token, torch::Tensor tile_desc) {
+                         torch::Tensor pair_token, torch::Tensor tile_desc,
+                         int64_t bm) {
   const int H = x.size(-1), N = out.size(-1);
   const int G = tile_desc.size(0);
   TORCH_CHECK(H % G2_BK == 0 && N % G2_BN == 0);
@@ -666,18 +676,32 @@ void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
     const char* v = getenv("ROOMAMD_MOE_BN128");
     return v && v[0] == '1';
   }();
+  const int BM = (int)bm;
+  TORCH_CHECK(BM == 128 || BM == 256, "bm must be 128 or 256");
   if (wide && N % 128 == 0) {
     dim3 grid(G, N / 128), block(512);
-    hipLaunchKernelGGL(moe_grouped_gemm128_kernel<128>, grid, block, 0, s,
-                       (short*)out.data_ptr(), (const short*)x.data_ptr(),
-                       (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
-                       tile_desc.data_ptr<int>(), H, N);
+    if (BM == 256)
+      hipLaunchKernelGGL((moe_grouped_gemm128_kernel<256, 128>), grid, block, 0, s,
+                         (short*)out.data_ptr(), (const short*)x.data_ptr(),
+                         (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
+                         tile_desc.data_ptr<int>(), H, N);
+    else
+      hipLaunchKernelGGL((moe_grouped_gemm128_kernel<128, 128>), grid, block, 0, s,
+                         (short*)out.data_ptr(), (const short*)x.data_ptr(),
+                         (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
+                         tile_desc.data_ptr<int>(), H, N);
   } else {
     dim3 grid(G, N / G2_BN), block(512);
-    hipLaunchKernelGGL(moe_grouped_gemm128_kernel<G2_BN>, grid, block, 0, s,
-                       (short*)out.data_ptr(), (const short*)x.data_ptr(),
-                       (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
-                       tile_desc.data_ptr<int>(), H, N);
+    if (BM == 256)
+      hipLaunchKernelGGL((moe_grouped_gemm128_kernel<256, G2_BN>), grid, block, 0, s,
+                         (short*)out.data_ptr(), (const short*)x.data_ptr(),
+                         (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
+                         tile_desc.data_ptr<int>(), H, N);
+    else
+      hipLaunchKernelGGL((moe_grouped_gemm128_kernel<128, G2_BN>), grid, block, 0, s,
+                         (short*)out.data_ptr(), (const short*)x.data_ptr(),
+                         (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
+                         tile_desc.data_ptr<int>(), H, N);
   }
   HIP_CHECK_KERNEL();
 }

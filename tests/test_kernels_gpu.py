@@ -519,3 +519,23 @@ def test_router_addnorm_matches_composition(B, E):
     xn_t = s * torch.rsqrt(s.pow(2).mean(-1, keepdim=True) + 1e-6) * gamma.float()
     y_t = xn_t @ w.float().T
     assert bf16_close(y, y_t, atol=0.3, rtol=6e-2)
+
+
+def test_moe_grouped_gemm_bm256_vs_matmul():
+    """BM=256 m-tiles (half the expert-panel re-reads) match the reference."""
+    torch.manual_seed(25)
+    E, H, N = 4, 2048, 1536
+    T = 700  # >1 256-row tile per expert plus ragged tails
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV) * 0.3
+    w = torch.randn(E, N, H, dtype=torch.bfloat16, device=DEV) * 0.05
+    pair_expert = torch.randint(0, E, (T,), device=DEV).int()
+    order = torch.argsort(pair_expert)
+    pair_expert = pair_expert[order].contiguous()
+    pair_token = order.int().contiguous()
+    tile_desc = ops.moe_build_desc_device(pair_expert, E, bm=256)
+    out = torch.empty(T, N, dtype=torch.bfloat16, device=DEV)
+    ops.moe_grouped_gemm128(out, x, w, pair_token, tile_desc, bm=256)
+    for p in [0, 1, 200, 399, 550, 699]:
+        t, e = int(pair_token[p]), int(pair_expert[p])
+        expect = (x[t].float() @ w[e].float().T).to(torch.bfloat16)
+        assert bf16_close(out[p], expect, atol=6e-2, rtol=6e-2), f"pair {p}"
