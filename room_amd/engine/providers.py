@@ -101,21 +101,31 @@ def register_engine(model: str, engine: ChatEngine) -> None:
 
 
 def get_model_provider(model: str) -> str:
-    """Model string → provider family (shape of model-provider.ts:31-41)."""
+    """Model string → provider family (model-provider.ts:31-41): cloud
+    prefixes route to HTTP providers; everything else runs in-process."""
     if model.startswith("stub"):
         return "stub"
-    if model in (LOCAL_MODEL_TAG, "local") or model.startswith("local:") \
-            or model.startswith("qwen"):
-        return "local"
-    return "local"  # everything runs in-process on this framework
+    from .cloud_providers import provider_of
+    cloud = provider_of(model)
+    if cloud:
+        return cloud
+    return "local"
 
 
-def resolve_engine(model: str) -> ChatEngine:
+def resolve_engine(model: str, ldb=None, room_id: int | None = None) -> ChatEngine:
     if model in _engines:
         return _engines[model]
     provider = get_model_provider(model)
     if provider == "stub":
         eng = StubEngine()
+        _engines[model] = eng
+        return eng
+    if provider != "local":
+        from .cloud_providers import (HttpChatEngine,
+                                      resolve_api_key_for_model)
+        key = (resolve_api_key_for_model(ldb, model, room_id)
+               if ldb is not None else None)
+        eng = HttpChatEngine(model, api_key=key)
         _engines[model] = eng
         return eng
     # lazy-build the GPU engine

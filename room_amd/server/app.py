@@ -1241,10 +1241,13 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.get("/api/providers/status")
     async def providers_status(role: str = Depends(get_role)):
         import torch as _torch
+
+        from ..engine.cloud_providers import get_model_auth_status
         return {"providers": [{
             "id": "local", "name": "room_amd in-process engine",
             "model": "qwen3-coder-30b", "ready": True,
-            "gpu": _torch.cuda.is_available()}]}
+            "gpu": _torch.cuda.is_available()}],
+            "auth": get_model_auth_status(ldb)}
 
     @app.get("/api/local-model/status")
     async def local_model_status(role: str = Depends(get_role)):
