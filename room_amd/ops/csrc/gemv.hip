@@ -115,7 +115,12 @@ static void gemv_launch(void* y, const short* x, const short* w, int B, int H,
                         int N, hipStream_t s) {
   dim3 grid((N + 3) / 4), block(256);
   const size_t lds = (size_t)B * H * sizeof(short);
-  const bool use_lds = gemv_ldsx_enabled() && B >= 3 && lds <= 64 * 1024;
+  // LDS staging wins only on very wide-N shapes (lm_head: 152k rows,
+  // 176 -> 151 us at B=5) where the grid's aggregate L2 x-traffic is the
+  // bottleneck; on the 2-5k-row projections the barrier+occupancy cost
+  // outweighs it (9.4 -> 9.9 us measured). Gate on N.
+  const bool use_lds = gemv_ldsx_enabled() && B >= 3 && lds <= 64 * 1024
+                       && N >= 16384;
   switch (B) {
 #define GEMV_CASE(BN) \
     case BN: \
