@@ -1,6 +1,7 @@
 """End-to-end engine tests on GPU (tiny Qwen3-MoE config: same architecture,
 4 layers) — generation, session KV reuse, concurrent batching, agent cycle."""
 import threading
+import time
 
 import pytest
 import torch
@@ -110,5 +111,44 @@ def test_session_lru_eviction():
         # oldest sessions were evicted; the newest still has its slot
         assert "lru-5" in eng.sessions
         assert "lru-0" not in eng.sessions
+    finally:
+        eng.shutdown()
+
+
+def test_concurrency_storm_with_eviction_pressure():
+    """32 threads × repeated generates against 8 KV slots: every request
+    completes, no slot/block leak afterwards (the round-1 leak class)."""
+    import queue as _q
+
+    from room_amd.engine.llm import LocalEngine
+    from room_amd.models.qwen3_moe import Qwen3MoEConfig
+    eng = LocalEngine(cfg=Qwen3MoEConfig.tiny(), kv_gb=1.0, max_seqs=8)
+    errors: _q.Queue = _q.Queue()
+
+    def run(i):
+        try:
+            for rep in range(3):
+                p = tok.encode(f"storm agent {i} rep {rep} " * 6)
+                skey = f"storm-{i}" if i % 2 == 0 else None  # half session-less
+                r = eng.generate(p, max_new_tokens=4, session_key=skey,
+                                 timeout=120)
+                assert len(r.out_tokens) == 4
+        except Exception as e:  # pragma: no cover - surfaced below
+            errors.put(f"{i}: {e}")
+
+    threads = [threading.Thread(target=run, args=(i,)) for i in range(32)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(240)
+    try:
+        assert errors.empty(), errors.get()
+        # drain: no active requests; all session-less slots returned
+        time.sleep(0.3)
+        with eng._lock:
+            assert not eng.admitter.active_slots
+            used = 1 + len(eng.sessions)          # pad slot + live sessions
+            free = len(eng.cache.free_slots)
+            assert used + free == eng.cache.max_seqs, (used, free)
     finally:
         eng.shutdown()
