@@ -19,6 +19,12 @@ from typing import Callable, Iterable, Optional
 from .kv_cache import BLOCK_SIZE, PagedKVCache
 
 
+class CacheFull(RuntimeError):
+    """Every KV slot is held by an ACTIVE request (nothing evictable).
+    Retryable: the scheduler defers the admission until a request
+    completes, instead of failing the caller."""
+
+
 class Session:
     __slots__ = ("slot", "tokens", "last_used")
 
@@ -102,7 +108,7 @@ class SessionAdmitter:
         while cache.blocks_free() < need:
             try:
                 self.evict_lru(exclude_slot=req.slot)
-            except RuntimeError:
+            except CacheFull:
                 break  # nothing evictable — ensure_capacity will raise
 
         req.prefill_tokens_run = len(req.pending_prefill)
@@ -151,7 +157,7 @@ class SessionAdmitter:
         candidates = [(s.last_used, key) for key, s in self.sessions.items()
                       if s.slot not in blocked]
         if not candidates:
-            raise RuntimeError("KV cache: no evictable sessions")
+            raise CacheFull("KV cache: no evictable sessions")
         _, key = min(candidates)
         s = self.sessions.pop(key)
         self.cache.free_seq(s.slot)

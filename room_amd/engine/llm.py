@@ -177,6 +177,7 @@ class LocalEngine:
         self.admitter = SessionAdmitter(self.cache, cfg.max_position)
         self._queue: "queue.Queue[GenRequest]" = queue.Queue()
         self._active: list[GenRequest] = []
+        self._waiting: list[GenRequest] = []   # deferred: all slots active
         self._lock = threading.Lock()
         self._stop = False
         self.stats = {"decode_steps": 0, "decode_tokens": 0, "prefill_tokens": 0,
@@ -267,10 +268,34 @@ class LocalEngine:
                     r.error = f"engine error: {e}"
                     self._complete(r, ok=False)
 
+    def _try_admit(self, req: GenRequest) -> bool:
+        """True = admitted; False = deferred (all slots held by ACTIVE
+        requests — retry when one completes). Other failures resolve the
+        future with an error (the caller blocks on req.done)."""
+        from .admission import CacheFull
+        try:
+            with self._lock:
+                self._admit(req)
+            self._active.append(req)
+            return True
+        except CacheFull:
+            self._waiting.append(req)
+            return False
+        except Exception as e:
+            req.error = f"admit failed: {e}"
+            req.done.set()
+            return True  # resolved, don't retry
+
     def _scheduler_iteration(self) -> None:
-        # admit new requests; an admit failure must resolve that future, never
-        # strand it (the caller blocks on req.done)
-        block = not self._active
+        # re-try deferred admissions first (slots may have freed)
+        waiting, self._waiting = self._waiting, []
+        for req in waiting:
+            if req.cancelled:
+                req.done.set()
+                continue
+            self._try_admit(req)
+        # admit new requests
+        block = not (self._active or self._waiting)
         try:
             while True:
                 req = self._queue.get(timeout=0.05 if block else 0)
@@ -278,13 +303,7 @@ class LocalEngine:
                 if req.cancelled:  # caller gave up before admission
                     req.done.set()
                     continue
-                try:
-                    with self._lock:
-                        self._admit(req)
-                    self._active.append(req)
-                except Exception as e:
-                    req.error = f"admit failed: {e}"
-                    req.done.set()
+                self._try_admit(req)
         except queue.Empty:
             pass
         # drop requests whose caller timed out (a timed-out

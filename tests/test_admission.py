@@ -6,6 +6,7 @@ device (the admitter touches no GPU-specific state).
 Session semantics mirror the reference's agent_sessions continuity
 (src/shared/agent-loop.ts:462-532): prefix reuse, divergence reset, durable
 re-prefill after eviction."""
+import pytest
 import torch
 
 from room_amd.engine.admission import SessionAdmitter
@@ -179,3 +180,20 @@ def test_context_window_truncation_keeps_head_and_tail():
     assert req.prompt_tokens[0] == 0             # head kept (system prompt)
     assert req.prompt_tokens[-1] == 1999         # recent tail kept
     assert req.prefill_tokens_run == budget
+
+
+def test_slot_exhaustion_by_active_requests_raises_cachefull():
+    """All slots held by ACTIVE requests → CacheFull (the scheduler defers
+    the admission rather than failing the caller)."""
+    from room_amd.engine.admission import CacheFull
+    cache, adm = make(max_seqs=2)
+    r1, r2 = Req([1, 2, 3]), Req([4, 5, 6])
+    adm.admit(r1)
+    adm.admit(r2)                    # both slots now active
+    r3 = Req([7, 8, 9])
+    with pytest.raises(CacheFull):
+        adm.admit(r3)
+    # a completion frees a slot and the deferred admit succeeds
+    adm.finish(r1, None)
+    adm.admit(r3)
+    assert r3.slot >= 0
