@@ -968,6 +968,74 @@ class McpServer:
                               if high else "Normal load")
             return out
 
+        # ---- reference-name aliases: every quoroom_* tool of the reference
+        # MCP registry (src/mcp/server.ts, 76 tools) resolves here too, so
+        # clients configured against the reference's tool names work
+        # unchanged. Trivial renames map quoroom_X → room_X; the rest map to
+        # the semantically-equivalent tool below.
+        REF_ALIAS = {
+            "browser": "room_browser_action",
+            "create_subgoal": "room_set_goal",
+            "credentials_get": "room_get_credential",
+            "credentials_list": "room_list_credentials",
+            "edit_skill": "room_modify_skill",
+            "export_worker_prompts": "room_export_prompts",
+            "import_worker_prompts": "room_import_prompts",
+            "identity_register": "room_register_identity",
+            "inbox_list": "room_list_escalations",
+            "inbox_reply": "room_answer_escalation",
+            "inbox_send_room": "room_send_room_message",
+            "list_goals": "room_goal_tree",
+            "propose": "room_announce",
+            "resources_get": "room_resources",
+            "room_activity": "room_activity_feed",
+            "room_status": "room_get_status",
+            "schedule": "room_create_task",
+            "self_mod_edit": "room_modify_skill",
+            "self_mod_history": "room_self_mod_audit",
+            "self_mod_revert": "room_revert_modification",
+            "send_message": "room_send_room_message",
+            "task_history": "room_task_runs",
+            "task_progress": "room_task_runs",
+            "unwatch": "room_delete_watch",
+            "update_progress": "room_update_goal_progress",
+            "wallet_create": "room_wallet_address",
+            "wallet_history": "room_payment_audit",
+            "watch": "room_watch_path",
+        }
+        REF_SUFFIXES = [
+            "abandon_goal", "activate_skill", "browser", "complete_goal",
+            "configure_room", "create_room", "create_skill", "create_subgoal",
+            "create_worker", "credentials_get", "credentials_list",
+            "deactivate_skill", "decision_detail", "delegate_task",
+            "delete_room", "delete_skill", "delete_task", "delete_worker",
+            "edit_skill", "export_worker_prompts", "forget", "get_setting",
+            "identity_get", "identity_register", "identity_update",
+            "import_worker_prompts", "inbox_list", "inbox_reply",
+            "inbox_send_room", "invite_create", "invite_list",
+            "invite_network", "list_decisions", "list_goals", "list_rooms",
+            "list_skills", "list_tasks", "list_watches", "list_workers",
+            "memory_list", "pause_room", "pause_task", "pause_watch",
+            "propose", "recall", "remember", "reset_session",
+            "resources_get", "restart_room", "resume_task", "resume_watch",
+            "room_activity", "room_status", "run_task", "save_wip",
+            "schedule", "self_mod_edit", "self_mod_history",
+            "self_mod_revert", "send_message", "set_goal", "set_setting",
+            "task_history", "task_progress", "unwatch", "update_progress",
+            "update_worker", "vote", "wallet_address", "wallet_balance",
+            "wallet_create", "wallet_history", "wallet_send", "wallet_topup",
+            "watch", "webhook_url",
+        ]
+        for suffix in REF_SUFFIXES:
+            target = REF_ALIAS.get(suffix, f"room_{suffix}")
+            if target not in self.tools:
+                continue  # defensive; covered by tests
+            schema, fn = self.tools[target]
+            alias = f"quoroom_{suffix}"
+            self.tools[alias] = ({**schema, "name": alias,
+                                  "description": schema["description"]
+                                  + f" (alias of {target})"}, fn)
+
     # ------------------------------------------------------------ JSON-RPC
 
     def handle(self, msg: dict) -> dict | None:

@@ -199,3 +199,22 @@ def test_parity_surface_tools(mcp):
     # the full surface is now >= the reference's 76 tools
     lst = mcp.handle({"jsonrpc": "2.0", "id": 9, "method": "tools/list"})
     assert len(lst["result"]["tools"]) >= 90
+
+
+def test_reference_tool_name_aliases(mcp):
+    """Every quoroom_* tool name the reference MCP server registers
+    (src/mcp/server.ts, 76 tools) resolves here, so clients configured
+    against the reference work unchanged."""
+    names = set(mcp.tools)
+    ref_suffixes = ["create_room", "list_rooms", "propose", "vote",
+                    "delegate_task", "remember", "recall", "schedule",
+                    "self_mod_history", "wallet_address", "browser",
+                    "inbox_reply", "export_worker_prompts", "webhook_url"]
+    for s in ref_suffixes:
+        assert f"quoroom_{s}" in names, s
+    assert sum(1 for n in names if n.startswith("quoroom_")) == 76
+    # aliases execute the same handler
+    out = call(mcp, "quoroom_create_room", {"name": "alias-room"})
+    assert out["room_id"]
+    rooms = call(mcp, "quoroom_list_rooms", {})
+    assert any(r["name"] == "alias-room" for r in rooms)
