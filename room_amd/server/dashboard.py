@@ -304,7 +304,7 @@ const RENDER = {
     press start. Watch live activity on the right; answer escalations in
     the messages tab; fund the wallet from the wallet tab.</div>
     <div class="ev"><b>Interfaces:</b> this dashboard, the REST API
-    (docs/API.md), the MCP stdio server (<code>room-amd mcp</code>, 95+
+    (docs/API.md), the MCP stdio server (<code>room-amd mcp</code>, 95 native + 76 reference-name alias
     tools) and webhooks (tasks + queen wake).</div>
     <div class="ev"><b>Keeper contact:</b> verify an email under
     contacts so escalations reach you; or read the outbox at
