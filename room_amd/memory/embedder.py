@@ -44,12 +44,31 @@ def _mode() -> str:
         return "hash"
 
 
+_EMBED_CACHE: dict = {}
+_EMBED_CACHE_MAX = 4096
+
+
 def embed(text: str) -> list[float]:
-    """384-dim L2-normalized embedding (deterministic per host mode)."""
-    if _mode() == "encoder":
+    """384-dim L2-normalized embedding (deterministic per host mode).
+
+    LRU-cached: agent cycles re-embed near-identical recall queries (the
+    room objective, the bench's fixed probe) every cycle, and the encoder
+    forward costs ~40 ms of event-loop time per call — cache hits remove
+    the prompt-build stagger that delays agents joining the decode batch."""
+    mode = _mode()
+    key = (mode, text)
+    hit = _EMBED_CACHE.get(key)
+    if hit is not None:
+        return list(hit)
+    if mode == "encoder":
         from . import encoder
-        return encoder.encode_texts([text])[0]
-    return embed_hash(text)
+        vec = encoder.encode_texts([text])[0]
+    else:
+        vec = embed_hash(text)
+    if len(_EMBED_CACHE) >= _EMBED_CACHE_MAX:
+        _EMBED_CACHE.pop(next(iter(_EMBED_CACHE)))
+    _EMBED_CACHE[key] = tuple(vec)
+    return vec
 
 
 def embed_hash(text: str) -> list[float]:

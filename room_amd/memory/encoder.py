@@ -164,13 +164,16 @@ class MiniEncoder:
         x = torch.zeros(B, T, d)
         mask = torch.zeros(B, T)
         for b, pieces in enumerate(piece_lists):
-            for i, p in enumerate(pieces):
+            vecs = []
+            for p in pieces:
                 vec = self._piece_cache.get(p)
                 if vec is None:
                     vec = _piece_vector(p)
                     if len(self._piece_cache) < 100_000:
                         self._piece_cache[p] = vec
-                x[b, i] = vec
+                vecs.append(vec)
+            # one stack+copy per text (per-piece __setitem__ was ~10 ms/text)
+            x[b, : len(pieces)] = torch.stack(vecs)
             mask[b, : len(pieces)] = 1.0
         x = x.to(self.device) + self.pos[:T]
         mask = mask.to(self.device)
