@@ -18,6 +18,7 @@ batches concurrent agents' prefill/decode into fused kernels
 """
 from __future__ import annotations
 
+import os
 import asyncio
 import json
 import time
@@ -241,6 +242,11 @@ class AgentLoopManager:
             worker_id=worker_id, room_id=room_id,
             on_log=log_buffer.append,
         )
+        if os.environ.get("ROOMAMD_CYCLE_PROF") == "1":
+            import sys as _sys
+            print(f"[cycleprof] w{worker_id} prep {1000*(time.time()-start):.0f}"
+                  f" ms before execute (t={time.time()%100:.3f})",
+                  file=_sys.stderr)
         result = await asyncio.to_thread(execute_agent, options)
 
         # context-overflow retry with fresh session (:773-782)

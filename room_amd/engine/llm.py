@@ -202,6 +202,10 @@ class LocalEngine:
         req = GenRequest(prompt_tokens=list(prompt_tokens),
                          max_new_tokens=max_new_tokens, temperature=temperature,
                          top_p=top_p, top_k=top_k, session_key=session_key)
+        if os.environ.get("ROOMAMD_CYCLE_PROF") == "1":
+            import sys as _sys
+            print(f"[cycleprof] enqueue {session_key} {len(prompt_tokens)}tok "
+                  f"(t={time.time()%100:.3f})", file=_sys.stderr)
         self._queue.put(req)
         if not req.done.wait(timeout):
             # mark cancelled: the scheduler drops it (admit or next step
