@@ -232,7 +232,8 @@ async def _cycle(mgr, room_id: int, worker_id: int, pad: str,
         w = q.get_worker(db, worker_id)
         if not w["wip"]:
             q.set_worker_wip(db, worker_id, pad)
-    out = await mgr.run_cycle(room_id, worker_id, max_turns=1)
+    out = await mgr.run_cycle(room_id, worker_id, max_turns=1,
+                              max_new_tokens=decode_tokens)
     res = out["result"]
     if not res.success:
         raise RuntimeError(f"cycle failed: {res.error}")

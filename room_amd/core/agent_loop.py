@@ -168,7 +168,8 @@ class AgentLoopManager:
 
     async def run_cycle(self, room_id: int, worker_id: int,
                         state: LoopState | None = None,
-                        max_turns: int | None = None) -> dict:
+                        max_turns: int | None = None,
+                        max_new_tokens: int | None = None) -> dict:
         """One full observe→prompt→execute→persist cycle. Async so many agents
         interleave; the blocking engine call runs in a worker thread and the
         GPU scheduler batches across agents."""
@@ -242,6 +243,8 @@ class AgentLoopManager:
             worker_id=worker_id, room_id=room_id,
             on_log=log_buffer.append,
         )
+        if max_new_tokens is not None:
+            options.max_new_tokens = max_new_tokens
         if os.environ.get("ROOMAMD_CYCLE_PROF") == "1":
             import sys as _sys
             print(f"[cycleprof] w{worker_id} prep {1000*(time.time()-start):.0f}"
