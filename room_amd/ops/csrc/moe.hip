@@ -569,7 +569,7 @@ void moe_combine(torch::Tensor out, torch::Tensor z, torch::Tensor pair_w,
 
 // BN templated: wider N-tiles cut the X-tile re-read factor (X is re-read
 // once per N-tile; gateup N=1536 at BN=64 → 24×, at BN=128 → 12×).
-template <int BM, int BN>
+template <int BM, int BN, int BK = G2_BK>
 __global__ __launch_bounds__(512)
 void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
                                 const short* __restrict__ x,   // [T, H]
@@ -588,8 +588,8 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
   const int wid = tid >> 6;       // wave index: MF 16-row m-slices each
   const int lane = tid & 63;
 
-  __shared__ short xs[BM][G2_BK + G2_PAD];
-  __shared__ short ws[BN][G2_BK + G2_PAD];
+  __shared__ short xs[BM][BK + G2_PAD];
+  __shared__ short ws[BN][BK + G2_PAD];
 
   const short* wbase = w + (long)e * N * H;
   constexpr int NF = BN / 16;             // n-fragments per wave
@@ -601,13 +601,13 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
     #pragma unroll
     for (int nf = 0; nf < NF; ++nf) acc[mf][nf] = cfrag_t{0.f, 0.f, 0.f, 0.f};
 
-  for (int k0 = 0; k0 < H; k0 += G2_BK) {
+  for (int k0 = 0; k0 < H; k0 += BK) {
     // stage X tile: BM rows × 64 k = BM*8 vec8 → BM/64 per thread
     #pragma unroll
-    for (int it = 0; it < BM / 64; ++it) {
+    for (int it = 0; it < BM * BK / 64 / G2_BK; ++it) {
       const int idx = tid + it * 512;
-      const int r = (idx * 8) / G2_BK;
-      const int c = (idx * 8) % G2_BK;
+      const int r = (idx * 8) / BK;
+      const int c = (idx * 8) % BK;
       bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
       if (r < msize) {
         const int tok = pair_token[row0 + r];
@@ -617,10 +617,10 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
     }
     // stage W tile: BN rows × 64 k → BN/64 vec8 per thread
     #pragma unroll
-    for (int it = 0; it < BN / 64; ++it) {
+    for (int it = 0; it < BN * BK / 64 / G2_BK; ++it) {
       const int idx = tid + it * 512;
-      const int r = (idx * 8) / G2_BK;
-      const int c = (idx * 8) % G2_BK;
+      const int r = (idx * 8) / BK;
+      const int c = (idx * 8) % BK;
       bf16x8 v = *reinterpret_cast<const bf16x8*>(
           wbase + (long)(n0 + r) * H + k0 + c);
       *reinterpret_cast<bf16x8*>(&ws[r][c]) = v;
@@ -628,7 +628,7 @@ void moe_grouped_gemm128_kernel(short* __restrict__ out,      // [P, N]
     __syncthreads();
 
     #pragma unroll
-    for (int kk = 0; kk < G2_BK / 32; ++kk) {
+    for (int kk = 0; kk < BK / 32; ++kk) {
       const int akoff = kk * 32 + (lane >> 4) * 8;
       #pragma unroll
       for (int mf = 0; mf < MF; ++mf) {
@@ -692,8 +692,17 @@ void moe_grouped_gemm128(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                          tile_desc.data_ptr<int>(), H, N);
   } else {
     dim3 grid(G, N / G2_BN), block(512);
+    static const bool bk128 = []() {
+      const char* v = getenv("ROOMAMD_MOE_BK128");
+      return v && v[0] == '1';
+    }();
     if (BM == 256)
       hipLaunchKernelGGL((moe_grouped_gemm128_kernel<256, G2_BN>), grid, block, 0, s,
+                         (short*)out.data_ptr(), (const short*)x.data_ptr(),
+                         (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
+                         tile_desc.data_ptr<int>(), H, N);
+    else if (bk128 && H % 128 == 0)
+      hipLaunchKernelGGL((moe_grouped_gemm128_kernel<128, G2_BN, 128>), grid, block, 0, s,
                          (short*)out.data_ptr(), (const short*)x.data_ptr(),
                          (const short*)w.data_ptr(), pair_token.data_ptr<int>(),
                          tile_desc.data_ptr<int>(), H, N);
