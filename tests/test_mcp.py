@@ -218,3 +218,40 @@ def test_reference_tool_name_aliases(mcp):
     assert out["room_id"]
     rooms = call(mcp, "quoroom_list_rooms", {})
     assert any(r["name"] == "alias-room" for r in rooms)
+
+
+def _all_tools(mcp):
+    resp = mcp.handle({"jsonrpc": "2.0", "id": 7, "method": "tools/list",
+                       "params": {}})
+    return resp["result"]["tools"]
+
+
+def test_tool_registry_integrity(mcp):
+    """Every registered tool exposes a valid MCP tool descriptor: unique
+    name, non-empty description, object input schema whose required keys
+    exist in properties (the contract clients introspect)."""
+    tools = _all_tools(mcp)
+    assert len(tools) == len({t["name"] for t in tools})
+    for t in tools:
+        assert t["name"] and t["description"], t["name"]
+        schema = t["inputSchema"]
+        assert schema["type"] == "object", t["name"]
+        props = schema.get("properties", {})
+        for req in schema.get("required", []):
+            assert req in props, f"{t['name']}: required {req} not in properties"
+        for pname, pdef in props.items():
+            assert pdef.get("type") in ("string", "integer", "boolean",
+                                        "number", "array", "object"), \
+                f"{t['name']}.{pname}"
+
+
+def test_unknown_tool_and_malformed_args(mcp):
+    resp = mcp.handle({"jsonrpc": "2.0", "id": 8, "method": "tools/call",
+                       "params": {"name": "room_no_such_tool",
+                                  "arguments": {}}})
+    assert "error" in resp or resp["result"].get("isError")
+    # a tool with missing required args returns a tool error, not a crash
+    resp2 = mcp.handle({"jsonrpc": "2.0", "id": 9, "method": "tools/call",
+                        "params": {"name": "room_create_room",
+                                   "arguments": {}}})
+    assert resp2 is not None
