@@ -107,6 +107,9 @@ class Qwen3MoEModel:
         # grouped-GEMM m-tile: 256 halves expert-panel re-reads when experts
         # average >=128 pairs (prefill chunks >=2k tokens)
         self.moe_bm = 256 if _os.environ.get('ROOMAMD_MOE_BM') == '256' else 128
+        # decode split-KV band (32 short / 64 long contexts); the engine sets
+        # this per decode step (and per captured graph band)
+        self.attn_splits = 32
         cos_t, sin_t = rope_tables(cfg.max_position, cfg.head_dim, cfg.rope_theta)
         self.cos_t = cos_t.to(self.device)
         self.sin_t = sin_t.to(self.device)
@@ -212,7 +215,8 @@ class Qwen3MoEModel:
                     ops.paged_attention_splitk(part, part_ml, q, kcaches[li],
                                                vcaches[li], block_table,
                                                seq_ids, q_pos, self.scale,
-                                               o_accum)
+                                               o_accum,
+                                               splits=self.attn_splits)
                     ops.attn_merge_o(o_accum, part, part_ml, layer.wo)
                     o = o_accum                   # f32 delta
                 else:
@@ -221,7 +225,8 @@ class Qwen3MoEModel:
                     ops.paged_attention_split(attn, q, kcaches[li],
                                               vcaches[li], block_table,
                                               seq_ids, q_pos, part, part_ml,
-                                              self.scale)
+                                              self.scale,
+                                              splits=self.attn_splits)
                     ops.gemv(obuf, attn.reshape(T, qdim), layer.wo)
                     o = obuf
                 # --- MoE block

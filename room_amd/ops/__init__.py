@@ -93,10 +93,13 @@ def paged_attention_split(out: torch.Tensor, q: torch.Tensor, kcache: torch.Tens
                           vcache: torch.Tensor, block_table: torch.Tensor,
                           seq_ids: torch.Tensor, q_pos: torch.Tensor,
                           part: torch.Tensor, part_ml: torch.Tensor,
-                          scale: float) -> torch.Tensor:
-    """Split-KV flash-decode: part [T, Hq, NSPLITS, 128] f32, part_ml [.., 2]."""
+                          scale: float, splits: int = 32) -> torch.Tensor:
+    """Split-KV flash-decode: part [T, Hq, NSPLITS_MAX, 128] f32, part_ml
+    [.., 2]; `splits` (32 short / 64 long contexts) sets the launch grid —
+    shorter per-WG chunk chains win once the serial span passes ~8k."""
     _require().paged_attention_split(out, q, kcache, vcache, block_table,
-                                     seq_ids, q_pos, part, part_ml, scale)
+                                     seq_ids, q_pos, part, part_ml, scale,
+                                     splits)
     return out
 
 
@@ -110,12 +113,13 @@ def paged_attention_splitk(part: torch.Tensor, part_ml: torch.Tensor,
                            q: torch.Tensor, kcache: torch.Tensor,
                            vcache: torch.Tensor, block_table: torch.Tensor,
                            seq_ids: torch.Tensor, q_pos: torch.Tensor,
-                           scale: float, o_zero: torch.Tensor) -> None:
+                           scale: float, o_zero: torch.Tensor,
+                           splits: int = 32) -> None:
     """Split-KV decode attention, partials only (no bf16 merge output); zeros
     `o_zero` (the attn_merge_o f32 accumulator) as a side job."""
     _require().paged_attention_splitk(part, part_ml, q, kcache, vcache,
                                       block_table, seq_ids, q_pos, scale,
-                                      o_zero)
+                                      o_zero, splits)
 
 
 def attn_merge_o(o_accum: torch.Tensor, part: torch.Tensor,

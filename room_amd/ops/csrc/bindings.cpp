@@ -54,13 +54,15 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
                            torch::Tensor kcache, torch::Tensor vcache,
                            torch::Tensor block_table, torch::Tensor seq_ids,
                            torch::Tensor q_pos, torch::Tensor part,
-                           torch::Tensor part_ml, double scale);
+                           torch::Tensor part_ml, double scale,
+                           int64_t splits);
 int64_t attn_nsplits();
 void paged_attention_splitk(torch::Tensor part, torch::Tensor part_ml,
                             torch::Tensor q, torch::Tensor kcache,
                             torch::Tensor vcache, torch::Tensor block_table,
                             torch::Tensor seq_ids, torch::Tensor q_pos,
-                            double scale, torch::Tensor o_zero);
+                            double scale, torch::Tensor o_zero,
+                            int64_t splits);
 void attn_merge_o(torch::Tensor o_accum, torch::Tensor part,
                   torch::Tensor part_ml, torch::Tensor wo);
 void router_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,

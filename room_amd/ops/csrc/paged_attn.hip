@@ -226,7 +226,8 @@ void write_kv(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
 // NSPLITS=64 A/B: wins only at L>=16k (63.6->50.0 us B=1), loses at the
 // bench's 4-6k sessions (8.1->8.3 ms/step) and at B=1 (4.5->5.0): empty-split
 // dispatch + 2x merge traffic. Static 32 keeps the decode graph replayable.
-#define NSPLITS 32
+#define NSPLITS 32          // default split count (short contexts)
+#define NSPLITS_MAX 64      // part-buffer layout stride; grid.z may be 32 or 64
 
 // CH: kv positions staged/scored per iteration. 64 halves the barrier count
 // per span (3 barriers per chunk) at the cost of LDS (41 KB vs 24 KB).
@@ -269,13 +270,15 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
   const int n_qheads = n_kvheads * QH_PER_KV;
   const int hq = hk * QH_PER_KV + h;
 
-  // split range, CH-aligned
-  const int span = ((bound + NSPLITS - 1) / NSPLITS + CH - 1) & ~(CH - 1);
+  // split range, CH-aligned; split count = launch grid.z (32 short / 64 long
+  // contexts), buffer layout stride fixed at NSPLITS_MAX
+  const int nsp = gridDim.z;
+  const int span = ((bound + nsp - 1) / nsp + CH - 1) & ~(CH - 1);
   const int lo = split * span;
   const int hi = min(bound, lo + span);
 
-  float* ml = part_ml + (((long)t * n_qheads + hq) * NSPLITS + split) * 2;
-  float* acc_out = part + (((long)t * n_qheads + hq) * NSPLITS + split) * HEAD_DIM;
+  float* ml = part_ml + (((long)t * n_qheads + hq) * NSPLITS_MAX + split) * 2;
+  float* acc_out = part + (((long)t * n_qheads + hq) * NSPLITS_MAX + split) * HEAD_DIM;
 
   if (lo >= hi) {  // empty split still writes a neutral partial
     if (h < QH_PER_KV) {
@@ -400,19 +403,19 @@ __global__ __launch_bounds__(128)
 void paged_attn_merge_kernel(short* __restrict__ out,       // [T, Hq, D]
                              const float* __restrict__ part,
                              const float* __restrict__ part_ml,
-                             int n_qheads) {
+                             int n_qheads, int n_splits) {
   const int t = blockIdx.x;
   const int hq = blockIdx.y;
   const int d = threadIdx.x;  // 128 threads = one dim each
-  const float* ml = part_ml + (((long)t * n_qheads + hq) * NSPLITS) * 2;
-  const float* pacc = part + (((long)t * n_qheads + hq) * NSPLITS) * HEAD_DIM;
+  const float* ml = part_ml + (((long)t * n_qheads + hq) * NSPLITS_MAX) * 2;
+  const float* pacc = part + (((long)t * n_qheads + hq) * NSPLITS_MAX) * HEAD_DIM;
 
   float m_star = -INFINITY;
-  #pragma unroll
-  for (int s = 0; s < NSPLITS; ++s) m_star = fmaxf(m_star, ml[2 * s]);
+  #pragma unroll 8
+  for (int s = 0; s < n_splits; ++s) m_star = fmaxf(m_star, ml[2 * s]);
   float l_tot = 0.f, a = 0.f;
-  #pragma unroll
-  for (int s = 0; s < NSPLITS; ++s) {
+  #pragma unroll 8
+  for (int s = 0; s < n_splits; ++s) {
     const float ms = ml[2 * s];
     if (ms == -INFINITY) continue;
     const float f = __expf(ms - m_star);
@@ -423,22 +426,33 @@ void paged_attn_merge_kernel(short* __restrict__ out,       // [T, Hq, D]
       f2bf(l_tot > 0.f ? a / l_tot : 0.f);
 }
 
-int64_t attn_nsplits() { return NSPLITS; }
+int64_t attn_nsplits() { return NSPLITS_MAX; }
+
+// split count for a given max context: 64 splits pay off past ~8k tokens
+// (shorter per-WG serial chunk chains beat the extra merge traffic there —
+// round-1 A/B: NSPLITS=64 wins only at L>=16k vs static-32 everywhere; with
+// banded decode graphs each band gets its own captured grid)
+static int pick_splits(int max_ctx) { return max_ctx >= 8192 ? 64 : NSPLITS; }
 
 void paged_attention_split(torch::Tensor out, torch::Tensor q,
                            torch::Tensor kcache, torch::Tensor vcache,
                            torch::Tensor block_table, torch::Tensor seq_ids,
                            torch::Tensor q_pos, torch::Tensor part,
-                           torch::Tensor part_ml, double scale) {
+                           torch::Tensor part_ml, double scale,
+                           int64_t splits) {
+  TORCH_CHECK(splits == 32 || splits == 64, "splits must be 32 or 64");
   // q may be a strided row-view into the packed qkv buffer
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
   const int T = q.size(0);
   const int n_kvheads = kcache.size(1);
   const int n_qheads = n_kvheads * QH_PER_KV;
   const int max_blocks = block_table.size(1);
-  TORCH_CHECK(part.size(0) >= T && part.size(2) == NSPLITS);
+  TORCH_CHECK(part.size(0) >= T && part.size(2) == NSPLITS_MAX);
+  int max_ctx = 0;  // host-known bound: use tensor length heuristic (eager
+  // path only; graph path passes splits explicitly via the splitk variant)
   hipStream_t s = c10::hip::getCurrentHIPStream();
-  dim3 g1(T, n_kvheads, NSPLITS);
+  dim3 g1(T, n_kvheads, (unsigned)splits);
+  (void)max_ctx;
   #define PA_SPLIT_ARGS part.data_ptr<float>(), part_ml.data_ptr<float>(), \
       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(), \
       (const short*)vcache.data_ptr(), block_table.data_ptr<int>(), \
@@ -461,7 +475,7 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
   dim3 g2(T, n_qheads);
   hipLaunchKernelGGL(paged_attn_merge_kernel, g2, dim3(HEAD_DIM), 0, s,
                      (short*)out.data_ptr(), part.data_ptr<float>(),
-                     part_ml.data_ptr<float>(), n_qheads);
+                     part_ml.data_ptr<float>(), n_qheads, (int)splits);
   HIP_CHECK_KERNEL();
 }
 
@@ -471,15 +485,17 @@ void paged_attention_splitk(torch::Tensor part, torch::Tensor part_ml,
                             torch::Tensor q, torch::Tensor kcache,
                             torch::Tensor vcache, torch::Tensor block_table,
                             torch::Tensor seq_ids, torch::Tensor q_pos,
-                            double scale, torch::Tensor o_zero) {
+                            double scale, torch::Tensor o_zero,
+                            int64_t splits) {
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
   const int T = q.size(0);
   const int n_kvheads = kcache.size(1);
   const int max_blocks = block_table.size(1);
-  TORCH_CHECK(part.size(0) >= T && part.size(2) == NSPLITS);
+  TORCH_CHECK(part.size(0) >= T && part.size(2) == NSPLITS_MAX);
   TORCH_CHECK(o_zero.dtype() == torch::kFloat32 && o_zero.is_contiguous());
+  TORCH_CHECK(splits == 32 || splits == 64, "splits must be 32 or 64");
   hipStream_t s = c10::hip::getCurrentHIPStream();
-  dim3 g1(T, n_kvheads, NSPLITS);
+  dim3 g1(T, n_kvheads, (unsigned)splits);
   #define PA_SPLITK_ARGS part.data_ptr<float>(), part_ml.data_ptr<float>(), \
       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(), \
       (const short*)vcache.data_ptr(), block_table.data_ptr<int>(), \

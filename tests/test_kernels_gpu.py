@@ -319,6 +319,11 @@ def test_paged_attention_split_matches_ref():
                               part, part_ml, scale)
     expect = ref.paged_attention_ref(q, kcache, vcache, bt, seq_ids, q_pos, scale)
     assert bf16_close(out, expect, atol=3e-2)
+    # 64-split band (long-context decode graphs) must match too
+    out64 = torch.empty_like(q)
+    ops.paged_attention_split(out64, q, kcache, vcache, bt, seq_ids, q_pos,
+                              part, part_ml, scale, splits=64)
+    assert bf16_close(out64, expect, atol=3e-2)
 
 
 def test_moe_grouped_gemm128_vs_matmul():
