@@ -389,7 +389,7 @@ class LocalEngine:
             # context band: 64 attention splits once any sequence's KV span
             # passes 8k (shorter per-WG serial chunk chains); banded graphs
             # keep each capture's grid fixed
-            band = 64 if max(positions) + steps >= 8192 else 32
+            band = 64 if max(positions) + steps >= 16384 else 32
             self.model.attn_splits = band
             key = (bucket, band, round(r0.temperature, 3), round(r0.top_p, 3),
                    r0.top_k)
@@ -426,7 +426,7 @@ class LocalEngine:
                 self._graphs_broken = True
 
         # eager fallback: one step
-        self.model.attn_splits = 64 if max(positions) >= 8192 else 32
+        self.model.attn_splits = 64 if max(positions) >= 16384 else 32
         for r in reqs:
             self.cache.ensure_capacity(r.slot, r.pos + 1)
         for r in reqs:

@@ -217,7 +217,8 @@ class Qwen3MoEModel:
                                                seq_ids, q_pos, self.scale,
                                                o_accum,
                                                splits=self.attn_splits)
-                    ops.attn_merge_o(o_accum, part, part_ml, layer.wo)
+                    ops.attn_merge_o(o_accum, part, part_ml, layer.wo,
+                                     splits=self.attn_splits)
                     o = o_accum                   # f32 delta
                 else:
                     attn = torch.empty(T, cfg.num_q_heads, cfg.head_dim,

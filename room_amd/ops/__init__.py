@@ -123,11 +123,12 @@ def paged_attention_splitk(part: torch.Tensor, part_ml: torch.Tensor,
 
 
 def attn_merge_o(o_accum: torch.Tensor, part: torch.Tensor,
-                 part_ml: torch.Tensor, wo: torch.Tensor) -> torch.Tensor:
+                 part_ml: torch.Tensor, wo: torch.Tensor,
+                 splits: int = 32) -> torch.Tensor:
     """Fused split-partial merge + O-projection: o_accum[B, H] (f32,
     pre-zeroed) += Wo @ merged-attention. Replaces the standalone merge
     kernel + O GEMV on the decode path."""
-    _require().attn_merge_o(o_accum, part, part_ml, wo)
+    _require().attn_merge_o(o_accum, part, part_ml, wo, splits)
     return o_accum
 
 

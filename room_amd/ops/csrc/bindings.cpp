@@ -64,7 +64,7 @@ void paged_attention_splitk(torch::Tensor part, torch::Tensor part_ml,
                             double scale, torch::Tensor o_zero,
                             int64_t splits);
 void attn_merge_o(torch::Tensor o_accum, torch::Tensor part,
-                  torch::Tensor part_ml, torch::Tensor wo);
+                  torch::Tensor part_ml, torch::Tensor wo, int64_t splits);
 void router_addnorm(torch::Tensor y, torch::Tensor x, torch::Tensor delta,
                     torch::Tensor x_out, torch::Tensor xn_out,
                     torch::Tensor gamma, torch::Tensor w, double eps);

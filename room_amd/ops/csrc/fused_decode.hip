@@ -21,7 +21,7 @@
 #include "common.h"
 
 #define HEAD_DIM 128
-#define NSPLITS 32          // must match paged_attn.hip
+#define NSPLITS_MAX 64      // partial-buffer stride (paged_attn.hip)
 #define AMO_NT 128          // output-tile width of attn_merge_o
 
 // ------------------------------------------------------------ attn_merge_o
@@ -33,7 +33,7 @@ void attn_merge_o_kernel(float* __restrict__ o_accum,      // [BN, N] (pre-zeroe
                          const float* __restrict__ part,    // [BN, Hq, NS, D]
                          const float* __restrict__ part_ml, // [BN, Hq, NS, 2]
                          const short* __restrict__ wo,      // [N, Hq*D]
-                         int n_qheads, int N) {
+                         int n_qheads, int N, int n_splits) {
   const int n0 = blockIdx.x * AMO_NT;
   const int hq = blockIdx.y;
   const int tid = threadIdx.x;
@@ -67,16 +67,15 @@ void attn_merge_o_kernel(float* __restrict__ o_accum,      // [BN, N] (pre-zeroe
     const int sg = tid >> 5;                       // split group 0..7
     const int d32 = tid & 31;
     const int d4 = d32 * 4;
-    constexpr int SPG = NSPLITS / 8;               // splits per group
+    const int SPG = n_splits / 8;                  // splits per group
     #pragma unroll
     for (int b = 0; b < BN; ++b) {
-      const float* ml = part_ml + (((long)b * n_qheads + hq) * NSPLITS) * 2;
-      const float* pacc = part + (((long)b * n_qheads + hq) * NSPLITS) * HEAD_DIM;
+      const float* ml = part_ml + (((long)b * n_qheads + hq) * NSPLITS_MAX) * 2;
+      const float* pacc = part + (((long)b * n_qheads + hq) * NSPLITS_MAX) * HEAD_DIM;
       float m_star = -INFINITY;
-      #pragma unroll
-      for (int s = 0; s < NSPLITS; ++s) m_star = fmaxf(m_star, ml[2 * s]);
+      #pragma unroll 8
+      for (int s = 0; s < n_splits; ++s) m_star = fmaxf(m_star, ml[2 * s]);
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-      #pragma unroll
       for (int i = 0; i < SPG; ++i) {
         const int s = sg * SPG + i;
         const float ms = ml[2 * s];
@@ -93,8 +92,8 @@ void attn_merge_o_kernel(float* __restrict__ o_accum,      // [BN, N] (pre-zeroe
       // l_tot: every thread recomputes the full sum from ml (L1-hot, cheap)
       if (sg == 0) {
         float lt = 0.f;
-        #pragma unroll
-        for (int s = 0; s < NSPLITS; ++s) {
+        #pragma unroll 8
+        for (int s = 0; s < n_splits; ++s) {
           const float ms = ml[2 * s];
           if (ms != -INFINITY) lt += ml[2 * s + 1] * __expf(ms - m_star);
         }
@@ -150,13 +149,14 @@ void attn_merge_o_kernel(float* __restrict__ o_accum,      // [BN, N] (pre-zeroe
 }
 
 void attn_merge_o(torch::Tensor o_accum, torch::Tensor part,
-                  torch::Tensor part_ml, torch::Tensor wo) {
+                  torch::Tensor part_ml, torch::Tensor wo, int64_t splits) {
   const int B = o_accum.size(0), N = o_accum.size(1);
   const int n_qheads = part.size(1);
   TORCH_CHECK(B >= 1 && B <= 8, "attn_merge_o handles B<=8 (decode)");
   TORCH_CHECK(o_accum.dtype() == torch::kFloat32 && part.dtype() == torch::kFloat32);
   TORCH_CHECK(wo.dtype() == torch::kBFloat16);
-  TORCH_CHECK(part.size(2) == NSPLITS && part.size(3) == HEAD_DIM);
+  TORCH_CHECK(part.size(2) == NSPLITS_MAX && part.size(3) == HEAD_DIM);
+  TORCH_CHECK(splits == 32 || splits == 64);
   TORCH_CHECK(N % AMO_NT == 0);
   TORCH_CHECK(wo.size(0) == N && wo.size(1) == n_qheads * HEAD_DIM);
   dim3 grid(N / AMO_NT, n_qheads), block(256);
@@ -166,7 +166,7 @@ void attn_merge_o(torch::Tensor o_accum, torch::Tensor part,
     case BN: hipLaunchKernelGGL((attn_merge_o_kernel<BN>), grid, block, 0, s, \
         o_accum.data_ptr<float>(), part.data_ptr<float>(), \
         part_ml.data_ptr<float>(), (const short*)wo.data_ptr(), \
-        n_qheads, N); break;
+        n_qheads, N, (int)splits); break;
     AMO_CASE(1) AMO_CASE(2) AMO_CASE(3) AMO_CASE(4)
     AMO_CASE(5) AMO_CASE(6) AMO_CASE(7) AMO_CASE(8)
 #undef AMO_CASE
