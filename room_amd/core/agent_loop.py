@@ -38,6 +38,14 @@ from .log_buffer import CycleLogBuffer
 from .rate_limit import RateLimitError, abortable_sleep, detect_rate_limit
 from .skills import load_skills_for_agent
 
+# Model-B (soft) queen policy (reference agent-loop.ts:22-28): the queen is
+# control-plane; direct web/browser execution is a logged deviation
+QUEEN_EXECUTION_TOOLS = {"room_web_search", "room_web_fetch", "room_browser"}
+QUEEN_POLICY_WIP_HINT = (
+    "[policy] Queen control-plane mode: delegate execution tasks to workers "
+    "with room_delegate_task, then monitor, unblock, and report outcomes. "
+    "Avoid direct web/browser execution.")
+
 
 @dataclass
 class LoopState:
@@ -228,7 +236,11 @@ class AgentLoopManager:
 
         tools = agent_tools.tools_for_role("queen" if is_queen else worker.get("role"))
 
+        execution_tools_used: set[str] = set()
+
         def tool_executor(call: ToolCall) -> str:
+            if is_queen and call.name in QUEEN_EXECUTION_TOOLS:
+                execution_tools_used.add(call.name)
             with self.ldb as db:
                 return agent_tools.execute_agent_tool(db, room_id, worker_id, call,
                                                       embed_fn=self.embed_fn,
@@ -279,6 +291,20 @@ class AgentLoopManager:
             q.log_room_activity(db, room_id, "cycle",
                                 f"{worker['name']}: {summary}",
                                 actor_id=worker_id, is_public=False)
+            # Model-B queen policy deviation (agent-loop.ts:707-728): log it
+            # and pin the control-plane hint into WIP
+            if is_queen and execution_tools_used:
+                used = ", ".join(sorted(execution_tools_used))
+                q.log_room_activity(
+                    db, room_id, "system",
+                    f"Queen policy deviation: execution tool use detected "
+                    f"({used}).", actor_id=worker_id)
+                fresh = q.get_worker(db, worker_id)
+                existing = ((fresh or {}).get("wip") or "").strip()
+                if QUEEN_POLICY_WIP_HINT not in existing:
+                    nxt = (f"{existing}\n\n{QUEEN_POLICY_WIP_HINT}"
+                           if existing else QUEEN_POLICY_WIP_HINT)
+                    q.set_worker_wip(db, worker_id, nxt[:2000])
             # auto-WIP fallback: if the agent didn't save WIP and produced text,
             # keep a trace so the next cycle continues (:854-863)
             w = q.get_worker(db, worker_id)

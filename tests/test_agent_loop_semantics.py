@@ -127,3 +127,41 @@ def test_trigger_agent_wakes_waiting_loop(mgr):
     state.wake_event.clear()
     m.wake_room_workers(room["id"], exclude=queen["id"])
     assert not state.wake_event.is_set()
+
+
+def test_queen_policy_deviation_logged(mgr):
+    """Model-B soft policy (agent-loop.ts:22-28, 707-728): a queen using an
+    execution tool gets a deviation activity row + the control-plane WIP
+    hint; workers don't."""
+    import asyncio
+
+    from room_amd.core.agent_loop import QUEEN_POLICY_WIP_HINT
+    from room_amd.engine.providers import register_engine
+
+    class WebUsingEngine:
+        def chat(self, messages, tools, options):
+            return ('<tool_call>{"name": "room_web_search", '
+                    '"arguments": {"query": "x"}}</tool_call>', 10, 5)
+
+    m, ldb = mgr
+    register_engine("web-stub", WebUsingEngine())
+    room, queen = _mkroom(ldb)
+    with ldb as db:
+        q.update_room(db, room["id"], worker_model="web-stub")
+        q.update_worker(db, queen["id"], model="web-stub")
+        w = q.create_worker(db, "researcher", "p", role="researcher",
+                            room_id=room["id"], model="web-stub")
+    out = asyncio.run(m.run_cycle(room["id"], queen["id"], max_turns=2))
+    assert out["result"].success
+    with ldb as db:
+        acts = q.get_room_activity(db, room["id"])
+        assert any("Queen policy deviation" in a["summary"] for a in acts)
+        fresh = q.get_worker(db, queen["id"])
+        assert QUEEN_POLICY_WIP_HINT in (fresh.get("wip") or "")
+        # worker using the same tool: no deviation row added
+        n_dev = sum(1 for a in acts if "deviation" in a["summary"])
+    asyncio.run(m.run_cycle(room["id"], w["id"], max_turns=2))
+    with ldb as db:
+        acts2 = q.get_room_activity(db, room["id"])
+        assert sum(1 for a in acts2
+                   if "deviation" in a["summary"]) == n_dev
