@@ -129,7 +129,10 @@ def test_escalation_alert_relay(tmp_path, monkeypatch):
     async def run():
         rt = ServerRuntime(ldb, TaskRunner(ldb), bus=EventBus())
         task = asyncio.create_task(rt._alert_relay_loop())
-        await asyncio.sleep(0.3)
+        for _ in range(60):                 # poll: loop start can lag under load
+            await asyncio.sleep(0.05)
+            if (tmp_path / "outbox.jsonl").exists():
+                break
         rt._stop.set()
         task.cancel()
 
