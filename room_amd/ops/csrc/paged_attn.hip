@@ -399,11 +399,12 @@ void paged_attn_split_kernel(float* __restrict__ part,   // [T, Hq, NSPLITS, D]
 }
 
 // merge: one wave per (t, hq); lane d-pairs combine the NSPLITS partials
+template <int NS>
 __global__ __launch_bounds__(128)
 void paged_attn_merge_kernel(short* __restrict__ out,       // [T, Hq, D]
                              const float* __restrict__ part,
                              const float* __restrict__ part_ml,
-                             int n_qheads, int n_splits) {
+                             int n_qheads) {
   const int t = blockIdx.x;
   const int hq = blockIdx.y;
   const int d = threadIdx.x;  // 128 threads = one dim each
@@ -411,11 +412,11 @@ void paged_attn_merge_kernel(short* __restrict__ out,       // [T, Hq, D]
   const float* pacc = part + (((long)t * n_qheads + hq) * NSPLITS_MAX) * HEAD_DIM;
 
   float m_star = -INFINITY;
-  #pragma unroll 8
-  for (int s = 0; s < n_splits; ++s) m_star = fmaxf(m_star, ml[2 * s]);
+  #pragma unroll
+  for (int s = 0; s < NS; ++s) m_star = fmaxf(m_star, ml[2 * s]);
   float l_tot = 0.f, a = 0.f;
-  #pragma unroll 8
-  for (int s = 0; s < n_splits; ++s) {
+  #pragma unroll
+  for (int s = 0; s < NS; ++s) {
     const float ms = ml[2 * s];
     if (ms == -INFINITY) continue;
     const float f = __expf(ms - m_star);
@@ -473,9 +474,14 @@ void paged_attention_split(torch::Tensor out, torch::Tensor q,
   #undef PA_SPLIT_ARGS
   HIP_CHECK_KERNEL();
   dim3 g2(T, n_qheads);
-  hipLaunchKernelGGL(paged_attn_merge_kernel, g2, dim3(HEAD_DIM), 0, s,
-                     (short*)out.data_ptr(), part.data_ptr<float>(),
-                     part_ml.data_ptr<float>(), n_qheads, (int)splits);
+  if (splits == 64)
+    hipLaunchKernelGGL(paged_attn_merge_kernel<64>, g2, dim3(HEAD_DIM), 0, s,
+                       (short*)out.data_ptr(), part.data_ptr<float>(),
+                       part_ml.data_ptr<float>(), n_qheads);
+  else
+    hipLaunchKernelGGL(paged_attn_merge_kernel<32>, g2, dim3(HEAD_DIM), 0, s,
+                       (short*)out.data_ptr(), part.data_ptr<float>(),
+                       part_ml.data_ptr<float>(), n_qheads);
   HIP_CHECK_KERNEL();
 }
 
