@@ -73,6 +73,14 @@ class SessionAdmitter:
                                  + req.prompt_tokens[-tail:])
 
         sess = self.sessions.get(req.session_key) if req.session_key else None
+        if sess is not None and sess.slot in self.active_slots:
+            # another request is mid-flight on this session's slot (e.g. a
+            # timed-out caller retrying before its cancelled request reached
+            # a scheduler boundary). Never share or free a live slot: drop
+            # the session binding and admit fresh — the in-flight finish
+            # sees the missing session and frees the old slot itself.
+            self.sessions.pop(req.session_key, None)
+            sess = None
         if sess is not None:
             # reuse the longest common token prefix: the session keeps its
             # slot and blocks; position rolls back to the divergence point.

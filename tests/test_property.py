@@ -90,3 +90,82 @@ def test_cron_step_interval_spacing(step):
     assert a.minute % step == 0 and b.minute % step == 0
     assert c.matches(a) and c.matches(b)
     assert (b - a).total_seconds() <= step * 60 + 3600  # step or rollover
+
+
+# ---------------------------------------------------------------- admission
+# Stateful property test: arbitrary admit/finish/release/evict interleavings
+# must preserve the slot/block accounting invariants (the class of bug that
+# broke round 1 shipped because only example-based GPU tests covered this).
+
+from hypothesis.stateful import (Bundle, RuleBasedStateMachine, invariant,
+                                 rule)
+
+
+class AdmissionMachine(RuleBasedStateMachine):
+    def __init__(self):
+        super().__init__()
+        import torch
+
+        from room_amd.engine.admission import CacheFull, SessionAdmitter
+        from room_amd.engine.kv_cache import PagedKVCache
+        self.CacheFull = CacheFull
+        self.cache = PagedKVCache(num_layers=1, num_kv_heads=1, head_dim=8,
+                                  num_blocks=40, max_seqs=5,
+                                  max_blocks_per_seq=64,
+                                  device=torch.device("cpu"))
+        self.adm = SessionAdmitter(self.cache, max_position=512)
+        self.active: list = []
+        self.counter = 0
+
+    reqs = Bundle("reqs")
+
+    @rule(target=reqs, sess=st.booleans(), plen=st.integers(1, 60))
+    def admit(self, sess, plen):
+        class R:
+            pass
+        r = R()
+        self.counter += 1
+        r.prompt_tokens = list(range(self.counter, self.counter + plen))
+        r.session_key = f"s{self.counter % 3}" if sess else None
+        r.max_new_tokens = 8
+        r.out_tokens = []
+        r.slot, r.pos, r.pending_prefill, r.prefill_tokens_run = -1, 0, [], 0
+        try:
+            self.adm.admit(r)
+        except self.CacheFull:
+            r.slot = -1
+            return r
+        assert r.pending_prefill, "admitted request must have prefill work"
+        self.cache.ensure_capacity(r.slot, r.pos + len(r.pending_prefill) + 2)
+        self.active.append(r)
+        return r
+
+    @rule(r=reqs, ok=st.booleans())
+    def finish(self, r, ok):
+        if r not in self.active:
+            return
+        self.active.remove(r)
+        r.out_tokens = [1, 2, 3]
+        self.adm.finish(r, (r.prompt_tokens + r.out_tokens[:-1]) if ok else None)
+
+    @rule(key=st.sampled_from(["s0", "s1", "s2", "nope"]))
+    def release(self, key):
+        self.adm.release(key)
+
+    @invariant()
+    def accounting_holds(self):
+        c = self.cache
+        # every slot is exactly one of: free, or allocated (tracked in seq_len)
+        assert len(c.free_slots) + len(c.seq_len) == c.max_seqs
+        assert set(c.free_slots).isdisjoint(c.seq_len.keys())
+        # block conservation (block 0 reserved)
+        used = sum(c.seq_nblocks.values())
+        assert used + c.blocks_free() == c.num_blocks - 1
+        # sessions and active requests only reference live slots
+        for s in self.adm.sessions.values():
+            assert s.slot in c.seq_len
+        for slot in self.adm.active_slots:
+            assert slot in c.seq_len
+
+
+TestAdmissionMachine = AdmissionMachine.TestCase
