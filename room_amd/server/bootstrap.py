@@ -75,6 +75,13 @@ def serve(port: int = 3700, host: str = "127.0.0.1",
     import uvicorn
 
     app, components = build_server(db_path=db_path)
+    # register the MCP server into AI-client configs that exist on this
+    # machine (reference index.ts:729-864; ROOMAMD_SKIP_MCP_REGISTER=1 skips)
+    from .mcp_register import register_mcp_globally
+    patched = register_mcp_globally(db_path or default_db_path())
+    hit = [k for k, v in patched.items() if v]
+    if hit:
+        print(f"MCP registered into: {', '.join(hit)}")
     # persist the port for MCP cross-process nudges (reference: api.port file)
     try:
         (data_dir() / "api.port").write_text(str(port))
