@@ -11,6 +11,7 @@ from __future__ import annotations
 
 import asyncio
 import json
+import os
 import time
 from collections import defaultdict, deque
 from typing import Optional
@@ -1389,6 +1390,59 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
             return {"code": q.get_setting(db, "keeper_referral_code")}
 
     # -------------------------------------------------------------- UI stub
+
+    # ------------------------------------------------- server self-restart
+    # (reference index.ts:526-576: /api/server/restart re-execs the process;
+    # /api/server/update-restart stages the update first; the download
+    # endpoint triggers staging without restarting)
+
+    def _schedule_restart(delay_s: float = 0.5) -> None:
+        import sys as _sys
+        import threading as _threading
+
+        def _go():
+            time.sleep(delay_s)
+            os.execv(_sys.executable, [_sys.executable, "-m", "room_amd.cli",
+                                       "serve"])
+        t = _threading.Thread(target=_go, daemon=True)
+        t.start()
+
+    @app.post("/api/server/restart")
+    async def server_restart(role: str = Depends(rw)):
+        if os.environ.get("ROOMAMD_ALLOW_RESTART") != "1":
+            # in-place re-exec is opt-in (test/CI processes must not re-exec)
+            return {"ok": False,
+                    "error": "restart disabled (set ROOMAMD_ALLOW_RESTART=1)"}
+        _schedule_restart()
+        return {"ok": True, "restarting": True}
+
+    def _update_checker():
+        uc = getattr(state, "update_checker", None)
+        if uc is None:
+            from ..core.update_checker import UpdateChecker
+            from .auth import data_dir
+            uc = state.update_checker = UpdateChecker(
+                "0.1.0", data_dir=data_dir())
+        return uc
+
+    @app.get("/api/status/update/download")
+    async def update_download(role: str = Depends(rw)):
+        uc = _update_checker()
+        if uc.latest_version is None:
+            uc.check()
+        return uc.stage_update()
+
+    @app.post("/api/server/update-restart")
+    async def server_update_restart(role: str = Depends(rw)):
+        uc = _update_checker()
+        staged = uc.stage_update()
+        if not staged.get("staged"):
+            return {"ok": False, "staged": staged}
+        if os.environ.get("ROOMAMD_ALLOW_RESTART") != "1":
+            return {"ok": False, "staged": staged,
+                    "error": "restart disabled (set ROOMAMD_ALLOW_RESTART=1)"}
+        _schedule_restart()
+        return {"ok": True, "staged": staged, "restarting": True}
 
     # -------------------------------------------- UI client-contract routes
     # Completes the endpoint set the reference SPA's typed client calls

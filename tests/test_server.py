@@ -441,3 +441,15 @@ def test_contacts_badges_and_settings_key(server):
                        headers=h).json()["connected"] is False
     assert client.post("/api/contacts/telegram/start",
                        headers=h).status_code == 503
+
+
+def test_server_restart_and_update_endpoints(server):
+    """Self-restart surface (index.ts:526-576): disabled by default (no
+    re-exec in tests), update download stages-or-reports offline."""
+    client, h, *_ = server
+    r = client.post("/api/server/restart", headers=h).json()
+    assert r["ok"] is False and "disabled" in r["error"]
+    d = client.get("/api/status/update/download", headers=h).json()
+    assert d["staged"] is False          # offline: nothing to download
+    ur = client.post("/api/server/update-restart", headers=h).json()
+    assert ur["ok"] is False
