@@ -186,6 +186,7 @@ def main() -> None:
         s = engine.stats
         print(f"[rank {ctx.rank}] engine stats: "
               f"prefill {s['prefill_tokens']} tok in {s['prefill_time']:.2f}s "
+              f"(prep {s['prefill_prep_time']:.2f}s enq {s['prefill_enq_time']:.2f}s) "
               f"({s['prefill_tokens']/max(s['prefill_time'],1e-9):.0f} tok/s), "
               f"decode {s['decode_tokens']} tok in {s['decode_steps']} steps "
               f"{s['decode_time']:.2f}s "

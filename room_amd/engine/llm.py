@@ -182,6 +182,7 @@ class LocalEngine:
         self._stop = False
         self.stats = {"decode_steps": 0, "decode_tokens": 0, "prefill_tokens": 0,
                       "decode_time": 0.0, "prefill_time": 0.0,
+                      "prefill_prep_time": 0.0, "prefill_enq_time": 0.0,
                       "graphs_captured": 0}
         self._thread = threading.Thread(target=self._scheduler_loop, daemon=True,
                                         name="room-amd-engine")
@@ -329,6 +330,7 @@ class LocalEngine:
 
     def _prefill_step(self, reqs: list[GenRequest]) -> None:
         t0 = time.time()
+        host_t0 = t0
         budget = PREFILL_CHUNK
         tokens, seq_ids, q_pos, last_rows, sampled_reqs = [], [], [], [], []
         segments = []  # (row0, count) per request — host-side q-tile info
@@ -355,9 +357,11 @@ class LocalEngine:
         rows_t = torch.tensor(last_rows, dtype=torch.int64, device=dev)
         qtiles = (ops.build_qtile_desc(segments, dev)
                   if len(tokens) > 8 else None)
+        self.stats["prefill_prep_time"] += time.time() - host_t0
         logits = self.model.forward(tokens_t, seq_t, pos_t, self.cache.block_table,
                                     self.cache.kcaches, self.cache.vcaches,
                                     logits_rows=rows_t, qtile_desc=qtiles)
+        self.stats["prefill_enq_time"] += time.time() - host_t0
         self.stats["prefill_tokens"] += len(tokens)
         if sampled_reqs:
             self._sample_and_append(sampled_reqs, logits)
