@@ -137,6 +137,88 @@ class _DecodeGraph:
         return self.hist_host[:k * b].tolist()
 
 
+PREFILL_BUCKETS = (512, 1024, 2048, 4096)   # graph-captured chunk shapes
+PREFILL_MAX_ROWS = 8                         # fixed logits-row slots
+
+
+class _PrefillGraph:
+    """One captured hipGraph per prefill bucket size.
+
+    The prefill forward was built host-sync-free (device-side MoE tile
+    descriptors, tensor-driven rope/attention), but launching its ~700
+    python-level ops costs ~30 ms per chunk — more than the GPU compute it
+    enqueues, so prefill ran launch-bound. Chunks are padded to a few fixed
+    bucket shapes (pad rows target the engine's scrap pad slot; zero-count
+    q-tiles no-op in flash_prefill) and each bucket's whole forward replays
+    as one graph. logits are always computed for PREFILL_MAX_ROWS fixed row
+    slots; the scheduler slices the real ones."""
+
+    def __init__(self, engine: "LocalEngine", tb: int):
+        dev = engine.device
+        self.tb = tb
+        self.pad_slot = engine.pad_slot
+        qt = 32  # flash_prefill q-tile rows (FP_QTOK)
+        # worst case: 8 request segments + 1 pad segment, each wasting <1 tile
+        self.gmax = tb // qt + PREFILL_MAX_ROWS + 2
+        self.tok = torch.zeros(tb, dtype=torch.int64, device=dev)
+        self.seq = torch.full((tb,), engine.pad_slot, dtype=torch.int32,
+                              device=dev)
+        self.pos = torch.zeros(tb, dtype=torch.int32, device=dev)
+        self.rows = torch.zeros(PREFILL_MAX_ROWS, dtype=torch.int64, device=dev)
+        self.qtiles = torch.zeros(self.gmax, 2, dtype=torch.int32, device=dev)
+        # pinned staging
+        self.h_tok = torch.zeros(tb, dtype=torch.int64, pin_memory=True)
+        self.h_seq = torch.full((tb,), engine.pad_slot, dtype=torch.int32,
+                                pin_memory=True)
+        self.h_pos = torch.zeros(tb, dtype=torch.int32, pin_memory=True)
+        self.h_rows = torch.zeros(PREFILL_MAX_ROWS, dtype=torch.int64,
+                                  pin_memory=True)
+        self.h_qtiles = torch.zeros(self.gmax, 2, dtype=torch.int32,
+                                    pin_memory=True)
+        model, cache = engine.model, engine.cache
+
+        def run_fwd():
+            return model.forward(self.tok, self.seq, self.pos,
+                                 cache.block_table, cache.kcaches,
+                                 cache.vcaches, logits_rows=self.rows,
+                                 qtile_desc=self.qtiles)
+
+        strm = torch.cuda.Stream(dev)
+        strm.wait_stream(torch.cuda.current_stream(dev))
+        with torch.cuda.stream(strm):
+            for _ in range(2):
+                run_fwd()
+        torch.cuda.current_stream(dev).wait_stream(strm)
+        torch.cuda.synchronize(dev)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.logits = run_fwd()            # [PREFILL_MAX_ROWS, vocab] f32
+
+    def run(self, tokens: list[int], seq_ids: list[int], q_pos: list[int],
+            last_rows: list[int], qtiles_host: list) -> torch.Tensor:
+        n = len(tokens)
+        tb = self.tb
+        self.h_tok[:n] = torch.tensor(tokens, dtype=torch.int64)
+        self.h_tok[n:] = 0
+        self.h_seq[:n] = torch.tensor(seq_ids, dtype=torch.int32)
+        self.h_seq[n:] = self.pad_slot
+        self.h_pos[:n] = torch.tensor(q_pos, dtype=torch.int32)
+        self.h_pos[n:] = 0
+        k = len(last_rows)
+        self.h_rows[:k] = torch.tensor(last_rows, dtype=torch.int64)
+        self.h_rows[k:] = 0
+        g = len(qtiles_host)
+        self.h_qtiles[:g] = torch.tensor(qtiles_host, dtype=torch.int32)
+        self.h_qtiles[g:] = 0                  # zero-count tiles: no-op
+        self.tok.copy_(self.h_tok, non_blocking=True)
+        self.seq.copy_(self.h_seq, non_blocking=True)
+        self.pos.copy_(self.h_pos, non_blocking=True)
+        self.rows.copy_(self.h_rows, non_blocking=True)
+        self.qtiles.copy_(self.h_qtiles, non_blocking=True)
+        self.graph.replay()
+        return self.logits
+
+
 class LocalEngine:
     """ChatEngine implementation running qwen3-coder-30b on this GPU."""
 
@@ -173,6 +255,7 @@ class LocalEngine:
         self.cache.ensure_capacity(self.pad_slot, 1)
         self.graphs_enabled = (os.environ.get("ROOMAMD_NO_GRAPHS") != "1")
         self._graphs: dict[tuple, _DecodeGraph] = {}
+        self._prefill_graphs: dict[int, _PrefillGraph] = {}
         self._graphs_broken = False
         self.admitter = SessionAdmitter(self.cache, cfg.max_position)
         self._queue: "queue.Queue[GenRequest]" = queue.Queue()
@@ -351,6 +434,36 @@ class LocalEngine:
                 last_rows.append(len(tokens) - 1)
                 sampled_reqs.append(r)
         dev = self.device
+        n = len(tokens)
+        graph_ok = (self.graphs_enabled and not self._graphs_broken
+                    and n > 8 and len(sampled_reqs) <= PREFILL_MAX_ROWS)
+        if graph_ok:
+            # graph path: pad the chunk to a fixed bucket shape and replay
+            # the captured forward (the eager forward is python-launch-bound
+            # at ~30 ms/chunk — more than the GPU work it enqueues)
+            tb = next(b for b in PREFILL_BUCKETS if b >= n)
+            if n < tb:  # pad rows live in their own 1-token-context segment
+                segments.append((n, tb - n))
+            qtiles_host = self._qtile_list(segments)
+            self.stats["prefill_prep_time"] += time.time() - host_t0
+            try:
+                g = self._prefill_graphs.get(tb)
+                if g is None:
+                    g = _PrefillGraph(self, tb)
+                    self._prefill_graphs[tb] = g
+                logits = g.run(tokens, seq_ids, q_pos, last_rows, qtiles_host)
+                self.stats["prefill_enq_time"] += time.time() - host_t0
+                self.stats["prefill_tokens"] += n
+                if sampled_reqs:
+                    self._sample_and_append(sampled_reqs,
+                                            logits[:len(sampled_reqs)])
+                self.stats["prefill_time"] += time.time() - t0
+                return
+            except Exception as e:
+                import sys
+                print(f"[room_amd] hipGraph prefill disabled: {e}",
+                      file=sys.stderr)
+                self._graphs_broken = True
         tokens_t = torch.tensor(tokens, dtype=torch.int64, device=dev)
         seq_t = torch.tensor(seq_ids, dtype=torch.int32, device=dev)
         pos_t = torch.tensor(q_pos, dtype=torch.int32, device=dev)
@@ -366,6 +479,18 @@ class LocalEngine:
         if sampled_reqs:
             self._sample_and_append(sampled_reqs, logits)
         self.stats["prefill_time"] += time.time() - t0
+
+    @staticmethod
+    def _qtile_list(segments: list) -> list:
+        """(row0, count) segments → ≤32-row q-tile descriptors (host list)."""
+        out = []
+        for row0, count in segments:
+            r = row0
+            while r < row0 + count:
+                take = min(32, row0 + count - r)
+                out.append((r, take))
+                r += take
+        return out
 
     def _decode_step(self, reqs: list[GenRequest]) -> None:
         if not reqs:
