@@ -176,6 +176,7 @@ class _PrefillGraph:
         self.h_qtiles = torch.zeros(self.gmax, 2, dtype=torch.int32,
                                     pin_memory=True)
         model, cache = engine.model, engine.cache
+        model.capture_gemm = True   # hipBLASLt is not capture-safe
 
         def run_fwd():
             return model.forward(self.tok, self.seq, self.pos,
@@ -191,8 +192,11 @@ class _PrefillGraph:
         torch.cuda.current_stream(dev).wait_stream(strm)
         torch.cuda.synchronize(dev)
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
-            self.logits = run_fwd()            # [PREFILL_MAX_ROWS, vocab] f32
+        try:
+            with torch.cuda.graph(self.graph):
+                self.logits = run_fwd()        # [PREFILL_MAX_ROWS, vocab] f32
+        finally:
+            model.capture_gemm = False
 
     def run(self, tokens: list[int], seq_ids: list[int], q_pos: list[int],
             last_rows: list[int], qtiles_host: list) -> torch.Tensor:
@@ -254,9 +258,12 @@ class LocalEngine:
         self.pad_slot = self.cache.alloc_seq()
         self.cache.ensure_capacity(self.pad_slot, 1)
         self.graphs_enabled = (os.environ.get("ROOMAMD_NO_GRAPHS") != "1")
+        self.prefill_graphs_enabled = (
+            os.environ.get("ROOMAMD_NO_PREFILL_GRAPHS") != "1")
         self._graphs: dict[tuple, _DecodeGraph] = {}
         self._prefill_graphs: dict[int, _PrefillGraph] = {}
         self._graphs_broken = False
+        self._prefill_graphs_broken = False
         self.admitter = SessionAdmitter(self.cache, cfg.max_position)
         self._queue: "queue.Queue[GenRequest]" = queue.Queue()
         self._active: list[GenRequest] = []
@@ -435,16 +442,20 @@ class LocalEngine:
                 sampled_reqs.append(r)
         dev = self.device
         n = len(tokens)
-        graph_ok = (self.graphs_enabled and not self._graphs_broken
+        graph_ok = (self.graphs_enabled and self.prefill_graphs_enabled
+                    and not self._graphs_broken
+                    and not self._prefill_graphs_broken
                     and n > 8 and len(sampled_reqs) <= PREFILL_MAX_ROWS)
         if graph_ok:
             # graph path: pad the chunk to a fixed bucket shape and replay
             # the captured forward (the eager forward is python-launch-bound
             # at ~30 ms/chunk — more than the GPU work it enqueues)
             tb = next(b for b in PREFILL_BUCKETS if b >= n)
-            if n < tb:  # pad rows live in their own 1-token-context segment
-                segments.append((n, tb - n))
-            qtiles_host = self._qtile_list(segments)
+            # pad rows live in their own 1-token-context segment — in a COPY:
+            # the eager fallback below reuses `segments` against n-row tensors,
+            # so a leaked pad segment would read q rows past the tensor end
+            seg_graph = segments + ([(n, tb - n)] if n < tb else [])
+            qtiles_host = self._qtile_list(seg_graph)
             self.stats["prefill_prep_time"] += time.time() - host_t0
             try:
                 g = self._prefill_graphs.get(tb)
@@ -463,7 +474,13 @@ class LocalEngine:
                 import sys
                 print(f"[room_amd] hipGraph prefill disabled: {e}",
                       file=sys.stderr)
-                self._graphs_broken = True
+                self._prefill_graphs_broken = True
+                # a capture that aborted mid-graph can leave stale stream
+                # state; fence the device before falling back to eager
+                try:
+                    torch.cuda.synchronize(dev)
+                except Exception:
+                    pass
         tokens_t = torch.tensor(tokens, dtype=torch.int64, device=dev)
         seq_t = torch.tensor(seq_ids, dtype=torch.int32, device=dev)
         pos_t = torch.tensor(q_pos, dtype=torch.int32, device=dev)

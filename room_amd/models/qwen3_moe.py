@@ -110,6 +110,11 @@ class Qwen3MoEModel:
         # decode split-KV band (32 short / 64 long contexts); the engine sets
         # this per decode step (and per captured graph band)
         self.attn_splits = 32
+        # capture-safe GEMMs: hipBLASLt aborts hipGraph capture, so captured
+        # prefill forwards route dense projections through the grouped MFMA
+        # kernel (single-expert). The engine flips this around capture.
+        self.capture_gemm = False
+        self._dense_desc: dict = {}   # T -> (desc, pair_token) for E=1 GEMMs
         cos_t, sin_t = rope_tables(cfg.max_position, cfg.head_dim, cfg.rope_theta)
         self.cos_t = cos_t.to(self.device)
         self.sin_t = sin_t.to(self.device)
@@ -121,6 +126,20 @@ class Qwen3MoEModel:
                      + self.layers[0].router_w.numel() + self.layers[0].w13.numel()
                      + self.layers[0].w2.numel())
         return self.embed.numel() + self.lm_head.numel() + cfg.num_layers * per_layer
+
+    def _dense(self, x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+        """Dense projection: hipBLASLt eager; grouped MFMA when capturing."""
+        if not self.capture_gemm:
+            return F.linear(x, w)
+        T = x.size(0)
+        dp = self._dense_desc.get(T)
+        if dp is None:
+            pe = torch.zeros(T, dtype=torch.int32, device=x.device)
+            desc = ops.moe_build_desc_device(pe, 1, bm=128)
+            pt = torch.arange(T, dtype=torch.int32, device=x.device)
+            dp = self._dense_desc[T] = (desc, pt)
+        y = torch.empty(T, w.size(0), dtype=torch.bfloat16, device=x.device)
+        return ops.dense_grouped_gemm(y, x, w, dp[0], dp[1])
 
     @torch.inference_mode()
     def forward(self, tokens: torch.Tensor, seq_ids: torch.Tensor,
@@ -199,7 +218,7 @@ class Qwen3MoEModel:
                 else:
                     ops.fused_add_rmsnorm(hbuf, x, moe_out, layer.input_norm_w,
                                           cfg.rms_eps)
-                qkv = F.linear(hbuf, layer.wqkv)   # hipBLASLt GEMM
+                qkv = self._dense(hbuf, layer.wqkv)
             # q stays a strided view into qkv (the attention kernels take a
             # row stride; .contiguous() copies profiled at 0.67 ms/step)
             q = qkv[:, :qdim].view(T, cfg.num_q_heads, cfg.head_dim)
@@ -251,7 +270,7 @@ class Qwen3MoEModel:
             if qtile_desc is not None:
                 ops.flash_prefill(attn, q, kcaches[li], vcaches[li], block_table,
                                   seq_ids, q_pos, qtile_desc, self.scale)
-                o = F.linear(attn.reshape(T, qdim), layer.wo)
+                o = self._dense(attn.reshape(T, qdim), layer.wo)
             else:
                 ops.paged_attention(attn, q, kcaches[li], vcaches[li], block_table,
                                     seq_ids, q_pos, self.scale)
@@ -259,7 +278,7 @@ class Qwen3MoEModel:
 
             # --- MoE block (prefill; decode handled above)
             ops.fused_add_rmsnorm(hbuf, x, o, layer.post_attn_norm_w, cfg.rms_eps)
-            router_logits = F.linear(hbuf, layer.router_w).float()
+            router_logits = self._dense(hbuf, layer.router_w).float()
             topk_ids, topk_w = ops.moe_router(router_logits,
                                               cfg.num_experts_per_tok)
             # f32 accumulator feeds fused_add_rmsnorm directly (templated

@@ -288,6 +288,17 @@ def moe_build_desc_device(pair_expert_sorted: torch.Tensor, num_experts: int,
     return desc
 
 
+def dense_grouped_gemm(y: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
+                       desc: torch.Tensor, pair_token: torch.Tensor
+                       ) -> torch.Tensor:
+    """Capture-safe dense GEMM y[T,N] = x[T,H] @ w[N,H]^T via the grouped
+    MFMA kernel with a single expert. hipBLASLt is faster eager but its
+    internal calls abort hipGraph capture; inside captured prefill forwards
+    this path runs instead (desc/pair_token prebuilt per T by the model)."""
+    _require().moe_grouped_gemm128(y, x, w.unsqueeze(0), pair_token, desc, 128)
+    return y
+
+
 def moe_combine_gather(out: torch.Tensor, z: torch.Tensor, topk_w: torch.Tensor,
                        inv_order: torch.Tensor) -> torch.Tensor:
     _require().moe_combine_gather(out, z, topk_w, inv_order)
