@@ -115,6 +115,7 @@ class Qwen3MoEModel:
         # kernel (single-expert). The engine flips this around capture.
         self.capture_gemm = False
         self._dense_desc: dict = {}   # T -> (desc, pair_token) for E=1 GEMMs
+        self._dense_wpad: dict = {}   # id(w) -> N-padded weight (N % 64 != 0)
         cos_t, sin_t = rope_tables(cfg.max_position, cfg.head_dim, cfg.rope_theta)
         self.cos_t = cos_t.to(self.device)
         self.sin_t = sin_t.to(self.device)
@@ -138,8 +139,18 @@ class Qwen3MoEModel:
             desc = ops.moe_build_desc_device(pe, 1, bm=128)
             pt = torch.arange(T, dtype=torch.int32, device=x.device)
             dp = self._dense_desc[T] = (desc, pt)
+        N = w.size(0)
+        if N % 64:  # e.g. tiny-config router (16 experts): pad N, slice out
+            wp = self._dense_wpad.get(id(w))
+            if wp is None:
+                wp = torch.zeros((N + 63) // 64 * 64, w.size(1),
+                                 dtype=w.dtype, device=w.device)
+                wp[:N] = w
+                self._dense_wpad[id(w)] = wp
+            w = wp
         y = torch.empty(T, w.size(0), dtype=torch.bfloat16, device=x.device)
-        return ops.dense_grouped_gemm(y, x, w, dp[0], dp[1])
+        ops.dense_grouped_gemm(y, x, w, dp[0], dp[1])
+        return y[:, :N] if N != w.size(0) else y
 
     @torch.inference_mode()
     def forward(self, tokens: torch.Tensor, seq_ids: torch.Tensor,
