@@ -281,7 +281,13 @@ def moe_build_desc_device(pair_expert_sorted: torch.Tensor, num_experts: int,
     the host-known pair count, empty tiles zero-sized."""
     P = pair_expert_sorted.numel()
     gmax = num_experts + (P + bm - 1) // bm
-    counts = torch.bincount(pair_expert_sorted.long(), minlength=num_experts)
+    # NOT torch.bincount: its nbins probe (.max().item()) host-syncs, which
+    # aborts hipGraph capture — scatter_add_ is sync-free
+    counts = torch.zeros(num_experts, dtype=torch.int64,
+                         device=pair_expert_sorted.device)
+    counts.scatter_add_(0, pair_expert_sorted.long(),
+                        torch.ones(P, dtype=torch.int64,
+                                   device=pair_expert_sorted.device))
     desc = torch.empty(gmax, 3, dtype=torch.int32,
                        device=pair_expert_sorted.device)
     _require().moe_build_desc(desc, counts, bm)
@@ -332,7 +338,10 @@ def build_moe_tile_desc(pair_expert_sorted: torch.Tensor, n_tiles_n: int,
     (expert, row_start, m_size, n_tile) per (M-tile × N-tile). Fully
     vectorized on the device — no host sync (runs once per layer per step)."""
     dev = pair_expert_sorted.device
-    counts = torch.bincount(pair_expert_sorted.long(), minlength=num_experts)
+    counts = torch.zeros(num_experts, dtype=torch.int64, device=dev)
+    counts.scatter_add_(0, pair_expert_sorted.long(),
+                        torch.ones(pair_expert_sorted.numel(),
+                                   dtype=torch.int64, device=dev))
     mtiles = (counts + bm - 1) // bm                      # [E]
     e_ids = torch.repeat_interleave(
         torch.arange(num_experts, device=dev), mtiles)    # [G_m]
