@@ -154,6 +154,10 @@ class _PrefillGraph:
     slots; the scheduler slices the real ones."""
 
     def __init__(self, engine: "LocalEngine", tb: int):
+        import sys as _sys
+        _t0 = time.time()
+        print(f"[room_amd] capturing prefill graph tb={tb} ...",
+              file=_sys.stderr, flush=True)
         dev = engine.device
         self.tb = tb
         self.pad_slot = engine.pad_slot
@@ -197,6 +201,8 @@ class _PrefillGraph:
                 self.logits = run_fwd()        # [PREFILL_MAX_ROWS, vocab] f32
         finally:
             model.capture_gemm = False
+        print(f"[room_amd] prefill graph tb={tb} captured "
+              f"in {time.time() - _t0:.1f}s", file=_sys.stderr, flush=True)
 
     def run(self, tokens: list[int], seq_ids: list[int], q_pos: list[int],
             last_rows: list[int], qtiles_host: list) -> torch.Tensor:
