@@ -179,6 +179,10 @@ class _PrefillGraph:
                                   pin_memory=True)
         self.h_qtiles = torch.zeros(self.gmax, 2, dtype=torch.int32,
                                     pin_memory=True)
+        # guards pinned-staging reuse: back-to-back same-bucket chunks (long
+        # prompts, nothing sampled between) would otherwise overwrite h_* on
+        # the host while the previous run's async H2D copies are in flight
+        self.copied = torch.cuda.Event()
         model, cache = engine.model, engine.cache
         model.capture_gemm = True   # hipBLASLt is not capture-safe
 
@@ -208,6 +212,7 @@ class _PrefillGraph:
             last_rows: list[int], qtiles_host: list) -> torch.Tensor:
         n = len(tokens)
         tb = self.tb
+        self.copied.synchronize()   # previous call's H2D copies done
         self.h_tok[:n] = torch.tensor(tokens, dtype=torch.int64)
         self.h_tok[n:] = 0
         self.h_seq[:n] = torch.tensor(seq_ids, dtype=torch.int32)
@@ -225,6 +230,7 @@ class _PrefillGraph:
         self.pos.copy_(self.h_pos, non_blocking=True)
         self.rows.copy_(self.h_rows, non_blocking=True)
         self.qtiles.copy_(self.h_qtiles, non_blocking=True)
+        self.copied.record()
         self.graph.replay()
         return self.logits
 
