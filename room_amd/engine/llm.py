@@ -111,7 +111,9 @@ class _DecodeGraph:
         torch.cuda.current_stream(dev).wait_stream(strm)
         torch.cuda.synchronize(dev)
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
+        # thread_local: agents' encoder/embedding work on other threads may
+        # call the allocator mid-capture; global mode corrupts those calls
+        with torch.cuda.graph(self.graph, capture_error_mode="thread_local"):
             run_step()
 
     def run_block(self, tokens: list[int], slots: list[int],
@@ -201,7 +203,8 @@ class _PrefillGraph:
         torch.cuda.synchronize(dev)
         self.graph = torch.cuda.CUDAGraph()
         try:
-            with torch.cuda.graph(self.graph):
+            with torch.cuda.graph(self.graph,
+                                  capture_error_mode="thread_local"):
                 self.logits = run_fwd()        # [PREFILL_MAX_ROWS, vocab] f32
         finally:
             model.capture_gemm = False
