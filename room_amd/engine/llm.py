@@ -289,6 +289,22 @@ class LocalEngine:
                       "decode_time": 0.0, "prefill_time": 0.0,
                       "prefill_prep_time": 0.0, "prefill_enq_time": 0.0,
                       "graphs_captured": 0}
+        # capture prefill-bucket graphs up front, before any agent/encoder
+        # thread can put concurrent work on the device mid-capture (captures
+        # are thread_local-mode too, but boot-time capture removes the window
+        # entirely); decode graphs stay lazy (keyed by sampling params)
+        if (self.graphs_enabled and self.prefill_graphs_enabled
+                and os.environ.get("ROOMAMD_NO_PRECAPTURE") != "1"):
+            try:
+                for tb in PREFILL_BUCKETS:
+                    self._prefill_graphs[tb] = _PrefillGraph(self, tb)
+            except Exception as e:
+                import sys as _sys
+                print(f"[room_amd] prefill graph pre-capture failed: {e}",
+                      file=_sys.stderr)
+                self._prefill_graphs.clear()
+                self._prefill_graphs_broken = True
+                torch.cuda.synchronize(self.device)
         self._thread = threading.Thread(target=self._scheduler_loop, daemon=True,
                                         name="room-amd-engine")
         self._thread.start()
