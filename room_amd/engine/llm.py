@@ -273,8 +273,14 @@ class LocalEngine:
         self.pad_slot = self.cache.alloc_seq()
         self.cache.ensure_capacity(self.pad_slot, 1)
         self.graphs_enabled = (os.environ.get("ROOMAMD_NO_GRAPHS") != "1")
+        # hipGraph-captured prefill is OPT-IN: capture-safe on the tiny
+        # config (multi-bucket, concurrent sessions, prefix reuse all pass —
+        # scripts/gpu_prefill_graph_probe*.py) and cuts enqueue 0.53→0.26 s,
+        # but on the 30b config the tb=512 bucket's warmup/capture hits an
+        # HSAIL memory fault not yet isolated (profiles/PERF_NOTES.md round-2
+        # "prefill graphs"); eager prefill is the default until it is.
         self.prefill_graphs_enabled = (
-            os.environ.get("ROOMAMD_NO_PREFILL_GRAPHS") != "1")
+            os.environ.get("ROOMAMD_PREFILL_GRAPHS") == "1")
         self._graphs: dict[tuple, _DecodeGraph] = {}
         self._prefill_graphs: dict[int, _PrefillGraph] = {}
         self._graphs_broken = False
