@@ -194,12 +194,13 @@ class _PrefillGraph:
                                  cache.vcaches, logits_rows=self.rows,
                                  qtile_desc=self.qtiles)
 
-        strm = torch.cuda.Stream(dev)
-        strm.wait_stream(torch.cuda.current_stream(dev))
-        with torch.cuda.stream(strm):
-            for _ in range(2):
-                run_fwd()
-        torch.cuda.current_stream(dev).wait_stream(strm)
+        # warmup on the CURRENT stream, not a side stream: the canonical
+        # side-stream warmup HANGS on the 30b config at tb=512 (bisected in
+        # scripts/gpu_pg_bisect.py — warmup-only on a side stream deadlocks;
+        # the identical forward on the default stream warms up, captures and
+        # replays green). torch.cuda.graph still captures on its own stream.
+        for _ in range(2):
+            run_fwd()
         torch.cuda.synchronize(dev)
         self.graph = torch.cuda.CUDAGraph()
         try:
