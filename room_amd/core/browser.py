@@ -60,6 +60,10 @@ class BrowserSession:
              "--remote-debugging-port=0",
              f"--user-data-dir={self.user_data_dir}", "about:blank"],
             stdout=subprocess.DEVNULL, stderr=subprocess.PIPE)
+        # Chromium forks renderers/zygotes: register for tree termination
+        # on shutdown (reference process-supervisor.ts semantics)
+        from .process_supervisor import register_managed_process
+        register_managed_process(self.proc.pid, f"browser:{session_id[:8]}")
         self.ws_url = self._wait_devtools_url()
         self.http_base = re.sub(r"^ws://([^/]+)/.*$", r"http://\1",
                                 self.ws_url)
@@ -185,14 +189,13 @@ class BrowserSession:
                 self._page_ws.close()
         except Exception:
             pass
+        from .process_supervisor import terminate_tree, unregister_managed_process
         try:
-            self.proc.terminate()
-            self.proc.wait(timeout=5)
+            terminate_tree(self.proc.pid, grace=5.0)
+            self.proc.wait(timeout=1)
         except Exception:
-            try:
-                self.proc.kill()
-            except Exception:
-                pass
+            pass
+        unregister_managed_process(self.proc.pid)
         shutil.rmtree(self.user_data_dir, ignore_errors=True)
 
 

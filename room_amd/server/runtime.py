@@ -71,6 +71,15 @@ class ServerRuntime:
             except (asyncio.CancelledError, Exception):
                 pass
         self._tasks = []
+        # reap any still-running managed children (browser trees etc.) —
+        # reference index.ts:974-1005 graceful shutdown → process supervisor
+        from ..core.browser import close_all_sessions
+        from ..core.process_supervisor import terminate_managed_processes
+        try:
+            close_all_sessions()
+        except Exception:
+            pass
+        terminate_managed_processes()
 
     async def _sleep(self, s: float) -> bool:
         try:
