@@ -7,6 +7,25 @@ import os
 import shutil
 import sys
 
+VERSION = "0.1.0"
+
+
+def _version_hint() -> None:
+    """Print a one-line hint when a newer staged version exists (reference
+    cli/version-hint.ts — non-blocking, fail-silent, suppressed for mcp
+    whose stdout is the protocol stream)."""
+    try:
+        from ..core.update_checker import UpdateChecker, compare_versions
+        from ..server.auth import data_dir
+        uc = UpdateChecker(VERSION, data_dir=data_dir())
+        staged = uc.staged_version()
+        if staged and (compare_versions(staged, VERSION) or 0) > 0:
+            print(f"[room-amd] v{staged} is staged — restart the server or "
+                  f"run `room-amd update` to apply (current v{VERSION})",
+                  file=sys.stderr)
+    except Exception:
+        pass
+
 
 def main(argv: list[str] | None = None) -> int:
     p = argparse.ArgumentParser(prog="room-amd",
@@ -28,6 +47,9 @@ def main(argv: list[str] | None = None) -> int:
     sub.add_parser("uninstall", help="remove ~/.roomamd data (asks first)")
 
     args = p.parse_args(argv)
+
+    if args.cmd in ("serve", "status", "update", None):
+        _version_hint()
 
     if args.cmd == "serve":
         from ..server.bootstrap import serve
@@ -62,7 +84,7 @@ def main(argv: list[str] | None = None) -> int:
         # report, and stage a user-space update when one is available
         from ..core.update_checker import UpdateChecker
         from ..server.auth import data_dir
-        uc = UpdateChecker("0.1.0", data_dir=data_dir())
+        uc = UpdateChecker(VERSION, data_dir=data_dir())
         uc.boot_health_check()
         st = uc.check()
         if st["state"] in ("offline", "error"):

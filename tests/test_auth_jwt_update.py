@@ -133,3 +133,28 @@ def test_update_checker_interval(monkeypatch):
     uc.last_checked_at -= 5 * 3600
     assert uc.maybe_check() is not None
     assert len(calls) == 2
+
+
+def test_cli_version_hint_prints_when_staged(tmp_path, monkeypatch, capsys):
+    """Reference cli/version-hint.ts: newer staged version → stderr hint."""
+    from room_amd.cli import __main__ as cli
+    from room_amd.core.update_checker import UpdateChecker
+
+    monkeypatch.setenv("ROOMAMD_DATA_DIR", str(tmp_path))
+    # stage a fake newer version the way stage_update lays it out
+    app = tmp_path / "app" / "9.9.9"
+    app.mkdir(parents=True)
+    (app / ".ready").write_text("")
+    uc = UpdateChecker("0.1.0", data_dir=tmp_path)
+    assert uc.staged_version() == "9.9.9"
+    cli._version_hint()
+    err = capsys.readouterr().err
+    assert "9.9.9" in err and "staged" in err
+
+
+def test_cli_version_hint_silent_when_current(tmp_path, monkeypatch, capsys):
+    from room_amd.cli import __main__ as cli
+
+    monkeypatch.setenv("ROOMAMD_DATA_DIR", str(tmp_path))
+    cli._version_hint()
+    assert capsys.readouterr().err == ""
