@@ -86,3 +86,29 @@ def test_terminate_tree_on_dead_pid_is_noop():
     proc = subprocess.Popen(["true"])
     proc.wait()
     assert ps.terminate_tree(proc.pid, grace=0.1) == []
+
+
+def test_runtime_stop_reaps_managed_children(tmp_path):
+    """ServerRuntime.stop() must terminate registered child trees
+    (reference index.ts:974-1005 graceful shutdown → process supervisor)."""
+    import asyncio
+
+    from room_amd.core.agent_loop import AgentLoopManager
+    from room_amd.core.tasks import TaskRunner
+    from room_amd.db import LockedDb, connect
+    from room_amd.server.runtime import ServerRuntime
+
+    ldb = LockedDb(connect(str(tmp_path / "t.db")))
+    proc = subprocess.Popen(["sleep", "30"])
+    ps.register_managed_process(proc.pid, "test-child")
+    rt = ServerRuntime(ldb, TaskRunner(ldb), loop_mgr=AgentLoopManager(ldb))
+
+    async def go():
+        await rt.start()
+        await asyncio.sleep(0.1)
+        await rt.stop()
+
+    asyncio.run(go())
+    proc.wait(timeout=5)
+    assert not _alive(proc.pid)
+    assert proc.pid not in ps.managed_pids()
