@@ -165,3 +165,124 @@ def test_queen_policy_deviation_logged(mgr):
         acts2 = q.get_room_activity(db, room["id"])
         assert sum(1 for a in acts2
                    if "deviation" in a["summary"]) == n_dev
+
+
+# --- session continuity: rotation at 20 cycles, compression at 30 msgs,
+# --- context-overflow retry (agent-loop.ts:462-532, :773-782)
+
+def _run(m, room, worker):
+    return asyncio.run(m.run_cycle(room["id"], worker["id"], max_turns=1))
+
+
+def test_session_rotation_after_20_cycles(mgr, monkeypatch):
+    m, ldb = mgr
+    room, queen = _mkroom(ldb)
+    seen = {}
+
+    def fake_execute(options):
+        from room_amd.engine.types import AgentExecutionResult
+        seen["messages"] = options.messages
+        return AgentExecutionResult(success=True, text="ok",
+                                    messages=[{"role": "user", "content": "x"}],
+                                    input_tokens=1, output_tokens=1)
+
+    import room_amd.core.agent_loop as al
+    monkeypatch.setattr(al, "execute_agent", fake_execute)
+    import json as _json
+    with ldb as db:
+        q.save_agent_session(db, queen["id"], session_id=None,
+                             messages_json=_json.dumps(
+                                 [{"role": "user", "content": "old"}]),
+                             model="stub", turn_count=al.SESSION_ROTATE_CYCLES)
+    _run(m, room, queen)
+    # rotation: the stored history must NOT be carried into the new cycle
+    assert seen["messages"] is None
+    with ldb as db:
+        sess = q.get_agent_session(db, queen["id"])
+    assert sess["turn_count"] == 1   # fresh session counter
+
+
+def test_session_continuity_below_rotation(mgr, monkeypatch):
+    m, ldb = mgr
+    room, queen = _mkroom(ldb)
+    seen = {}
+
+    def fake_execute(options):
+        from room_amd.engine.types import AgentExecutionResult
+        seen["messages"] = options.messages
+        return AgentExecutionResult(success=True, text="ok", messages=[],
+                                    input_tokens=1, output_tokens=1)
+
+    import room_amd.core.agent_loop as al
+    monkeypatch.setattr(al, "execute_agent", fake_execute)
+    import json as _json
+    hist = [{"role": "user", "content": "prior"}]
+    with ldb as db:
+        q.save_agent_session(db, queen["id"], session_id=None,
+                             messages_json=_json.dumps(hist),
+                             model="stub", turn_count=3)
+    _run(m, room, queen)
+    assert seen["messages"] == hist
+    with ldb as db:
+        sess = q.get_agent_session(db, queen["id"])
+    assert sess["turn_count"] == 4
+
+
+def test_session_compression_at_30_msgs(mgr, monkeypatch):
+    m, ldb = mgr
+    room, queen = _mkroom(ldb)
+    called = {}
+
+    def fake_compress(messages, model=None):
+        called["n"] = len(messages)
+        return messages[-2:]
+
+    def fake_execute(options):
+        from room_amd.engine.types import AgentExecutionResult
+        called["passed"] = options.messages
+        return AgentExecutionResult(success=True, text="ok", messages=[],
+                                    input_tokens=1, output_tokens=1)
+
+    import room_amd.core.agent_loop as al
+    monkeypatch.setattr(al, "compress_session", fake_compress)
+    monkeypatch.setattr(al, "execute_agent", fake_execute)
+    import json as _json
+    hist = [{"role": "user", "content": f"m{i}"}
+            for i in range(al.SESSION_COMPRESS_AT_MSGS)]
+    with ldb as db:
+        q.save_agent_session(db, queen["id"], session_id=None,
+                             messages_json=_json.dumps(hist),
+                             model="stub", turn_count=2)
+    _run(m, room, queen)
+    assert called["n"] == al.SESSION_COMPRESS_AT_MSGS
+    assert called["passed"] == hist[-2:]
+
+
+def test_context_overflow_retries_with_fresh_session(mgr, monkeypatch):
+    m, ldb = mgr
+    room, queen = _mkroom(ldb)
+    calls = []
+
+    def fake_execute(options):
+        from room_amd.engine.types import AgentExecutionResult
+        calls.append(options.messages)
+        if len(calls) == 1:
+            return AgentExecutionResult(success=False, text="",
+                                        error="Context length exceeded",
+                                        messages=[])
+        return AgentExecutionResult(success=True, text="recovered",
+                                    messages=[], input_tokens=1,
+                                    output_tokens=1)
+
+    import room_amd.core.agent_loop as al
+    monkeypatch.setattr(al, "execute_agent", fake_execute)
+    import json as _json
+    hist = [{"role": "user", "content": "big"}]
+    with ldb as db:
+        q.save_agent_session(db, queen["id"], session_id=None,
+                             messages_json=_json.dumps(hist),
+                             model="stub", turn_count=2)
+    out = _run(m, room, queen)
+    assert out["result"].success
+    assert calls[0] == hist      # first try with history
+    assert calls[1] is None      # retry with fresh session
