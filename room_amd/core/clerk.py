@@ -110,27 +110,114 @@ def execute_clerk_tool(ldb: LockedDb, call: ToolCall) -> str:
         return json.dumps({"error": str(e)})
 
 
+# Project-doc sync (reference: src/shared/clerk-profile-config.ts
+# CLERK_PROJECT_DOC_SPECS + hash setting keys): configured docs are mirrored
+# into clerk memory entities when their content hash changes, so the clerk
+# can answer questions about the system from its own docs.
+CLERK_DOC_SPECS = [
+    {"entity": "Project README", "rel": "README.md",
+     "hash_key": "clerk_project_doc_hash_readme"},
+    {"entity": "Project Dashboard Source", "rel": "room_amd/server/dashboard.py",
+     "hash_key": "clerk_project_doc_hash_dashboard"},
+]
+CLERK_DOC_CONTENT_MAX = 200_000
+CLERK_DOC_SYNC_MIN_S = 60.0
+
+
+def sync_project_docs(ldb: LockedDb, root: str | None = None,
+                      min_interval_s: float = CLERK_DOC_SYNC_MIN_S) -> int:
+    """Sync changed project docs into clerk memory entities; returns the
+    number of docs refreshed. Hash-gated (no-op when nothing changed) and
+    rate-limited by the last-sync timestamp setting."""
+    import hashlib
+    import time as _time
+    from pathlib import Path
+
+    base = Path(root) if root else Path(__file__).resolve().parents[2]
+    now = _time.time()
+    with ldb as db:
+        last = q.get_setting(db, "clerk_project_doc_last_sync")
+        if last and now - float(last) < min_interval_s:
+            return 0
+        q.set_setting(db, "clerk_project_doc_last_sync", str(now))
+        synced = 0
+        for spec in CLERK_DOC_SPECS:
+            path = base / spec["rel"]
+            try:
+                content = path.read_text()[:CLERK_DOC_CONTENT_MAX]
+            except OSError:
+                continue
+            digest = hashlib.sha256(content.encode()).hexdigest()[:16]
+            if q.get_setting(db, spec["hash_key"]) == digest:
+                continue
+            ent = q.get_entity_by_name(db, spec["entity"])
+            if ent is None:
+                ent = q.create_entity(db, spec["entity"], entity_type="document",
+                                      category="project_doc")
+            q.add_observation(db, ent["id"], content, source="doc_sync")
+            q.set_setting(db, spec["hash_key"], digest)
+            synced += 1
+    return synced
+
+
+def clerk_model_chain(ldb: LockedDb, model: str = "stub") -> list[str]:
+    """Model preference chain (reference clerk-profile-config.ts:
+    DEFAULT_CLERK_MODEL → subscription → OpenAI → Anthropic fallbacks):
+    clerk_model setting → requested model → stub (always available)."""
+    with ldb as db:
+        setting = q.get_setting(db, "clerk_model")
+    chain = []
+    for m in (setting, model, "stub"):
+        if m and m not in chain:
+            chain.append(m)
+    return chain
+
+
 def clerk_chat(ldb: LockedDb, content: str, memory=None,
                model: str = "stub") -> str:
-    """One clerk chat turn with tool execution + usage accounting."""
+    """One clerk chat turn with tool execution + usage accounting; walks
+    the model fallback chain when a model is unavailable (reference
+    clerk-profile.ts model fallback)."""
     with ldb as db:
         q.add_clerk_message(db, "user", content)
-        model_setting = q.get_setting(db, "clerk_model")
-        if model_setting:
-            model = model_setting
 
-    result = execute_agent(AgentExecutionOptions(
-        prompt=content, model=model,
-        system_prompt="You are the Clerk — the keeper's system-wide assistant. "
-                      "Manage rooms and tasks via tools; answer concisely.",
-        max_turns=5, tools=CLERK_TOOLS,
-        tool_executor=lambda call: execute_clerk_tool(ldb, call)))
+    chain = clerk_model_chain(ldb, model)
+    result = None
+    for m in chain:
+        model = m
+        try:
+            result = execute_agent(AgentExecutionOptions(
+                prompt=content, model=m,
+                system_prompt="You are the Clerk — the keeper's system-wide "
+                              "assistant. Manage rooms and tasks via tools; "
+                              "answer concisely.",
+                max_turns=5, tools=CLERK_TOOLS,
+                tool_executor=lambda call: execute_clerk_tool(ldb, call)))
+        except Exception as e:
+            result = None
+            last_err = str(e)
+            continue
+        if result.success or not _model_unavailable(result.error):
+            break
+    if result is None:
+        from ..engine.types import AgentExecutionResult
+        result = AgentExecutionResult(success=False, text="",
+                                      error=last_err)
     reply = result.text or result.error or ""
     with ldb as db:
         q.add_clerk_message(db, "assistant", reply, source="chat")
         q.log_clerk_usage(db, "chat", model, result.input_tokens,
                           result.output_tokens, success=result.success)
     return reply
+
+
+def _model_unavailable(error: str | None) -> bool:
+    if not error:
+        return False
+    e = error.lower()
+    return any(s in e for s in ("model not found", "no api key", "unavailable",
+                                "not registered", "connection", "unreachable",
+                                "no egress", "api key"))
 
 
 class CommentaryEngine:

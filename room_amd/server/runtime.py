@@ -127,6 +127,9 @@ class ServerRuntime:
                 # release poll on the updateChecker.ts 4h cadence (offline →
                 # state recorded, never raises)
                 await asyncio.to_thread(self.update_checker.maybe_check)
+                # project docs → clerk memory (hash-gated, 60s min interval)
+                from ..core.clerk import sync_project_docs
+                await asyncio.to_thread(sync_project_docs, self.ldb)
             except Exception:
                 pass
             if await self._sleep(MAINTENANCE_S):

@@ -140,3 +140,73 @@ def test_escalation_alert_relay(tmp_path, monkeypatch):
     outbox = (tmp_path / "outbox.jsonl").read_text().strip().splitlines()
     entries = [json.loads(l) for l in outbox]
     assert any("escalation" in e["subject"] for e in entries), entries
+
+
+def test_project_doc_sync_hash_gated(tmp_path):
+    """clerk-profile-config.ts DOC_SPECS semantics: docs land in memory
+    entities once per content hash; rewrite → refreshed observation."""
+    from room_amd.core.clerk import sync_project_docs
+    from room_amd.db import LockedDb, init_test_db
+    from room_amd.db import queries as q
+
+    ldb = LockedDb(init_test_db())
+    (tmp_path / "README.md").write_text("# proj v1")
+    (tmp_path / "room_amd" / "server").mkdir(parents=True)
+    (tmp_path / "room_amd" / "server" / "dashboard.py").write_text("# dash")
+
+    assert sync_project_docs(ldb, root=str(tmp_path), min_interval_s=0) == 2
+    with ldb as db:
+        ent = q.get_entity_by_name(db, "Project README")
+        obs = q.get_observations(db, ent["id"])
+    assert obs and "# proj v1" in obs[0]["content"]
+
+    # unchanged → no new observations
+    assert sync_project_docs(ldb, root=str(tmp_path), min_interval_s=0) == 0
+
+    # changed content → refreshed
+    (tmp_path / "README.md").write_text("# proj v2")
+    assert sync_project_docs(ldb, root=str(tmp_path), min_interval_s=0) == 1
+    with ldb as db:
+        obs = q.get_observations(db, ent["id"])
+    assert any("# proj v2" in o["content"] for o in obs)
+
+
+def test_project_doc_sync_rate_limited(tmp_path):
+    from room_amd.core.clerk import sync_project_docs
+    from room_amd.db import LockedDb, init_test_db
+
+    ldb = LockedDb(init_test_db())
+    (tmp_path / "README.md").write_text("x")
+    assert sync_project_docs(ldb, root=str(tmp_path), min_interval_s=3600) == 1
+    # second call inside the interval: rate-limited no-op even with changes
+    (tmp_path / "README.md").write_text("y")
+    assert sync_project_docs(ldb, root=str(tmp_path), min_interval_s=3600) == 0
+
+
+def test_clerk_model_fallback_chain(tmp_path):
+    from room_amd.core.clerk import clerk_model_chain
+    from room_amd.db import LockedDb, init_test_db
+    from room_amd.db import queries as q
+
+    ldb = LockedDb(init_test_db())
+    assert clerk_model_chain(ldb, "stub") == ["stub"]
+    with ldb as db:
+        q.set_setting(db, "clerk_model", "openai:gpt-4o-mini")
+    assert clerk_model_chain(ldb, "stub") == ["openai:gpt-4o-mini", "stub"]
+
+
+def test_clerk_chat_falls_back_to_stub(tmp_path, monkeypatch):
+    """An unavailable configured model must not break clerk chat — the
+    chain ends at the always-available stub."""
+    from room_amd.core.clerk import clerk_chat
+    from room_amd.db import LockedDb, init_test_db
+    from room_amd.db import queries as q
+
+    ldb = LockedDb(init_test_db())
+    with ldb as db:
+        q.set_setting(db, "clerk_model", "openai:gpt-4o-mini")  # no key/offline
+    reply = clerk_chat(ldb, "hello")
+    assert reply                      # stub answered
+    with ldb as db:
+        rows = db.execute("SELECT model, success FROM clerk_usage").fetchall()
+    assert rows[-1]["model"] == "stub"
