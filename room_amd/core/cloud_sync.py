@@ -137,3 +137,64 @@ def fetch_cloud_room_messages(ldb: LockedDb, room_id: int) -> int:
                                   from_room_id=m.get("from"))
             n += 1
     return n
+
+
+# --- activity push (reference: src/server/cloud.ts — event-bus subscriber
+# maps internal event types to cloud activity kinds and pushes them to the
+# relay, max 1 push/sec per room, fail-silent)
+
+# internal bus event type → cloud activity kind (the reference's map,
+# cloud.ts CLOUD_EVENT_MAP, keyed by ITS event vocabulary; keys here are
+# this framework's bus types for the same moments)
+CLOUD_EVENT_MAP = {
+    "decision": "decision_created",
+    "escalation": "escalation",
+    "escalation:resolved": "escalation_resolved",
+    "message": "room_message",
+    "worker_created": "worker_created",
+    "room_started": "room_started",
+    "cycle_finished": "cycle",
+}
+
+
+class ActivityPusher:
+    """Wildcard bus subscriber that relays mapped room events to the cloud.
+    Inert without ROOMAMD_CLOUD_API; a `sender` hook makes it testable."""
+
+    def __init__(self, bus, ldb: LockedDb, sender=None,
+                 min_gap_s: float = 1.0):
+        self.ldb = ldb
+        self.min_gap_s = min_gap_s
+        self._last_push: dict[int, float] = {}
+        self._sender = sender
+        self._unsub = bus.on("*", self._on_event)
+
+    def stop(self) -> None:
+        if self._unsub:
+            self._unsub()
+            self._unsub = None
+
+    def _on_event(self, channel: str, event: dict) -> None:
+        if self._sender is None and not cloud_api():
+            return
+        kind = CLOUD_EVENT_MAP.get(event.get("type") or "")
+        if kind is None or not channel.startswith("room:"):
+            return
+        try:
+            room_id = int(channel.split(":", 1)[1])
+        except ValueError:
+            return
+        now = time.time()
+        if now - self._last_push.get(room_id, 0.0) < self.min_gap_s:
+            return
+        self._last_push[room_id] = now
+        token = load_room_tokens().get(str(room_id))
+        payload = {"type": kind, "data": event.get("data"),
+                   "timestamp": event.get("timestamp")}
+        try:
+            if self._sender is not None:
+                self._sender(room_id, payload)
+            elif token:
+                _post(f"/rooms/{room_id}/activity", payload, token=token)
+        except Exception:
+            pass

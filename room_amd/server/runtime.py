@@ -57,6 +57,11 @@ class ServerRuntime:
         ]
         if self.commentary is not None:
             self._tasks.append(asyncio.create_task(self._commentary_loop()))
+        # cloud activity push (reference cloud.ts): inert without
+        # ROOMAMD_CLOUD_API, 1 push/s/room rate limit, fail-silent
+        if self.bus is not None:
+            from ..core.cloud_sync import ActivityPusher
+            self._activity_pusher = ActivityPusher(self.bus, self.ldb)
         self._tasks.append(asyncio.create_task(self._alert_relay_loop()))
         self._tasks.append(asyncio.create_task(self._watcher_loop()))
         self._watch_mtimes: dict[int, float] = {}
@@ -71,6 +76,9 @@ class ServerRuntime:
             except (asyncio.CancelledError, Exception):
                 pass
         self._tasks = []
+        if getattr(self, "_activity_pusher", None) is not None:
+            self._activity_pusher.stop()
+            self._activity_pusher = None
         # reap any still-running managed children (browser trees etc.) —
         # reference index.ts:974-1005 graceful shutdown → process supervisor
         from ..core.browser import close_all_sessions

@@ -244,3 +244,38 @@ async def _stop_after(rt, coro):
         return await coro
     finally:
         await rt.stop()
+
+
+def test_cloud_activity_pusher(db):
+    """cloud.ts activity push: mapped room events relayed, 1/s/room rate
+    limit, unmapped events and non-room channels ignored."""
+    import time as _time
+
+    from room_amd.core.cloud_sync import ActivityPusher
+    from room_amd.core.events import EventBus
+
+    ldb = db
+    bus = EventBus()
+    sent = []
+    p = ActivityPusher(bus, ldb, sender=lambda rid, pl: sent.append((rid, pl)),
+                       min_gap_s=0.2)
+    bus.emit("room:7", "decision", {"id": 1})
+    assert sent == [(7, {"type": "decision_created", "data": {"id": 1},
+                         "timestamp": sent[0][1]["timestamp"]})]
+    # rate limited within the gap
+    bus.emit("room:7", "message", {"id": 2})
+    assert len(sent) == 1
+    # different room not limited
+    bus.emit("room:8", "message", {"id": 3})
+    assert len(sent) == 2 and sent[1][1]["type"] == "room_message"
+    # unmapped type and non-room channel ignored
+    bus.emit("room:9", "cycle_started", {})
+    bus.emit("runs", "run_finished", {})
+    assert len(sent) == 2
+    # after the gap the room can push again
+    _time.sleep(0.25)
+    bus.emit("room:7", "escalation", {"id": 4})
+    assert len(sent) == 3
+    p.stop()
+    bus.emit("room:7", "decision", {"id": 5})
+    assert len(sent) == 3  # unsubscribed
