@@ -255,3 +255,70 @@ def test_unknown_tool_and_malformed_args(mcp):
                         "params": {"name": "room_create_room",
                                    "arguments": {}}})
     assert resp2 is not None
+
+
+def test_every_tool_smoke_callable(mcp):
+    """Call EVERY registered tool once with schema-synthesized minimal
+    arguments: each must return a JSON-RPC result (ok or a structured tool
+    error) — never a handler crash without a response. Guards all ~170
+    handlers against signature drift."""
+    from room_amd.core import room as room_mod
+    from room_amd.db import queries as q
+
+    with mcp.ldb as db:
+        r = room_mod.create_room(db, "smoke-room", goal="g",
+                                 worker_model="stub")
+        room_id = r["id"]
+        worker_id = r["queen_worker_id"]
+        g = q.create_goal(db, room_id, "smoke goal")
+        d = q.create_decision(db, room_id, worker_id, "smoke d", "low_impact")
+        ent = q.create_entity(db, "smoke entity", room_id=room_id,
+                              observations=["obs"])
+        task = q.create_task(db, "smoke task", "do", trigger_type="manual",
+                             room_id=room_id)
+
+    known_ids = {"room_id": room_id, "worker_id": worker_id,
+                 "target_worker_id": worker_id, "goal_id": g["id"],
+                 "decision_id": d["id"], "entity_id": ent["id"],
+                 "task_id": task["id"], "parent_goal_id": g["id"],
+                 "escalation_id": 1, "message_id": 1, "skill_id": 1,
+                 "watch_id": 1, "credential_id": 1, "session_id": "s1"}
+
+    def synth(name, schema):
+        args = {}
+        props = schema.get("properties", {})
+        for key in schema.get("required", []):
+            pdef = props.get(key, {})
+            if key in known_ids:
+                args[key] = known_ids[key]
+            elif pdef.get("enum"):
+                args[key] = pdef["enum"][0]
+            elif pdef.get("type") == "integer":
+                args[key] = 1
+            elif pdef.get("type") == "number":
+                args[key] = 1.0
+            elif pdef.get("type") == "boolean":
+                args[key] = False
+            elif pdef.get("type") == "array":
+                args[key] = []
+            elif pdef.get("type") == "object":
+                args[key] = {}
+            else:
+                args[key] = "smoke"
+        return args
+
+    tools = _all_tools(mcp)
+    assert len(tools) > 150
+    failures = []
+    for i, t in enumerate(tools):
+        req = {"jsonrpc": "2.0", "id": 1000 + i, "method": "tools/call",
+               "params": {"name": t["name"],
+                          "arguments": synth(t["name"], t["inputSchema"])}}
+        try:
+            resp = mcp.handle(req)
+        except Exception as e:  # handler crashed through the dispatcher
+            failures.append((t["name"], f"raised {e!r}"))
+            continue
+        if resp is None or ("result" not in resp and "error" not in resp):
+            failures.append((t["name"], f"bad response {resp!r}"))
+    assert not failures, failures
