@@ -64,6 +64,24 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     app = FastAPI(title="room_amd", version="0.1.0")
     app.state.ctx = state
 
+    # malformed bodies and dangling references become structured 4xx, not
+    # unhandled 500s (reference routes return 400/404 on bad payloads):
+    # a KeyError from payload["field"] is a missing required field; an
+    # IntegrityError is an insert referencing a deleted row
+    import sqlite3 as _sqlite3
+
+    from fastapi.responses import JSONResponse
+
+    @app.exception_handler(KeyError)
+    async def _missing_field(request, exc):
+        return JSONResponse(status_code=400,
+                            content={"detail": f"missing field: {exc}"})
+
+    @app.exception_handler(_sqlite3.IntegrityError)
+    async def _dangling_ref(request, exc):
+        return JSONResponse(status_code=409,
+                            content={"detail": f"constraint failed: {exc}"})
+
     # opt-in HTTP profiling (reference: QUOROOM_PROFILE_HTTP=1,
     # server/index.ts:289-320 — per-request hrtime, normalized endpoint
     # buckets, slow-request threshold log)
@@ -176,6 +194,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
 
     @app.post("/api/rooms")
     async def create_room(payload: dict = Body(...), role: str = Depends(rw)):
+        if not payload.get("name"):
+            raise HTTPException(400, "name required")
         with ldb as db:
             room = room_mod.create_room(
                 db, payload["name"], goal=payload.get("goal"),
@@ -252,6 +272,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/rooms/{room_id}/restart")
     async def restart_room(room_id: int, role: str = Depends(rw)):
         with ldb as db:
+            if q.get_room(db, room_id) is None:
+                raise HTTPException(404, "room not found")
             return room_mod.restart_room(db, room_id)
 
     @app.get("/api/rooms/{room_id}/activity")
@@ -412,7 +434,11 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/decisions/{decision_id}/vote")
     async def vote(decision_id: int, payload: dict = Body(...),
                    role: str = Depends(rw)):
+        if "worker_id" not in payload or "vote" not in payload:
+            raise HTTPException(400, "worker_id and vote required")
         with ldb as db:
+            if q.get_decision(db, decision_id) is None:
+                raise HTTPException(404, "decision not found")
             try:
                 return quorum_mod.vote(db, decision_id, payload["worker_id"],
                                        payload["vote"], payload.get("reasoning"))
@@ -422,7 +448,11 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/decisions/{decision_id}/keeper-vote")
     async def keeper_vote(decision_id: int, payload: dict = Body(...),
                           role: str = Depends(rw)):
+        if "vote" not in payload:
+            raise HTTPException(400, "vote required")
         with ldb as db:
+            if q.get_decision(db, decision_id) is None:
+                raise HTTPException(404, "decision not found")
             try:
                 return quorum_mod.keeper_vote(db, decision_id, payload["vote"])
             except ValueError as e:
@@ -630,6 +660,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/rooms/{room_id}/escalations")
     async def create_escalation(room_id: int, payload: dict = Body(...),
                                 role: str = Depends(rw)):
+        if not payload.get("question"):
+            raise HTTPException(400, "question required")
         with ldb as db:
             e = q.create_escalation(db, room_id, payload["question"],
                                     from_agent_id=payload.get("from_agent_id"))
@@ -639,6 +671,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/escalations/{escalation_id}/answer")
     async def answer_escalation(escalation_id: int, payload: dict = Body(...),
                                 role: str = Depends(rw)):
+        if "answer" not in payload:
+            raise HTTPException(400, "answer required")
         with ldb as db:
             q.answer_escalation(db, escalation_id, payload["answer"])
             row = db.execute("SELECT * FROM escalations WHERE id = ?",
@@ -706,6 +740,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/rooms/{room_id}/credentials")
     async def set_credential(room_id: int, payload: dict = Body(...),
                              role: str = Depends(rw)):
+        if "name" not in payload or "value" not in payload:
+            raise HTTPException(400, "name and value required")
         with ldb as db:
             row = q.set_credential(db, room_id, payload["name"],
                                    encrypt_secret(payload["value"]),
@@ -755,6 +791,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/rooms/{room_id}/chat")
     async def room_chat(room_id: int, payload: dict = Body(...),
                         role: str = Depends(rw)):
+        if not payload.get("content"):
+            raise HTTPException(400, "content required")
         with ldb as db:
             q.add_chat_message(db, room_id, "user", payload["content"])
             e = q.create_escalation(db, room_id, payload["content"])
@@ -773,6 +811,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/clerk/chat")
     async def clerk_chat(payload: dict = Body(...), role: str = Depends(rw)):
         from ..core.clerk import clerk_chat as do_chat
+        if not payload.get("content"):
+            raise HTTPException(400, "content required")
         # keeper message → commentary engine pauses for its silence window
         bus.emit("clerk", "keeper_message", {})
         reply = await asyncio.to_thread(do_chat, ldb, payload["content"],
@@ -843,6 +883,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     async def room_from_template(payload: dict = Body(...),
                                  role: str = Depends(rw)):
         from ..core.templates import instantiate_room_template
+        if "template" not in payload or "name" not in payload:
+            raise HTTPException(400, "template and name required")
         with ldb as db:
             try:
                 room = instantiate_room_template(
@@ -928,6 +970,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     async def goal_add_update(goal_id: int, body: dict = Body(...),
                               role: str = Depends(rw)):
         with ldb as db:
+            if q.get_goal(db, goal_id) is None:
+                raise HTTPException(404, "goal not found")
             q.add_goal_update(db, goal_id, body.get("observation", ""),
                               metric_value=body.get("progress"))
             return q.get_goal(db, goal_id)
@@ -966,7 +1010,12 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     @app.post("/api/memory/relations")
     async def memory_relate(body: dict = Body(...),
                             role: str = Depends(rw)):
+        if "from_entity" not in body or "to_entity" not in body:
+            raise HTTPException(400, "from_entity and to_entity required")
         with ldb as db:
+            if (q.get_entity(db, body["from_entity"]) is None
+                    or q.get_entity(db, body["to_entity"]) is None):
+                raise HTTPException(404, "entity not found")
             return q.create_relation(db, body["from_entity"],
                                      body["to_entity"],
                                      body.get("relation_type", "related_to"))
@@ -1147,7 +1196,10 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
                                     role: str = Depends(rw)):
         from ..core import self_mod as self_mod_mod
         with ldb as db:
-            return self_mod_mod.revert_modification(db, audit_id)
+            try:
+                return self_mod_mod.revert_modification(db, audit_id)
+            except ValueError as e:
+                raise HTTPException(404, str(e))
 
     @app.post("/api/tasks/{task_id}/pause")
     async def task_pause(task_id: int, role: str = Depends(rw)):
