@@ -322,3 +322,43 @@ def test_every_tool_smoke_callable(mcp):
         if resp is None or ("result" not in resp and "error" not in resp):
             failures.append((t["name"], f"bad response {resp!r}"))
     assert not failures, failures
+
+
+def test_every_tool_survives_dangling_ids(mcp):
+    """Second sweep: every tool called with ids that don't exist must return
+    a structured response (ok-with-empty or error payload), never crash the
+    dispatcher (missing-resource robustness, mirrors the route smoke)."""
+    def synth(schema):
+        args = {}
+        props = schema.get("properties", {})
+        for key in schema.get("required", []):
+            pdef = props.get(key, {})
+            if pdef.get("enum"):
+                args[key] = pdef["enum"][0]
+            elif pdef.get("type") == "integer":
+                args[key] = 999_999          # dangling id
+            elif pdef.get("type") == "number":
+                args[key] = 999_999.0
+            elif pdef.get("type") == "boolean":
+                args[key] = True
+            elif pdef.get("type") == "array":
+                args[key] = []
+            elif pdef.get("type") == "object":
+                args[key] = {}
+            else:
+                args[key] = "no-such-thing"
+        return args
+
+    failures = []
+    for i, t in enumerate(_all_tools(mcp)):
+        req = {"jsonrpc": "2.0", "id": 2000 + i, "method": "tools/call",
+               "params": {"name": t["name"],
+                          "arguments": synth(t["inputSchema"])}}
+        try:
+            resp = mcp.handle(req)
+        except Exception as e:
+            failures.append((t["name"], f"raised {e!r}"))
+            continue
+        if resp is None or ("result" not in resp and "error" not in resp):
+            failures.append((t["name"], f"bad response {resp!r}"))
+    assert not failures, failures
