@@ -453,3 +453,43 @@ def test_server_restart_and_update_endpoints(server):
     assert d["staged"] is False          # offline: nothing to download
     ur = client.post("/api/server/update-restart", headers=h).json()
     assert ur["ok"] is False
+
+
+def test_websocket_rejects_bad_token(server):
+    client, *_ = server
+    import pytest as _pytest
+    with _pytest.raises(Exception):
+        with client.websocket_connect("/ws?token=invalid") as ws:
+            ws.receive_json()
+
+
+def test_websocket_channel_filtering_and_unsubscribe(server):
+    """Events only reach subscribed channels; unsubscribe stops delivery
+    (reference ws.ts:1-10 subscribe/unsubscribe protocol)."""
+    client, h, *_ = server
+    auth = server[4]
+    with client.websocket_connect(f"/ws?token={auth.agent_token}") as ws:
+        ws.send_json({"type": "subscribe", "channel": "rooms"})
+        ws.send_json({"type": "ping"})
+        assert ws.receive_json()["type"] == "pong"
+
+        # an event on an UNsubscribed channel must not be delivered: create
+        # a room (rooms channel, subscribed) and then emit activity in it
+        # (room:<id> channel, not subscribed) — only the first arrives
+        r = client.post("/api/rooms", json={"name": "f1",
+                                            "worker_model": "stub"},
+                        headers=h).json()
+        assert ws.receive_json()["type"] == "room_created"
+        client.post(f"/api/rooms/{r['id']}/escalations",
+                    json={"question": "q?"}, headers=h)
+        ws.send_json({"type": "ping"})         # fence: next frame is pong,
+        assert ws.receive_json()["type"] == "pong"  # escalation was filtered
+
+        # unsubscribe: further rooms events are not delivered either
+        ws.send_json({"type": "unsubscribe", "channel": "rooms"})
+        ws.send_json({"type": "ping"})
+        assert ws.receive_json()["type"] == "pong"
+        client.post("/api/rooms", json={"name": "f2",
+                                        "worker_model": "stub"}, headers=h)
+        ws.send_json({"type": "ping"})
+        assert ws.receive_json()["type"] == "pong"
