@@ -279,3 +279,47 @@ def test_cloud_activity_pusher(db):
     p.stop()
     bus.emit("room:7", "decision", {"id": 5})
     assert len(sent) == 3  # unsubscribed
+
+
+def test_migrations_idempotent_and_versioned(db):
+    """db-migrations.ts semantics: additive, version-gated, idempotent,
+    duplicate-column tolerant."""
+    import sqlite3
+
+    from room_amd.db import migrations as mig
+
+    ldb = db
+    with ldb as conn:
+        # no pending migrations on a fresh schema
+        assert mig.run_migrations(conn) == 0
+        base = conn.execute(
+            "SELECT MAX(version) AS v FROM schema_version").fetchone()["v"]
+
+    fake = [(base + 1, ["ALTER TABLE workers ADD COLUMN test_col INTEGER"]),
+            (base + 2, ["CREATE TABLE IF NOT EXISTS test_mig (id INTEGER)"])]
+    old = mig.MIGRATIONS
+    mig.MIGRATIONS = old + fake
+    try:
+        with ldb as conn:
+            assert mig.run_migrations(conn) == 2
+            assert mig._has_column(conn, "workers", "test_col")
+            conn.execute("INSERT INTO test_mig (id) VALUES (1)")
+            # re-run: version-gated no-op
+            assert mig.run_migrations(conn) == 0
+            v = conn.execute(
+                "SELECT MAX(version) AS v FROM schema_version").fetchone()["v"]
+            assert v == base + 2
+        # a duplicate-column statement at a NEW version must not raise
+        mig.MIGRATIONS = old + fake + [
+            (base + 3, ["ALTER TABLE workers ADD COLUMN test_col INTEGER"])]
+        with ldb as conn:
+            assert mig.run_migrations(conn) == 1
+        # but a genuinely broken statement must raise
+        mig.MIGRATIONS = old + fake + [
+            (base + 4, ["ALTER TABLE no_such_table ADD COLUMN x INTEGER"])]
+        import pytest as _pytest
+        with ldb as conn:
+            with _pytest.raises(sqlite3.OperationalError):
+                mig.run_migrations(conn)
+    finally:
+        mig.MIGRATIONS = old
