@@ -493,3 +493,16 @@ def test_websocket_channel_filtering_and_unsubscribe(server):
                                         "worker_model": "stub"}, headers=h)
         ws.send_json({"type": "ping"})
         assert ws.receive_json()["type"] == "pong"
+
+
+def test_dashboard_covers_reference_tabs(server):
+    """The served dashboard must expose the reference SPA's surfaces
+    (App.tsx:549-581 tab list; transactions lives inside wallet here)."""
+    client, *_ = server
+    html = client.get("/").text
+    for tab in ["overview", "goals", "votes", "workers", "tasks", "skills",
+                "memory", "messages", "wallet", "credentials",
+                "room-settings", "settings", "status", "help"]:
+        assert f"'{tab}'" in html, f"tab {tab} missing"
+    # live updates + auth flow wired
+    assert "/ws?token=" in html and "handshake" in html
