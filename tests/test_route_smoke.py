@@ -101,3 +101,24 @@ def test_every_route_handled(env):
         if resp.status_code == 500 and resp.text.strip() == "Internal Server Error":
             failures.append((method, path, "unhandled 500"))
     assert not failures, failures
+
+
+def test_missing_field_returns_400(env):
+    """App-level KeyError handler: a bare payload["x"] miss is a 400 with
+    the field named, not an unhandled 500."""
+    app, client, h, ids = env
+    r = client.post("/api/clerk/chat", headers=h, json={})
+    assert r.status_code == 400
+    r2 = client.post(f"/api/rooms/{ids['room_id']}/tasks", headers=h, json={})
+    assert r2.status_code == 400
+    assert "missing field" in r2.text or "required" in r2.text
+
+
+def test_dangling_reference_is_4xx(env):
+    app, client, h, ids = env
+    r = client.post("/api/goals/999999/updates", headers=h,
+                    json={"observation": "x"})
+    assert r.status_code == 404
+    r2 = client.post("/api/decisions/999999/vote", headers=h,
+                     json={"worker_id": 1, "vote": "yes"})
+    assert r2.status_code == 404
