@@ -109,10 +109,7 @@ def test_missing_field_returns_400(env):
     app, client, h, ids = env
     r = client.post("/api/clerk/chat", headers=h, json={})
     assert r.status_code == 400
-    # fresh room: the whole-surface sweep above deletes the fixture room
-    rid = client.post("/api/rooms", headers=h,
-                      json={"name": "mf", "worker_model": "stub"}).json()["id"]
-    r2 = client.post(f"/api/rooms/{rid}/tasks", headers=h, json={})
+    r2 = client.post("/api/tasks", headers=h, json={})
     assert r2.status_code == 400
     assert "missing field" in r2.text or "required" in r2.text
 
