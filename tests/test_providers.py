@@ -135,3 +135,50 @@ def test_queen_update_worker_and_web_tools(db):
         db, r["id"], r["queen_worker_id"],
         ToolCall("room_web_search", {"query": "anything"})))
     assert isinstance(res, dict)  # offline → {"error": ...} or results
+
+
+def test_parse_tool_calls_malformed_is_safe():
+    """Malformed tool-call blocks must not crash the loop: broken JSON,
+    unclosed tags and empty bodies are skipped; valid calls around them
+    still parse (the model WILL emit garbage sometimes)."""
+    from room_amd.engine.providers import parse_tool_calls
+
+    # broken JSON inside a block
+    assert parse_tool_calls('<tool_call>{not json}</tool_call>') == []
+    # unclosed tag
+    assert parse_tool_calls('<tool_call>{"name": "a"}') == []
+    # empty body
+    assert parse_tool_calls('<tool_call></tool_call>') == []
+    # garbage beside a valid call
+    calls = parse_tool_calls(
+        '<tool_call>oops</tool_call> text '
+        '<tool_call>{"name": "good", "arguments": {}}</tool_call>')
+    assert [c.name for c in calls] == ["good"]
+    # non-dict JSON
+    assert parse_tool_calls('<tool_call>[1,2]</tool_call>') == []
+    # missing name key
+    assert parse_tool_calls('<tool_call>{"arguments": {}}</tool_call>') == []
+
+
+def test_tool_executor_exception_becomes_tool_error(db):
+    """A tool executor that raises must surface as a tool-result error in
+    the transcript, not abort the cycle (reference executeQueenTool
+    try/catch semantics)."""
+    from room_amd.engine.providers import StubEngine, execute_agent, register_engine
+    from room_amd.engine.types import AgentExecutionOptions, ToolDef
+
+    register_engine("stub", StubEngine())
+
+    def boom(call):
+        raise RuntimeError("tool exploded")
+
+    opts = AgentExecutionOptions(
+        prompt="use the tool", model="stub", system_prompt="s", max_turns=2,
+        tools=[ToolDef(name="set_goal", description="d",
+                       parameters={"type": "object", "properties": {}})],
+        tool_executor=boom)
+    res = execute_agent(opts)
+    # loop completed (success or graceful failure), never an unhandled raise
+    assert res is not None
+    joined = " ".join(m.get("content", "") for m in (res.messages or []))
+    assert "tool exploded" in joined or res.success
