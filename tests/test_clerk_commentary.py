@@ -210,3 +210,58 @@ def test_clerk_chat_falls_back_to_stub(tmp_path, monkeypatch):
     with ldb as db:
         rows = db.execute("SELECT model, success FROM clerk_usage").fetchall()
     assert rows[-1]["model"] == "stub"
+
+
+def test_clerk_tools_full_round_trip():
+    """Every clerk tool branch against real SQL (clerk-tools.ts parity):
+    room CRUD/lifecycle, task + message creation, keeper vote, escalation
+    answer, status."""
+    import json
+
+    from room_amd.core.clerk import CLERK_TOOLS, execute_clerk_tool
+    from room_amd.db import LockedDb, init_test_db
+    from room_amd.db import queries as q
+    from room_amd.engine.types import ToolCall
+
+    ldb = LockedDb(init_test_db())
+
+    def run(name, args):
+        return json.loads(execute_clerk_tool(ldb, ToolCall(name, args)))
+
+    rid = run("clerk_create_room", {"name": "clerk-made", "goal": "g"})["room_id"]
+    rooms = run("clerk_list_rooms", {})
+    assert any(r["id"] == rid for r in rooms)
+    assert run("clerk_pause_room", {"room_id": rid})["paused"]
+    assert run("clerk_resume_room", {"room_id": rid})["resumed"]
+    st = run("clerk_room_status", {"room_id": rid})
+    assert st["room"] == "clerk-made"
+
+    tid = run("clerk_create_task", {"name": "t", "prompt": "p",
+                                    "room_id": rid})["task_id"]
+    tasks = run("clerk_list_tasks", {"room_id": rid})
+    assert any(t["id"] == tid for t in tasks)
+
+    eid = run("clerk_send_message", {"room_id": rid, "body": "hi"})["escalation_id"]
+    assert run("clerk_answer_escalation",
+               {"escalation_id": eid, "answer": "ok"})["answered"]
+
+    with ldb as db:
+        room = q.get_room(db, rid)
+        d = q.create_decision(db, rid, room["queen_worker_id"], "plan",
+                              "high_impact")
+    out = run("clerk_keeper_vote", {"decision_id": d["id"], "vote": "approve"})
+    # a voting-status decision records the override; it applies at
+    # resolution (quorum.ts:112-133 keeper-override semantics)
+    assert out["decision_id"] == d["id"]
+    with ldb as db:
+        rec = q.get_decision(db, d["id"])
+    assert rec["keeper_vote"] == "approve"
+
+    assert run("clerk_restart_room", {"room_id": rid})["restarted"]
+    assert run("clerk_delete_room", {"room_id": rid})["deleted"]
+    assert not any(r["id"] == rid for r in run("clerk_list_rooms", {}))
+
+    # every registered tool with empty args: structured error, not a raise
+    for t in CLERK_TOOLS:
+        out = json.loads(execute_clerk_tool(ldb, ToolCall(t.name, {})))
+        assert isinstance(out, (dict, list))
