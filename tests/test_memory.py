@@ -84,3 +84,30 @@ def test_index_pending(db):
     assert svc.store.size == 1
     with ldb as conn:
         assert q.get_unembedded_entities(conn) == []
+
+
+def test_hybrid_fusion_weights(db):
+    """Fusion contract (db-queries.ts:1021-1059): FTS reciprocal-rank
+    (k=60) × 0.4 + cosine × 0.6 — a strong semantic hit outranks a
+    rank-1 FTS hit, and scores combine additively for entities in both."""
+    from room_amd.core import room as room_mod
+    from room_amd.db import queries as q
+
+    r = room_mod.create_room(db, "fuse", worker_model="stub")
+    kw = q.create_entity(db, "keyword hit", room_id=r["id"],
+                         observations=["alpha beta gamma"])
+    sem = q.create_entity(db, "semantic hit", room_id=r["id"],
+                          observations=["unrelated words entirely"])
+    both = q.create_entity(db, "both hit", room_id=r["id"],
+                           observations=["alpha beta overlap"])
+
+    semantic = [(sem["id"], 0.95), (both["id"], 0.5)]
+    hits = q.hybrid_search(db, "alpha beta", None, limit=3, room_id=r["id"],
+                           semantic_hits=semantic)
+    by_id = {h["id"]: h["score"] for h in hits}
+    # rank-1 FTS alone ≈ 0.4/61 = 0.0066; cosine 0.95 × 0.6 = 0.57
+    assert by_id[sem["id"]] > by_id[kw["id"]]
+    # "both" accumulates fts + semantic, beating its semantic part alone
+    assert by_id[both["id"]] > 0.5 * 0.6
+    # exact weight spot check for the pure-semantic entity
+    assert abs(by_id[sem["id"]] - 0.95 * 0.6) < 1e-9
