@@ -435,3 +435,30 @@ def test_sealed_ballot_min_voters_voter_health(db):
     by_id = {h["worker_id"]: h for h in health}
     assert by_id[w1["id"]]["votes_cast"] == 1
     assert by_id[w1["id"]]["is_healthy"]
+
+
+def test_rate_limit_reset_time_formats():
+    """All three reset-time formats parse (rate-limit.ts:65-94): clock
+    time, unix timestamp, and the relative forms already covered above."""
+    import time as _time
+
+    # "resets at H:MM AM/PM" — future clock time within the clamp window
+    future = _time.localtime(_time.time() + 10 * 60)
+    ampm = "AM" if future.tm_hour < 12 else "PM"
+    h12 = future.tm_hour % 12 or 12
+    info = detect_rate_limit(
+        f"usage limit reached, resets at {h12}:{future.tm_min:02d} {ampm}")
+    assert info.detected
+    assert 30_000 <= info.wait_ms <= 3_600_000
+
+    # unix timestamp ~7 minutes out
+    ts = int(_time.time()) + 420
+    info2 = detect_rate_limit(f"rate limited, resets at {ts}")
+    assert info2.detected
+    assert 6 * 60_000 < info2.wait_ms <= 8 * 60_000
+
+    # detection patterns without any parsable time: default backoff, clamped
+    info3 = detect_rate_limit("quota exceeded")
+    assert info3.detected and info3.wait_ms >= 30_000
+    assert detect_rate_limit("model overloaded, try later").detected
+    assert not detect_rate_limit(None).detected
