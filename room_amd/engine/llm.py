@@ -253,6 +253,12 @@ class LocalEngine:
                    else Qwen3MoEConfig.qwen3_coder_30b())
         self.cfg = cfg
         self.device = torch.device(device)
+        if self.device.type == "cuda" and self.device.index is None:
+            # resolve NOW, in the constructing thread: the CUDA current
+            # device is per-thread, so the scheduler thread would otherwise
+            # see device 0 on every rank of a one-process-per-GPU job and
+            # launch kernels on rank-0's streams against rank-k's memory
+            self.device = torch.device("cuda", torch.cuda.current_device())
         t0 = time.time()
         self.model = Qwen3MoEModel(cfg, self.device, seed=seed)
         self.load_seconds = time.time() - t0
@@ -387,9 +393,7 @@ class LocalEngine:
         req.done.set()
 
     def _scheduler_loop(self) -> None:
-        idx = self.device.index if self.device.index is not None \
-            else torch.cuda.current_device()
-        torch.cuda.set_device(idx)
+        torch.cuda.set_device(self.device.index)   # resolved in __init__
         while not self._stop:
             try:
                 self._scheduler_iteration()
