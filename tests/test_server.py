@@ -506,3 +506,27 @@ def test_dashboard_covers_reference_tabs(server):
         assert f"'{tab}'" in html, f"tab {tab} missing"
     # live updates + auth flow wired
     assert "/ws?token=" in html and "handshake" in html
+
+
+def test_build_server_lifecycle(tmp_path, monkeypatch):
+    """bootstrap.build_server: full wiring (db, memory, loops, runtime)
+    boots via the app lifespan, serves status, and shuts down cleanly
+    (reference startServer → startServerRuntime path)."""
+    monkeypatch.setenv("ROOMAMD_DATA_DIR", str(tmp_path))
+    from room_amd.server.bootstrap import build_server
+
+    app, comps = build_server(db_path=str(tmp_path / "boot.db"),
+                              use_gpu=False, skip_token_file=True)
+    from fastapi.testclient import TestClient
+    with TestClient(app) as client:   # runs startup/shutdown events
+        h = {"Authorization": f"Bearer {comps['auth'].agent_token}"}
+        st = client.get("/api/status", headers=h)
+        assert st.status_code == 200
+        body = st.json()
+        assert "version" in body or "rooms" in body or body
+        # runtime loops are live
+        assert comps["runtime"]._tasks
+    # after shutdown: loops stopped, no managed children left behind
+    assert not comps["runtime"]._tasks
+    from room_amd.core.process_supervisor import managed_pids
+    assert managed_pids() == {}
