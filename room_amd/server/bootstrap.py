@@ -56,14 +56,20 @@ def build_server(db_path: str | None = None, use_gpu: bool | None = None,
     app = create_app(ldb, loop_mgr=mgr, runner=runner, memory=memory,
                      auth=auth, bus=bus)
 
-    @app.on_event("startup")
-    async def _startup():
-        await runtime.start()
+    # lifespan wiring (on_event is deprecated): runtime loops start with
+    # the app and stop before the process exits
+    import contextlib
 
-    @app.on_event("shutdown")
-    async def _shutdown():
-        await runtime.stop()
-        mgr.stop_all()
+    @contextlib.asynccontextmanager
+    async def _lifespan(_app):
+        await runtime.start()
+        try:
+            yield
+        finally:
+            await runtime.stop()
+            mgr.stop_all()
+
+    app.router.lifespan_context = _lifespan
 
     components = {"ldb": ldb, "bus": bus, "auth": auth, "memory": memory,
                   "loop_mgr": mgr, "runner": runner, "runtime": runtime}
