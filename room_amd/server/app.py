@@ -1637,7 +1637,8 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
                     if bal and bal.get("balance") is not None else {})
         return {"totalBalance": sum(v for v in by_chain.values() if v) or 0,
                 "byChain": by_chain, "address": w["address"],
-                "fetchedAt": _dt.datetime.utcnow().isoformat() + "Z"}
+                "fetchedAt": _dt.datetime.now(_dt.timezone.utc)
+                .isoformat().replace("+00:00", "Z")}
 
     @app.get("/api/rooms/{room_id}/wallet/onramp-url")
     async def wallet_onramp_url(room_id: int, role: str = Depends(get_role)):
@@ -1840,7 +1841,7 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
         await ws.accept()
         subscribed: set[str] = set()
         queue: asyncio.Queue = asyncio.Queue(maxsize=1000)
-        loop = asyncio.get_event_loop()
+        loop = asyncio.get_running_loop()
 
         def on_event(channel: str, event: dict) -> None:
             if channel in subscribed or "*" in subscribed:
