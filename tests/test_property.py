@@ -169,3 +169,77 @@ class AdmissionMachine(RuleBasedStateMachine):
 
 
 TestAdmissionMachine = AdmissionMachine.TestCase
+
+
+class KVCacheMachine(RuleBasedStateMachine):
+    """Block-pool accounting invariants under arbitrary alloc/extend/free
+    interleavings (engine/kv_cache.py, CPU tensors): no block owned twice,
+    conservation of blocks, capacity math, slot reuse safety."""
+
+    def __init__(self):
+        super().__init__()
+        import torch
+
+        from room_amd.engine.kv_cache import BLOCK_SIZE, PagedKVCache
+        self.BS = BLOCK_SIZE
+        self.c = PagedKVCache(num_layers=1, num_kv_heads=1, head_dim=8,
+                              num_blocks=64, max_seqs=8, max_blocks_per_seq=16,
+                              device=torch.device("cpu"))
+        self.live: set[int] = set()
+
+    @rule()
+    def alloc(self):
+        if self.c.free_slots:
+            slot = self.c.alloc_seq()
+            assert slot not in self.live
+            self.live.add(slot)
+        else:
+            import pytest as _pytest
+            with _pytest.raises(RuntimeError):
+                self.c.alloc_seq()
+
+    @rule(tokens=st.integers(min_value=1, max_value=260))
+    def extend(self, tokens):
+        if not self.live:
+            return
+        slot = sorted(self.live)[0]
+        need = (tokens + self.BS - 1) // self.BS
+        if need > self.c.max_blocks_per_seq:
+            import pytest as _pytest
+            with _pytest.raises(RuntimeError):
+                self.c.ensure_capacity(slot, tokens)
+        elif need - self.c.seq_nblocks[slot] > len(self.c.free_blocks):
+            import pytest as _pytest
+            with _pytest.raises(RuntimeError):
+                self.c.ensure_capacity(slot, tokens)
+        else:
+            self.c.ensure_capacity(slot, tokens)
+            assert self.c.seq_nblocks[slot] >= need
+
+    @rule()
+    def free(self):
+        if self.live:
+            slot = sorted(self.live)[-1]
+            self.c.free_seq(slot)
+            self.live.discard(slot)
+
+    @invariant()
+    def no_block_shared_and_conserved(self):
+        owned = []
+        for slot in self.live:
+            n = self.c.seq_nblocks.get(slot, 0)
+            owned.extend(self.c._bt_host[slot][:n])
+        assert len(owned) == len(set(owned)), "block owned twice"
+        assert not (set(owned) & set(self.c.free_blocks)), "owned AND free"
+        assert 0 not in owned, "reserved scrap block handed out"
+        # conservation: free + owned == all blocks minus the reserved one
+        assert len(self.c.free_blocks) + len(owned) == self.c.num_blocks - 1
+
+    @invariant()
+    def slots_conserved(self):
+        assert len(self.c.free_slots) + len(self.live) == self.c.max_seqs
+
+
+KVCacheTest = KVCacheMachine.TestCase
+KVCacheTest.settings = settings(max_examples=60, stateful_step_count=40,
+                                deadline=None)
