@@ -126,14 +126,33 @@ def resolve_voting_decision(db: sqlite3.Connection, decision_id: int) -> dict:
             db.execute("UPDATE workers SET votes_missed = votes_missed + 1"
                        " WHERE id = ?", (w["id"],))
     src = t.get("source", "local")
-    if t["yes"] > t["no"]:
+    room = q.get_room(db, decision["room_id"])
+    threshold = (room["config"].get("threshold", "majority")
+                 if room else "majority")
+    yes, no = t["yes"], t["no"]
+    if threshold == "unanimous":
+        # every cast yes/no vote must be yes (README "unanimous")
+        if yes > 0 and no == 0:
+            q.resolve_decision(db, decision_id, "approved",
+                               f"{yes}-0 unanimous ({src})")
+        else:
+            q.resolve_decision(db, decision_id, "rejected",
+                               f"{yes}-{no} not unanimous ({src})")
+    elif threshold == "supermajority":
+        # ≥2/3 of cast yes/no votes
+        if (yes + no) > 0 and 3 * yes >= 2 * (yes + no):
+            q.resolve_decision(db, decision_id, "approved",
+                               f"{yes}-{no} supermajority ({src})")
+        else:
+            q.resolve_decision(db, decision_id, "rejected",
+                               f"{yes}-{no} below 2/3 ({src})")
+    elif yes > no:
         q.resolve_decision(db, decision_id, "approved",
-                           f"{t['yes']}-{t['no']} ({src})")
-    elif t["no"] > t["yes"]:
+                           f"{yes}-{no} ({src})")
+    elif no > yes:
         q.resolve_decision(db, decision_id, "rejected",
-                           f"{t['yes']}-{t['no']} ({src})")
+                           f"{yes}-{no} ({src})")
     else:
-        room = q.get_room(db, decision["room_id"])
         tie_breaker = room["config"].get("tieBreaker", "queen") if room else "queen"
         if tie_breaker == "queen" and decision["proposer_id"] == (
                 room or {}).get("queen_worker_id"):

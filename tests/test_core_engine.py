@@ -462,3 +462,43 @@ def test_rate_limit_reset_time_formats():
     assert info3.detected and info3.wait_ms >= 30_000
     assert detect_rate_limit("model overloaded, try later").detected
     assert not detect_rate_limit(None).detected
+
+
+def test_quorum_threshold_modes(db):
+    """Room-config vote thresholds (reference README "Quorum Voting"):
+    majority (default, queen tie-break), supermajority (≥2/3 of cast
+    yes/no), unanimous (any no rejects)."""
+    import json as _json
+
+    from room_amd.core import quorum
+
+    def mkroom(threshold):
+        r = room.create_room(db, f"th-{threshold}", worker_model="stub")
+        cfg = dict(r["config"])
+        cfg["threshold"] = threshold
+        q.update_room(db, r["id"], config=_json.dumps(cfg))
+        ws = [q.create_worker(db, f"w{i}", "p", room_id=r["id"])
+              for i in range(3)]
+        return q.get_room(db, r["id"]), ws
+
+    def decide(r, ws, votes):
+        d = q.create_decision(db, r["id"], r["queen_worker_id"], "p",
+                              "high_impact")
+        for w, v in zip(ws, votes):
+            quorum.vote(db, d["id"], w["id"], v)
+        return quorum.resolve_voting_decision(db, d["id"])
+
+    # supermajority: 2/3 yes passes, 2/4... with 3 voters: 2 yes 1 no = 2/3 ✓
+    r, ws = mkroom("supermajority")
+    assert decide(r, ws, ["yes", "yes", "no"])["status"] == "approved"
+    assert decide(r, ws, ["yes", "no", "no"])["status"] == "rejected"
+    # simple majority would pass 2-1 either way; supermajority rejects 1-2
+
+    # unanimous: one no rejects even with majority yes
+    r2, ws2 = mkroom("unanimous")
+    assert decide(r2, ws2, ["yes", "yes", "no"])["status"] == "rejected"
+    assert decide(r2, ws2, ["yes", "yes", "yes"])["status"] == "approved"
+
+    # majority (default): 2-1 approves
+    r3, ws3 = mkroom("majority")
+    assert decide(r3, ws3, ["yes", "yes", "no"])["status"] == "approved"
