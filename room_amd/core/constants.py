@@ -80,6 +80,14 @@ QUEEN_DEFAULTS_BY_PLAN = {
     "api": {"queenCycleGapMs": 2 * 60 * 1000, "queenMaxTurns": 50},
 }
 
+# constants.ts:168-174 — codex/ChatGPT plan pacing map
+CHATGPT_DEFAULTS_BY_PLAN = {
+    "none": {"queenCycleGapMs": 10 * 60 * 1000, "queenMaxTurns": 50},
+    "plus": {"queenCycleGapMs": 5 * 60 * 1000, "queenMaxTurns": 50},
+    "pro": {"queenCycleGapMs": 2 * 60 * 1000, "queenMaxTurns": 50},
+    "api": {"queenCycleGapMs": 2 * 60 * 1000, "queenMaxTurns": 50},
+}
+
 WORKER_ROLE_PRESETS = {
     "guardian": {
         "cycleGapMs": 30_000,

@@ -196,11 +196,33 @@ def create_app(ldb: LockedDb, loop_mgr=None, runner=None, memory=None,
     async def create_room(payload: dict = Body(...), role: str = Depends(rw)):
         if not payload.get("name"):
             raise HTTPException(400, "name required")
+        from ..core.constants import (CHATGPT_DEFAULTS_BY_PLAN,
+                                      QUEEN_DEFAULTS_BY_PLAN)
         with ldb as db:
             room = room_mod.create_room(
                 db, payload["name"], goal=payload.get("goal"),
                 worker_model=payload.get("worker_model", "qwen3-coder-30b"),
                 queen_cycle_gap_ms=payload.get("queen_cycle_gap_ms"))
+            # plan-aware queen pacing defaults (reference rooms.ts:131-147):
+            # explicit gap in the payload wins; otherwise the keeper's
+            # queen_model + plan settings pick the pacing map entry
+            if payload.get("queen_cycle_gap_ms") is None:
+                queen_model = q.get_setting(db, "queen_model")
+                if queen_model == "codex":
+                    raw = q.get_setting(db, "chatgpt_plan") or ""
+                    plan = CHATGPT_DEFAULTS_BY_PLAN.get(
+                        raw, CHATGPT_DEFAULTS_BY_PLAN["none"])
+                else:
+                    raw = q.get_setting(db, "claude_plan") or ""
+                    plan = QUEEN_DEFAULTS_BY_PLAN.get(
+                        raw, QUEEN_DEFAULTS_BY_PLAN["none"])
+                q.update_room(db, room["id"],
+                              queen_cycle_gap_ms=plan["queenCycleGapMs"],
+                              queen_max_turns=plan["queenMaxTurns"])
+                if queen_model:
+                    q.update_worker(db, room["queen_worker_id"],
+                                    model=queen_model)
+                room = q.get_room(db, room["id"])
         bus.emit("rooms", "room_created", {"id": room["id"]})
         return room
 

@@ -530,3 +530,34 @@ def test_build_server_lifecycle(tmp_path, monkeypatch):
     assert not comps["runtime"]._tasks
     from room_amd.core.process_supervisor import managed_pids
     assert managed_pids() == {}
+
+
+def test_room_creation_plan_aware_pacing(server):
+    """rooms.ts:131-147: queen pacing defaults from the keeper's plan
+    settings; explicit payload gap wins; queen_model setting applied."""
+    client, h, *_ = server
+    # no settings → plan "none" → 10-min gap
+    r1 = client.post("/api/rooms", json={"name": "p1",
+                                         "worker_model": "stub"},
+                     headers=h).json()
+    assert r1["queen_cycle_gap_ms"] == 10 * 60 * 1000
+    # claude_plan=max → 30s gap; queen_model setting lands on the queen
+    client.put("/api/settings/claude_plan", json={"value": "max"}, headers=h)
+    client.put("/api/settings/queen_model", json={"value": "stub"}, headers=h)
+    r2 = client.post("/api/rooms", json={"name": "p2",
+                                         "worker_model": "stub"},
+                     headers=h).json()
+    assert r2["queen_cycle_gap_ms"] == 30 * 1000
+    # explicit gap wins over the plan
+    r3 = client.post("/api/rooms", json={"name": "p3", "worker_model": "stub",
+                                         "queen_cycle_gap_ms": 77_000},
+                     headers=h).json()
+    assert r3["queen_cycle_gap_ms"] == 77_000
+    # codex branch uses the ChatGPT map
+    client.put("/api/settings/queen_model", json={"value": "codex"},
+               headers=h)
+    client.put("/api/settings/chatgpt_plan", json={"value": "pro"}, headers=h)
+    r4 = client.post("/api/rooms", json={"name": "p4",
+                                         "worker_model": "stub"},
+                     headers=h).json()
+    assert r4["queen_cycle_gap_ms"] == 2 * 60 * 1000
